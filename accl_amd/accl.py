@@ -1,0 +1,139 @@
+"""High-level Python API (the PyACCL-equivalent surface).
+
+Construction mirrors the reference driver's bring-up: process launch and
+rank discovery are the caller's job (the reference uses MPI only for that —
+test/host/xrt/include/fixture.hpp:127); here the bootstrap allgather of the
+arena IPC handles runs over torch.distributed when it is initialized, and is
+computed locally for the emulator (shm names are deterministic).
+"""
+import os
+
+from . import _core
+
+DataType = _core.DataType
+ReduceFunction = _core.ReduceFunction
+
+
+def emu_job_name(seed=None):
+    """A job name every rank can derive identically (pass a shared seed)."""
+    seed = seed if seed is not None else os.environ.get("ACCL_EMU_JOB", "accljob")
+    return f"accl_{seed}"
+
+
+_TORCH_DT = None
+
+
+def _torch_dtype_map():
+    global _TORCH_DT
+    if _TORCH_DT is None:
+        import torch
+        _TORCH_DT = {
+            DataType.float32: torch.float32,
+            DataType.float64: torch.float64,
+            DataType.float16: torch.float16,
+            DataType.bfloat16: torch.bfloat16,
+            DataType.int32: torch.int32,
+            DataType.int64: torch.int64,
+            DataType.int8: torch.int8,
+        }
+    return _TORCH_DT
+
+
+class ACCL:
+    """One rank of the collective engine.
+
+    backend: "emu" (CPU emulator over shm), "gpu" (MI355X persistent engine),
+    or "auto" (gpu if a HIP device is visible, else emu).
+    """
+
+    def __init__(self, nranks=None, rank=None, backend="auto", job=None,
+                 device=None, heap_bytes=None, bootstrap="auto", **opts):
+        if nranks is None:
+            nranks = int(os.environ.get("WORLD_SIZE", "1"))
+        if rank is None:
+            rank = int(os.environ.get("RANK", "0"))
+        if backend == "auto":
+            backend = "gpu" if self._has_gpu() else "emu"
+        self.backend_name = backend
+        if backend == "emu":
+            hb = heap_bytes or (256 << 20)
+            self._a = _core.create_emu(nranks, rank, emu_job_name(job), hb, opts)
+        elif backend == "gpu":
+            if device is None:
+                device = int(os.environ.get("LOCAL_RANK", "0"))
+            hb = heap_bytes or (8 << 30)
+            self._a = _core.create_gpu(nranks, rank, device, hb,
+                                       int(os.environ.get("ACCL_ENGINE_WGS", "0")),
+                                       opts)
+        else:
+            raise ValueError(f"unknown backend {backend!r}")
+        self.device_index = device or 0
+        self._connect(bootstrap, nranks, rank, job)
+
+    @staticmethod
+    def _has_gpu():
+        try:
+            import torch
+            return torch.cuda.is_available()
+        except Exception:
+            return False
+
+    def _connect(self, bootstrap, nranks, rank, job):
+        blob = self._a.local_blob()
+        if nranks == 1:
+            self._a.connect([blob])
+            return
+        if self.backend_name == "emu":
+            # shm names are deterministic — compute peers' blobs locally
+            blobs = [f"/{emu_job_name(job)}_r{r}".encode() for r in range(nranks)]
+            blobs[rank] = blob
+            self._a.connect(blobs)
+            return
+        if bootstrap in ("auto", "torch"):
+            import torch.distributed as dist
+            if not dist.is_initialized():
+                raise RuntimeError(
+                    "accl_amd: torch.distributed must be initialized (gloo is "
+                    "fine) to bootstrap a multi-rank GPU engine")
+            objs = [None] * nranks
+            dist.all_gather_object(objs, bytes(blob))
+            self._a.connect(list(objs))
+        else:
+            raise ValueError(f"unknown bootstrap {bootstrap!r}")
+
+    # ---------------- buffers ----------------
+    def create_buffer(self, count, dtype=DataType.float32, device_only=False):
+        return self._a.create_buffer(count, dtype, device_only)
+
+    def tensor(self, buf):
+        """Zero-copy torch view of a buffer (device tensor on the gpu
+        backend, cpu tensor view is not supported for emu — use read/write)."""
+        import torch
+        cap = self._a.buffer_dlpack(buf, self.device_index)
+        return torch.from_dlpack(cap)
+
+    def buffer_like(self, tensor, device_only=True):
+        dt = {v: k for k, v in _torch_dtype_map().items()}[tensor.dtype]
+        return self.create_buffer(tensor.numel(), dt, device_only)
+
+    # ---------------- properties ----------------
+    @property
+    def rank(self):
+        return self._a.rank
+
+    @property
+    def nranks(self):
+        return self._a.nranks
+
+    # ---------------- ops (delegate) ----------------
+    def __getattr__(self, name):
+        if name in ("copy", "combine", "send", "recv", "bcast", "scatter",
+                    "gather", "allgather", "reduce", "allreduce",
+                    "reduce_scatter", "alltoall", "barrier", "nop",
+                    "create_communicator", "split_communicator",
+                    "free_request", "deinit"):
+            return getattr(self._a, name)
+        raise AttributeError(name)
+
+    def close(self):
+        self._a.deinit()

@@ -1,0 +1,273 @@
+// Python bindings for accl_amd — the PyACCL-equivalent surface.
+// Buffers interoperate zero-copy with torch via DLPack (__dlpack__), so the
+// bench path touches no host memory.
+#include <pybind11/functional.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+#include <cstring>
+#include "../core/accl.hpp"
+#include "../emu/emudevice.hpp"
+#include "../gpu/gpudevice.hpp"
+
+namespace py = pybind11;
+using namespace accl;
+
+// ------------------------------- minimal DLPack (no external header) ------
+namespace dlpack {
+struct DLDevice { int32_t device_type; int32_t device_id; };
+struct DLDataType { uint8_t code; uint8_t bits; uint16_t lanes; };
+struct DLTensor {
+  void* data; DLDevice device; int32_t ndim; DLDataType dtype;
+  int64_t* shape; int64_t* strides; uint64_t byte_offset;
+};
+struct DLManagedTensor {
+  DLTensor dl_tensor; void* manager_ctx;
+  void (*deleter)(DLManagedTensor*);
+};
+constexpr int kDLCPU = 1, kDLROCM = 10;
+}  // namespace dlpack
+
+static dlpack::DLDataType dl_dtype(DataType dt) {
+  switch (dt) {
+    case DataType::float32: return {2, 32, 1};
+    case DataType::float64: return {2, 64, 1};
+    case DataType::float16: return {2, 16, 1};
+    case DataType::bfloat16: return {4, 16, 1};
+    case DataType::int32: return {0, 32, 1};
+    case DataType::int64: return {0, 64, 1};
+    case DataType::int8: return {0, 8, 1};
+    default: return {2, 32, 1};
+  }
+}
+
+struct DLHolder {
+  int64_t shape[1];
+  dlpack::DLManagedTensor mt;
+};
+
+static py::capsule make_dlpack(BaseBuffer& b, bool gpu, int device_id) {
+  auto* h = new DLHolder();
+  h->shape[0] = int64_t(b.count());
+  h->mt.dl_tensor.data = b.device_ptr();
+  h->mt.dl_tensor.device = {gpu ? dlpack::kDLROCM : dlpack::kDLCPU,
+                            gpu ? device_id : 0};
+  h->mt.dl_tensor.ndim = 1;
+  h->mt.dl_tensor.dtype = dl_dtype(b.dtype());
+  h->mt.dl_tensor.shape = h->shape;
+  h->mt.dl_tensor.strides = nullptr;
+  h->mt.dl_tensor.byte_offset = 0;
+  h->mt.manager_ctx = h;
+  h->mt.deleter = [](dlpack::DLManagedTensor* t) {
+    delete (DLHolder*)t->manager_ctx;
+  };
+  return py::capsule(&h->mt, "dltensor", [](PyObject* cap) {
+    if (PyCapsule_IsValid(cap, "dltensor")) {
+      auto* mt = (dlpack::DLManagedTensor*)PyCapsule_GetPointer(cap, "dltensor");
+      if (mt && mt->deleter) mt->deleter(mt);
+    }
+  });
+}
+
+// ------------------------------------------------------------ module ----
+PYBIND11_MODULE(_core, m) {
+  m.doc() = "accl_amd core: MI355X-native ACCL-class collective engine";
+
+  py::enum_<DataType>(m, "DataType")
+      .value("none", DataType::none)
+      .value("float16", DataType::float16)
+      .value("float32", DataType::float32)
+      .value("float64", DataType::float64)
+      .value("int32", DataType::int32)
+      .value("int64", DataType::int64)
+      .value("bfloat16", DataType::bfloat16)
+      .value("int8", DataType::int8);
+
+  py::enum_<ReduceFunction>(m, "ReduceFunction")
+      .value("SUM", ReduceFunction::SUM)
+      .value("MAX", ReduceFunction::MAX);
+
+  m.attr("TAG_ANY") = py::int_(u64(TAG_ANY));
+  m.attr("GLOBAL_COMM") = py::int_(GLOBAL_COMM);
+
+  py::class_<BaseBuffer>(m, "Buffer")
+      .def_property_readonly("count", &BaseBuffer::count)
+      .def_property_readonly("bytes", &BaseBuffer::bytes)
+      .def_property_readonly("dtype", &BaseBuffer::dtype)
+      .def_property_readonly("arena_offset", &BaseBuffer::arena_offset)
+      .def("sync_to_device", &BaseBuffer::sync_to_device)
+      .def("sync_from_device", &BaseBuffer::sync_from_device)
+      .def("slice",
+           [](BaseBuffer& b, u64 start, u64 end) { return b.slice(start, end); })
+      .def("write",
+           [](BaseBuffer& b, py::buffer data, bool to_device) {
+             py::buffer_info info = data.request();
+             u64 n = u64(info.size) * u64(info.itemsize);
+             if (n > b.bytes()) throw std::runtime_error("write too large");
+             if (b.host_ptr()) std::memcpy(b.host_ptr(), info.ptr, n);
+             if (to_device) {
+               if (b.host_ptr()) b.sync_to_device();
+             }
+           },
+           py::arg("data"), py::arg("to_device") = true)
+      .def("read",
+           [](BaseBuffer& b, py::buffer out, bool from_device) {
+             py::buffer_info info = out.request(true);
+             u64 n = u64(info.size) * u64(info.itemsize);
+             if (n > b.bytes()) throw std::runtime_error("read too large");
+             if (from_device) b.sync_from_device();
+             std::memcpy(info.ptr, b.host_ptr(), n);
+           },
+           py::arg("out"), py::arg("from_device") = true);
+
+  py::class_<Request>(m, "Request")
+      .def("wait", &Request::wait, py::arg("timeout_ms") = 120000,
+           py::call_guard<py::gil_scoped_release>())
+      .def("test", &Request::test)
+      .def("retcode", &Request::retcode)
+      .def("duration_us", &Request::duration_us,
+           py::call_guard<py::gil_scoped_release>());
+
+  py::class_<ACCL>(m, "ACCL")
+      .def_property_readonly("rank", &ACCL::rank)
+      .def_property_readonly("nranks", &ACCL::nranks)
+      .def("local_blob",
+           [](ACCL& a) {
+             auto b = a.local_blob();
+             return py::bytes(b.data(), b.size());
+           })
+      .def("connect",
+           [](ACCL& a, const std::vector<py::bytes>& blobs) {
+             std::vector<std::vector<char>> v;
+             for (auto& b : blobs) {
+               std::string s = b;
+               v.emplace_back(s.begin(), s.end());
+             }
+             a.connect(v);
+           },
+           py::call_guard<py::gil_scoped_release>())
+      .def("deinit", &ACCL::deinit, py::call_guard<py::gil_scoped_release>())
+      .def("create_buffer",
+           [](ACCL& a, u64 count, DataType dt, bool device_only) {
+             return device_only ? a.create_buffer_device(count, dt)
+                                : a.create_buffer(count, dt);
+           },
+           py::arg("count"), py::arg("dtype"), py::arg("device_only") = false,
+           py::keep_alive<0, 1>())
+      .def("buffer_dlpack",
+           [](ACCL& a, BaseBuffer& b, int device_id) {
+             return make_dlpack(b, a.backend()->is_gpu(), device_id);
+           },
+           py::arg("buffer"), py::arg("device_id") = 0)
+      .def("create_communicator", &ACCL::create_communicator)
+      .def("split_communicator", &ACCL::split_communicator)
+      .def("copy", &ACCL::copy, py::arg("src"), py::arg("dst"), py::arg("count"),
+           py::arg("from_device") = false, py::arg("to_device") = false,
+           py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
+           py::call_guard<py::gil_scoped_release>())
+      .def("combine", &ACCL::combine, py::arg("count"), py::arg("function"),
+           py::arg("op0"), py::arg("op1"), py::arg("result"),
+           py::arg("from_device") = false, py::arg("to_device") = false,
+           py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
+           py::call_guard<py::gil_scoped_release>())
+      .def("send", &ACCL::send, py::arg("src"), py::arg("count"), py::arg("dst"),
+           py::arg("tag") = u32(TAG_ANY), py::arg("comm") = GLOBAL_COMM,
+           py::arg("from_device") = false,
+           py::arg("compress_dtype") = DataType::none,
+           py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
+           py::call_guard<py::gil_scoped_release>())
+      .def("recv", &ACCL::recv, py::arg("dst"), py::arg("count"), py::arg("src"),
+           py::arg("tag") = u32(TAG_ANY), py::arg("comm") = GLOBAL_COMM,
+           py::arg("to_device") = false,
+           py::arg("compress_dtype") = DataType::none,
+           py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
+           py::call_guard<py::gil_scoped_release>())
+      .def("bcast", &ACCL::bcast, py::arg("buf"), py::arg("count"),
+           py::arg("root"), py::arg("comm") = GLOBAL_COMM,
+           py::arg("from_device") = false, py::arg("to_device") = false,
+           py::arg("compress_dtype") = DataType::none,
+           py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
+           py::call_guard<py::gil_scoped_release>())
+      .def("scatter", &ACCL::scatter, py::arg("src"), py::arg("dst"),
+           py::arg("count"), py::arg("root"), py::arg("comm") = GLOBAL_COMM,
+           py::arg("from_device") = false, py::arg("to_device") = false,
+           py::arg("compress_dtype") = DataType::none,
+           py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
+           py::call_guard<py::gil_scoped_release>())
+      .def("gather", &ACCL::gather, py::arg("src"), py::arg("dst"),
+           py::arg("count"), py::arg("root"), py::arg("comm") = GLOBAL_COMM,
+           py::arg("from_device") = false, py::arg("to_device") = false,
+           py::arg("compress_dtype") = DataType::none,
+           py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
+           py::call_guard<py::gil_scoped_release>())
+      .def("allgather", &ACCL::allgather, py::arg("src"), py::arg("dst"),
+           py::arg("count"), py::arg("comm") = GLOBAL_COMM,
+           py::arg("from_device") = false, py::arg("to_device") = false,
+           py::arg("compress_dtype") = DataType::none,
+           py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
+           py::call_guard<py::gil_scoped_release>())
+      .def("reduce", &ACCL::reduce, py::arg("src"), py::arg("dst"),
+           py::arg("count"), py::arg("root"), py::arg("function"),
+           py::arg("comm") = GLOBAL_COMM, py::arg("from_device") = false,
+           py::arg("to_device") = false,
+           py::arg("compress_dtype") = DataType::none,
+           py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
+           py::call_guard<py::gil_scoped_release>())
+      .def("allreduce", &ACCL::allreduce, py::arg("src"), py::arg("dst"),
+           py::arg("count"), py::arg("function"), py::arg("comm") = GLOBAL_COMM,
+           py::arg("from_device") = false, py::arg("to_device") = false,
+           py::arg("compress_dtype") = DataType::none,
+           py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
+           py::call_guard<py::gil_scoped_release>())
+      .def("reduce_scatter", &ACCL::reduce_scatter, py::arg("src"),
+           py::arg("dst"), py::arg("count"), py::arg("function"),
+           py::arg("comm") = GLOBAL_COMM, py::arg("from_device") = false,
+           py::arg("to_device") = false,
+           py::arg("compress_dtype") = DataType::none,
+           py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
+           py::call_guard<py::gil_scoped_release>())
+      .def("alltoall", &ACCL::alltoall, py::arg("src"), py::arg("dst"),
+           py::arg("count"), py::arg("comm") = GLOBAL_COMM,
+           py::arg("from_device") = false, py::arg("to_device") = false,
+           py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
+           py::call_guard<py::gil_scoped_release>())
+      .def("barrier", &ACCL::barrier, py::arg("comm") = GLOBAL_COMM,
+           py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
+           py::call_guard<py::gil_scoped_release>())
+      .def("nop", &ACCL::nop, py::arg("run_async") = false,
+           py::return_value_policy::reference)
+      .def("free_request", &ACCL::free_request);
+
+  auto mk_cfg = [](u32 nranks, u32 rank, py::dict opts) {
+    ProtoConfig c = default_proto_config(nranks, rank);
+    if (opts.contains("n_slots")) c.n_slots = opts["n_slots"].cast<u32>();
+    if (opts.contains("slot_bytes")) c.slot_bytes = opts["slot_bytes"].cast<u32>();
+    if (opts.contains("max_eager")) c.max_eager = opts["max_eager"].cast<u64>();
+    if (opts.contains("timeout_us")) c.timeout_us = opts["timeout_us"].cast<u64>();
+    return c;
+  };
+
+  m.def("create_emu",
+        [mk_cfg](u32 nranks, u32 rank, const std::string& job, u64 heap_bytes,
+                 py::dict opts) {
+          ProtoConfig c = mk_cfg(nranks, rank, opts);
+          auto be = std::unique_ptr<Backend>(
+              new EmuDevice(nranks, rank, job, &c, heap_bytes));
+          return new ACCL(std::move(be));
+        },
+        py::arg("nranks"), py::arg("rank"), py::arg("job"),
+        py::arg("heap_bytes") = u64(256u << 20), py::arg("opts") = py::dict());
+
+  m.def("create_gpu",
+        [mk_cfg](u32 nranks, u32 rank, int device, u64 heap_bytes, int wgs,
+                 py::dict opts) {
+          ProtoConfig c = mk_cfg(nranks, rank, opts);
+          auto be = std::unique_ptr<Backend>(
+              new GpuDevice(nranks, rank, device, &c, heap_bytes, wgs));
+          return new ACCL(std::move(be));
+        },
+        py::arg("nranks"), py::arg("rank"), py::arg("device") = 0,
+        py::arg("heap_bytes") = u64(8ull << 30), py::arg("wgs") = 0,
+        py::arg("opts") = py::dict());
+
+  m.def("error_to_string", &error_to_string);
+}

@@ -1,0 +1,281 @@
+// accl_amd transport protocol — the arena layout + publish/consume primitives
+// shared by the host runtime, the CPU emulator and the GPU persistent engine.
+//
+// Semantics reproduced from the reference's eager/rendezvous dual transport:
+//   - eager rx slot rings   ~ rx-buffer table + packetizer/depacketizer + rxbuf
+//     offload engines (reference: kernels/cclo/hls/rxbuf_offload/*.cpp,
+//     eth_intf/udp_packetizer.cpp:24-83; table layout ccl_offload_control.h:267-295)
+//   - rendezvous addr/done  ~ RNDZV machinery (reference:
+//     ccl_offload_control.c:142-408, rdma_sq_handler.cpp:23-142)
+// but re-designed for xGMI peer HBM: the sender writes payload directly into the
+// receiver's arena (no wire, no reassembly); ordering comes from system-scope
+// release/acquire publication instead of in-order AXI streams.
+//
+// Compiles as C++17 host code and as HIP device code (single-source property,
+// reference: ccl_offload_control.h:229-264 MB_FW_EMULATION).
+#pragma once
+#include "types.hpp"
+
+namespace accl {
+
+// ------------------------------------------------------------ atomics shim
+// Host side uses GCC builtins (shm across processes); device side uses HIP
+// scoped atomics. System scope everywhere: peers are other GPUs (xGMI) or
+// other processes (shm).
+// Each shim is ACCL_HD with the body selected per compilation pass — the
+// device pass uses HIP scoped atomics / gfx950 fences, every host pass the
+// GCC builtins (standard single-source pattern).
+u64 wallclock_host_ns();  // defined in core/util.cpp (steady_clock)
+
+ACCL_HD inline u64 ld_sys(const volatile u64* p) {
+#if defined(__HIP_DEVICE_COMPILE__)
+  return __hip_atomic_load((const u64*)p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+#else
+  return __atomic_load_n((const u64*)p, __ATOMIC_RELAXED);
+#endif
+}
+ACCL_HD inline u32 ld_sys32(const volatile u32* p) {
+#if defined(__HIP_DEVICE_COMPILE__)
+  return __hip_atomic_load((const u32*)p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+#else
+  return __atomic_load_n((const u32*)p, __ATOMIC_RELAXED);
+#endif
+}
+ACCL_HD inline void st_sys(volatile u64* p, u64 v) {
+#if defined(__HIP_DEVICE_COMPILE__)
+  __hip_atomic_store((u64*)p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+#else
+  __atomic_store_n((u64*)p, v, __ATOMIC_RELAXED);
+#endif
+}
+ACCL_HD inline void st_sys32(volatile u32* p, u32 v) {
+#if defined(__HIP_DEVICE_COMPILE__)
+  __hip_atomic_store((u32*)p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+#else
+  __atomic_store_n((u32*)p, v, __ATOMIC_RELAXED);
+#endif
+}
+// Release all prior (plain) stores, then publish a flag. ROCm 7.2 can drop
+// the s_waitcnt after buffer_wbl2 when the wave's scoreboard looks empty —
+// restate it in asm, always (MI355X_MICROARCH.md §Workgroup dispatch,
+// compiler hazard).
+ACCL_HD inline void fence_release_sys() {
+#if defined(__HIP_DEVICE_COMPILE__)
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+#else
+  __atomic_thread_fence(__ATOMIC_RELEASE);
+#endif
+}
+ACCL_HD inline void fence_acquire_sys() {
+#if defined(__HIP_DEVICE_COMPILE__)
+  __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+#else
+  __atomic_thread_fence(__ATOMIC_ACQUIRE);
+#endif
+}
+ACCL_HD inline void cpu_pause() {
+#if defined(__HIP_DEVICE_COMPILE__)
+  __builtin_amdgcn_s_sleep(8);
+#elif defined(__x86_64__)
+  __builtin_ia32_pause();
+#endif
+}
+ACCL_HD inline u64 wallclock() {  // 100 MHz ticks on both engines
+#if defined(__HIP_DEVICE_COMPILE__)
+  return __builtin_amdgcn_s_memrealtime();
+#else
+  return wallclock_host_ns() / 10;
+#endif
+}
+constexpr u64 TICKS_PER_US = 100;
+
+// -------------------------------------------------------------- parameters
+// Runtime-configurable (ACCL: rx-buffer count/size + max eager/rendezvous
+// sizes, accl.hpp:102-104); these are the compile-time maxima / defaults.
+struct ProtoConfig {
+  u32 n_slots;        // eager rx slots per (src,dst) pair        [default 8]
+  u32 slot_bytes;     // payload bytes per slot                   [default 1 MiB]
+  u32 n_rndzv;        // rendezvous ring depth per pair           [default 8]
+  u32 n_stream;       // device-stream ring slots per rank        [default 8]
+  u32 stream_bytes;   // device-stream slot payload
+  u64 max_eager;      // bytes; above this send/recv go rendezvous when legal
+  u64 timeout_us;     // spin bound for every wait
+  u32 nranks;
+  u32 rank;
+};
+
+inline ProtoConfig default_proto_config(u32 nranks, u32 rank) {
+  ProtoConfig c{};
+  c.n_slots = 8; c.slot_bytes = 1u << 20; c.n_rndzv = 8;
+  c.n_stream = 8; c.stream_bytes = 1u << 20;
+  c.max_eager = 4u << 20;      // beyond this, arena-resident transfers go direct
+  c.timeout_us = 10u * 1000 * 1000;
+  c.nranks = nranks; c.rank = rank;
+  return c;
+}
+
+// ------------------------------------------------------------ arena layout
+// One arena per rank. Peers map the whole arena (hipIpc / shm) and address
+// everything as arena_base[peer] + offset. All control words are 8-byte
+// aligned and written by exactly one rank at a time (SPSC rings per pair).
+//
+// Layout (offsets in ArenaHdr, computed by arena_layout()):
+//   [0]                ArenaHdr
+//   hdr.eager_off      per src-peer EagerChan control (headers + credit)
+//   hdr.rndzv_off      per peer RndzvRing pair (addr ring + done ring)
+//   hdr.stream_off     stream rings (device-initiated collectives, stream_put)
+//   hdr.slots_off      eager payload slots [src_peer][slot]
+//   hdr.heap_off       buffer heap until hdr.arena_bytes
+
+struct alignas(64) SlotHdr {
+  u64 seq;        // published last; seq is 1-based message-segment number
+  u32 tag;
+  u32 bytes;      // payload bytes in this slot (wire bytes)
+  u64 msg_count;  // total element count of the message (first segment only)
+  u32 arith;      // packed wire dtype info (same packing as CallDesc.arith)
+  u32 flags;      // SEG_FIRST / SEG_LAST
+  u64 _pad[4];
+};
+static_assert(sizeof(SlotHdr) == 64, "");
+
+enum SegFlags : u32 { SEG_FIRST = 1, SEG_LAST = 2 };
+
+struct alignas(64) RndzvRec {
+  u64 seq;        // published last, 1-based per (pair, ring)
+  u32 tag;
+  u32 arith;
+  u64 offset;     // destination arena offset (addr ring) / echo (done ring)
+  u64 count;      // elements
+  u64 _pad[4];
+};
+static_assert(sizeof(RndzvRec) == 64, "");
+
+// Control block per ordered pair, resident in the CONSUMER's arena so every
+// poll is a local read. Writers are remote (xGMI / peer shm).
+struct alignas(64) EagerChanCtl {
+  // written by the sender (remote), polled by receiver:
+  //   SlotHdr hdr[n_slots] follows this struct.
+  // written by the receiver of the OPPOSITE direction (remote), polled locally
+  // by this rank acting as sender towards `peer`:
+  u64 credit;     // number of slots this rank's messages to `peer`... see note
+  u64 _pad[7];
+};
+// NOTE on credit placement: for channel (s -> r), slot headers+payload live in
+// r's arena at index [s]; the credit word lives in s's arena at index [r]
+// (field `credit` of s's EagerChanCtl[r]) and is advanced by r. Each side
+// polls only its own HBM.
+
+struct alignas(64) ArenaHdr {
+  u32 magic;            // 'ACCL'
+  u32 version;
+  u32 rank, nranks;
+  u32 n_slots, slot_bytes, n_rndzv, n_stream;
+  u32 stream_bytes, _pad0;
+  u64 arena_bytes;
+  u64 eager_off;        // EagerChanCtl[nranks] + SlotHdr[nranks][n_slots]
+  u64 rndzv_addr_off;   // RndzvRec[nranks][n_rndzv]  (this rank = sender side)
+  u64 rndzv_done_off;   // RndzvRec[nranks][n_rndzv]  (this rank = receiver side)
+  u64 stream_off;       // stream rings
+  u64 slots_off;        // payload [nranks][n_slots][slot_bytes]
+  u64 heap_off;
+  u64 barrier_off;      // per-peer barrier tokens u64[nranks]
+  u64 direct_off;       // per-peer cumulative direct-write progress u64[nranks]
+  u64 spare_off;        // staging region for non-arena rendezvous targets
+  u64 spare_bytes;      //   (reference: spare buffers, accl.cpp:1174-1196)
+};
+constexpr u32 ARENA_MAGIC = 0x4143434Cu;  // "ACCL"
+
+struct ArenaLayout {
+  u64 eager_off, rndzv_addr_off, rndzv_done_off, stream_off, slots_off,
+      barrier_off, direct_off, spare_off, spare_bytes, heap_off,
+      total_ctl_bytes;
+};
+
+inline ArenaLayout arena_layout(const ProtoConfig& c, u64 spare_bytes = 32u << 20) {
+  auto align_up = [](u64 x, u64 a) { return (x + a - 1) & ~(a - 1); };
+  ArenaLayout L{};
+  u64 off = align_up(sizeof(ArenaHdr), 256);
+  L.eager_off = off;
+  off += u64(c.nranks) * (sizeof(EagerChanCtl) + u64(c.n_slots) * sizeof(SlotHdr));
+  L.rndzv_addr_off = off = align_up(off, 256);
+  off += u64(c.nranks) * c.n_rndzv * sizeof(RndzvRec);
+  L.rndzv_done_off = off = align_up(off, 256);
+  off += u64(c.nranks) * c.n_rndzv * sizeof(RndzvRec);
+  L.stream_off = off = align_up(off, 256);
+  off += u64(c.n_stream) * (sizeof(SlotHdr) + c.stream_bytes);
+  L.barrier_off = off = align_up(off, 256);
+  off += u64(c.nranks) * sizeof(u64);
+  L.direct_off = off = align_up(off, 256);
+  off += u64(c.nranks) * sizeof(u64);
+  L.slots_off = off = align_up(off, 4096);
+  off += u64(c.nranks) * c.n_slots * u64(c.slot_bytes);
+  L.spare_off = off = align_up(off, 4096);
+  L.spare_bytes = spare_bytes;
+  off += spare_bytes;
+  L.heap_off = off = align_up(off, 4096);
+  L.total_ctl_bytes = off;
+  return L;
+}
+
+// --------------------------------------------------- per-rank runtime view
+// Everything an engine (GPU or CPU) needs to run the transport: its own arena
+// plus the mapped base pointer of every peer's arena. Filled by the host at
+// initialize() (the accl_network_utils analogue — SURVEY §3.1).
+struct TransportView {
+  ProtoConfig cfg;
+  char* arena[MAX_RANKS];   // arena[r] = base of rank r's arena in MY address
+                            // space (own rank: local pointer). null if absent.
+  ACCL_HD ArenaHdr* hdr(u32 r) const { return (ArenaHdr*)arena[r]; }
+  // eager channel (s -> r): control in r's arena, indexed by s
+  ACCL_HD EagerChanCtl* chan_ctl(u32 r, u32 s) const {
+    return (EagerChanCtl*)(arena[r] + hdr(r)->eager_off +
+                           u64(s) * (sizeof(EagerChanCtl) + u64(hdr(r)->n_slots) * sizeof(SlotHdr)));
+  }
+  ACCL_HD SlotHdr* slot_hdr(u32 r, u32 s, u32 slot) const {
+    return (SlotHdr*)((char*)chan_ctl(r, s) + sizeof(EagerChanCtl)) + slot;
+  }
+  ACCL_HD char* slot_payload(u32 r, u32 s, u32 slot) const {
+    const ArenaHdr* h = hdr(r);
+    return arena[r] + h->slots_off +
+           (u64(s) * h->n_slots + slot) * u64(h->slot_bytes);
+  }
+  // rendezvous addr ring for pair (sender = s): lives in s's arena, lane [r]
+  // (r = the receiver that posts into it).
+  ACCL_HD RndzvRec* rndzv_addr(u32 s, u32 r, u32 i) const {
+    return (RndzvRec*)(arena[s] + hdr(s)->rndzv_addr_off) +
+           u64(r) * hdr(s)->n_rndzv + i;
+  }
+  // rendezvous done ring: lives in receiver r's arena, lane [s].
+  ACCL_HD RndzvRec* rndzv_done(u32 r, u32 s, u32 i) const {
+    return (RndzvRec*)(arena[r] + hdr(r)->rndzv_done_off) +
+           u64(s) * hdr(r)->n_rndzv + i;
+  }
+  ACCL_HD u64* barrier_word(u32 r, u32 peer) const {
+    return (u64*)(arena[r] + hdr(r)->barrier_off) + peer;
+  }
+  // cumulative direct-write progress: in r's arena, written by peer s
+  ACCL_HD volatile u64* direct_word(u32 r, u32 s) const {
+    return (volatile u64*)(arena[r] + hdr(r)->direct_off) + s;
+  }
+  ACCL_HD char* heap_ptr(u32 r, u64 off) const { return arena[r] + off; }
+};
+
+// Per-pair sequence state, PRIVATE to one engine (not shared): lives in the
+// engine's own state block. Mirrors the communicator's inbound/outbound seq
+// counters (reference: communicator.hpp:34-39 rank_t session/seq).
+struct PairSeq {
+  u64 eager_tx[MAX_RANKS];     // segments sent to peer
+  u64 eager_rx[MAX_RANKS];     // segments consumed from peer
+  u64 credit_ret[MAX_RANKS];   // last credit value written back to peer
+  u64 rndzv_addr_tx[MAX_RANKS];
+  u64 rndzv_addr_rx[MAX_RANKS];
+  u64 rndzv_done_tx[MAX_RANKS];
+  u64 rndzv_done_rx[MAX_RANKS];
+  u64 direct_tx[MAX_RANKS];    // cumulative bytes direct-written to peer
+  u64 direct_rx[MAX_RANKS];    // cumulative bytes direct-received from peer
+  u64 barrier_epoch[MAX_RANKS];// per-PAIR barrier epoch (must match both ends)
+};
+
+}  // namespace accl

@@ -1,0 +1,858 @@
+// accl_amd collective scheduler — single source for the GPU persistent engine
+// and the CPU emulator engine.
+//
+// This is the rebuild of the reference's collective microcode
+// (reference: kernels/cclo/fw/sw_apps/ccl_offload_control/src/
+// ccl_offload_control.c — run() dispatch :2375-2459, send :573-649,
+// recv :653-710, bcast :796-988, scatter :992-1123, gather :1128-1294,
+// allgather :1297-1503, reduce :1507-1744, reduce_scatter :1748-1852,
+// allreduce :1855-2075, barrier :2078-2120, all_to_all :2123-2218), built
+// MI355X-first: instead of emitting AXI DataMover instructions it runs a set
+// of *flows* (segmented, credit-throttled, gate-chained data streams) over
+// xGMI peer memory, and instead of ring-only schedules it uses fullmesh
+// schedules that drive all 7 xGMI links of a GPU at once where P <= the
+// flow budget.
+//
+// Template parameter `Mover` is the data plane:
+//   u32  submit(const MoveDesc&)  — async; returns token (monotonic)
+//   bool poll(u32 token)          — completed? (acquire semantics on true)
+// GPU: movers = all engine workgroups (gpu/engine.hip). CPU: synchronous
+// execution (emu/emudevice.cpp).
+#pragma once
+#include "types.hpp"
+#include "proto.hpp"
+#include "move.hpp"
+
+namespace accl {
+
+constexpr u32 MAX_FLOWS = 72;        // >= 4*(P-1)+1 for fullmesh at P<=16
+constexpr u32 FLOW_INFLIGHT = 4;     // outstanding segments per flow
+                                     // (reference keeps <=3 eager moves in
+                                     // flight, ccl_offload_control.c:626-648)
+
+enum FlowKind : u8 {
+  FLOW_IDLE = 0,
+  FLOW_LOCAL,      // local move src(+red) -> dst
+  FLOW_TX,         // eager send to gpeer via rx slots
+  FLOW_RX,         // eager recv from gpeer (copy or fused reduce)
+  FLOW_TX_DIRECT,  // direct write into peer arena (rendezvous data phase)
+  FLOW_RX_DIRECT,  // wait on peer's direct-write progress word
+};
+
+struct PendSeg {
+  u32 token;
+  u32 slot;       // eager: slot index; direct: unused
+  u64 elems;
+  u64 seq;        // eager: segment seq to publish / credit value to return
+  u32 first_last; // SEG_FIRST/SEG_LAST for hdr publish
+};
+
+struct Flow {
+  u8 kind;
+  u8 func;        // ReduceFunction+1; 0 = plain copy
+  u8 sdt, wdt, ddt, bdt;  // src, wire, dst, reduce-operand dtypes
+  u32 gpeer;
+  u32 tag;        // expected tag (rx) / stamped tag (tx); TAG_ANY allowed on rx
+  u32 matched_tag;
+  const char* src;
+  char* dst;
+  const char* red;        // rx fused-reduce second operand (may alias dst)
+  u64 count;              // total elements
+  u64 submitted;          // elements handed to the mover
+  u64 done;               // elements retired IN ORDER (gates dependents)
+  const u64* gate;        // if set: submitted may not pass *gate
+  // direct mode:
+  u64 peer_off;           // destination offset in peer arena (tx_direct)
+  u64 prog_base;          // cumulative-bytes baseline (both direct kinds)
+  // pending segment fifo:
+  PendSeg pend[FLOW_INFLIGHT];
+  u32 ph, pt;             // head/tail (pt-ph = in flight)
+  u64 msg_count_hdr;      // value for SlotHdr.msg_count (tx)
+};
+
+// ---------------------------------------------------------------- engine
+template <class Mover>
+struct Cclo {
+  TransportView tv;
+  ProtoConfig cfg;
+  PairSeq sq;
+  Mover* mv;
+  CommView comms[MAX_COMMS];
+  u32 ncomms;
+  u64 timeout_ticks;
+  u64 max_eager_bytes;     // above this, arena<->arena transfers go direct
+  u32 err;                 // error bits of the current call
+  Flow flows[MAX_FLOWS];
+
+  ACCL_HD u32 me() const { return cfg.rank; }
+
+  // ---------------- tiny helpers ----------------
+  ACCL_HD char* local_ptr(u64 addr, bool arena) {
+    return arena ? tv.arena[me()] + addr : (char*)addr;
+  }
+  ACCL_HD static u64 min64(u64 a, u64 b) { return a < b ? a : b; }
+
+  ACCL_HD bool wait_pred_tick(u64& deadline) {
+    cpu_pause();
+    if (wallclock() > deadline) { err |= E_TIMEOUT; return false; }
+    return true;
+  }
+  ACCL_HD u64 deadline_now() { return wallclock() + timeout_ticks; }
+
+  // ------------- 64-byte record publish / consume (one lane) -------------
+  // Publication rule per MI355X_MICROARCH §Workgroup dispatch: plain field
+  // stores, release fence (system), then the seq word.
+  ACCL_HD void publish_rec(volatile u64* rec8, const u64* val8) {
+    for (int i = 1; i < 8; ++i) rec8[i] = val8[i];
+    fence_release_sys();
+    st_sys(&rec8[0], val8[0]);
+  }
+
+  // ---------------- eager slot helpers ----------------
+  // Channel (s -> r): headers+payload in r's arena lane [s]; credit word in
+  // s's arena (chan_ctl(s, r)->credit), advanced by r.
+  ACCL_HD u64 tx_credit(u32 peer) {  // slots consumed by peer (cumulative)
+    return ld_sys(&tv.chan_ctl(me(), peer)->credit);
+  }
+
+  // ---------------- flow stepping ----------------
+  ACCL_HD bool flow_done(const Flow& f) {
+    return f.kind == FLOW_IDLE || (f.done >= f.count && f.ph == f.pt);
+  }
+
+  ACCL_HD u64 gate_limit(const Flow& f) {
+    if (!f.gate) return f.count;
+    return min64(f.count, ld_sys((const volatile u64*)f.gate));
+  }
+
+  // retire completed segments in order; apply post actions
+  ACCL_HD bool flow_retire(Flow& f) {
+    bool any = false;
+    while (f.ph != f.pt) {
+      PendSeg& p = f.pend[f.ph % FLOW_INFLIGHT];
+      if (!mv->poll(p.token)) break;
+      switch (f.kind) {
+        case FLOW_TX: {
+          // payload complete -> publish the slot header (release chain:
+          // mover released on completion, poll() acquired).
+          SlotHdr* h = tv.slot_hdr(f.gpeer, me(), p.slot);
+          h->tag = f.tag;
+          h->bytes = u32(p.elems * dtype_size(DataType(f.wdt)));
+          h->msg_count = f.msg_count_hdr;
+          h->arith = u32(f.wdt);
+          h->flags = p.first_last;
+          fence_release_sys();
+          st_sys(&h->seq, p.seq);
+          break;
+        }
+        case FLOW_RX: {
+          // payload consumed -> return credit to the sender (cumulative).
+          sq.credit_ret[f.gpeer] = p.seq;
+          st_sys(&tv.chan_ctl(f.gpeer, me())->credit, p.seq);
+          break;
+        }
+        case FLOW_TX_DIRECT: {
+          // advance the cumulative progress word in the peer's arena
+          fence_release_sys();
+          st_sys(tv.direct_word(f.gpeer, me()),
+                 f.prog_base + (f.done + p.elems) * dtype_size(DataType(f.ddt)));
+          break;
+        }
+        default: break;
+      }
+      f.done += p.elems;
+      f.ph++;
+      any = true;
+    }
+    return any;
+  }
+
+  ACCL_HD bool flow_submit(Flow& f) {
+    bool any = false;
+    while (f.submitted < f.count && (f.pt - f.ph) < FLOW_INFLIGHT) {
+      u64 avail = gate_limit(f);
+      if (f.submitted >= avail) break;
+      MoveDesc m{};
+      switch (f.kind) {
+        case FLOW_LOCAL: {
+          const u64 seg_cap = (4u << 20) / dtype_size(DataType(f.ddt));
+          u64 n = min64(avail - f.submitted, seg_cap);
+          m.dst = (u64)(f.dst + f.submitted * dtype_size(DataType(f.ddt)));
+          m.dst_dt = f.ddt;
+          m.src[0] = (u64)(f.src + f.submitted * dtype_size(DataType(f.sdt)));
+          m.src_dt[0] = f.sdt;
+          m.nsrc = 1;
+          if (f.func) {
+            m.src[1] = (u64)(f.red + f.submitted * dtype_size(DataType(f.bdt)));
+            m.src_dt[1] = f.bdt;
+            m.nsrc = 2;
+            m.func = f.func - 1;
+          }
+          m.count = n;
+          u32 tok = mv->submit(m);
+          f.pend[f.pt % FLOW_INFLIGHT] = {tok, 0, n, 0, 0};
+          f.pt++; f.submitted += n; any = true;
+          break;
+        }
+        case FLOW_TX: {
+          u64 next = sq.eager_tx[f.gpeer];          // segments sent so far
+          const u32 n_slots = cfg.n_slots;
+          if (next - tx_credit(f.gpeer) >= n_slots) return any;  // no credit
+          const u64 seg_cap = u64(cfg.slot_bytes) / dtype_size(DataType(f.wdt));
+          u64 n = min64(avail - f.submitted, seg_cap);
+          u32 slot = u32(next % n_slots);
+          m.dst = (u64)tv.slot_payload(f.gpeer, me(), slot);
+          m.dst_dt = f.wdt;
+          m.src[0] = (u64)(f.src + f.submitted * dtype_size(DataType(f.sdt)));
+          m.src_dt[0] = f.sdt;
+          m.nsrc = 1;
+          m.count = n;
+          u32 tok = mv->submit(m);
+          u32 fl = (f.submitted == 0 ? SEG_FIRST : 0) |
+                   (f.submitted + n >= f.count ? SEG_LAST : 0);
+          f.pend[f.pt % FLOW_INFLIGHT] = {tok, slot, n, next + 1, fl};
+          f.pt++; f.submitted += n;
+          sq.eager_tx[f.gpeer] = next + 1;
+          any = true;
+          break;
+        }
+        case FLOW_RX: {
+          u64 next = sq.eager_rx[f.gpeer];          // segments consumed
+          u32 slot = u32(next % cfg.n_slots);
+          SlotHdr* h = tv.slot_hdr(me(), f.gpeer, slot);
+          if (ld_sys(&h->seq) != next + 1) return any;   // not arrived
+          fence_acquire_sys();
+          // Per-segment tag demultiplex: several rx flows may drain one
+          // pair channel concurrently (e.g. allreduce phase 1 + phase 2);
+          // a head segment that is not ours is left for the sibling flow.
+          // (reference analogue: rxbuf_seek matching by (tag, src, seqn),
+          // kernels/cclo/hls/rxbuf_offload/rxbuf_seek.cpp:53-72)
+          u32 want = (f.submitted == 0) ? f.tag : f.matched_tag;
+          if (want != TAG_ANY && h->tag != want) return any;
+          if (f.submitted == 0) f.matched_tag = h->tag;
+          u32 wsz = dtype_size(DataType(f.wdt));
+          if (h->arith != u32(f.wdt)) { err |= E_COMPRESSION; return any; }
+          u64 n = u64(h->bytes) / wsz;
+          if (n > f.count - f.submitted) { err |= E_SEGMENT; return any; }
+          m.dst = (u64)(f.dst + f.submitted * dtype_size(DataType(f.ddt)));
+          m.dst_dt = f.ddt;
+          m.src[0] = (u64)tv.slot_payload(me(), f.gpeer, slot);
+          m.src_dt[0] = f.wdt;
+          m.nsrc = 1;
+          if (f.func) {
+            m.src[1] = (u64)(f.red + f.submitted * dtype_size(DataType(f.bdt)));
+            m.src_dt[1] = f.bdt;
+            m.nsrc = 2;
+            m.func = f.func - 1;
+          }
+          m.count = n;
+          u32 tok = mv->submit(m);
+          f.pend[f.pt % FLOW_INFLIGHT] = {tok, slot, n, next + 1, 0};
+          f.pt++; f.submitted += n;
+          sq.eager_rx[f.gpeer] = next + 1;
+          any = true;
+          break;
+        }
+        case FLOW_TX_DIRECT: {
+          const u64 seg_cap = (8u << 20) / dtype_size(DataType(f.ddt));
+          u64 n = min64(avail - f.submitted, seg_cap);
+          u32 dsz = dtype_size(DataType(f.ddt));
+          m.dst = (u64)(tv.arena[f.gpeer] + f.peer_off + f.submitted * dsz);
+          m.dst_dt = f.ddt;
+          m.src[0] = (u64)(f.src + f.submitted * dtype_size(DataType(f.sdt)));
+          m.src_dt[0] = f.sdt;
+          m.nsrc = 1;
+          m.count = n;
+          u32 tok = mv->submit(m);
+          f.pend[f.pt % FLOW_INFLIGHT] = {tok, 0, n, 0, 0};
+          f.pt++; f.submitted += n; any = true;
+          break;
+        }
+        case FLOW_RX_DIRECT: {
+          // pure wait: done advances with the peer's progress word
+          u64 w = ld_sys(tv.direct_word(me(), f.gpeer));
+          u64 avail_elems = min64(f.count, (w - f.prog_base) / dtype_size(DataType(f.ddt)));
+          if (avail_elems > f.done) {
+            fence_acquire_sys();
+            f.done = avail_elems;
+            f.submitted = avail_elems;
+            any = true;
+          }
+          return any;
+        }
+        default: return any;
+      }
+    }
+    return any;
+  }
+
+  // run a set of flows to completion (the engine inner loop)
+  ACCL_HD u32 run_flows(u32 n) {
+    u64 deadline = deadline_now();
+    for (;;) {
+      bool any = false, alldone = true;
+      for (u32 i = 0; i < n; ++i) {
+        Flow& f = flows[i];
+        if (flow_done(f)) continue;
+        alldone = false;
+        any |= flow_retire(f);
+        if (err) return err;
+        any |= flow_submit(f);
+        if (err) return err;
+      }
+      if (alldone) return E_OK;
+      if (any) deadline = deadline_now();
+      else if (!wait_pred_tick(deadline)) return err;
+    }
+  }
+
+  // ---------------- flow constructors ----------------
+  ACCL_HD Flow& fl(u32 i) { Flow& f = flows[i]; f = Flow{}; return f; }
+
+  ACCL_HD void mk_local(u32 i, const char* src, DataType sdt, char* dst,
+                        DataType ddt, u64 count, const char* red = nullptr,
+                        DataType bdt = DataType::none, int func = -1,
+                        const u64* gate = nullptr) {
+    Flow& f = fl(i);
+    f.kind = FLOW_LOCAL; f.src = src; f.dst = dst; f.red = red;
+    f.sdt = u8(sdt); f.ddt = u8(ddt); f.bdt = u8(bdt);
+    f.func = u8(func + 1); f.count = count; f.gate = gate;
+  }
+  ACCL_HD void mk_tx(u32 i, u32 gpeer, const char* src, DataType sdt,
+                     DataType wdt, u64 count, u32 tag,
+                     const u64* gate = nullptr) {
+    Flow& f = fl(i);
+    f.kind = FLOW_TX; f.gpeer = gpeer; f.src = src;
+    f.sdt = u8(sdt); f.wdt = u8(wdt); f.count = count; f.tag = tag;
+    f.gate = gate; f.msg_count_hdr = count;
+  }
+  ACCL_HD void mk_rx(u32 i, u32 gpeer, char* dst, DataType ddt, DataType wdt,
+                     u64 count, u32 tag, const char* red = nullptr,
+                     DataType bdt = DataType::none, int func = -1,
+                     const u64* gate = nullptr) {
+    Flow& f = fl(i);
+    f.kind = FLOW_RX; f.gpeer = gpeer; f.dst = dst; f.red = red;
+    f.ddt = u8(ddt); f.wdt = u8(wdt); f.bdt = u8(bdt);
+    f.func = u8(func + 1); f.count = count; f.tag = tag; f.gate = gate;
+  }
+  ACCL_HD void mk_tx_direct(u32 i, u32 gpeer, const char* src, DataType sdt,
+                            DataType ddt, u64 count, u64 peer_off,
+                            const u64* gate = nullptr) {
+    Flow& f = fl(i);
+    f.kind = FLOW_TX_DIRECT; f.gpeer = gpeer; f.src = src;
+    f.sdt = u8(sdt); f.ddt = u8(ddt); f.count = count;
+    f.peer_off = peer_off; f.gate = gate;
+    f.prog_base = sq.direct_tx[gpeer];
+    sq.direct_tx[gpeer] += count * dtype_size(ddt);
+  }
+  ACCL_HD void mk_rx_direct(u32 i, u32 gpeer, u64 count, DataType ddt) {
+    Flow& f = fl(i);
+    f.kind = FLOW_RX_DIRECT; f.gpeer = gpeer; f.ddt = u8(ddt); f.count = count;
+    f.prog_base = sq.direct_rx[gpeer];
+    sq.direct_rx[gpeer] += count * dtype_size(ddt);
+  }
+
+  // ---------------- rendezvous record rings ----------------
+  // post {offset,count,tag} into PEER's addr ring (peer = the sender that
+  // will write to us).  reference: rendezvous_send_addr
+  // (ccl_offload_control.c:142-150).
+  ACCL_HD void post_addr(u32 gpeer, u64 offset, u64 count, u32 tag, u32 arith) {
+    u64 seq = ++sq.rndzv_addr_tx[gpeer];
+    RndzvRec* r = tv.rndzv_addr(gpeer, me(), u32((seq - 1) % cfg.n_rndzv));
+    u64 val[8] = {seq, (u64(arith) << 32) | tag, offset, count, 0, 0, 0, 0};
+    publish_rec((volatile u64*)r, val);
+  }
+  // wait for the next addr record from gpeer (the receiver), local poll.
+  // reference: rendezvous_get_addr (ccl_offload_control.c:154-212).
+  ACCL_HD bool wait_addr(u32 gpeer, RndzvRec& out) {
+    u64 seq = sq.rndzv_addr_rx[gpeer] + 1;
+    RndzvRec* r = tv.rndzv_addr(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
+    u64 deadline = deadline_now();
+    while (ld_sys(&r->seq) != seq)
+      if (!wait_pred_tick(deadline)) return false;
+    fence_acquire_sys();
+    out.seq = seq; out.tag = r->tag; out.arith = r->arith;
+    out.offset = r->offset; out.count = r->count;
+    sq.rndzv_addr_rx[gpeer] = seq;
+    return true;
+  }
+  ACCL_HD void post_done(u32 gpeer, u32 tag) {
+    u64 seq = ++sq.rndzv_done_tx[gpeer];
+    RndzvRec* r = tv.rndzv_done(gpeer, me(), u32((seq - 1) % cfg.n_rndzv));
+    u64 val[8] = {seq, tag, 0, 0, 0, 0, 0, 0};
+    publish_rec((volatile u64*)r, val);
+  }
+  ACCL_HD bool wait_done(u32 gpeer) {
+    u64 seq = sq.rndzv_done_rx[gpeer] + 1;
+    RndzvRec* r = tv.rndzv_done(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
+    u64 deadline = deadline_now();
+    while (ld_sys(&r->seq) != seq)
+      if (!wait_pred_tick(deadline)) return false;
+    fence_acquire_sys();
+    sq.rndzv_done_rx[gpeer] = seq;
+    return true;
+  }
+
+  // ================= collectives =================
+  // All take global rank ids resolved through CommView. `d` fields follow
+  // CallDesc. Returns error bits (0 = ok).
+
+  ACCL_HD u32 op_copy(const CallDesc& d) {
+    u64 n = desc_count(d);
+    mk_local(0, local_ptr(d.addr0, d.flags & F_SRC_ARENA), desc_dtype(d),
+             local_ptr(d.addr2, d.flags & F_DST_ARENA), desc_dtype(d), n);
+    return run_flows(1);
+  }
+
+  ACCL_HD u32 op_combine(const CallDesc& d) {
+    // reference: combine (ccl_offload_control.c:551-569)
+    u64 n = desc_count(d);
+    mk_local(0, local_ptr(d.addr0, d.flags & F_SRC_ARENA), desc_dtype(d),
+             local_ptr(d.addr2, d.flags & F_DST_ARENA), desc_dtype(d), n,
+             local_ptr(d.addr1, d.flags & F_OP1_ARENA), desc_dtype(d),
+             int(d.function));
+    return run_flows(1);
+  }
+
+  // protocol selection shared by both sides (reference: eager if bytes <=
+  // max_eager or compressed, ccl_offload_control.c:587-610)
+  ACCL_HD bool use_rndzv(u64 count, DataType dt, DataType wdt) {
+    return dt == wdt && count * dtype_size(dt) > max_eager_bytes;
+  }
+
+  ACCL_HD u32 op_send(const CallDesc& d, const CommView& c) {
+    u64 n = desc_count(d);
+    u32 peer = c.global(d.root_src_dst);
+    DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    if (peer == me()) {  // self-send: defer to matching recv via loopback slot
+      mk_tx(0, peer, src, dt, wdt, n, d.tag);
+      return run_flows(1);
+    }
+    if (use_rndzv(n, dt, wdt)) {
+      // follow the receiver's posted windows
+      u64 sent = 0;
+      while (sent < n) {
+        RndzvRec rec{};
+        if (!wait_addr(peer, rec)) return err;
+        if (rec.tag != d.tag && rec.tag != TAG_ANY) { err |= E_MATCH; return err; }
+        u64 w = min64(n - sent, rec.count);
+        mk_tx_direct(0, peer, src + sent * dtype_size(dt), dt,
+                     DataType(rec.arith ? rec.arith : u32(dt)), w, rec.offset);
+        u32 e = run_flows(1);
+        if (e) return e;
+        sent += w;
+      }
+      post_done(peer, d.tag);
+      return E_OK;
+    }
+    mk_tx(0, peer, src, dt, wdt, n, d.tag);
+    return run_flows(1);
+  }
+
+  ACCL_HD u32 op_recv(const CallDesc& d, const CommView& c) {
+    u64 n = desc_count(d);
+    u32 peer = c.global(d.root_src_dst);
+    DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    if (use_rndzv(n, dt, wdt) && peer != me()) {
+      if (d.flags & F_DST_ARENA) {
+        post_addr(peer, d.addr2, n, d.tag, u32(dt));
+        mk_rx_direct(0, peer, n, dt);
+        u32 e = run_flows(1);
+        if (e) return e;
+        return wait_done(peer) ? E_OK : err;
+      }
+      // stage through the spare region in windows (double-buffered)
+      ArenaHdr* h = tv.hdr(me());
+      u64 half = (h->spare_bytes / 2) / dtype_size(dt);
+      if (!half) { err |= E_INVALID_ARG; return err; }
+      u64 got = 0; int cur = 0; u64 pending = 0; u64 pend_off = 0;
+      u64 posted = 0;
+      // pipeline: post window k+1 while copying window k out
+      while (got < n || pending) {
+        if (posted < n && (posted - got) < 2 * half) {
+          u64 w = min64(n - posted, half);
+          post_addr(peer, h->spare_off + u64(cur) * half * dtype_size(dt), w,
+                    d.tag, u32(dt));
+          posted += w; cur ^= 1;
+        }
+        if (pending) {
+          mk_local(0, tv.arena[me()] + pend_off, dt,
+                   dst + got * dtype_size(dt), dt, pending);
+          u32 e = run_flows(1);
+          if (e) return e;
+          got += pending; pending = 0;
+        }
+        if (got < posted && !pending) {
+          // wait for the oldest posted window to land
+          u64 w = min64(min64(n, got + half) - got, half);
+          mk_rx_direct(0, peer, w, dt);
+          u32 e = run_flows(1);
+          if (e) return e;
+          pending = w;
+          pend_off = h->spare_off + u64((got / half) % 2) * half * dtype_size(dt);
+        }
+      }
+      return wait_done(peer) ? E_OK : err;
+    }
+    mk_rx(0, peer, dst, dt, wdt, n, d.tag);
+    return run_flows(1);
+  }
+
+  // fullmesh bcast: root pushes to every peer; direct when dst offsets can
+  // be exchanged (arena), else eager. reference: broadcast
+  // (ccl_offload_control.c:796-988 — binary/flat tree; xGMI is all-to-all
+  // so a flat push uses P-1 links at once and needs no relay).
+  ACCL_HD u32 op_bcast(const CallDesc& d, const CommView& c) {
+    u64 n = desc_count(d);
+    u32 root = d.root_src_dst;
+    DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    bool rndzv = use_rndzv(n, dt, wdt) && (d.flags & F_DST_ARENA);
+    u32 tag = TAG_COLL | (u32(Op::bcast) << 16) | d.comm_id;
+    if (c.size == 1) return E_OK;
+    if (c.rank == root) {
+      const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+      u32 nf = 0;
+      for (u32 p = 0; p < c.size; ++p) {
+        if (p == root) continue;
+        if (rndzv) {
+          RndzvRec rec{};
+          if (!wait_addr(c.global(p), rec)) return err;
+          mk_tx_direct(nf++, c.global(p), src, dt, dt, n, rec.offset);
+        } else {
+          mk_tx(nf++, c.global(p), src, dt, wdt, n, tag);
+        }
+      }
+      return run_flows(nf);
+    }
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    if (rndzv) {
+      post_addr(c.global(root), d.addr2, n, tag, u32(dt));
+      mk_rx_direct(0, c.global(root), n, dt);
+      return run_flows(1);
+    }
+    mk_rx(0, c.global(root), dst, dt, wdt, n, tag);
+    return run_flows(1);
+  }
+
+  // reference: scatter (ccl_offload_control.c:992-1123)
+  ACCL_HD u32 op_scatter(const CallDesc& d, const CommView& c) {
+    u64 n = desc_count(d);  // per-rank count
+    u32 root = d.root_src_dst;
+    DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    u32 tag = TAG_COLL | (u32(Op::scatter) << 16) | d.comm_id;
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    if (c.rank == root) {
+      const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+      u32 nf = 0;
+      for (u32 p = 0; p < c.size; ++p) {
+        const char* s = src + u64(p) * n * dtype_size(dt);
+        if (p == root) mk_local(nf++, s, dt, dst, dt, n);
+        else mk_tx(nf++, c.global(p), s, dt, wdt, n, tag);
+      }
+      return run_flows(nf);
+    }
+    mk_rx(0, c.global(root), dst, dt, wdt, n, tag);
+    return run_flows(1);
+  }
+
+  // reference: gather (ccl_offload_control.c:1128-1294; ring relay there —
+  // here: direct fan-in, every inbound link concurrently)
+  ACCL_HD u32 op_gather(const CallDesc& d, const CommView& c) {
+    u64 n = desc_count(d);
+    u32 root = d.root_src_dst;
+    DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    u32 tag = TAG_COLL | (u32(Op::gather) << 16) | d.comm_id;
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    if (c.rank == root) {
+      char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+      u32 nf = 0;
+      for (u32 p = 0; p < c.size; ++p) {
+        char* dp = dst + u64(p) * n * dtype_size(dt);
+        if (p == root) mk_local(nf++, src, dt, dp, dt, n);
+        else mk_rx(nf++, c.global(p), dp, dt, wdt, n, tag);
+      }
+      return run_flows(nf);
+    }
+    mk_tx(0, c.global(root), src, dt, wdt, n, tag);
+    return run_flows(1);
+  }
+
+  // fullmesh allgather (reference: ring store-and-forward,
+  // ccl_offload_control.c:1297-1503; fullmesh pushes each chunk once over
+  // its own link instead of P-1 relay hops)
+  ACCL_HD u32 op_allgather(const CallDesc& d, const CommView& c) {
+    u64 n = desc_count(d);
+    DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    u32 tag = TAG_COLL | (u32(Op::allgather) << 16) | d.comm_id;
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    u32 r = c.rank;
+    u32 nf = 0;
+    mk_local(nf++, src, dt, dst + u64(r) * n * dtype_size(dt), dt, n);
+    bool direct = use_rndzv(n, dt, wdt) && (d.flags & F_DST_ARENA) &&
+                  (d.flags & F_SRC_ARENA) && c.size <= 9;
+    if (direct) {
+      for (u32 p = 0; p < c.size; ++p)
+        if (p != r) post_addr(c.global(p), d.addr2 + u64(r) * n * dtype_size(dt),
+                              n, tag, u32(dt));
+      for (u32 p = 0; p < c.size; ++p) {
+        if (p == r) continue;
+        RndzvRec rec{};
+        if (!wait_addr(c.global(p), rec)) return err;
+        mk_tx_direct(nf++, c.global(p), src, dt, dt, n, rec.offset);
+        mk_rx_direct(nf++, c.global(p), n, dt);
+      }
+      return run_flows(nf);
+    }
+    for (u32 p = 0; p < c.size; ++p) {
+      if (p == r) continue;
+      mk_tx(nf++, c.global(p), src, dt, wdt, n, tag);
+      mk_rx(nf++, c.global(p), dst + u64(p) * n * dtype_size(dt), dt, wdt, n, tag);
+    }
+    return run_flows(nf);
+  }
+
+  // reduce at root: fan-in with a serialized reduce chain per segment
+  // (reference: reduce, ccl_offload_control.c:1507-1744)
+  ACCL_HD u32 op_reduce(const CallDesc& d, const CommView& c) {
+    u64 n = desc_count(d);
+    u32 root = d.root_src_dst;
+    DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    u32 tag = TAG_COLL | (u32(Op::reduce) << 16) | d.comm_id;
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    if (c.size == 1) {
+      mk_local(0, src, dt, local_ptr(d.addr2, d.flags & F_DST_ARENA), dt, n);
+      return run_flows(1);
+    }
+    if (c.rank != root) {
+      mk_tx(0, c.global(root), src, dt, wdt, n, tag);
+      return run_flows(1);
+    }
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    u32 nf = 0;
+    mk_local(nf, src, dt, dst, dt, n);
+    const u64* gate = &flows[nf].done;
+    nf++;
+    for (u32 p = 0; p < c.size; ++p) {
+      if (p == root) continue;
+      mk_rx(nf, c.global(p), dst, dt, wdt, n, tag, dst, dt, int(d.function), gate);
+      gate = &flows[nf].done;
+      nf++;
+    }
+    return run_flows(nf);
+  }
+
+  // fullmesh reduce_scatter (reference ring: ccl_offload_control.c:1748-1852)
+  // Every rank owns chunk r: peers push their chunk r, owner chains reduces.
+  ACCL_HD u32 op_reduce_scatter(const CallDesc& d, const CommView& c) {
+    u64 n = desc_count(d);  // per-rank result count
+    DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    u32 tag = TAG_COLL | (u32(Op::reduce_scatter) << 16) | d.comm_id;
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    u32 r = c.rank;
+    if (c.size == 1) {
+      mk_local(0, src, dt, dst, dt, n);
+      return run_flows(1);
+    }
+    u32 nf = 0;
+    // outbound: my chunk p -> rank p, all links at once
+    for (u32 p = 0; p < c.size; ++p)
+      if (p != r)
+        mk_tx(nf++, c.global(p), src + u64(p) * n * dtype_size(dt), dt, wdt, n, tag);
+    // inbound chain into dst
+    mk_local(nf, src + u64(r) * n * dtype_size(dt), dt, dst, dt, n);
+    const u64* gate = &flows[nf].done;
+    nf++;
+    for (u32 p = 0; p < c.size; ++p) {
+      if (p == r) continue;
+      mk_rx(nf, c.global(p), dst, dt, wdt, n, tag, dst, dt, int(d.function), gate);
+      gate = &flows[nf].done;
+      nf++;
+    }
+    return run_flows(nf);
+  }
+
+  // allreduce = fullmesh reduce_scatter + fullmesh allgather, all phases as
+  // one flow set so phase 2 streams out as phase 1 chunks retire.
+  // (reference: segmented ring RS+AG, ccl_offload_control.c:1855-2075.)
+  ACCL_HD u32 op_allreduce(const CallDesc& d, const CommView& c) {
+    u64 total = desc_count(d);
+    DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    u32 tag = TAG_COLL | (u32(Op::allreduce) << 16) | d.comm_id;
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    const u32 P = c.size, r = c.rank;
+    if (P == 1) {
+      mk_local(0, src, dt, dst, dt, total);
+      return run_flows(1);
+    }
+    if (P > 9) return ring_allreduce(d, c);  // flow budget: fullmesh P<=9
+    const u32 dsz = dtype_size(dt);
+    // chunk partition: chunk i = [off(i), off(i+1)), balanced
+    u64 base = total / P, rem = total % P;
+    u64 off[MAX_RANKS + 1];
+    off[0] = 0;
+    for (u32 i = 0; i < P; ++i) off[i + 1] = off[i] + base + (i < rem ? 1 : 0);
+    u64 myn = off[r + 1] - off[r];
+    u32 nf = 0;
+    // phase 1 outbound: src chunk p -> rank p
+    for (u32 p = 0; p < P; ++p)
+      if (p != r)
+        mk_tx(nf++, c.global(p), src + off[p] * dsz, dt, wdt,
+              off[p + 1] - off[p], tag);
+    // phase 1 inbound chain into dst chunk r
+    mk_local(nf, src + off[r] * dsz, dt, dst + off[r] * dsz, dt, myn);
+    const u64* gate = &flows[nf].done;
+    nf++;
+    for (u32 p = 0; p < P; ++p) {
+      if (p == r) continue;
+      mk_rx(nf, c.global(p), dst + off[r] * dsz, dt, wdt, myn, tag,
+            dst + off[r] * dsz, dt, int(d.function), gate);
+      gate = &flows[nf].done;
+      nf++;
+    }
+    const u64* phase1_done = gate;
+    // phase 2: broadcast my reduced chunk, receive everyone else's
+    u32 tag2 = tag + 1;
+    for (u32 p = 0; p < P; ++p) {
+      if (p == r) continue;
+      mk_tx(nf++, c.global(p), dst + off[r] * dsz, dt, wdt, myn, tag2, phase1_done);
+      mk_rx(nf++, c.global(p), dst + off[p] * dsz, dt, wdt,
+            off[p + 1] - off[p], tag2);
+    }
+    return run_flows(nf);
+  }
+
+  // ring allreduce for P > fullmesh budget (reference schedule shape:
+  // ccl_offload_control.c:1888-2071). Works on dst as the accumulator.
+  ACCL_HD u32 ring_allreduce(const CallDesc& d, const CommView& c) {
+    u64 total = desc_count(d);
+    DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    u32 tag = TAG_COLL | (u32(Op::allreduce) << 16) | 0x8000 | d.comm_id;
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    const u32 P = c.size, r = c.rank;
+    const u32 dsz = dtype_size(dt);
+    u64 base = total / P, rem = total % P;
+    u64 off[MAX_RANKS + 1];
+    off[0] = 0;
+    for (u32 i = 0; i < P; ++i) off[i + 1] = off[i] + base + (i < rem ? 1 : 0);
+    u32 next = c.global((r + 1) % P), prev = c.global((r + P - 1) % P);
+    // local init: dst = src
+    mk_local(0, src, dt, dst, dt, total);
+    u32 e = run_flows(1);
+    if (e) return e;
+    // P-1 reduce-scatter steps: send dst chunk (r-s), recv+reduce chunk (r-s-1)
+    for (u32 s = 0; s < P - 1; ++s) {
+      u32 ct = (r + P - s) % P;        // chunk to send
+      u32 cr = (r + P - s - 1) % P;    // chunk to receive+reduce
+      mk_tx(0, next, dst + off[ct] * dsz, dt, wdt, off[ct + 1] - off[ct], tag);
+      mk_rx(1, prev, dst + off[cr] * dsz, dt, wdt, off[cr + 1] - off[cr], tag,
+            dst + off[cr] * dsz, dt, int(d.function));
+      if ((e = run_flows(2))) return e;
+    }
+    // P-1 allgather steps
+    for (u32 s = 0; s < P - 1; ++s) {
+      u32 ct = (r + 1 + P - s) % P;
+      u32 cr = (r + P - s) % P;
+      mk_tx(0, next, dst + off[ct] * dsz, dt, wdt, off[ct + 1] - off[ct], tag);
+      mk_rx(1, prev, dst + off[cr] * dsz, dt, wdt, off[cr + 1] - off[cr], tag);
+      if ((e = run_flows(2))) return e;
+    }
+    return E_OK;
+  }
+
+  // alltoall: P-1 pairwise exchanges + local copy (reference: fused flat
+  // broadcasts, ccl_offload_control.c:2123-2218)
+  ACCL_HD u32 op_alltoall(const CallDesc& d, const CommView& c) {
+    u64 n = desc_count(d);  // per-pair count
+    DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    u32 tag = TAG_COLL | (u32(Op::alltoall) << 16) | d.comm_id;
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    u32 r = c.rank;
+    u32 nf = 0;
+    mk_local(nf++, src + u64(r) * n * dtype_size(dt), dt,
+             dst + u64(r) * n * dtype_size(dt), dt, n);
+    for (u32 p = 0; p < c.size; ++p) {
+      if (p == r) continue;
+      mk_tx(nf++, c.global(p), src + u64(p) * n * dtype_size(dt), dt, wdt, n, tag);
+      mk_rx(nf++, c.global(p), dst + u64(p) * n * dtype_size(dt), dt, wdt, n, tag);
+    }
+    return run_flows(nf);
+  }
+
+  // dissemination-free fullmesh barrier: bump my token in every peer's
+  // arena, wait for all peers' tokens locally.
+  // (reference: gather+scatter notification barrier,
+  // ccl_offload_control.c:2078-2120)
+  ACCL_HD u32 op_barrier(const CallDesc&, const CommView& c) {
+    // epochs are per PAIR: two ranks' common-barrier counts always agree,
+    // while a global per-rank epoch would not across subgroup communicators.
+    fence_release_sys();
+    for (u32 p = 0; p < c.size; ++p) {
+      if (p == c.rank) continue;
+      u32 g = c.global(p);
+      st_sys(tv.barrier_word(g, me()), ++sq.barrier_epoch[g]);
+    }
+    u64 deadline = deadline_now();
+    for (u32 p = 0; p < c.size; ++p) {
+      if (p == c.rank) continue;
+      u32 g = c.global(p);
+      volatile u64* w = tv.barrier_word(me(), g);
+      while (ld_sys(w) < sq.barrier_epoch[g])
+        if (!wait_pred_tick(deadline)) return err;
+    }
+    fence_acquire_sys();
+    return E_OK;
+  }
+
+  // ---------------- dispatch ----------------
+  // reference: run() scenario switch (ccl_offload_control.c:2375-2459)
+  ACCL_HD u32 run_call(const CallDesc& d) {
+    err = 0;
+    if (d.comm_id >= ncomms && Op(d.scenario) != Op::copy &&
+        Op(d.scenario) != Op::combine && Op(d.scenario) != Op::config &&
+        Op(d.scenario) != Op::nop)
+      return E_COMM;
+    const CommView& c = comms[d.comm_id < ncomms ? d.comm_id : 0];
+    switch (Op(d.scenario)) {
+      case Op::nop: return E_OK;
+      case Op::copy: return op_copy(d);
+      case Op::combine: return op_combine(d);
+      case Op::send: return op_send(d, c);
+      case Op::recv: return op_recv(d, c);
+      case Op::bcast: return op_bcast(d, c);
+      case Op::scatter: return op_scatter(d, c);
+      case Op::gather: return op_gather(d, c);
+      case Op::allgather: return op_allgather(d, c);
+      case Op::reduce: return op_reduce(d, c);
+      case Op::allreduce: return op_allreduce(d, c);
+      case Op::reduce_scatter: return op_reduce_scatter(d, c);
+      case Op::alltoall: return op_alltoall(d, c);
+      case Op::barrier: return op_barrier(d, c);
+      case Op::config: return run_config(d);
+      default: return E_INVALID_OP;
+    }
+  }
+
+  ACCL_HD u32 run_config(const CallDesc& d) {
+    switch (CfgFunc(d.function)) {
+      case CfgFunc::set_timeout:
+        timeout_ticks = desc_count(d) * 1000 * TICKS_PER_US;  // ms -> ticks
+        return E_OK;
+      case CfgFunc::set_max_eager_size:
+        max_eager_bytes = desc_count(d);
+        return E_OK;
+      case CfgFunc::reset:
+      case CfgFunc::enable_pkt:
+        return E_OK;
+      default: return E_OK;
+    }
+  }
+};
+
+}  // namespace accl

@@ -1,0 +1,314 @@
+#include "accl.hpp"
+#include <algorithm>
+#include <cstdlib>
+#include <cstring>
+
+namespace accl {
+
+// ------------------------------------------------------------- BaseBuffer
+BaseBuffer::BaseBuffer(ACCL* owner, u64 arena_off, u64 count, DataType dt,
+                       bool own_host, void* host_ptr, u64 root_off)
+    : owner_(owner), off_(arena_off), count_(count), dt_(dt),
+      own_host_(own_host), host_(host_ptr), root_off_(root_off),
+      own_arena_(root_off != 0) {}
+
+BaseBuffer::~BaseBuffer() {
+  if (own_arena_ && owner_) owner_->backend()->free_block(root_off_);
+  if (own_host_ && host_) std::free(host_);
+}
+
+void BaseBuffer::sync_to_device() {
+  if (host_) owner_->backend()->write_arena(off_, host_, bytes());
+}
+void BaseBuffer::sync_from_device() {
+  if (host_) owner_->backend()->read_arena(off_, host_, bytes());
+}
+void* BaseBuffer::device_ptr() const {
+  return owner_->backend()->arena_local() + off_;
+}
+std::unique_ptr<BaseBuffer> BaseBuffer::slice(u64 start, u64 end) {
+  if (end > count_ || start > end) throw accl_error("buffer slice out of range");
+  u64 esz = dtype_size(dt_);
+  void* h = host_ ? (char*)host_ + start * esz : nullptr;
+  return std::unique_ptr<BaseBuffer>(new BaseBuffer(
+      owner_, off_ + start * esz, end - start, dt_, false, h, 0));
+}
+
+// ---------------------------------------------------------------- Request
+u32 Request::wait(u64 timeout_ms) {
+  if (!done_) {
+    be_->wait(seq_, &ret_, timeout_ms);
+    done_ = true;
+  }
+  return ret_.errcode;
+}
+bool Request::test() {
+  if (done_) return true;
+  done_ = be_->test(seq_, &ret_);
+  return done_;
+}
+u32 Request::retcode() { return done_ ? ret_.errcode : 0; }
+double Request::duration_us() {
+  wait();
+  return double(ret_.t_end - ret_.t_start) / TICKS_PER_US;
+}
+
+// ------------------------------------------------------------------- ACCL
+ACCL::ACCL(std::unique_ptr<Backend> backend) : be_(std::move(backend)) {
+  comm_sizes_.push_back(be_->cfg().nranks);
+  comm_ranks_.push_back(be_->cfg().rank);
+}
+
+ACCL::~ACCL() {
+  for (Request* r : reqs_) delete r;
+  deinit();
+}
+
+void ACCL::deinit() {
+  if (be_) be_->shutdown();
+}
+
+std::unique_ptr<BaseBuffer> ACCL::create_buffer(u64 count, DataType dt) {
+  u64 bytes = count * dtype_size(dt);
+  u64 off = be_->alloc(bytes);
+  void* host = std::malloc(bytes ? bytes : 1);
+  return std::unique_ptr<BaseBuffer>(
+      new BaseBuffer(this, off, count, dt, true, host, off));
+}
+std::unique_ptr<BaseBuffer> ACCL::create_buffer(void* host, u64 count, DataType dt) {
+  u64 off = be_->alloc(count * dtype_size(dt));
+  return std::unique_ptr<BaseBuffer>(
+      new BaseBuffer(this, off, count, dt, false, host, off));
+}
+std::unique_ptr<BaseBuffer> ACCL::create_buffer_device(u64 count, DataType dt) {
+  u64 off = be_->alloc(count * dtype_size(dt));
+  return std::unique_ptr<BaseBuffer>(
+      new BaseBuffer(this, off, count, dt, false, nullptr, off));
+}
+
+u32 ACCL::create_communicator(const std::vector<u32>& global_ranks, u32 my_local) {
+  u32 id = be_->add_comm(global_ranks, my_local);
+  comm_sizes_.resize(id + 1);
+  comm_ranks_.resize(id + 1);
+  comm_sizes_[id] = u32(global_ranks.size());
+  comm_ranks_[id] = my_local;
+  return id;
+}
+u32 ACCL::split_communicator(const std::vector<u32>& global_ranks) {
+  auto it = std::find(global_ranks.begin(), global_ranks.end(), rank());
+  if (it == global_ranks.end())
+    throw accl_error("split_communicator: caller not in group");
+  return create_communicator(global_ranks, u32(it - global_ranks.begin()));
+}
+
+CallDesc ACCL::make_desc(Op op, u64 count, DataType dt, DataType wire) {
+  CallDesc d{};
+  d.scenario = u32(op);
+  d.count_lo = u32(count & 0xFFFFFFFFu);
+  d.count_hi = u32(count >> 32);
+  if (wire == DataType::none) wire = dt;
+  d.arith = u32(dt) | (u32(wire) << 8);
+  return d;
+}
+
+Request* ACCL::finish(CallDesc d, bool run_async, BaseBuffer* sync_out,
+                      u64 out_count, BaseBuffer* sync_in0, u64 in0_count,
+                      BaseBuffer* sync_in1, u64 in1_count) {
+  if (sync_in0 && sync_in0->host_ptr())
+    be_->write_arena(sync_in0->arena_offset(), sync_in0->host_ptr(),
+                     in0_count * dtype_size(sync_in0->dtype()));
+  if (sync_in1 && sync_in1->host_ptr())
+    be_->write_arena(sync_in1->arena_offset(), sync_in1->host_ptr(),
+                     in1_count * dtype_size(sync_in1->dtype()));
+  u64 seq = be_->submit(d);
+  Request* r = new Request(be_.get(), seq);
+  reqs_.push_back(r);
+  if (!run_async) {
+    u32 e = r->wait();
+    if (e) throw accl_error("accl op failed: " + error_to_string(e), e);
+    if (sync_out && sync_out->host_ptr())
+      be_->read_arena(sync_out->arena_offset(), sync_out->host_ptr(),
+                      out_count * dtype_size(sync_out->dtype()));
+  }
+  return r;
+}
+
+void ACCL::free_request(Request* r) {
+  auto it = std::find(reqs_.begin(), reqs_.end(), r);
+  if (it != reqs_.end()) {
+    reqs_.erase(it);
+    delete r;
+  }
+}
+
+// --------------------------------------------------------------- op calls
+Request* ACCL::copy(BaseBuffer& src, BaseBuffer& dst, u64 count,
+                    bool from_device, bool to_device, bool run_async) {
+  CallDesc d = make_desc(Op::copy, count, src.dtype(), src.dtype());
+  d.addr0 = src.arena_offset();
+  d.addr2 = dst.arena_offset();
+  d.flags = F_SRC_ARENA | F_DST_ARENA;
+  return finish(d, run_async, to_device ? nullptr : &dst, count,
+                from_device ? nullptr : &src, count);
+}
+
+Request* ACCL::combine(u64 count, ReduceFunction f, BaseBuffer& op0,
+                       BaseBuffer& op1, BaseBuffer& res, bool from_device,
+                       bool to_device, bool run_async) {
+  CallDesc d = make_desc(Op::combine, count, op0.dtype(), op0.dtype());
+  d.addr0 = op0.arena_offset();
+  d.addr1 = op1.arena_offset();
+  d.addr2 = res.arena_offset();
+  d.function = u32(f);
+  d.flags = F_SRC_ARENA | F_DST_ARENA | F_OP1_ARENA;
+  return finish(d, run_async, to_device ? nullptr : &res, count,
+                from_device ? nullptr : &op0, count,
+                from_device ? nullptr : &op1, count);
+}
+
+Request* ACCL::send(BaseBuffer& src, u64 count, u32 dst, u32 tag, u32 comm,
+                    bool from_device, DataType compress, bool run_async) {
+  if (tag != TAG_ANY && tag > MAX_USER_TAG)
+    throw accl_error("send: tag out of range");
+  CallDesc d = make_desc(Op::send, count, src.dtype(), compress);
+  d.addr0 = src.arena_offset();
+  d.root_src_dst = dst;
+  d.tag = tag;
+  d.comm_id = comm;
+  d.flags = F_SRC_ARENA;
+  return finish(d, run_async, nullptr, 0, from_device ? nullptr : &src, count);
+}
+
+Request* ACCL::recv(BaseBuffer& dst, u64 count, u32 src, u32 tag, u32 comm,
+                    bool to_device, DataType compress, bool run_async) {
+  CallDesc d = make_desc(Op::recv, count, dst.dtype(), compress);
+  d.addr2 = dst.arena_offset();
+  d.root_src_dst = src;
+  d.tag = tag;
+  d.comm_id = comm;
+  d.flags = F_DST_ARENA;
+  return finish(d, run_async, to_device ? nullptr : &dst, count);
+}
+
+Request* ACCL::bcast(BaseBuffer& buf, u64 count, u32 root, u32 comm,
+                     bool from_device, bool to_device, DataType compress,
+                     bool run_async) {
+  CallDesc d = make_desc(Op::bcast, count, buf.dtype(), compress);
+  d.addr0 = buf.arena_offset();
+  d.addr2 = buf.arena_offset();
+  d.root_src_dst = root;
+  d.comm_id = comm;
+  d.flags = F_SRC_ARENA | F_DST_ARENA;
+  bool is_root = comm_rank(comm) == root;
+  return finish(d, run_async, (!to_device && !is_root) ? &buf : nullptr, count,
+                (!from_device && is_root) ? &buf : nullptr, count);
+}
+
+Request* ACCL::scatter(BaseBuffer& src, BaseBuffer& dst, u64 count, u32 root,
+                       u32 comm, bool from_device, bool to_device,
+                       DataType compress, bool run_async) {
+  CallDesc d = make_desc(Op::scatter, count, src.dtype(), compress);
+  d.addr0 = src.arena_offset();
+  d.addr2 = dst.arena_offset();
+  d.root_src_dst = root;
+  d.comm_id = comm;
+  d.flags = F_SRC_ARENA | F_DST_ARENA;
+  bool is_root = comm_rank(comm) == root;
+  return finish(d, run_async, to_device ? nullptr : &dst, count,
+                (!from_device && is_root) ? &src : nullptr,
+                count * comm_size(comm));
+}
+
+Request* ACCL::gather(BaseBuffer& src, BaseBuffer& dst, u64 count, u32 root,
+                      u32 comm, bool from_device, bool to_device,
+                      DataType compress, bool run_async) {
+  CallDesc d = make_desc(Op::gather, count, src.dtype(), compress);
+  d.addr0 = src.arena_offset();
+  d.addr2 = dst.arena_offset();
+  d.root_src_dst = root;
+  d.comm_id = comm;
+  d.flags = F_SRC_ARENA | F_DST_ARENA;
+  bool is_root = comm_rank(comm) == root;
+  return finish(d, run_async,
+                (!to_device && is_root) ? &dst : nullptr,
+                count * comm_size(comm), from_device ? nullptr : &src, count);
+}
+
+Request* ACCL::allgather(BaseBuffer& src, BaseBuffer& dst, u64 count, u32 comm,
+                         bool from_device, bool to_device, DataType compress,
+                         bool run_async) {
+  CallDesc d = make_desc(Op::allgather, count, src.dtype(), compress);
+  d.addr0 = src.arena_offset();
+  d.addr2 = dst.arena_offset();
+  d.comm_id = comm;
+  d.flags = F_SRC_ARENA | F_DST_ARENA;
+  return finish(d, run_async, to_device ? nullptr : &dst,
+                count * comm_size(comm), from_device ? nullptr : &src, count);
+}
+
+Request* ACCL::reduce(BaseBuffer& src, BaseBuffer& dst, u64 count, u32 root,
+                      ReduceFunction f, u32 comm, bool from_device,
+                      bool to_device, DataType compress, bool run_async) {
+  CallDesc d = make_desc(Op::reduce, count, src.dtype(), compress);
+  d.addr0 = src.arena_offset();
+  d.addr2 = dst.arena_offset();
+  d.root_src_dst = root;
+  d.function = u32(f);
+  d.comm_id = comm;
+  d.flags = F_SRC_ARENA | F_DST_ARENA;
+  bool is_root = comm_rank(comm) == root;
+  return finish(d, run_async, (!to_device && is_root) ? &dst : nullptr, count,
+                from_device ? nullptr : &src, count);
+}
+
+Request* ACCL::allreduce(BaseBuffer& src, BaseBuffer& dst, u64 count,
+                         ReduceFunction f, u32 comm, bool from_device,
+                         bool to_device, DataType compress, bool run_async) {
+  CallDesc d = make_desc(Op::allreduce, count, src.dtype(), compress);
+  d.addr0 = src.arena_offset();
+  d.addr2 = dst.arena_offset();
+  d.function = u32(f);
+  d.comm_id = comm;
+  d.flags = F_SRC_ARENA | F_DST_ARENA;
+  return finish(d, run_async, to_device ? nullptr : &dst, count,
+                from_device ? nullptr : &src, count);
+}
+
+Request* ACCL::reduce_scatter(BaseBuffer& src, BaseBuffer& dst, u64 count,
+                              ReduceFunction f, u32 comm, bool from_device,
+                              bool to_device, DataType compress,
+                              bool run_async) {
+  CallDesc d = make_desc(Op::reduce_scatter, count, src.dtype(), compress);
+  d.addr0 = src.arena_offset();
+  d.addr2 = dst.arena_offset();
+  d.function = u32(f);
+  d.comm_id = comm;
+  d.flags = F_SRC_ARENA | F_DST_ARENA;
+  return finish(d, run_async, to_device ? nullptr : &dst, count,
+                from_device ? nullptr : &src, count * comm_size(comm));
+}
+
+Request* ACCL::alltoall(BaseBuffer& src, BaseBuffer& dst, u64 count, u32 comm,
+                        bool from_device, bool to_device, bool run_async) {
+  CallDesc d = make_desc(Op::alltoall, count, src.dtype(), src.dtype());
+  d.addr0 = src.arena_offset();
+  d.addr2 = dst.arena_offset();
+  d.comm_id = comm;
+  d.flags = F_SRC_ARENA | F_DST_ARENA;
+  return finish(d, run_async, to_device ? nullptr : &dst,
+                count * comm_size(comm), from_device ? nullptr : &src,
+                count * comm_size(comm));
+}
+
+Request* ACCL::barrier(u32 comm, bool run_async) {
+  CallDesc d = make_desc(Op::barrier, 0, DataType::float32, DataType::float32);
+  d.comm_id = comm;
+  return finish(d, run_async, nullptr, 0);
+}
+
+Request* ACCL::nop(bool run_async) {
+  CallDesc d = make_desc(Op::nop, 0, DataType::float32, DataType::float32);
+  return finish(d, run_async, nullptr, 0);
+}
+
+}  // namespace accl

@@ -1,0 +1,129 @@
+// Backend = one rank's device engine as seen by the host runtime: a call
+// descriptor ring (the hostctrl/CMD_CALL analogue, reference:
+// kernels/plugins/hostctrl/hostctrl.cpp:22-63 and driver/xrt/src/
+// xrtdevice.cpp:36-192), a return ring (RETVAL readback,
+// ccl_offload_control.c:2291-2306), and the arena (exchange memory + rx
+// buffers + heap). Concrete backends: EmuDevice (CPU engine thread over shm)
+// and GpuDevice (persistent HIP kernel over HBM).
+#pragma once
+#include <cstring>
+#include <memory>
+#include <string>
+#include <vector>
+#include "../common/proto.hpp"
+#include "alloc.hpp"
+#include "util.hpp"
+
+namespace accl {
+
+// Control words shared host<->engine (pinned host memory on GPU).
+struct alignas(64) CtrlPage {
+  volatile u64 doorbell;   // count of descriptors published by host
+  volatile u64 shutdown;   // host sets 1; engine exits its loop
+  volatile u64 comm_gen;   // bump after rewriting the comm mirror
+  volatile u64 ncomms;
+  volatile u64 heartbeat;  // engine bumps when idle (liveness)
+  volatile u64 engine_up;  // engine sets 1 once running
+  u64 _pad[2];
+};
+
+constexpr u32 RING_CAP = 64;
+
+struct RingPage {
+  CtrlPage ctrl;
+  CommView comm_mirror[MAX_COMMS];
+  CallDesc descs[RING_CAP];
+  RetEntry rets[RING_CAP];
+};
+
+class Backend {
+ public:
+  virtual ~Backend() = default;
+
+  // --- bring-up (two-phase; the caller runs the bootstrap allgather) ---
+  virtual std::vector<char> local_blob() = 0;
+  virtual void connect(const std::vector<std::vector<char>>& blobs) = 0;
+  virtual void shutdown() = 0;
+  virtual bool is_gpu() const = 0;
+
+  const ProtoConfig& cfg() const { return cfg_; }
+  char* arena_local() const { return arena_base_; }
+
+  // --- calls ---
+  u64 submit(CallDesc d) {
+    // in-order ring; throttle on ring occupancy
+    u64 seq = head_;
+    while (seq - tail_retired() >= RING_CAP) cpu_pause();
+    d.seq = u32(seq);
+    ring_->descs[seq % RING_CAP] = d;
+    head_ = seq + 1;
+    __atomic_store_n((u64*)&ring_->ctrl.doorbell, head_, __ATOMIC_RELEASE);
+    return seq;
+  }
+  bool test(u64 seq, RetEntry* out) {
+    RetEntry& r = ring_->rets[seq % RING_CAP];
+    if (__atomic_load_n(&r.seq, __ATOMIC_ACQUIRE) != u32(seq + 1)) return false;
+    if (out) *out = r;
+    return true;
+  }
+  // returns error bits; throws on host-side timeout
+  u32 wait(u64 seq, RetEntry* out = nullptr, u64 timeout_ms = 120000) {
+    RetEntry r{};
+    u64 t0 = wallclock_host_ns();
+    while (!test(seq, &r)) {
+      cpu_pause();
+      if (wallclock_host_ns() - t0 > timeout_ms * 1000000ull)
+        throw accl_error("accl: host wait timeout on call seq " +
+                         std::to_string(seq));
+    }
+    if (out) *out = r;
+    return r.errcode;
+  }
+  u64 call(const CallDesc& d, RetEntry* out = nullptr) {
+    u64 s = submit(d);
+    u32 e = wait(s, out);
+    if (e) throw accl_error("accl call failed: " + error_to_string(e), e);
+    return s;
+  }
+
+  // --- arena memory ---
+  u64 alloc(u64 bytes) { return heap_.alloc(bytes); }
+  void free_block(u64 off) { heap_.free_block(off); }
+  virtual void write_arena(u64 off, const void* src, u64 bytes) = 0;
+  virtual void read_arena(u64 off, void* dst, u64 bytes) = 0;
+
+  // --- communicators (quiesce, rewrite mirror, bump generation) ---
+  // reference: Communicator rank-table write, driver/xrt/src/
+  // communicator.cpp:25-52; subgroup creation accl.cpp:955-962.
+  u32 add_comm(const std::vector<u32>& members, u32 my_local) {
+    quiesce();
+    u32 id = u32(__atomic_load_n((u64*)&ring_->ctrl.ncomms, __ATOMIC_RELAXED));
+    if (id >= MAX_COMMS) throw accl_error("accl: too many communicators");
+    CommView& c = ring_->comm_mirror[id];
+    c.id = id;
+    c.rank = my_local;
+    c.size = u32(members.size());
+    for (u32 i = 0; i < members.size(); ++i) c.members[i] = members[i];
+    __atomic_store_n((u64*)&ring_->ctrl.ncomms, u64(id + 1), __ATOMIC_RELEASE);
+    __atomic_fetch_add((u64*)&ring_->ctrl.comm_gen, 1, __ATOMIC_RELEASE);
+    return id;
+  }
+  void quiesce() {
+    if (head_) wait(head_ - 1);
+  }
+
+ protected:
+  u64 tail_retired() {
+    // ring slots free once their RetEntry is published
+    while (retired_ < head_ && test(retired_, nullptr)) retired_++;
+    return retired_;
+  }
+
+  ProtoConfig cfg_{};
+  char* arena_base_ = nullptr;
+  RingPage* ring_ = nullptr;   // host-visible (pinned on GPU)
+  HeapAlloc heap_;
+  u64 head_ = 0, retired_ = 0;
+};
+
+}  // namespace accl

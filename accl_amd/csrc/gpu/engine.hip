@@ -1,0 +1,318 @@
+// The persistent CCLO engine kernel for MI355X (gfx950).
+//
+// One cooperative-by-construction grid per GPU (grid <= 1 workgroup per CU,
+// so residency is guaranteed by size — cdna_hip_programming.md §1): lane 0 of
+// workgroup 0 runs the collective scheduler (common/sched.hpp, the microcode
+// analogue of reference ccl_offload_control.c); every other wave is a mover
+// executing MoveDescs — vectorized copy / cast / n-ary reduce between local
+// HBM and peer HBM over xGMI (the data plane: reference dma_mover +
+// reduce_ops + hp_compression, kernels/cclo/hls, kernels/plugins).
+//
+// Memory-ordering discipline (MI355X_MICROARCH.md §Workgroup dispatch):
+//  * movers: payload stores -> s_waitcnt vmcnt(0) (asm) -> system release
+//    fence -> tiles_done release-add (agent) — the release chain the
+//    scheduler extends to peers when it publishes slot headers.
+//  * all polls are relaxed + s_sleep; one acquire fence after a match.
+#include <hip/hip_runtime.h>
+#include <hip/hip_fp16.h>
+#include "engine.hpp"
+
+namespace accl {
+
+#define AGENT __HIP_MEMORY_SCOPE_AGENT
+
+// ------------------------------------------------------------- GpuMover
+__device__ u32 GpuMover::submit(const MoveDesc& m) {
+  u64 h = head_cache;
+  u32 slot = u32(h % MOVE_RING);
+  if (h >= MOVE_RING) {
+    // wait for the slot's previous occupant to finish (movers never block,
+    // so this always terminates)
+    MoveState& prev = st[slot];
+    while (__hip_atomic_load(&prev.tiles_done, __ATOMIC_RELAXED, AGENT) <
+           prev.tiles_total)
+      __builtin_amdgcn_s_sleep(1);
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  }
+  MoveDesc& d = ring[slot];
+  d = m;
+  d.epoch = h + 1;
+  MoveState& s = st[slot];
+  s.tiles_total = move_tiles(m);
+  __hip_atomic_store(&s.tiles_claimed, 0u, __ATOMIC_RELAXED, AGENT);
+  __hip_atomic_store(&s.tiles_done, 0u, __ATOMIC_RELAXED, AGENT);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __hip_atomic_store(head, h + 1, __ATOMIC_RELAXED, AGENT);
+  head_cache = h + 1;
+  return u32(h);
+}
+
+__device__ bool GpuMover::poll(u32 token) {
+  u32 slot = token % MOVE_RING;
+  // slot recycled past this token => long complete
+  if (u32(ring[slot].epoch) != token + 1) return true;
+  MoveState& s = st[slot];
+  u32 done = __hip_atomic_load(&s.tiles_done, __ATOMIC_RELAXED, AGENT);
+  if (done < s.tiles_total) return false;
+  __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  return true;
+}
+
+// ------------------------------------------------------------ mover tiles
+template <typename T>
+struct SumOp { __device__ static T apply(T a, T b) { return a + b; } };
+template <typename T>
+struct MaxOp { __device__ static T apply(T a, T b) { return a > b ? a : b; } };
+
+__device__ __forceinline__ float ld_f32(const void* p, u64 i, DataType dt) {
+  switch (dt) {
+    case DataType::float32: return ((const float*)p)[i];
+    case DataType::float16: return __half2float(((const __half*)p)[i]);
+    case DataType::bfloat16: {
+      u16 h = ((const u16*)p)[i];
+      u32 b = u32(h) << 16;
+      return __uint_as_float(b);
+    }
+    default: return 0.f;
+  }
+}
+__device__ __forceinline__ void st_f32(void* p, u64 i, DataType dt, float v) {
+  switch (dt) {
+    case DataType::float32: ((float*)p)[i] = v; break;
+    case DataType::float16: ((__half*)p)[i] = __float2half(v); break;
+    case DataType::bfloat16: {
+      u32 x = __float_as_uint(v);
+      if ((x & 0x7F800000u) == 0x7F800000u && (x & 0x7FFFFFu)) {
+        ((u16*)p)[i] = u16((x >> 16) | 0x40);
+      } else {
+        x += 0x7FFFu + ((x >> 16) & 1);
+        ((u16*)p)[i] = u16(x >> 16);
+      }
+      break;
+    }
+    default: break;
+  }
+}
+
+__device__ __forceinline__ bool aligned16(const void* p) {
+  return (u64(p) & 15) == 0;
+}
+
+// pure same-dtype copy, 16B vectorized when aligned
+__device__ void tile_copy(const MoveDesc& m, u64 lo, u64 hi, int lane) {
+  u32 esz = dtype_size(DataType(m.dst_dt));
+  u64 bytes = (hi - lo) * esz;
+  const char* s = (const char*)m.src[0] + lo * esz;
+  char* d = (char*)m.dst + lo * esz;
+  if (aligned16(s) && aligned16(d) && (bytes & 15) == 0) {
+    const uint4* s4 = (const uint4*)s;
+    uint4* d4 = (uint4*)d;
+    u64 n = bytes / 16;
+    for (u64 i = lane; i < n; i += 64) d4[i] = s4[i];
+  } else if ((u64(s) & 3) == 0 && (u64(d) & 3) == 0 && (bytes & 3) == 0) {
+    const u32* s1 = (const u32*)s;
+    u32* d1 = (u32*)d;
+    u64 n = bytes / 4;
+    for (u64 i = lane; i < n; i += 64) d1[i] = s1[i];
+  } else {
+    for (u64 i = lane; i < bytes; i += 64) d[i] = s[i];
+  }
+}
+
+// f32 n-ary reduce, 16B vectorized (the reduce_ops hot path:
+// reference kernels/plugins/reduce_ops/reduce_ops.cpp:83-106)
+template <template <class> class OP>
+__device__ void tile_reduce_f32(const MoveDesc& m, u64 lo, u64 hi, int lane) {
+  float* d = (float*)m.dst + lo;
+  const float* s0 = (const float*)m.src[0] + lo;
+  const float* s1 = (const float*)m.src[1] + lo;
+  u64 n = hi - lo;
+  bool v16 = aligned16(d) && aligned16(s0) && aligned16(s1) && (n & 3) == 0 &&
+             m.nsrc == 2;
+  if (v16) {
+    const float4* a = (const float4*)s0;
+    const float4* b = (const float4*)s1;
+    float4* o = (float4*)d;
+    u64 n4 = n / 4;
+    for (u64 i = lane; i < n4; i += 64) {
+      float4 x = a[i], y = b[i];
+      o[i] = make_float4(OP<float>::apply(x.x, y.x), OP<float>::apply(x.y, y.y),
+                         OP<float>::apply(x.z, y.z), OP<float>::apply(x.w, y.w));
+    }
+    return;
+  }
+  for (u64 i = lane; i < n; i += 64) {
+    float acc = ((const float*)m.src[0])[lo + i];
+    for (u32 k = 1; k < m.nsrc; ++k)
+      acc = OP<float>::apply(acc, ((const float*)m.src[k])[lo + i]);
+    d[i] = acc;
+  }
+}
+
+// float-domain path for any f32/f16/bf16 mix (cast + reduce fused — the
+// hp_compression + reduce_ops lanes in one pass)
+__device__ void tile_float_generic(const MoveDesc& m, u64 lo, u64 hi, int lane) {
+  const bool is_max = m.nsrc >= 2 && ReduceFunction(m.func) == ReduceFunction::MAX;
+  for (u64 i = lo + lane; i < hi; i += 64) {
+    float acc = ld_f32((const void*)m.src[0], i, DataType(m.src_dt[0]));
+    for (u32 k = 1; k < m.nsrc; ++k) {
+      float v = ld_f32((const void*)m.src[k], i, DataType(m.src_dt[k]));
+      acc = is_max ? (acc > v ? acc : v) : acc + v;
+    }
+    st_f32((void*)m.dst, i, DataType(m.dst_dt), acc);
+  }
+}
+
+__device__ bool dtype_is_floatish(DataType d) {
+  return d == DataType::float32 || d == DataType::float16 ||
+         d == DataType::bfloat16;
+}
+
+__device__ void run_tile(const MoveDesc& m, u32 t, int lane) {
+  u64 te = move_tile_elems(m);
+  u64 lo = u64(t) * te;
+  u64 hi = lo + te;
+  if (hi > m.count) hi = m.count;
+  if (lo >= hi) return;
+  // dispatch
+  if (m.nsrc == 1 && m.src_dt[0] == m.dst_dt) {
+    tile_copy(m, lo, hi, lane);
+    return;
+  }
+  bool all_f32 = m.dst_dt == u8(DataType::float32);
+  bool floatish = dtype_is_floatish(DataType(m.dst_dt));
+  for (u32 k = 0; k < m.nsrc; ++k) {
+    all_f32 = all_f32 && m.src_dt[k] == u8(DataType::float32);
+    floatish = floatish && dtype_is_floatish(DataType(m.src_dt[k]));
+  }
+  if (all_f32 && m.nsrc >= 2) {
+    if (ReduceFunction(m.func) == ReduceFunction::SUM)
+      tile_reduce_f32<SumOp>(m, lo, hi, lane);
+    else
+      tile_reduce_f32<MaxOp>(m, lo, hi, lane);
+    return;
+  }
+  if (floatish) {
+    tile_float_generic(m, lo, hi, lane);
+    return;
+  }
+  // exact scalar fallback (f64 / int dtypes / mixed)
+  for (u64 i = lo + lane; i < hi; i += 64) execute_move_range(m, i, i + 1);
+}
+
+// ------------------------------------------------------------- mover main
+__device__ void mover_main(GpuEngineState* S) {
+  const int lane = int(threadIdx.x) & 63;
+  MoveDesc* ring = S->mover.ring;
+  MoveState* st = S->mover.st;
+  u64* head = S->mover.head;
+  u64 cursor = 0;
+  for (;;) {
+    u64 h = __hip_atomic_load(head, __ATOMIC_RELAXED, AGENT);
+    if (cursor == h) {
+      if (__hip_atomic_load(S->mover.stop, __ATOMIC_RELAXED, AGENT)) return;
+      __builtin_amdgcn_s_sleep(8);
+      continue;
+    }
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    bool did = false;
+    for (u64 mi = cursor; mi < h; ++mi) {
+      u32 slot = u32(mi % MOVE_RING);
+      if (u32(ring[slot].epoch) != u32(mi + 1)) {
+        if (mi == cursor) cursor++;
+        continue;  // recycled past us
+      }
+      MoveState& ms = st[slot];
+      u32 total = ms.tiles_total;
+      if (__hip_atomic_load(&ms.tiles_claimed, __ATOMIC_RELAXED, AGENT) >= total) {
+        if (mi == cursor) cursor++;
+        continue;
+      }
+      u32 t = 0xFFFFFFFFu;
+      if (lane == 0)
+        t = __hip_atomic_fetch_add(&ms.tiles_claimed, 1u, __ATOMIC_RELAXED, AGENT);
+      t = u32(__shfl(int(t), 0, 64));
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+      // after the claim+acquire, desc/epoch/total are coherent for whichever
+      // epoch owns the slot now; re-read them.
+      u32 cur_total = ms.tiles_total;
+      if (t < cur_total) {
+        run_tile(ring[slot], t, lane);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        if (lane == 0)
+          __hip_atomic_fetch_add(&ms.tiles_done, 1u, __ATOMIC_RELEASE, AGENT);
+        did = true;
+        break;  // rescan from cursor for freshest work
+      }
+      if (mi == cursor) cursor++;
+    }
+    if (!did) __builtin_amdgcn_s_sleep(2);
+  }
+}
+
+// --------------------------------------------------------- scheduler main
+__device__ void scheduler_main(GpuEngineState* S) {
+  Cclo<GpuMover>& C = S->cclo;
+  C.mv = &S->mover;
+  S->mover.ring = S->mq;
+  S->mover.st = S->mst;
+  S->mover.head = &S->mq_head;
+  S->mover.stop = &S->stop;
+  CtrlPage* ctrl = S->ctrl;
+  st_sys(&ctrl->engine_up, 1);
+  u64 consumed = 0, cached_gen = 0, beat = 0;
+  for (;;) {
+    u64 db = ld_sys(&ctrl->doorbell);
+    if (consumed == db) {
+      if (ld_sys(&ctrl->shutdown)) break;
+      if ((++beat & 0x3FF) == 0) st_sys(&ctrl->heartbeat, beat);
+      __builtin_amdgcn_s_sleep(16);
+      continue;
+    }
+    fence_acquire_sys();
+    u64 gen = ld_sys(&ctrl->comm_gen);
+    if (gen != cached_gen) {
+      C.ncomms = u32(ld_sys(&ctrl->ncomms));
+      for (u32 i = 0; i < C.ncomms; ++i) C.comms[i] = S->comm_mirror[i];
+      cached_gen = gen;
+    }
+    bool halted = false;
+    while (consumed < db) {
+      CallDesc d = S->descs[consumed % RING_CAP];
+      RetEntry& r = S->rets[consumed % RING_CAP];
+      u64 t0 = wallclock();
+      u32 e;
+      if (Op(d.scenario) == Op::halt) { e = E_OK; halted = true; }
+      else e = C.run_call(d);
+      r.errcode = e;
+      r.t_start = t0;
+      r.t_end = wallclock();
+      fence_release_sys();
+      st_sys32(&r.seq, u32(consumed + 1));
+      consumed++;
+      if (halted) break;
+    }
+    if (halted) break;
+  }
+  // tell movers to exit, then leave
+  __hip_atomic_store(S->mover.stop, 1u, __ATOMIC_RELEASE, AGENT);
+}
+
+__global__ void __launch_bounds__(256, 1) accl_engine_kernel(GpuEngineState* S) {
+  if (blockIdx.x == 0 && threadIdx.x < 64) {
+    if (threadIdx.x == 0) scheduler_main(S);
+    return;  // lanes 1..63 of the scheduler wave idle under the exec mask
+  }
+  mover_main(S);
+}
+
+void gpu_engine_launch(GpuEngineState* state_dev, int n_wgs, void* hip_stream) {
+  hipLaunchKernelGGL(accl_engine_kernel, dim3(n_wgs), dim3(256), 0,
+                     (hipStream_t)hip_stream, state_dev);
+}
+
+}  // namespace accl

@@ -1,0 +1,48 @@
+// GPU engine state — shared between gpudevice.cpp (host setup) and
+// engine.hip (device code). Both TUs are compiled by hipcc.
+//
+// The persistent-kernel analogue of the CCLO block design: scheduler lane =
+// MicroBlaze control loop (ccl_offload_control.c run()), mover waves = the
+// DMA/arith/segmenter data plane (reference: kernels/cclo/hls/dma_mover/,
+// kernels/plugins/reduce_ops/), the pinned descriptor ring = hostctrl
+// (kernels/plugins/hostctrl/hostctrl.cpp:22-63).
+#pragma once
+#include "../common/sched.hpp"
+#include "../core/backend.hpp"
+
+namespace accl {
+
+// Device-side mover handle. Methods (submit/poll) are device-only and live
+// in engine.hip; the POD fields are set up by the host.
+struct GpuMover {
+  MoveDesc* ring;          // device, MOVE_RING entries
+  MoveState* st;           // device
+  u64* head;               // device: published move count (scheduler writes)
+  u64 head_cache;          // scheduler-private mirror
+  u32* stop;               // device: scheduler tells movers to exit
+
+#if defined(__HIPCC__)
+  __device__ u32 submit(const MoveDesc& m);
+  __device__ bool poll(u32 token);
+#endif
+};
+
+struct GpuEngineState {
+  Cclo<GpuMover> cclo;     // trivially-copyable; host fills, device runs
+  GpuMover mover;
+  MoveDesc mq[MOVE_RING];
+  MoveState mst[MOVE_RING];
+  u64 mq_head;
+  u32 stop;
+  u32 _pad;
+  // pinned-host pointers (device-accessible):
+  CallDesc* descs;
+  RetEntry* rets;
+  CtrlPage* ctrl;
+  CommView* comm_mirror;
+};
+
+// launches the persistent engine kernel (defined in engine.hip)
+void gpu_engine_launch(GpuEngineState* state_dev, int n_wgs, void* hip_stream);
+
+}  // namespace accl

@@ -1,0 +1,188 @@
+#include "gpudevice.hpp"
+#include <hip/hip_runtime.h>
+#include <unistd.h>
+#include <cstdlib>
+#include <cstring>
+#include "engine.hpp"
+
+namespace accl {
+
+static void hip_check(hipError_t e, const char* what) {
+  if (e != hipSuccess)
+    throw accl_error(std::string("hip error in ") + what + ": " +
+                     hipGetErrorString(e));
+}
+
+struct GpuBlob {
+  hipIpcMemHandle_t handle;
+  u64 arena_bytes;
+  u32 rank;
+  u32 device;
+  i32 pid;
+  u32 _pad;
+};
+
+GpuDevice::GpuDevice(u32 nranks, u32 rank, int device_index,
+                     const ProtoConfig* cfg_override, u64 heap_bytes,
+                     int engine_wgs) {
+  dev_ = device_index;
+  cfg_ = cfg_override ? *cfg_override : default_proto_config(nranks, rank);
+  cfg_.nranks = nranks;
+  cfg_.rank = rank;
+  if (const char* e = std::getenv("ACCL_ENGINE_WGS")) engine_wgs = atoi(e);
+  if (engine_wgs > 0) engine_wgs_ = engine_wgs;
+
+  hip_check(hipSetDevice(dev_), "hipSetDevice");
+  layout_ = arena_layout(cfg_);
+  arena_bytes_ = layout_.total_ctl_bytes + heap_bytes;
+
+  // Fine-grained HBM so peer stores + system-scope flags are coherent over
+  // xGMI without kernel boundaries. ACCL_COARSE_ARENA=1 falls back to
+  // coarse (measurement escape hatch).
+  bool coarse = std::getenv("ACCL_COARSE_ARENA") != nullptr;
+  hipError_t e = hipErrorUnknown;
+  if (!coarse) {
+    e = hipExtMallocWithFlags((void**)&arena_base_, arena_bytes_,
+                              hipDeviceMallocFinegrained);
+    fine_grained_ = (e == hipSuccess);
+  }
+  if (coarse || e != hipSuccess) {
+    hip_check(hipMalloc((void**)&arena_base_, arena_bytes_), "hipMalloc arena");
+    fine_grained_ = false;
+  }
+  hip_check(hipMemset(arena_base_, 0, layout_.total_ctl_bytes), "memset ctl");
+
+  ArenaHdr h{};
+  h.version = 1;
+  h.rank = rank; h.nranks = nranks;
+  h.n_slots = cfg_.n_slots; h.slot_bytes = cfg_.slot_bytes;
+  h.n_rndzv = cfg_.n_rndzv; h.n_stream = cfg_.n_stream;
+  h.stream_bytes = cfg_.stream_bytes;
+  h.arena_bytes = arena_bytes_;
+  h.eager_off = layout_.eager_off;
+  h.rndzv_addr_off = layout_.rndzv_addr_off;
+  h.rndzv_done_off = layout_.rndzv_done_off;
+  h.stream_off = layout_.stream_off;
+  h.slots_off = layout_.slots_off;
+  h.heap_off = layout_.heap_off;
+  h.barrier_off = layout_.barrier_off;
+  h.direct_off = layout_.direct_off;
+  h.spare_off = layout_.spare_off;
+  h.spare_bytes = layout_.spare_bytes;
+  h.magic = ARENA_MAGIC;
+  hip_check(hipMemcpy(arena_base_, &h, sizeof(h), hipMemcpyHostToDevice),
+            "write ArenaHdr");
+
+  heap_.init(layout_.heap_off, arena_bytes_ - layout_.heap_off);
+
+  hip_check(hipHostMalloc(&ring_pinned_, sizeof(RingPage), 0), "hipHostMalloc ring");
+  std::memset(ring_pinned_, 0, sizeof(RingPage));
+  ring_ = (RingPage*)ring_pinned_;
+
+  hip_check(hipMalloc(&state_dev_, sizeof(GpuEngineState)), "hipMalloc state");
+  hipStream_t s;
+  hip_check(hipStreamCreateWithFlags(&s, hipStreamNonBlocking), "stream");
+  stream_ = s;
+}
+
+GpuDevice::~GpuDevice() {
+  try { shutdown(); } catch (...) {}
+  for (u32 r = 0; r < cfg_.nranks; ++r)
+    if (peer_base_[r] && r != cfg_.rank) (void)hipIpcCloseMemHandle(peer_base_[r]);
+  if (state_dev_) (void)hipFree(state_dev_);
+  if (arena_base_) (void)hipFree(arena_base_);
+  if (ring_pinned_) (void)hipHostFree(ring_pinned_);
+  if (stream_) (void)hipStreamDestroy((hipStream_t)stream_);
+}
+
+std::vector<char> GpuDevice::local_blob() {
+  GpuBlob b{};
+  hip_check(hipIpcGetMemHandle(&b.handle, arena_base_), "hipIpcGetMemHandle");
+  b.arena_bytes = arena_bytes_;
+  b.rank = cfg_.rank;
+  b.device = u32(dev_);
+  b.pid = i32(getpid());
+  std::vector<char> out(sizeof(b));
+  std::memcpy(out.data(), &b, sizeof(b));
+  return out;
+}
+
+void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
+  if (blobs.size() != cfg_.nranks) throw accl_error("gpu: blob count != nranks");
+  for (u32 r = 0; r < cfg_.nranks; ++r) {
+    if (r == cfg_.rank) { peer_base_[r] = arena_base_; continue; }
+    GpuBlob b{};
+    std::memcpy(&b, blobs[r].data(), sizeof(b));
+    void* p = nullptr;
+    hip_check(hipIpcOpenMemHandle(&p, b.handle, hipIpcMemLazyEnablePeerAccess),
+              "hipIpcOpenMemHandle");
+    peer_base_[r] = (char*)p;
+  }
+
+  // build the engine state on host, copy to device
+  auto* st = new GpuEngineState();
+  std::memset((void*)st, 0, sizeof(GpuEngineState));
+  Cclo<GpuMover>& C = st->cclo;
+  C.cfg = cfg_;
+  C.tv.cfg = cfg_;
+  for (u32 r = 0; r < cfg_.nranks; ++r) C.tv.arena[r] = peer_base_[r];
+  C.timeout_ticks = cfg_.timeout_us * TICKS_PER_US;
+  C.max_eager_bytes = cfg_.max_eager;
+  // device pointers inside the state buffer itself
+  auto* dstate = (GpuEngineState*)state_dev_;
+  st->mover.ring = dstate->mq;          // address arithmetic on device ptr
+  st->mover.st = dstate->mst;
+  st->mover.head = &dstate->mq_head;
+  st->mover.stop = &dstate->stop;
+  // pinned pointers as seen by the device
+  RingPage* rp = (RingPage*)ring_pinned_;
+  void* dev_ptr = nullptr;
+  hip_check(hipHostGetDevicePointer(&dev_ptr, rp, 0), "hostGetDevicePointer");
+  auto* rp_dev = (RingPage*)dev_ptr;
+  st->descs = rp_dev->descs;
+  st->rets = rp_dev->rets;
+  st->ctrl = &rp_dev->ctrl;
+  st->comm_mirror = rp_dev->comm_mirror;
+  hip_check(hipMemcpy(state_dev_, st, sizeof(GpuEngineState),
+                      hipMemcpyHostToDevice), "state upload");
+  delete st;
+
+  // global communicator 0 (before launch: engine reads mirror at gen bump)
+  std::vector<u32> members(cfg_.nranks);
+  for (u32 i = 0; i < cfg_.nranks; ++i) members[i] = i;
+  // write mirror directly (engine not yet running; add_comm would quiesce)
+  CommView& c = ring_->comm_mirror[0];
+  c.id = 0; c.rank = cfg_.rank; c.size = cfg_.nranks;
+  for (u32 i = 0; i < cfg_.nranks; ++i) c.members[i] = i;
+  __atomic_store_n((u64*)&ring_->ctrl.ncomms, 1, __ATOMIC_RELEASE);
+  __atomic_store_n((u64*)&ring_->ctrl.comm_gen, 1, __ATOMIC_RELEASE);
+
+  gpu_engine_launch((GpuEngineState*)state_dev_, engine_wgs_, stream_);
+  launched_ = true;
+  // wait for the engine to come up
+  u64 t0 = wallclock_host_ns();
+  while (!__atomic_load_n((u64*)&ring_->ctrl.engine_up, __ATOMIC_ACQUIRE)) {
+    if (wallclock_host_ns() - t0 > 30ull * 1000000000)
+      throw accl_error("gpu: engine kernel never came up");
+    usleep(100);
+  }
+}
+
+void GpuDevice::shutdown() {
+  if (!launched_) return;
+  __atomic_store_n((u64*)&ring_->ctrl.shutdown, 1, __ATOMIC_RELEASE);
+  hipError_t e = hipStreamSynchronize((hipStream_t)stream_);
+  launched_ = false;
+  hip_check(e, "engine shutdown");
+}
+
+void GpuDevice::write_arena(u64 off, const void* src, u64 bytes) {
+  hip_check(hipMemcpy(arena_base_ + off, src, bytes, hipMemcpyHostToDevice),
+            "write_arena");
+}
+void GpuDevice::read_arena(u64 off, void* dst, u64 bytes) {
+  hip_check(hipMemcpy(dst, arena_base_ + off, bytes, hipMemcpyDeviceToHost),
+            "read_arena");
+}
+
+}  // namespace accl

@@ -1,0 +1,45 @@
+// GpuDevice — host-side backend owning one MI355X GPU: fine-grained HBM
+// arena (IPC-shared with peer ranks over xGMI), pinned host descriptor ring,
+// and the persistent engine kernel.
+//
+// Analogue of the reference's XRTDevice + accl_network_utils bring-up
+// (reference: driver/xrt/src/xrtdevice.cpp:36-303;
+// driver/utils/accl_network_utils/ — POE configuration becomes IPC handle
+// exchange + peer mapping).
+#pragma once
+#include "../core/backend.hpp"
+
+namespace accl {
+
+class GpuDevice : public Backend {
+ public:
+  GpuDevice(u32 nranks, u32 rank, int device_index,
+            const ProtoConfig* cfg_override = nullptr,
+            u64 heap_bytes = 8ull << 30, int engine_wgs = 0);
+  ~GpuDevice() override;
+
+  std::vector<char> local_blob() override;
+  void connect(const std::vector<std::vector<char>>& blobs) override;
+  void shutdown() override;
+  bool is_gpu() const override { return true; }
+
+  void write_arena(u64 off, const void* src, u64 bytes) override;
+  void read_arena(u64 off, void* dst, u64 bytes) override;
+
+  int device_index() const { return dev_; }
+  u64 arena_bytes() const { return arena_bytes_; }
+
+ private:
+  int dev_ = 0;
+  int engine_wgs_ = 16;
+  u64 arena_bytes_ = 0;
+  ArenaLayout layout_{};
+  bool fine_grained_ = true;
+  char* peer_base_[MAX_RANKS] = {};
+  void* ring_pinned_ = nullptr;     // RingPage, hipHostMalloc
+  void* state_dev_ = nullptr;       // GpuEngineState, hipMalloc
+  void* stream_ = nullptr;          // hipStream_t
+  bool launched_ = false;
+};
+
+}  // namespace accl

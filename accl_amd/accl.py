@@ -89,7 +89,13 @@ class ACCL:
             blobs[rank] = blob
             self._a.connect(blobs)
             return
-        if bootstrap in ("auto", "torch"):
+        if bootstrap == "auto":
+            try:
+                import torch.distributed as dist
+                bootstrap = "torch" if dist.is_initialized() else "file"
+            except Exception:
+                bootstrap = "file"
+        if bootstrap == "torch":
             import torch.distributed as dist
             if not dist.is_initialized():
                 raise RuntimeError(
@@ -98,6 +104,28 @@ class ACCL:
             objs = [None] * nranks
             dist.all_gather_object(objs, bytes(blob))
             self._a.connect(list(objs))
+        elif bootstrap == "file":
+            # filesystem rendezvous under a job-derived directory (single node)
+            import tempfile
+            import time
+            d = os.path.join(tempfile.gettempdir(),
+                             f"accl_bootstrap_{emu_job_name(job)}")
+            os.makedirs(d, exist_ok=True)
+            with open(os.path.join(d, f"r{rank}.tmp"), "wb") as f:
+                f.write(bytes(blob))
+            os.rename(os.path.join(d, f"r{rank}.tmp"),
+                      os.path.join(d, f"r{rank}.blob"))
+            blobs = []
+            deadline = time.time() + 60
+            for r in range(nranks):
+                p = os.path.join(d, f"r{r}.blob")
+                while not os.path.exists(p):
+                    if time.time() > deadline:
+                        raise RuntimeError(f"bootstrap: rank {r} blob missing")
+                    time.sleep(0.01)
+                with open(p, "rb") as f:
+                    blobs.append(f.read())
+            self._a.connect(blobs)
         else:
             raise ValueError(f"unknown bootstrap {bootstrap!r}")
 

@@ -1,0 +1,163 @@
+#!/usr/bin/env python3
+"""accl_amd flagship benchmark — all-reduce bus bandwidth (BASELINE.json).
+
+One "step" = one blocking fp32 allreduce of the flagship message size
+(default 256 MiB) through the persistent GPU engine over xGMI.
+
+Run:  python bench.py --gpus N --steps K --warmup W
+The driver launches N>1 as one rank per GPU via torch.distributed.run;
+torch.distributed (gloo) is used ONLY to bootstrap (IPC handle allgather)
+and to reduce timings across ranks — the data plane is accl_amd's own.
+
+Bus bandwidth = 2(P-1)/P * bytes / t  (the standard allreduce factor;
+reference metric protocol: BASELINE.md, test/host/Coyote/test.cpp:523-534
+computes Gbit/s the same way for its collectives).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+BASELINE_GBPS = 12.5  # reference anchor: 100 Gbps line rate
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=int(os.environ.get("WORLD_SIZE", "1")))
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--bytes", type=int, default=256 << 20)
+    p.add_argument("--sweep", action="store_true",
+                   help="also print the 4KB..1GB busbw curve (rank 0)")
+    p.add_argument("--backend", default="auto", choices=["auto", "gpu", "emu"])
+    p.add_argument("--out-csv", default=None)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    rank = int(os.environ.get("RANK", "0"))
+
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        dist = tdist
+        if not dist.is_initialized():
+            dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    import accl_amd as A
+    import numpy as np
+
+    backend = args.backend
+    if backend == "auto":
+        backend = "gpu" if A.ACCL._has_gpu() else "emu"
+    slot_mb = 4 if backend == "gpu" else 1
+    heap = max(4 * args.bytes + (64 << 20), 1 << 30) if backend == "gpu" else None
+    a = A.ACCL(nranks=world, rank=rank, backend=backend,
+               heap_bytes=heap,
+               opts={"slot_bytes": slot_mb << 20, "n_slots": 8})
+
+    count = args.bytes // 4
+    src = a.create_buffer(count, A.DataType.float32, device_only=True)
+    dst = a.create_buffer(count, A.DataType.float32, device_only=True)
+    if backend == "gpu":
+        import torch
+        t = a.tensor(src)
+        t.copy_(torch.randn(count, device=t.device))
+        torch.cuda.synchronize()
+    else:
+        tmp = a.create_buffer(min(count, 1 << 20), A.DataType.float32)
+        tmp.write(np.random.default_rng(0).standard_normal(
+            min(count, 1 << 20), dtype=np.float32))
+        a.copy(tmp, src, min(count, 1 << 20))
+
+    def one_allreduce(s=src, d=dst, n=count):
+        a.allreduce(s, d, n, A.ReduceFunction.SUM,
+                    from_device=True, to_device=True)
+
+    def timed(fn, steps, warmup):
+        for _ in range(warmup):
+            fn()
+        a.barrier()
+        t0 = time.perf_counter()
+        for _ in range(steps):
+            fn()
+        a.barrier()
+        t1 = time.perf_counter()
+        el = (t1 - t0) / steps
+        if dist is not None:
+            import torch
+            t = torch.tensor([el], dtype=torch.float64)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            el = float(t.item())
+        return el
+
+    def busbw(nbytes, sec, P):
+        factor = 2.0 * (P - 1) / P if P > 1 else 1.0
+        return factor * nbytes / sec / 1e9
+
+    el = timed(one_allreduce, args.steps, args.warmup)
+    value = busbw(args.bytes, el, world)
+
+    sweep_rows = []
+    if args.sweep:
+        sz = 4096
+        while sz <= (1 << 30):
+            c = sz // 4
+            if c <= count:
+                s2, d2 = src.slice(0, c), dst.slice(0, c)
+                e = timed(lambda s=s2, d=d2, c=c: a.allreduce(
+                    s, d, c, A.ReduceFunction.SUM, from_device=True,
+                    to_device=True), max(3, min(20, (1 << 26) // sz)), 2)
+                sweep_rows.append((sz, e * 1e6, busbw(sz, e, world)))
+            sz *= 4
+        if rank == 0:
+            lines = ["bytes,usec,busbw_GBps"] + [
+                f"{b},{u:.2f},{g:.2f}" for b, u, g in sweep_rows]
+            csv = "\n".join(lines)
+            print(csv, file=sys.stderr)
+            if args.out_csv:
+                with open(args.out_csv, "w") as f:
+                    f.write(csv + "\n")
+
+    if rank == 0:
+        out = {
+            "metric": "allreduce_busbw_GBps",
+            "value": round(value, 3),
+            "unit": "GB/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(el * 1e3, 4),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(value / BASELINE_GBPS, 3),
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "allreduce",
+                "collective": "allreduce",
+                "message_bytes": args.bytes,
+                "global_batch": 1,
+                "seq_len": args.bytes // 4,
+                "parallelism": f"fullmesh_p{world}",
+                "backend": backend,
+                "metric_desc": ("all-reduce bus bandwidth (GB/s), fp32, "
+                                "per BASELINE.json; vs_baseline anchors the "
+                                "reference's 100 Gbps (12.5 GB/s) line rate"),
+            },
+        }
+        print(json.dumps(out))
+    a.close()
+    if dist is not None:
+        dist.barrier()
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,70 @@
+"""Multi-process harness for emulator (and single-GPU) collective tests.
+
+Mirrors the reference's MPI-launched gtest + emulator-process model
+(reference: test/host/xrt/src/test.cpp run under mpirun with --startemu,
+test/host/xrt/src/utility.cpp:25-70): here each rank is a forked process
+running one scenario function; shm arenas replace the ZMQ ethernet.
+"""
+import multiprocessing as mp
+import os
+import traceback
+
+import numpy as np
+
+_JOB_COUNTER = 0
+
+
+def fresh_job():
+    global _JOB_COUNTER
+    _JOB_COUNTER += 1
+    return f"t{os.getpid()}_{_JOB_COUNTER}"
+
+
+def _worker(fn, rank, nranks, job, opts, q, backend):
+    try:
+        import accl_amd as A
+        a = A.ACCL(nranks=nranks, rank=rank, backend=backend, job=job, **(
+            {"opts": opts} if opts else {}))
+        try:
+            fn(a, rank, nranks)
+        finally:
+            a.close()
+        q.put((rank, None))
+    except Exception:
+        q.put((rank, traceback.format_exc()))
+
+
+def run_ranks(fn, nranks, opts=None, timeout=120, backend="emu"):
+    """Run `fn(accl, rank, nranks)` on every rank; raise on any failure."""
+    ctx = mp.get_context("fork")
+    q = ctx.Queue()
+    job = fresh_job()
+    ps = [ctx.Process(target=_worker, args=(fn, r, nranks, job, opts, q, backend))
+          for r in range(nranks)]
+    for p in ps:
+        p.start()
+    errs = []
+    try:
+        for _ in range(nranks):
+            rank, err = q.get(timeout=timeout)
+            if err:
+                errs.append(f"rank {rank}:\n{err}")
+    finally:
+        for p in ps:
+            p.join(timeout=10)
+            if p.is_alive():
+                p.terminate()
+                errs.append(f"rank {p.pid} hung; terminated")
+    if errs:
+        raise AssertionError("\n".join(errs))
+
+
+def rd(buf, count, dtype=np.float32):
+    out = np.zeros(count, dtype)
+    buf.read(out)
+    return out
+
+
+def pattern(count, rank, dtype=np.float32, seed=0):
+    # deterministic per-rank data, bounded magnitude (safe for f16 sums)
+    return (((np.arange(count) * 7 + rank * 13 + seed) % 61) - 30).astype(dtype)

@@ -1,0 +1,320 @@
+"""Emulator-backend collective tests — the reference gtest matrix
+(reference: test/host/xrt/src/test.cpp: copy/combine/sendrcv incl.
+segmentation edges, bcast/scatter/gather/allgather/reduce/allreduce/
+reduce_scatter/alltoall/barrier, multicomm, compression) run against the CPU
+engine (same scheduler source as the GPU engine)."""
+import numpy as np
+import pytest
+
+import accl_amd as A
+from emu_util import pattern, rd, run_ranks
+
+DT = A.DataType
+RF = A.ReduceFunction
+
+# small slots to exercise segmentation + credit throttling cheaply
+SMALL = {"n_slots": 4, "slot_bytes": 4096, "timeout_us": 20_000_000}
+
+
+def _mk(a, count, dtype=DT.float32):
+    return a.create_buffer(count, dtype)
+
+
+# --------------------------------------------------------------- local ops
+def _copy(a, rank, n):
+    cnt = 5000
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    x = pattern(cnt, rank)
+    s.write(x)
+    a.copy(s, d, cnt)
+    assert np.array_equal(rd(d, cnt), x)
+
+
+def _combine(a, rank, n):
+    cnt = 3000
+    s1, s2, d = _mk(a, cnt), _mk(a, cnt), _mk(a, cnt)
+    x, y = pattern(cnt, 1), pattern(cnt, 2)
+    s1.write(x); s2.write(y)
+    a.combine(cnt, RF.SUM, s1, s2, d)
+    assert np.array_equal(rd(d, cnt), x + y)
+    a.combine(cnt, RF.MAX, s1, s2, d)
+    assert np.array_equal(rd(d, cnt), np.maximum(x, y))
+
+
+def test_copy_combine():
+    run_ranks(_copy, 1)
+    run_ranks(_combine, 1)
+
+
+# ----------------------------------------------------------------- sendrecv
+def _sendrecv(a, rank, n):
+    cnt = 3000
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    s.write(pattern(cnt, rank))
+    prev, nxt = (rank - 1) % n, (rank + 1) % n
+    if rank % 2 == 0:
+        a.send(s, cnt, dst=nxt, tag=9)
+        a.recv(d, cnt, src=prev, tag=9)
+    else:
+        a.recv(d, cnt, src=prev, tag=9)
+        a.send(s, cnt, dst=nxt, tag=9)
+    assert np.array_equal(rd(d, cnt), pattern(cnt, prev))
+
+
+def _sendrecv_seg_edges(a, rank, n):
+    # counts straddling slot boundaries (reference: segmentation tests,
+    # test.cpp:345-393: rxbuf_size multiples +/- offsets)
+    slot_elems = 4096 // 4
+    for cnt in (slot_elems - 1, slot_elems, slot_elems + 1,
+                3 * slot_elems, 5 * slot_elems + 7):
+        s, d = _mk(a, cnt), _mk(a, cnt)
+        s.write(pattern(cnt, rank, seed=cnt))
+        if rank == 0:
+            a.send(s, cnt, dst=1, tag=cnt % 1000)
+            a.recv(d, cnt, src=1, tag=1 + cnt % 1000)
+            assert np.array_equal(rd(d, cnt), pattern(cnt, 1, seed=cnt))
+        elif rank == 1:
+            a.recv(d, cnt, src=0, tag=cnt % 1000)
+            assert np.array_equal(rd(d, cnt), pattern(cnt, 0, seed=cnt))
+            a.send(s, cnt, dst=0, tag=1 + cnt % 1000)
+
+
+def _sendrecv_tag_any(a, rank, n):
+    cnt = 100
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    s.write(pattern(cnt, rank))
+    if rank == 0:
+        a.send(s, cnt, dst=1, tag=42)
+    elif rank == 1:
+        a.recv(d, cnt, src=0, tag=A.TAG_ANY)
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 0))
+
+
+@pytest.mark.parametrize("n", [2, 4])
+def test_sendrecv(n):
+    run_ranks(_sendrecv, n, opts=SMALL)
+
+
+def test_sendrecv_segmentation():
+    run_ranks(_sendrecv_seg_edges, 2, opts=SMALL)
+
+
+def test_sendrecv_tag_any():
+    run_ranks(_sendrecv_tag_any, 2)
+
+
+def _rendezvous_sendrecv(a, rank, n):
+    # above max_eager with arena buffers -> direct rendezvous write
+    cnt = 300_000  # 1.2 MB > max_eager=256K
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    s.write(pattern(cnt, rank))
+    if rank == 0:
+        a.send(s, cnt, dst=1, tag=3)
+        a.recv(d, cnt, src=1, tag=4)
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 1))
+    elif rank == 1:
+        a.recv(d, cnt, src=0, tag=3)
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 0))
+        a.send(s, cnt, dst=0, tag=4)
+
+
+def test_rendezvous_sendrecv():
+    run_ranks(_rendezvous_sendrecv, 2,
+              opts={"max_eager": 256 * 1024, "timeout_us": 20_000_000})
+
+
+# --------------------------------------------------------------- collectives
+def _bcast(a, rank, n):
+    for root in range(min(n, 3)):
+        cnt = 4000
+        b = _mk(a, cnt)
+        if rank == root:
+            b.write(pattern(cnt, root, seed=7))
+        a.bcast(b, cnt, root=root)
+        assert np.array_equal(rd(b, cnt), pattern(cnt, root, seed=7))
+
+
+def _scatter(a, rank, n):
+    for root in range(min(n, 2)):
+        cnt = 1500
+        s, d = _mk(a, cnt * n), _mk(a, cnt)
+        if rank == root:
+            s.write(np.concatenate([pattern(cnt, r, seed=3) for r in range(n)]))
+        a.scatter(s, d, cnt, root=root)
+        assert np.array_equal(rd(d, cnt), pattern(cnt, rank, seed=3))
+
+
+def _gather(a, rank, n):
+    for root in range(min(n, 2)):
+        cnt = 1500
+        s, d = _mk(a, cnt), _mk(a, cnt * n)
+        s.write(pattern(cnt, rank, seed=4))
+        a.gather(s, d, cnt, root=root)
+        if rank == root:
+            exp = np.concatenate([pattern(cnt, r, seed=4) for r in range(n)])
+            assert np.array_equal(rd(d, cnt * n), exp)
+
+
+def _allgather(a, rank, n):
+    cnt = 1500
+    s, d = _mk(a, cnt), _mk(a, cnt * n)
+    s.write(pattern(cnt, rank, seed=5))
+    a.allgather(s, d, cnt)
+    exp = np.concatenate([pattern(cnt, r, seed=5) for r in range(n)])
+    assert np.array_equal(rd(d, cnt * n), exp)
+
+
+def _reduce(a, rank, n):
+    for root in range(min(n, 2)):
+        for f, npf in ((RF.SUM, np.sum), (RF.MAX, np.max)):
+            cnt = 2000
+            s, d = _mk(a, cnt), _mk(a, cnt)
+            s.write(pattern(cnt, rank, seed=6))
+            a.reduce(s, d, cnt, root=root, function=f)
+            if rank == root:
+                allv = np.stack([pattern(cnt, r, seed=6) for r in range(n)])
+                exp = allv.sum(0) if f == RF.SUM else allv.max(0)
+                assert np.allclose(rd(d, cnt), exp)
+
+
+def _allreduce(a, rank, n):
+    for cnt in (1, 63, 2000, 300_000):
+        s, d = _mk(a, cnt), _mk(a, cnt)
+        s.write(pattern(cnt, rank, seed=cnt))
+        a.allreduce(s, d, cnt, RF.SUM)
+        exp = np.stack([pattern(cnt, r, seed=cnt) for r in range(n)]).sum(0)
+        assert np.allclose(rd(d, cnt), exp), f"cnt={cnt}"
+
+
+def _reduce_scatter(a, rank, n):
+    cnt = 1700  # per-rank
+    s, d = _mk(a, cnt * n), _mk(a, cnt)
+    s.write(np.concatenate([pattern(cnt, 100 * rank + j, seed=8)
+                            for j in range(n)]))
+    a.reduce_scatter(s, d, cnt, RF.SUM)
+    exp = np.stack([pattern(cnt, 100 * r + rank, seed=8)
+                    for r in range(n)]).sum(0)
+    assert np.allclose(rd(d, cnt), exp)
+
+
+def _alltoall(a, rank, n):
+    cnt = 900
+    s, d = _mk(a, cnt * n), _mk(a, cnt * n)
+    s.write(np.concatenate([pattern(cnt, 100 * rank + j, seed=9)
+                            for j in range(n)]))
+    a.alltoall(s, d, cnt)
+    exp = np.concatenate([pattern(cnt, 100 * r + rank, seed=9)
+                          for r in range(n)])
+    assert np.array_equal(rd(d, cnt * n), exp)
+
+
+def _barrier(a, rank, n):
+    for _ in range(5):
+        a.barrier()
+
+
+COLLECTIVES = [_bcast, _scatter, _gather, _allgather, _reduce, _allreduce,
+               _reduce_scatter, _alltoall, _barrier]
+
+
+@pytest.mark.parametrize("fn", COLLECTIVES, ids=lambda f: f.__name__.strip("_"))
+@pytest.mark.parametrize("n", [2, 4])
+def test_collective(fn, n):
+    run_ranks(fn, n, opts=SMALL, timeout=180)
+
+
+@pytest.mark.parametrize("fn", [_allreduce, _allgather, _reduce_scatter])
+def test_collective_p3_odd(fn):
+    run_ranks(fn, 3, opts=SMALL, timeout=180)
+
+
+def test_single_rank_collectives():
+    def all_ops(a, rank, n):
+        _bcast(a, rank, n)
+        _allgather(a, rank, n)
+        _reduce(a, rank, n)
+        _allreduce(a, rank, n)
+        _reduce_scatter(a, rank, n)
+        _alltoall(a, rank, n)
+        _barrier(a, rank, n)
+    run_ranks(all_ops, 1)
+
+
+# -------------------------------------------------------------- dtypes
+def _allreduce_dtypes(a, rank, n):
+    for dt, npdt in ((DT.float64, np.float64), (DT.int32, np.int32),
+                     (DT.int64, np.int64), (DT.float16, np.float16)):
+        cnt = 1000
+        s, d = _mk(a, cnt, dt), _mk(a, cnt, dt)
+        x = pattern(cnt, rank, dtype=npdt, seed=11)
+        s.write(x)
+        a.allreduce(s, d, cnt, RF.SUM)
+        exp = np.stack([pattern(cnt, r, dtype=npdt, seed=11)
+                        for r in range(n)]).sum(0).astype(npdt)
+        assert np.allclose(rd(d, cnt, npdt).astype(np.float64),
+                           exp.astype(np.float64), atol=1e-2)
+
+
+def test_allreduce_dtypes():
+    run_ranks(_allreduce_dtypes, 2)
+
+
+# ---------------------------------------------------- compression (wire cast)
+def _compressed(a, rank, n):
+    # f32 buffers, f16 on the wire (reference: hp_compression lanes +
+    # compressed sendrcv/allreduce tests)
+    cnt = 2048
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    s.write(pattern(cnt, rank, seed=12))
+    a.allreduce(s, d, cnt, RF.SUM, compress_dtype=DT.float16)
+    exp = np.stack([pattern(cnt, r, seed=12) for r in range(n)]).sum(0)
+    assert np.allclose(rd(d, cnt), exp, atol=0.5)
+    if rank == 0:
+        a.send(s, cnt, dst=1, tag=2, compress_dtype=DT.bfloat16)
+    elif rank == 1:
+        a.recv(d, cnt, src=0, tag=2, compress_dtype=DT.bfloat16)
+        assert np.allclose(rd(d, cnt), pattern(cnt, 0, seed=12), atol=1.0)
+
+
+def test_compressed():
+    run_ranks(_compressed, 2, opts=SMALL)
+
+
+# ------------------------------------------------------------- multicomm
+def _multicomm(a, rank, n):
+    # reference: split communicator + collectives inside the subgroup
+    # (test.cpp:756-833)
+    half = [r for r in range(n) if r < n // 2]
+    cnt = 512
+    if rank in half:
+        cid = a.split_communicator(half)
+        s, d = _mk(a, cnt), _mk(a, cnt)
+        s.write(pattern(cnt, rank, seed=13))
+        a.allreduce(s, d, cnt, RF.SUM, comm=cid)
+        exp = np.stack([pattern(cnt, r, seed=13) for r in half]).sum(0)
+        assert np.allclose(rd(d, cnt), exp)
+        a.barrier(comm=cid)
+    a.barrier()
+
+
+def test_multicomm():
+    run_ranks(_multicomm, 4, opts=SMALL)
+
+
+# ------------------------------------------------------------ async + perf
+def _async_and_duration(a, rank, n):
+    cnt = 1000
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    s.write(pattern(cnt, rank))
+    s.sync_to_device()
+    r = a.allreduce(s, d, cnt, RF.SUM, from_device=True, to_device=True,
+                    run_async=True)
+    assert r.wait() == 0
+    assert r.duration_us() >= 0.0
+    d.sync_from_device()
+    exp = np.stack([pattern(cnt, q) for q in range(n)]).sum(0)
+    assert np.allclose(rd(d, cnt), exp)
+
+
+def test_async():
+    run_ranks(_async_and_duration, 2)

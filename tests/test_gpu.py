@@ -1,0 +1,158 @@
+"""GPU-engine tests (run on an MI355X via gpurun; every test is @gpu).
+
+The 2-rank tests run two processes sharing one GPU — same IPC/arena/engine
+protocol as the multi-GPU case, with xGMI replaced by local HBM; the 8-GPU
+path is exercised by the driver's round-end scaling bench."""
+import numpy as np
+import pytest
+
+import accl_amd as A
+from emu_util import pattern, rd, run_ranks
+
+DT = A.DataType
+RF = A.ReduceFunction
+
+gpu = pytest.mark.gpu
+
+
+def _has_gpu():
+    try:
+        import torch
+        return torch.cuda.is_available()
+    except Exception:
+        return False
+
+
+pytestmark = [gpu, pytest.mark.skipif(not _has_gpu(), reason="no HIP GPU")]
+
+
+@pytest.fixture(scope="module")
+def acc1():
+    a = A.ACCL(nranks=1, rank=0, backend="gpu", job="g1", heap_bytes=2 << 30)
+    yield a
+    a.close()
+
+
+def test_copy_engine(acc1):
+    a = acc1
+    for cnt in (17, 4096, 1 << 20):
+        s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt, DT.float32)
+        x = np.random.default_rng(cnt).standard_normal(cnt, dtype=np.float32)
+        s.write(x)
+        a.copy(s, d, cnt)
+        assert np.array_equal(rd(d, cnt), x)
+
+
+def test_combine_numerics_vs_torch(acc1):
+    """Engine reduce kernel vs a plain PyTorch fp32 reference."""
+    import torch
+    a = acc1
+    cnt = 1 << 20
+    s1, s2, d = (a.create_buffer(cnt, DT.float32) for _ in range(3))
+    x = np.random.default_rng(1).standard_normal(cnt, dtype=np.float32)
+    y = np.random.default_rng(2).standard_normal(cnt, dtype=np.float32)
+    s1.write(x); s2.write(y)
+    a.combine(cnt, RF.SUM, s1, s2, d)
+    ref = (torch.from_numpy(x).cuda() + torch.from_numpy(y).cuda()).cpu().numpy()
+    assert np.allclose(rd(d, cnt), ref, atol=0), "fp32 add must be exact"
+    a.combine(cnt, RF.MAX, s1, s2, d)
+    ref = torch.maximum(torch.from_numpy(x).cuda(),
+                        torch.from_numpy(y).cuda()).cpu().numpy()
+    assert np.array_equal(rd(d, cnt), ref)
+
+
+def test_combine_dtypes(acc1):
+    import torch
+    a = acc1
+    cnt = 8192
+    for dt, tdt, tol in ((DT.bfloat16, torch.bfloat16, 0.0),
+                         (DT.float16, torch.float16, 0.0),
+                         (DT.float64, torch.float64, 0.0),
+                         (DT.int32, torch.int32, 0.0)):
+        s1, s2, d = (a.create_buffer(cnt, dt) for _ in range(3))
+        t1 = torch.randn(cnt, dtype=torch.float32).to(tdt) if tdt.is_floating_point \
+            else torch.randint(-1000, 1000, (cnt,), dtype=tdt)
+        t2 = torch.randn(cnt, dtype=torch.float32).to(tdt) if tdt.is_floating_point \
+            else torch.randint(-1000, 1000, (cnt,), dtype=tdt)
+        s1.write(t1.view(torch.int8).numpy() if tdt == torch.bfloat16 else t1.numpy())
+        s2.write(t2.view(torch.int8).numpy() if tdt == torch.bfloat16 else t2.numpy())
+        a.combine(cnt, RF.SUM, s1, s2, d)
+        ref = (t1.cuda().float() + t2.cuda().float()).to(tdt).cpu()
+        got = np.zeros(cnt * t1.element_size(), np.int8)
+        d.read(got.view(np.int8))
+        got_t = torch.from_numpy(got).view(tdt)
+        fr = ref.float()
+        fg = got_t.float()
+        assert torch.allclose(fg, fr, atol=float(tol), rtol=1e-2), str(dt)
+
+
+def test_dlpack_tensor_view(acc1):
+    import torch
+    a = acc1
+    cnt = 4096
+    b = a.create_buffer(cnt, DT.float32, device_only=True)
+    t = a.tensor(b)
+    assert t.is_cuda and t.numel() == cnt
+    t.fill_(3.0)
+    torch.cuda.synchronize()
+    d = a.create_buffer(cnt, DT.float32)
+    a.copy(b, d, cnt, from_device=True)
+    assert np.allclose(rd(d, cnt), 3.0)
+
+
+def test_allreduce_single(acc1):
+    a = acc1
+    cnt = 1 << 22
+    s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt, DT.float32)
+    x = np.random.default_rng(3).standard_normal(cnt, dtype=np.float32)
+    s.write(x)
+    a.allreduce(s, d, cnt, RF.SUM)
+    assert np.array_equal(rd(d, cnt), x)
+
+
+# ---------------- 2 processes, 1 GPU: full protocol over IPC ----------------
+def _ar2(a, rank, n):
+    for cnt in (1000, 1 << 20):
+        s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt, DT.float32)
+        s.write(pattern(cnt, rank, seed=cnt))
+        a.allreduce(s, d, cnt, RF.SUM)
+        exp = np.stack([pattern(cnt, r, seed=cnt) for r in range(n)]).sum(0)
+        assert np.allclose(rd(d, cnt), exp), f"cnt={cnt}"
+
+
+def _sendrecv2(a, rank, n):
+    cnt = 200_000
+    s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt, DT.float32)
+    s.write(pattern(cnt, rank))
+    if rank == 0:
+        a.send(s, cnt, dst=1, tag=5)
+        a.recv(d, cnt, src=1, tag=6)
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 1))
+    else:
+        a.recv(d, cnt, src=0, tag=5)
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 0))
+        a.send(s, cnt, dst=0, tag=6)
+    a.barrier()
+
+
+def _coll2(a, rank, n):
+    cnt = 50_000
+    s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt * n, DT.float32)
+    s.write(pattern(cnt, rank, seed=5))
+    a.allgather(s, d, cnt)
+    exp = np.concatenate([pattern(cnt, r, seed=5) for r in range(n)])
+    assert np.array_equal(rd(d, cnt * n), exp)
+    s2 = a.create_buffer(cnt * n, DT.float32)
+    d2 = a.create_buffer(cnt, DT.float32)
+    s2.write(np.concatenate([pattern(cnt, 100 * rank + j, seed=8)
+                             for j in range(n)]))
+    a.reduce_scatter(s2, d2, cnt, RF.SUM)
+    exp = np.stack([pattern(cnt, 100 * r + rank, seed=8)
+                    for r in range(n)]).sum(0)
+    assert np.allclose(rd(d2, cnt), exp)
+
+
+@pytest.mark.parametrize("fn", [_ar2, _sendrecv2, _coll2],
+                         ids=["allreduce", "sendrecv", "allgather_rs"])
+def test_two_ranks_one_gpu(fn):
+    run_ranks(fn, 2, backend="gpu", timeout=180)

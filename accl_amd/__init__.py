@@ -9,6 +9,14 @@ for hardware-free testing. See docs/DESIGN.md.
 
 
 def _load_core():
+    # Load torch FIRST so its bundled HIP/ROCr libraries own the runtime and
+    # _core's libamdhip64.so.7 / libhsa-runtime64.so.1 needs resolve to them.
+    # With /opt/rocm's copies loaded first instead, two ROCr instances end up
+    # in the process and the second one to initialize sees zero GPUs.
+    try:
+        import torch  # noqa: F401
+    except Exception:
+        pass
     try:
         from . import _core
         return _core

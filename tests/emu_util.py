@@ -36,7 +36,10 @@ def _worker(fn, rank, nranks, job, opts, q, backend):
 
 def run_ranks(fn, nranks, opts=None, timeout=120, backend="emu"):
     """Run `fn(accl, rank, nranks)` on every rank; raise on any failure."""
-    ctx = mp.get_context("fork")
+    # fork is fine for the CPU emulator; GPU children must be spawned (a
+    # forked child inherits the parent's initialized HIP runtime, which does
+    # not survive fork)
+    ctx = mp.get_context("spawn" if backend == "gpu" else "fork")
     q = ctx.Queue()
     job = fresh_job()
     ps = [ctx.Process(target=_worker, args=(fn, r, nranks, job, opts, q, backend))

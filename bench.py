@@ -69,7 +69,9 @@ def main():
         import torch
         t = a.tensor(src)
         t.copy_(torch.randn(count, device=t.device))
-        torch.cuda.synchronize()
+        # NOT torch.cuda.synchronize(): hipDeviceSynchronize would wait on
+        # the persistent engine kernel's stream (which never ends).
+        torch.cuda.current_stream().synchronize()
     else:
         tmp = a.create_buffer(min(count, 1 << 20), A.DataType.float32)
         tmp.write(np.random.default_rng(0).standard_normal(

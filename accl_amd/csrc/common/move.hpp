@@ -38,7 +38,7 @@ struct alignas(64) MoveState {
 static_assert(sizeof(MoveState) == 64, "");
 
 constexpr u32 MOVE_RING = 64;          // in-flight move slots per engine
-constexpr u64 MOVE_TILE_BYTES = 1u << 18;  // 256 KiB per mover tile
+constexpr u64 MOVE_TILE_BYTES = 1u << 17;  // 128 KiB per mover tile
 
 ACCL_HD inline u64 move_bytes(const MoveDesc& m) {
   return m.count * dtype_size(DataType(m.dst_dt));

@@ -175,7 +175,12 @@ struct Cclo {
       MoveDesc m{};
       switch (f.kind) {
         case FLOW_LOCAL: {
-          const u64 seg_cap = (4u << 20) / dtype_size(DataType(f.ddt));
+          // Local moves need no wire segmentation: one move exposes every
+          // tile to the whole mover fleet at once (max data-plane
+          // parallelism). Gated flows still chase the gate in segments.
+          const u64 seg_cap = f.gate
+              ? (8u << 20) / dtype_size(DataType(f.ddt))
+              : f.count;
           u64 n = min64(avail - f.submitted, seg_cap);
           m.dst = (u64)(f.dst + f.submitted * dtype_size(DataType(f.ddt)));
           m.dst_dt = f.ddt;

@@ -100,7 +100,9 @@ __device__ __forceinline__ bool aligned16(const void* p) {
   return (u64(p) & 15) == 0;
 }
 
-// pure same-dtype copy, 16B vectorized when aligned
+// pure same-dtype copy, 16B vectorized when aligned. 8-deep software
+// pipeline: 8 loads in flight per lane (8 KiB per wave) so HBM (~1 us over
+// xGMI) latency is covered by MLP, not occupancy alone.
 __device__ void tile_copy(const MoveDesc& m, u64 lo, u64 hi, int lane) {
   u32 esz = dtype_size(DataType(m.dst_dt));
   u64 bytes = (hi - lo) * esz;
@@ -110,12 +112,30 @@ __device__ void tile_copy(const MoveDesc& m, u64 lo, u64 hi, int lane) {
     const uint4* s4 = (const uint4*)s;
     uint4* d4 = (uint4*)d;
     u64 n = bytes / 16;
-    for (u64 i = lane; i < n; i += 64) d4[i] = s4[i];
+    u64 i = lane;
+    for (; i + 7 * 64 < n; i += 8 * 64) {
+      uint4 v0 = s4[i], v1 = s4[i + 64], v2 = s4[i + 2 * 64], v3 = s4[i + 3 * 64];
+      uint4 v4 = s4[i + 4 * 64], v5 = s4[i + 5 * 64], v6 = s4[i + 6 * 64],
+            v7 = s4[i + 7 * 64];
+      d4[i] = v0; d4[i + 64] = v1; d4[i + 2 * 64] = v2; d4[i + 3 * 64] = v3;
+      d4[i + 4 * 64] = v4; d4[i + 5 * 64] = v5; d4[i + 6 * 64] = v6;
+      d4[i + 7 * 64] = v7;
+    }
+    for (; i < n; i += 64) d4[i] = s4[i];
   } else if ((u64(s) & 3) == 0 && (u64(d) & 3) == 0 && (bytes & 3) == 0) {
     const u32* s1 = (const u32*)s;
     u32* d1 = (u32*)d;
     u64 n = bytes / 4;
-    for (u64 i = lane; i < n; i += 64) d1[i] = s1[i];
+    u64 i = lane;
+    for (; i + 7 * 64 < n; i += 8 * 64) {
+      u32 v0 = s1[i], v1 = s1[i + 64], v2 = s1[i + 2 * 64], v3 = s1[i + 3 * 64];
+      u32 v4 = s1[i + 4 * 64], v5 = s1[i + 5 * 64], v6 = s1[i + 6 * 64],
+          v7 = s1[i + 7 * 64];
+      d1[i] = v0; d1[i + 64] = v1; d1[i + 2 * 64] = v2; d1[i + 3 * 64] = v3;
+      d1[i + 4 * 64] = v4; d1[i + 5 * 64] = v5; d1[i + 6 * 64] = v6;
+      d1[i + 7 * 64] = v7;
+    }
+    for (; i < n; i += 64) d1[i] = s1[i];
   } else {
     for (u64 i = lane; i < bytes; i += 64) d[i] = s[i];
   }
@@ -136,11 +156,22 @@ __device__ void tile_reduce_f32(const MoveDesc& m, u64 lo, u64 hi, int lane) {
     const float4* b = (const float4*)s1;
     float4* o = (float4*)d;
     u64 n4 = n / 4;
-    for (u64 i = lane; i < n4; i += 64) {
-      float4 x = a[i], y = b[i];
-      o[i] = make_float4(OP<float>::apply(x.x, y.x), OP<float>::apply(x.y, y.y),
-                         OP<float>::apply(x.z, y.z), OP<float>::apply(x.w, y.w));
+    u64 i = lane;
+    // 4-deep pipeline x 2 operand streams = 8 loads in flight per lane
+    for (; i + 3 * 64 < n4; i += 4 * 64) {
+      float4 x0 = a[i], x1 = a[i + 64], x2 = a[i + 2 * 64], x3 = a[i + 3 * 64];
+      float4 y0 = b[i], y1 = b[i + 64], y2 = b[i + 2 * 64], y3 = b[i + 3 * 64];
+#define ACCL_R4(x, y) make_float4(OP<float>::apply(x.x, y.x), \
+    OP<float>::apply(x.y, y.y), OP<float>::apply(x.z, y.z), \
+    OP<float>::apply(x.w, y.w))
+      o[i] = ACCL_R4(x0, y0); o[i + 64] = ACCL_R4(x1, y1);
+      o[i + 2 * 64] = ACCL_R4(x2, y2); o[i + 3 * 64] = ACCL_R4(x3, y3);
     }
+    for (; i < n4; i += 64) {
+      float4 x = a[i], y = b[i];
+      o[i] = ACCL_R4(x, y);
+    }
+#undef ACCL_R4
     return;
   }
   for (u64 i = lane; i < n; i += 64) {
@@ -205,6 +236,9 @@ __device__ void run_tile(const MoveDesc& m, u32 t, int lane) {
 // ------------------------------------------------------------- mover main
 __device__ void mover_main(GpuEngineState* S) {
   const int lane = int(threadIdx.x) & 63;
+  // global mover-wave index (wave 0 of block 0 is the scheduler)
+  const u32 gw = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64) - 1;
+  const u32 nwaves = gridDim.x * (blockDim.x / 64) - 1;
   MoveDesc* ring = S->mover.ring;
   MoveState* st = S->mover.st;
   u64* head = S->mover.head;
@@ -226,6 +260,16 @@ __device__ void mover_main(GpuEngineState* S) {
       }
       MoveState& ms = st[slot];
       u32 total = ms.tiles_total;
+      // Per-move rotated eligibility window: only ~tiles_total waves contend
+      // on tiles_claimed (bounds small-move atomic traffic with a 1000-wave
+      // fleet), rotated by move index so concurrent small moves land on
+      // disjoint wave subsets.
+      if (((gw + nwaves - u32(mi * 37) % nwaves) % nwaves) >= total) {
+        if (mi == cursor &&
+            __hip_atomic_load(&ms.tiles_claimed, __ATOMIC_RELAXED, AGENT) >= total)
+          cursor++;
+        continue;
+      }
       if (__hip_atomic_load(&ms.tiles_claimed, __ATOMIC_RELAXED, AGENT) >= total) {
         if (mi == cursor) cursor++;
         continue;

@@ -31,7 +31,7 @@ class GpuDevice : public Backend {
 
  private:
   int dev_ = 0;
-  int engine_wgs_ = 16;
+  int engine_wgs_ = 256;  // 1 WG per CU: full-fleet data plane, residency by size
   u64 arena_bytes_ = 0;
   ArenaLayout layout_{};
   bool fine_grained_ = true;

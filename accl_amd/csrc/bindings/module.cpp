@@ -146,6 +146,11 @@ PYBIND11_MODULE(_core, m) {
            },
            py::call_guard<py::gil_scoped_release>())
       .def("deinit", &ACCL::deinit, py::call_guard<py::gil_scoped_release>())
+      .def("debug_wave_tiles",
+           [](ACCL& a) {
+             auto* g = dynamic_cast<GpuDevice*>(a.backend());
+             return g ? g->debug_wave_tiles() : std::vector<u32>{};
+           })
       .def("create_buffer",
            [](ACCL& a, u64 count, DataType dt, bool device_only) {
              return device_only ? a.create_buffer_device(count, dt)

@@ -161,9 +161,9 @@ __device__ void tile_reduce_f32(const MoveDesc& m, u64 lo, u64 hi, int lane) {
     for (; i + 3 * 64 < n4; i += 4 * 64) {
       float4 x0 = a[i], x1 = a[i + 64], x2 = a[i + 2 * 64], x3 = a[i + 3 * 64];
       float4 y0 = b[i], y1 = b[i + 64], y2 = b[i + 2 * 64], y3 = b[i + 3 * 64];
-#define ACCL_R4(x, y) make_float4(OP<float>::apply(x.x, y.x), \
-    OP<float>::apply(x.y, y.y), OP<float>::apply(x.z, y.z), \
-    OP<float>::apply(x.w, y.w))
+#define ACCL_R4(a_, b_) make_float4(OP<float>::apply(a_.x, b_.x), \
+    OP<float>::apply(a_.y, b_.y), OP<float>::apply(a_.z, b_.z), \
+    OP<float>::apply(a_.w, b_.w))
       o[i] = ACCL_R4(x0, y0); o[i + 64] = ACCL_R4(x1, y1);
       o[i + 2 * 64] = ACCL_R4(x2, y2); o[i + 3 * 64] = ACCL_R4(x3, y3);
     }
@@ -283,6 +283,7 @@ __device__ void mover_main(GpuEngineState* S) {
       // epoch owns the slot now; re-read them.
       u32 cur_total = ms.tiles_total;
       if (t < cur_total) {
+        if (lane == 0) S->wave_tiles[gw & 4095] += 1;
         run_tile(ring[slot], t, lane);
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");

@@ -40,6 +40,8 @@ struct GpuEngineState {
   RetEntry* rets;
   CtrlPage* ctrl;
   CommView* comm_mirror;
+  // debug: tiles executed per mover wave (plain per-wave stores)
+  u32 wave_tiles[4096];
 };
 
 // launches the persistent engine kernel (defined in engine.hip)

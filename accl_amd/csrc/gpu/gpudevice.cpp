@@ -176,6 +176,15 @@ void GpuDevice::shutdown() {
   hip_check(e, "engine shutdown");
 }
 
+std::vector<u32> GpuDevice::debug_wave_tiles() {
+  std::vector<u32> v(4096);
+  hip_check(hipMemcpy(v.data(),
+                      (char*)state_dev_ + offsetof(GpuEngineState, wave_tiles),
+                      sizeof(u32) * v.size(), hipMemcpyDeviceToHost),
+            "read wave_tiles");
+  return v;
+}
+
 void GpuDevice::write_arena(u64 off, const void* src, u64 bytes) {
   hip_check(hipMemcpy(arena_base_ + off, src, bytes, hipMemcpyHostToDevice),
             "write_arena");

@@ -28,6 +28,8 @@ class GpuDevice : public Backend {
 
   int device_index() const { return dev_; }
   u64 arena_bytes() const { return arena_bytes_; }
+  // debug: per-wave executed-tile counters from the engine state
+  std::vector<u32> debug_wave_tiles();
 
  private:
   int dev_ = 0;

@@ -151,6 +151,11 @@ PYBIND11_MODULE(_core, m) {
              auto* g = dynamic_cast<GpuDevice*>(a.backend());
              return g ? g->debug_wave_tiles() : std::vector<u32>{};
            })
+      .def("debug_timeline",
+           [](ACCL& a) {
+             auto* g = dynamic_cast<GpuDevice*>(a.backend());
+             return g ? g->debug_timeline() : std::vector<u64>{};
+           })
       .def("create_buffer",
            [](ACCL& a, u64 count, DataType dt, bool device_only) {
              return device_only ? a.create_buffer_device(count, dt)

@@ -44,6 +44,7 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  if (dbg) dbg[0] = wallclock();
   __hip_atomic_store(head, h + 1, __ATOMIC_RELAXED, AGENT);
   head_cache = h + 1;
   return u32(h);
@@ -243,13 +244,18 @@ __device__ void mover_main(GpuEngineState* S) {
   MoveState* st = S->mover.st;
   u64* head = S->mover.head;
   u64 cursor = 0;
+  u32 idle = 0;
   for (;;) {
     u64 h = __hip_atomic_load(head, __ATOMIC_RELAXED, AGENT);
     if (cursor == h) {
       if (__hip_atomic_load(S->mover.stop, __ATOMIC_RELAXED, AGENT)) return;
-      __builtin_amdgcn_s_sleep(8);
+      // exponential idle backoff: cheap wake-up when busy, low L2/issue
+      // pressure on a quiet engine (co-resident compute kernels, config 5)
+      idle = idle < 240 ? idle + 8 : 240;
+      __builtin_amdgcn_s_sleep(8 + (idle >> 2));
       continue;
     }
+    idle = 0;
     __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     bool did = false;
     for (u64 mi = cursor; mi < h; ++mi) {
@@ -283,13 +289,18 @@ __device__ void mover_main(GpuEngineState* S) {
       // epoch owns the slot now; re-read them.
       u32 cur_total = ms.tiles_total;
       if (t < cur_total) {
-        if (lane == 0) S->wave_tiles[gw & 4095] += 1;
+        if (lane == 0) {
+          S->wave_tiles[gw & 4095] += 1;
+          if (t == 0) S->dbg[1] = wallclock();
+        }
         run_tile(ring[slot], t, lane);
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-        if (lane == 0)
+        if (lane == 0) {
           __hip_atomic_fetch_add(&ms.tiles_done, 1u, __ATOMIC_RELEASE, AGENT);
+          if (t == 0) S->dbg[2] = wallclock();
+        }
         did = true;
         break;  // rescan from cursor for freshest work
       }
@@ -307,6 +318,7 @@ __device__ void scheduler_main(GpuEngineState* S) {
   S->mover.st = S->mst;
   S->mover.head = &S->mq_head;
   S->mover.stop = &S->stop;
+  S->mover.dbg = S->dbg;
   CtrlPage* ctrl = S->ctrl;
   st_sys(&ctrl->engine_up, 1);
   u64 consumed = 0, cached_gen = 0, beat = 0;
@@ -347,7 +359,7 @@ __device__ void scheduler_main(GpuEngineState* S) {
   __hip_atomic_store(S->mover.stop, 1u, __ATOMIC_RELEASE, AGENT);
 }
 
-__global__ void __launch_bounds__(256, 1) accl_engine_kernel(GpuEngineState* S) {
+__global__ void __launch_bounds__(1024, 1) accl_engine_kernel(GpuEngineState* S) {
   if (blockIdx.x == 0 && threadIdx.x < 64) {
     if (threadIdx.x == 0) scheduler_main(S);
     return;  // lanes 1..63 of the scheduler wave idle under the exec mask
@@ -356,7 +368,7 @@ __global__ void __launch_bounds__(256, 1) accl_engine_kernel(GpuEngineState* S) 
 }
 
 void gpu_engine_launch(GpuEngineState* state_dev, int n_wgs, void* hip_stream) {
-  hipLaunchKernelGGL(accl_engine_kernel, dim3(n_wgs), dim3(256), 0,
+  hipLaunchKernelGGL(accl_engine_kernel, dim3(n_wgs), dim3(1024), 0,
                      (hipStream_t)hip_stream, state_dev);
 }
 

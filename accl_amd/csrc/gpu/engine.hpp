@@ -20,6 +20,7 @@ struct GpuMover {
   u64* head;               // device: published move count (scheduler writes)
   u64 head_cache;          // scheduler-private mirror
   u32* stop;               // device: scheduler tells movers to exit
+  u64* dbg;                // debug timeline (GpuEngineState::dbg)
 
 #if defined(__HIPCC__)
   __device__ u32 submit(const MoveDesc& m);
@@ -42,6 +43,9 @@ struct GpuEngineState {
   CommView* comm_mirror;
   // debug: tiles executed per mover wave (plain per-wave stores)
   u32 wave_tiles[4096];
+  // debug timeline of the most recent move: [0]=submit, [1]=first claim,
+  // [2]=first tile done, [3]=poll observed complete (wallclock ticks)
+  u64 dbg[16];
 };
 
 // launches the persistent engine kernel (defined in engine.hip)

@@ -134,6 +134,7 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
   st->mover.st = dstate->mst;
   st->mover.head = &dstate->mq_head;
   st->mover.stop = &dstate->stop;
+  st->mover.dbg = dstate->dbg;
   // pinned pointers as seen by the device
   RingPage* rp = (RingPage*)ring_pinned_;
   void* dev_ptr = nullptr;
@@ -174,6 +175,15 @@ void GpuDevice::shutdown() {
   hipError_t e = hipStreamSynchronize((hipStream_t)stream_);
   launched_ = false;
   hip_check(e, "engine shutdown");
+}
+
+std::vector<u64> GpuDevice::debug_timeline() {
+  std::vector<u64> v(16);
+  hip_check(hipMemcpy(v.data(),
+                      (char*)state_dev_ + offsetof(GpuEngineState, dbg),
+                      sizeof(u64) * v.size(), hipMemcpyDeviceToHost),
+            "read dbg");
+  return v;
 }
 
 std::vector<u32> GpuDevice::debug_wave_tiles() {

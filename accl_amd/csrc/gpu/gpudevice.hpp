@@ -30,6 +30,7 @@ class GpuDevice : public Backend {
   u64 arena_bytes() const { return arena_bytes_; }
   // debug: per-wave executed-tile counters from the engine state
   std::vector<u32> debug_wave_tiles();
+  std::vector<u64> debug_timeline();
 
  private:
   int dev_ = 0;

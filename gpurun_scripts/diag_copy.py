@@ -61,4 +61,7 @@ for _ in range(50):
     ts.append(time.perf_counter() - t0)
 print(f"accl copy 4KB: host {min(ts)*1e6:.1f} us, device {r.duration_us():.1f} us",
       flush=True)
+tl = a._a.debug_timeline()
+print(f"timeline ticks(10ns): submit->claim {tl[1]-tl[0]} "
+      f"claim->tile0done {tl[2]-tl[1]}", flush=True)
 a.close()

@@ -22,6 +22,10 @@ def fresh_job():
 
 def _worker(fn, rank, nranks, job, opts, q, backend):
     try:
+        if backend == "gpu":
+            # N engines share one test GPU: shrink each grid so all stay
+            # resident together (full-size grids could not co-schedule)
+            os.environ.setdefault("ACCL_ENGINE_WGS", str(max(16, 128 // nranks)))
         import accl_amd as A
         a = A.ACCL(nranks=nranks, rank=rank, backend=backend, job=job, **(
             {"opts": opts} if opts else {}))

@@ -252,7 +252,9 @@ __device__ void mover_main(GpuEngineState* S) {
       // exponential idle backoff: cheap wake-up when busy, low L2/issue
       // pressure on a quiet engine (co-resident compute kernels, config 5)
       idle = idle < 240 ? idle + 8 : 240;
-      __builtin_amdgcn_s_sleep(8 + (idle >> 2));
+      if (idle < 64) __builtin_amdgcn_s_sleep(8);
+      else if (idle < 160) __builtin_amdgcn_s_sleep(32);
+      else __builtin_amdgcn_s_sleep(64);
       continue;
     }
     idle = 0;

@@ -237,9 +237,8 @@ __device__ void run_tile(const MoveDesc& m, u32 t, int lane) {
 // ------------------------------------------------------------- mover main
 __device__ void mover_main(GpuEngineState* S) {
   const int lane = int(threadIdx.x) & 63;
-  // global mover-wave index (wave 0 of block 0 is the scheduler)
-  const u32 gw = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64) - 1;
-  const u32 nwaves = gridDim.x * (blockDim.x / 64) - 1;
+  const u32 gw = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+  const u32 nwaves = gridDim.x * (blockDim.x / 64);
   MoveDesc* ring = S->mover.ring;
   MoveState* st = S->mover.st;
   u64* head = S->mover.head;
@@ -361,17 +360,23 @@ __device__ void scheduler_main(GpuEngineState* S) {
   __hip_atomic_store(S->mover.stop, 1u, __ATOMIC_RELEASE, AGENT);
 }
 
-__global__ void __launch_bounds__(1024, 1) accl_engine_kernel(GpuEngineState* S) {
-  if (blockIdx.x == 0 && threadIdx.x < 64) {
-    if (threadIdx.x == 0) scheduler_main(S);
-    return;  // lanes 1..63 of the scheduler wave idle under the exec mask
-  }
+// Two kernels: the scheduler (1 wave, register-heavy control loop — its own
+// CU) and the mover fleet (lean copy/reduce waves, 8 of the 16 wave slots
+// per CU so co-resident compute kernels always have room — config 5).
+__global__ void __launch_bounds__(64, 1) accl_scheduler_kernel(GpuEngineState* S) {
+  if (threadIdx.x == 0) scheduler_main(S);
+}
+
+__global__ void __launch_bounds__(512, 2) accl_mover_kernel(GpuEngineState* S) {
   mover_main(S);
 }
 
-void gpu_engine_launch(GpuEngineState* state_dev, int n_wgs, void* hip_stream) {
-  hipLaunchKernelGGL(accl_engine_kernel, dim3(n_wgs), dim3(1024), 0,
-                     (hipStream_t)hip_stream, state_dev);
+void gpu_engine_launch(GpuEngineState* state_dev, int n_wgs, void* sched_stream,
+                       void* mover_stream) {
+  hipLaunchKernelGGL(accl_mover_kernel, dim3(n_wgs), dim3(512), 0,
+                     (hipStream_t)mover_stream, state_dev);
+  hipLaunchKernelGGL(accl_scheduler_kernel, dim3(1), dim3(64), 0,
+                     (hipStream_t)sched_stream, state_dev);
 }
 
 }  // namespace accl

@@ -48,7 +48,9 @@ struct GpuEngineState {
   u64 dbg[16];
 };
 
-// launches the persistent engine kernel (defined in engine.hip)
-void gpu_engine_launch(GpuEngineState* state_dev, int n_wgs, void* hip_stream);
+// launches the persistent engine kernels (defined in engine.hip):
+// mover fleet on mover_stream, scheduler on sched_stream
+void gpu_engine_launch(GpuEngineState* state_dev, int n_wgs, void* sched_stream,
+                       void* mover_stream);
 
 }  // namespace accl

@@ -83,6 +83,8 @@ GpuDevice::GpuDevice(u32 nranks, u32 rank, int device_index,
   hipStream_t s;
   hip_check(hipStreamCreateWithFlags(&s, hipStreamNonBlocking), "stream");
   stream_ = s;
+  hip_check(hipStreamCreateWithFlags(&s, hipStreamNonBlocking), "mover stream");
+  mover_stream_ = s;
 }
 
 GpuDevice::~GpuDevice() {
@@ -93,6 +95,7 @@ GpuDevice::~GpuDevice() {
   if (arena_base_) (void)hipFree(arena_base_);
   if (ring_pinned_) (void)hipHostFree(ring_pinned_);
   if (stream_) (void)hipStreamDestroy((hipStream_t)stream_);
+  if (mover_stream_) (void)hipStreamDestroy((hipStream_t)mover_stream_);
 }
 
 std::vector<char> GpuDevice::local_blob() {
@@ -158,7 +161,9 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
   __atomic_store_n((u64*)&ring_->ctrl.ncomms, 1, __ATOMIC_RELEASE);
   __atomic_store_n((u64*)&ring_->ctrl.comm_gen, 1, __ATOMIC_RELEASE);
 
-  gpu_engine_launch((GpuEngineState*)state_dev_, engine_wgs_, stream_);
+  gpu_engine_launch((GpuEngineState*)state_dev_, engine_wgs_, stream_,
+                    mover_stream_);
+  hip_check(hipGetLastError(), "engine kernel launch");
   launched_ = true;
   // wait for the engine to come up
   u64 t0 = wallclock_host_ns();
@@ -173,8 +178,10 @@ void GpuDevice::shutdown() {
   if (!launched_) return;
   __atomic_store_n((u64*)&ring_->ctrl.shutdown, 1, __ATOMIC_RELEASE);
   hipError_t e = hipStreamSynchronize((hipStream_t)stream_);
+  hipError_t e2 = hipStreamSynchronize((hipStream_t)mover_stream_);
   launched_ = false;
   hip_check(e, "engine shutdown");
+  hip_check(e2, "mover shutdown");
 }
 
 std::vector<u64> GpuDevice::debug_timeline() {

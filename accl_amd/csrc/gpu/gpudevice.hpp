@@ -38,6 +38,7 @@ class GpuDevice : public Backend {
   u64 arena_bytes_ = 0;
   ArenaLayout layout_{};
   bool fine_grained_ = true;
+  void* mover_stream_ = nullptr;
   char* peer_base_[MAX_RANKS] = {};
   void* ring_pinned_ = nullptr;     // RingPage, hipHostMalloc
   void* state_dev_ = nullptr;       // GpuEngineState, hipMalloc

@@ -46,6 +46,8 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   if (dbg) dbg[0] = wallclock();
   __hip_atomic_store(head, h + 1, __ATOMIC_RELAXED, AGENT);
+  for (u32 i = 0; i < DOORBELL_REPS; ++i)
+    __hip_atomic_store(&rep[i][0], h + 1, __ATOMIC_RELAXED, AGENT);
   head_cache = h + 1;
   return u32(h);
 }
@@ -235,19 +237,25 @@ __device__ void run_tile(const MoveDesc& m, u32 t, int lane) {
 }
 
 // ------------------------------------------------------------- mover main
+// Static tile partitioning: move mi's tile t belongs to wave
+// (t + rot(mi)) % nwaves == gw — no claim atomics, no shared hot line.
+// Completion = ONE tiles_done add per participating wave. The doorbell is
+// replicated (DOORBELL_REPS lines, ~nwaves/64 pollers each) so noticing a
+// new move costs ~L2-hit latency, not a fleet-wide serialized cacheline.
 __device__ void mover_main(GpuEngineState* S) {
   const int lane = int(threadIdx.x) & 63;
   const u32 gw = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
   const u32 nwaves = gridDim.x * (blockDim.x / 64);
   MoveDesc* ring = S->mover.ring;
   MoveState* st = S->mover.st;
-  u64* head = S->mover.head;
+  volatile u64* my_rep = &S->head_rep[gw % DOORBELL_REPS][0];
   u64 cursor = 0;
   u32 idle = 0;
   for (;;) {
-    u64 h = __hip_atomic_load(head, __ATOMIC_RELAXED, AGENT);
+    u64 h = __hip_atomic_load((const u64*)&my_rep[0], __ATOMIC_RELAXED, AGENT);
     if (cursor == h) {
-      if (__hip_atomic_load(S->mover.stop, __ATOMIC_RELAXED, AGENT)) return;
+      if (__hip_atomic_load((const u64*)&my_rep[1], __ATOMIC_RELAXED, AGENT))
+        return;
       // exponential idle backoff: cheap wake-up when busy, low L2/issue
       // pressure on a quiet engine (co-resident compute kernels, config 5)
       idle = idle < 240 ? idle + 8 : 240;
@@ -258,56 +266,30 @@ __device__ void mover_main(GpuEngineState* S) {
     }
     idle = 0;
     __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-    bool did = false;
-    for (u64 mi = cursor; mi < h; ++mi) {
-      u32 slot = u32(mi % MOVE_RING);
-      if (u32(ring[slot].epoch) != u32(mi + 1)) {
-        if (mi == cursor) cursor++;
-        continue;  // recycled past us
-      }
-      MoveState& ms = st[slot];
-      u32 total = ms.tiles_total;
-      // Per-move rotated eligibility window: only ~tiles_total waves contend
-      // on tiles_claimed (bounds small-move atomic traffic with a 1000-wave
-      // fleet), rotated by move index so concurrent small moves land on
-      // disjoint wave subsets.
-      if (((gw + nwaves - u32(mi * 37) % nwaves) % nwaves) >= total) {
-        if (mi == cursor &&
-            __hip_atomic_load(&ms.tiles_claimed, __ATOMIC_RELAXED, AGENT) >= total)
-          cursor++;
-        continue;
-      }
-      if (__hip_atomic_load(&ms.tiles_claimed, __ATOMIC_RELAXED, AGENT) >= total) {
-        if (mi == cursor) cursor++;
-        continue;
-      }
-      u32 t = 0xFFFFFFFFu;
-      if (lane == 0)
-        t = __hip_atomic_fetch_add(&ms.tiles_claimed, 1u, __ATOMIC_RELAXED, AGENT);
-      t = u32(__shfl(int(t), 0, 64));
-      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-      // after the claim+acquire, desc/epoch/total are coherent for whichever
-      // epoch owns the slot now; re-read them.
-      u32 cur_total = ms.tiles_total;
-      if (t < cur_total) {
-        if (lane == 0) {
-          S->wave_tiles[gw & 4095] += 1;
-          if (t == 0) S->dbg[1] = wallclock();
-        }
-        run_tile(ring[slot], t, lane);
+    while (cursor < h) {
+      u32 slot = u32(cursor % MOVE_RING);
+      const MoveDesc& m = ring[slot];
+      // desc publish happens-before the doorbell store; sanity-check epoch
+      if (u32(m.epoch) != u32(cursor + 1)) break;  // not visible yet: retry
+      u32 total = move_tiles(m);
+      u32 rot = u32(cursor * 37) % nwaves;
+      u32 first = (gw + nwaves - rot) % nwaves;
+      if (first < total) {
+        if (lane == 0 && first == 0) S->dbg[1] = wallclock();
+        u32 cnt = 0;
+        for (u32 t = first; t < total; t += nwaves, ++cnt) run_tile(m, t, lane);
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         if (lane == 0) {
-          __hip_atomic_fetch_add(&ms.tiles_done, 1u, __ATOMIC_RELEASE, AGENT);
-          if (t == 0) S->dbg[2] = wallclock();
+          S->wave_tiles[gw & 4095] += cnt;
+          __hip_atomic_fetch_add(&st[slot].tiles_done, cnt, __ATOMIC_RELEASE,
+                                 AGENT);
+          if (first == 0) S->dbg[2] = wallclock();
         }
-        did = true;
-        break;  // rescan from cursor for freshest work
       }
-      if (mi == cursor) cursor++;
+      cursor++;
     }
-    if (!did) __builtin_amdgcn_s_sleep(2);
   }
 }
 
@@ -320,6 +302,7 @@ __device__ void scheduler_main(GpuEngineState* S) {
   S->mover.head = &S->mq_head;
   S->mover.stop = &S->stop;
   S->mover.dbg = S->dbg;
+  S->mover.rep = S->head_rep;
   CtrlPage* ctrl = S->ctrl;
   st_sys(&ctrl->engine_up, 1);
   u64 consumed = 0, cached_gen = 0, beat = 0;
@@ -358,6 +341,8 @@ __device__ void scheduler_main(GpuEngineState* S) {
   }
   // tell movers to exit, then leave
   __hip_atomic_store(S->mover.stop, 1u, __ATOMIC_RELEASE, AGENT);
+  for (u32 i = 0; i < DOORBELL_REPS; ++i)
+    __hip_atomic_store(&S->head_rep[i][1], 1ull, __ATOMIC_RELEASE, AGENT);
 }
 
 // Two kernels: the scheduler (1 wave, register-heavy control loop — its own

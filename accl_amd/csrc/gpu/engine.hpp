@@ -21,12 +21,17 @@ struct GpuMover {
   u64 head_cache;          // scheduler-private mirror
   u32* stop;               // device: scheduler tells movers to exit
   u64* dbg;                // debug timeline (GpuEngineState::dbg)
+  // replicated doorbell lines: [i][0]=head, [i][1]=stop. Each mover wave
+  // polls replica [gw%64] so no single cacheline serves the whole fleet.
+  u64 (*rep)[8];
 
 #if defined(__HIPCC__)
   __device__ u32 submit(const MoveDesc& m);
   __device__ bool poll(u32 token);
 #endif
 };
+
+constexpr u32 DOORBELL_REPS = 64;
 
 struct GpuEngineState {
   Cclo<GpuMover> cclo;     // trivially-copyable; host fills, device runs
@@ -36,6 +41,7 @@ struct GpuEngineState {
   u64 mq_head;
   u32 stop;
   u32 _pad;
+  alignas(64) u64 head_rep[DOORBELL_REPS][8];
   // pinned-host pointers (device-accessible):
   CallDesc* descs;
   RetEntry* rets;

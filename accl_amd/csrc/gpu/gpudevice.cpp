@@ -138,6 +138,7 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
   st->mover.head = &dstate->mq_head;
   st->mover.stop = &dstate->stop;
   st->mover.dbg = dstate->dbg;
+  st->mover.rep = dstate->head_rep;
   // pinned pointers as seen by the device
   RingPage* rp = (RingPage*)ring_pinned_;
   void* dev_ptr = nullptr;

@@ -46,8 +46,12 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   if (dbg) dbg[0] = wallclock();
   __hip_atomic_store(head, h + 1, __ATOMIC_RELAXED, AGENT);
+  // doorbell word packs (head, latest move's tile count): a wave that is
+  // caught up decides ownership without touching the descriptor line, so a
+  // 1-tile move wakes 1 wave's worth of desc traffic, not the whole fleet's
+  u64 packed = ((h + 1) << 24) | (s.tiles_total & 0xFFFFFFu);
   for (u32 i = 0; i < DOORBELL_REPS; ++i)
-    __hip_atomic_store(&rep[i][0], h + 1, __ATOMIC_RELAXED, AGENT);
+    __hip_atomic_store(&rep[i][0], packed, __ATOMIC_RELAXED, AGENT);
   head_cache = h + 1;
   return u32(h);
 }
@@ -252,7 +256,9 @@ __device__ void mover_main(GpuEngineState* S) {
   u64 cursor = 0;
   u32 idle = 0;
   for (;;) {
-    u64 h = __hip_atomic_load((const u64*)&my_rep[0], __ATOMIC_RELAXED, AGENT);
+    u64 packed = __hip_atomic_load((const u64*)&my_rep[0], __ATOMIC_RELAXED, AGENT);
+    u64 h = packed >> 24;
+    u32 latest_total = u32(packed & 0xFFFFFFu);
     if (cursor == h) {
       if (__hip_atomic_load((const u64*)&my_rep[1], __ATOMIC_RELAXED, AGENT))
         return;
@@ -268,12 +274,15 @@ __device__ void mover_main(GpuEngineState* S) {
     __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     while (cursor < h) {
       u32 slot = u32(cursor % MOVE_RING);
+      u32 rot = u32(cursor * 37) % nwaves;
+      u32 first = (gw + nwaves - rot) % nwaves;
+      // when caught up, the doorbell word tells us our share without a
+      // descriptor read (the common small-move fast path)
+      if (cursor == h - 1 && first >= latest_total) { cursor++; continue; }
       const MoveDesc& m = ring[slot];
       // desc publish happens-before the doorbell store; sanity-check epoch
       if (u32(m.epoch) != u32(cursor + 1)) break;  // not visible yet: retry
       u32 total = move_tiles(m);
-      u32 rot = u32(cursor * 37) % nwaves;
-      u32 first = (gw + nwaves - rot) % nwaves;
       if (first < total) {
         if (lane == 0 && first == 0) S->dbg[1] = wallclock();
         u32 cnt = 0;
@@ -352,13 +361,13 @@ __global__ void __launch_bounds__(64, 1) accl_scheduler_kernel(GpuEngineState* S
   if (threadIdx.x == 0) scheduler_main(S);
 }
 
-__global__ void __launch_bounds__(512, 2) accl_mover_kernel(GpuEngineState* S) {
+__global__ void __launch_bounds__(256, 2) accl_mover_kernel(GpuEngineState* S) {
   mover_main(S);
 }
 
 void gpu_engine_launch(GpuEngineState* state_dev, int n_wgs, void* sched_stream,
                        void* mover_stream) {
-  hipLaunchKernelGGL(accl_mover_kernel, dim3(n_wgs), dim3(512), 0,
+  hipLaunchKernelGGL(accl_mover_kernel, dim3(n_wgs), dim3(256), 0,
                      (hipStream_t)mover_stream, state_dev);
   hipLaunchKernelGGL(accl_scheduler_kernel, dim3(1), dim3(64), 0,
                      (hipStream_t)sched_stream, state_dev);

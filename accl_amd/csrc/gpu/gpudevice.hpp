@@ -34,7 +34,7 @@ class GpuDevice : public Backend {
 
  private:
   int dev_ = 0;
-  int engine_wgs_ = 256;  // 1 WG per CU: full-fleet data plane, residency by size
+  int engine_wgs_ = 640;  // 256-thr WGs, ~10 mover waves per CU (6 slots left free)
   u64 arena_bytes_ = 0;
   ArenaLayout layout_{};
   bool fine_grained_ = true;

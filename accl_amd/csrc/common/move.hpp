@@ -23,7 +23,9 @@ struct alignas(128) MoveDesc {
   u32 func;               // ReduceFunction (ignored for nsrc==1)
   u8 src_dt[MOVE_MAX_SRC];// DataType per source
   u8 dst_dt;
-  u8 _pad[7];
+  u8 inline_done;         // GPU: executed inline by the scheduler WG's small
+                          // mover wave; fleet must skip (bookkeeping only)
+  u8 _pad[6];
   u64 epoch;              // published last (GPU queue); emulator ignores
 };
 static_assert(sizeof(MoveDesc) == 128, "");

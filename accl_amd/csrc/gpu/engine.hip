@@ -34,13 +34,27 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
       __builtin_amdgcn_s_sleep(1);
     __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
   }
+  bool small = small_mb && move_bytes(m) <= SMALL_INLINE_MAX;
   MoveDesc& d = ring[slot];
   d = m;
+  d.inline_done = small ? 1 : 0;
   d.epoch = h + 1;
   MoveState& s = st[slot];
-  s.tiles_total = move_tiles(m);
+  s.tiles_total = small ? 0 : move_tiles(m);
   __hip_atomic_store(&s.tiles_claimed, 0u, __ATOMIC_RELAXED, AGENT);
   __hip_atomic_store(&s.tiles_done, 0u, __ATOMIC_RELAXED, AGENT);
+  if (small) {
+    // hand to the sibling wave over LDS; block until its system-release
+    // completes (a sub-32KB move is ~1-2 us — cheaper than a fleet handoff)
+    SmallMb* mb = (SmallMb*)small_mb;
+    mb->d = m;
+    u64 sq = ++small_seq;
+    __hip_atomic_store(&mb->seq, sq, __ATOMIC_RELEASE,
+                       __HIP_MEMORY_SCOPE_WORKGROUP);
+    while (__hip_atomic_load(&mb->done, __ATOMIC_ACQUIRE,
+                             __HIP_MEMORY_SCOPE_WORKGROUP) != sq)
+      __builtin_amdgcn_s_sleep(1);
+  }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -128,7 +142,24 @@ __device__ void tile_copy(const MoveDesc& m, u64 lo, u64 hi, int lane) {
       d4[i + 4 * 64] = v4; d4[i + 5 * 64] = v5; d4[i + 6 * 64] = v6;
       d4[i + 7 * 64] = v7;
     }
-    for (; i < n; i += 64) d4[i] = s4[i];
+    if (i < n) {
+      // tail: clamp load indices (duplicate loads are free) and predicate
+      // stores, so all remaining loads issue together instead of a
+      // one-load-one-store dependent chain (a 4KB move IS this tail)
+      u64 last = n - 1;
+      u64 i1 = i + 64 < n ? i + 64 : last, i2 = i + 128 < n ? i + 128 : last;
+      u64 i3 = i + 192 < n ? i + 192 : last, i4 = i + 256 < n ? i + 256 : last;
+      u64 i5 = i + 320 < n ? i + 320 : last, i6 = i + 384 < n ? i + 384 : last;
+      uint4 v0 = s4[i], v1 = s4[i1], v2 = s4[i2], v3 = s4[i3];
+      uint4 v4 = s4[i4], v5 = s4[i5], v6 = s4[i6];
+      d4[i] = v0;
+      if (i + 64 < n) d4[i1] = v1;
+      if (i + 128 < n) d4[i2] = v2;
+      if (i + 192 < n) d4[i3] = v3;
+      if (i + 256 < n) d4[i4] = v4;
+      if (i + 320 < n) d4[i5] = v5;
+      if (i + 384 < n) d4[i6] = v6;
+    }
   } else if ((u64(s) & 3) == 0 && (u64(d) & 3) == 0 && (bytes & 3) == 0) {
     const u32* s1 = (const u32*)s;
     u32* d1 = (u32*)d;
@@ -174,9 +205,16 @@ __device__ void tile_reduce_f32(const MoveDesc& m, u64 lo, u64 hi, int lane) {
       o[i] = ACCL_R4(x0, y0); o[i + 64] = ACCL_R4(x1, y1);
       o[i + 2 * 64] = ACCL_R4(x2, y2); o[i + 3 * 64] = ACCL_R4(x3, y3);
     }
-    for (; i < n4; i += 64) {
-      float4 x = a[i], y = b[i];
-      o[i] = ACCL_R4(x, y);
+    if (i < n4) {  // clamped-tail (see tile_copy): all loads in flight
+      u64 last = n4 - 1;
+      u64 i1 = i + 64 < n4 ? i + 64 : last, i2 = i + 128 < n4 ? i + 128 : last;
+      u64 i3 = i + 192 < n4 ? i + 192 : last;
+      float4 x0 = a[i], x1 = a[i1], x2 = a[i2], x3 = a[i3];
+      float4 y0 = b[i], y1 = b[i1], y2 = b[i2], y3 = b[i3];
+      o[i] = ACCL_R4(x0, y0);
+      if (i + 64 < n4) o[i1] = ACCL_R4(x1, y1);
+      if (i + 128 < n4) o[i2] = ACCL_R4(x2, y2);
+      if (i + 192 < n4) o[i3] = ACCL_R4(x3, y3);
     }
 #undef ACCL_R4
     return;
@@ -282,7 +320,7 @@ __device__ void mover_main(GpuEngineState* S) {
       const MoveDesc& m = ring[slot];
       // desc publish happens-before the doorbell store; sanity-check epoch
       if (u32(m.epoch) != u32(cursor + 1)) break;  // not visible yet: retry
-      u32 total = move_tiles(m);
+      u32 total = m.inline_done ? 0 : move_tiles(m);
       if (first < total) {
         if (lane == 0 && first == 0) S->dbg[1] = wallclock();
         u32 cnt = 0;
@@ -354,11 +392,47 @@ __device__ void scheduler_main(GpuEngineState* S) {
     __hip_atomic_store(&S->head_rep[i][1], 1ull, __ATOMIC_RELEASE, AGENT);
 }
 
-// Two kernels: the scheduler (1 wave, register-heavy control loop — its own
-// CU) and the mover fleet (lean copy/reduce waves, 8 of the 16 wave slots
+// Small-mover wave: sibling wave of the scheduler, same workgroup. Executes
+// sub-32KB moves on an LDS handshake. The payload release is SYSTEM scope so
+// the scheduler's subsequent slot-header publish happens-after it.
+__device__ void small_mover(SmallMb* mb, int lane) {
+  u64 last = 0;
+  for (;;) {
+    u64 sq = __hip_atomic_load(&mb->seq, __ATOMIC_ACQUIRE,
+                               __HIP_MEMORY_SCOPE_WORKGROUP);
+    if (sq == last) {
+      if (__hip_atomic_load(&mb->quit, __ATOMIC_RELAXED,
+                            __HIP_MEMORY_SCOPE_WORKGROUP))
+        return;
+      __builtin_amdgcn_s_sleep(1);
+      continue;
+    }
+    run_tile(mb->d, 0, lane);  // a small move always fits one tile
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    if (lane == 0)
+      __hip_atomic_store(&mb->done, sq, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_WORKGROUP);
+    last = sq;
+  }
+}
+
+// Two kernels: the scheduler WG (control wave + small-mover wave — one CU)
+// and the mover fleet (lean copy/reduce waves, ~10 of the 16 wave slots
 // per CU so co-resident compute kernels always have room — config 5).
-__global__ void __launch_bounds__(64, 1) accl_scheduler_kernel(GpuEngineState* S) {
-  if (threadIdx.x == 0) scheduler_main(S);
+__global__ void __launch_bounds__(128, 1) accl_scheduler_kernel(GpuEngineState* S) {
+  __shared__ SmallMb mb;
+  if (threadIdx.x == 0) { mb.seq = 0; mb.done = 0; mb.quit = 0; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    S->mover.small_mb = (void*)&mb;
+    scheduler_main(S);
+    __hip_atomic_store(&mb.quit, 1ull, __ATOMIC_RELEASE,
+                       __HIP_MEMORY_SCOPE_WORKGROUP);
+  } else if (threadIdx.x >= 64) {
+    small_mover(&mb, int(threadIdx.x) & 63);
+  }
 }
 
 __global__ void __launch_bounds__(256, 2) accl_mover_kernel(GpuEngineState* S) {
@@ -369,7 +443,7 @@ void gpu_engine_launch(GpuEngineState* state_dev, int n_wgs, void* sched_stream,
                        void* mover_stream) {
   hipLaunchKernelGGL(accl_mover_kernel, dim3(n_wgs), dim3(256), 0,
                      (hipStream_t)mover_stream, state_dev);
-  hipLaunchKernelGGL(accl_scheduler_kernel, dim3(1), dim3(64), 0,
+  hipLaunchKernelGGL(accl_scheduler_kernel, dim3(1), dim3(128), 0,
                      (hipStream_t)sched_stream, state_dev);
 }
 

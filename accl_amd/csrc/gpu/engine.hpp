@@ -12,6 +12,18 @@
 
 namespace accl {
 
+// LDS mailbox between the scheduler wave and its sibling small-mover wave
+// (same workgroup): sub-32KB moves run on a ~100ns LDS handshake instead of
+// the fleet doorbell (which costs ~10us of wake/handoff).
+struct SmallMb {
+  u64 seq;    // published by scheduler (workgroup release)
+  u64 done;   // published by small mover after system-release of payload
+  u64 quit;
+  u64 _pad;
+  MoveDesc d;
+};
+constexpr u64 SMALL_INLINE_MAX = 32u << 10;  // bytes
+
 // Device-side mover handle. Methods (submit/poll) are device-only and live
 // in engine.hip; the POD fields are set up by the host.
 struct GpuMover {
@@ -24,6 +36,8 @@ struct GpuMover {
   // replicated doorbell lines: [i][0]=head, [i][1]=stop. Each mover wave
   // polls replica [gw%64] so no single cacheline serves the whole fleet.
   u64 (*rep)[8];
+  void* small_mb;          // LDS SmallMb (set by the scheduler kernel)
+  u64 small_seq;           // scheduler-private inline-move counter
 
 #if defined(__HIPCC__)
   __device__ u32 submit(const MoveDesc& m);

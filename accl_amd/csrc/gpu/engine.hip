@@ -46,6 +46,7 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
   if (small) {
     // hand to the sibling wave over LDS; block until its system-release
     // completes (a sub-32KB move is ~1-2 us — cheaper than a fleet handoff)
+    if (dbg) dbg[5] = wallclock();
     SmallMb* mb = (SmallMb*)small_mb;
     mb->d = m;
     u64 sq = ++small_seq;
@@ -54,6 +55,7 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
     while (__hip_atomic_load(&mb->done, __ATOMIC_ACQUIRE,
                              __HIP_MEMORY_SCOPE_WORKGROUP) != sq)
       __builtin_amdgcn_s_sleep(1);
+    if (dbg) { dbg[6] = wallclock(); dbg[7]++; }
   }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");

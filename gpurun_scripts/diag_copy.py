@@ -63,5 +63,6 @@ print(f"accl copy 4KB: host {min(ts)*1e6:.1f} us, device {r.duration_us():.1f} u
       flush=True)
 tl = a._a.debug_timeline()
 print(f"timeline ticks(10ns): submit->claim {tl[1]-tl[0]} "
-      f"claim->tile0done {tl[2]-tl[1]}", flush=True)
+      f"claim->tile0done {tl[2]-tl[1]} inline_wait {tl[6]-tl[5]} "
+      f"inline_count {tl[7]}", flush=True)
 a.close()

@@ -86,6 +86,19 @@ struct Cclo {
 
   ACCL_HD u32 me() const { return cfg.rank; }
 
+  // device-only micro-timeline (GPU: GpuMover::dbg; emulator mover: no-op)
+  template <class M>
+  ACCL_HD auto stamp_impl(M* m, int i, int) -> decltype((void)m->dbg) {
+#if defined(__HIP_DEVICE_COMPILE__)
+    m->dbg[i] = wallclock();
+#else
+    (void)m; (void)i;
+#endif
+  }
+  template <class M>
+  ACCL_HD void stamp_impl(M*, int, long) {}
+  ACCL_HD void stamp(int i) { stamp_impl(mv, i, 0); }
+
   // ---------------- tiny helpers ----------------
   ACCL_HD char* local_ptr(u64 addr, bool arena) {
     return arena ? tv.arena[me()] + addr : (char*)addr;
@@ -293,6 +306,7 @@ struct Cclo {
 
   // run a set of flows to completion (the engine inner loop)
   ACCL_HD u32 run_flows(u32 n) {
+    stamp(10);
     u64 deadline = deadline_now();
     for (;;) {
       bool any = false, alldone = true;
@@ -403,10 +417,14 @@ struct Cclo {
   // CallDesc. Returns error bits (0 = ok).
 
   ACCL_HD u32 op_copy(const CallDesc& d) {
+    stamp(8);
     u64 n = desc_count(d);
     mk_local(0, local_ptr(d.addr0, d.flags & F_SRC_ARENA), desc_dtype(d),
              local_ptr(d.addr2, d.flags & F_DST_ARENA), desc_dtype(d), n);
-    return run_flows(1);
+    stamp(9);
+    u32 e = run_flows(1);
+    stamp(11);
+    return e;
   }
 
   ACCL_HD u32 op_combine(const CallDesc& d) {

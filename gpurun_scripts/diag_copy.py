@@ -62,7 +62,8 @@ for _ in range(50):
 print(f"accl copy 4KB: host {min(ts)*1e6:.1f} us, device {r.duration_us():.1f} us",
       flush=True)
 tl = a._a.debug_timeline()
-print(f"timeline ticks(10ns): submit->claim {tl[1]-tl[0]} "
-      f"claim->tile0done {tl[2]-tl[1]} inline_wait {tl[6]-tl[5]} "
-      f"inline_count {tl[7]}", flush=True)
+print(f"timeline ticks(10ns): inline_wait {tl[6]-tl[5]} "
+      f"op->mk_local {tl[9]-tl[8]} mk->flows {tl[10]-tl[9]} "
+      f"flows->submit {tl[5]-tl[10]} submitdone->ret {tl[11]-tl[6]}",
+      flush=True)
 a.close()

@@ -319,7 +319,7 @@ struct Cclo {
         any |= flow_submit(f);
         if (err) return err;
       }
-      if (alldone) return E_OK;
+      if (alldone) { stamp(14); return E_OK; }
       if (any) deadline = deadline_now();
       else if (!wait_pred_tick(deadline)) return err;
     }

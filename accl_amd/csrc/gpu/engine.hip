@@ -69,6 +69,7 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
   for (u32 i = 0; i < DOORBELL_REPS; ++i)
     __hip_atomic_store(&rep[i][0], packed, __ATOMIC_RELAXED, AGENT);
   head_cache = h + 1;
+  if (dbg) dbg[13] = wallclock();
   return u32(h);
 }
 

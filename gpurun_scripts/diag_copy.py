@@ -64,6 +64,7 @@ print(f"accl copy 4KB: host {min(ts)*1e6:.1f} us, device {r.duration_us():.1f} u
 tl = a._a.debug_timeline()
 print(f"timeline ticks(10ns): inline_wait {tl[6]-tl[5]} "
       f"op->mk_local {tl[9]-tl[8]} mk->flows {tl[10]-tl[9]} "
-      f"flows->submit {tl[5]-tl[10]} submitdone->ret {tl[11]-tl[6]}",
+      f"flows->submit {tl[5]-tl[10]} inline->subret {tl[13]-tl[6]} "
+      f"subret->alldone {tl[14]-tl[13]} alldone->ret {tl[11]-tl[14]}",
       flush=True)
 a.close()

@@ -23,6 +23,7 @@ struct SmallMb {
   MoveDesc d;
 };
 constexpr u64 SMALL_INLINE_MAX = 32u << 10;  // bytes
+constexpr u32 INLINE_TOKEN = 0x80000000u;    // poll(): done at submit time
 
 // Device-side mover handle. Methods (submit/poll) are device-only and live
 // in engine.hip; the POD fields are set up by the host.

@@ -140,6 +140,17 @@ __device__ __forceinline__ bool aligned16(const void* p) {
   return (u64(p) & 15) == 0;
 }
 
+// Explicit global address space for the data plane: generic ("flat") pointers
+// compile to flat_load/flat_store, which tick BOTH vmcnt and lgkmcnt — and
+// lgkmcnt retires out of order, so every consumer needs s_waitcnt lgkmcnt(0),
+// serializing the whole software pipeline. addrspace(1) pointers compile to
+// global_load/global_store with in-order vmcnt-only tracking.
+#define GAS __attribute__((address_space(1)))
+// Builtin 16B vectors: HIP_vector_type ctors are not address-space-aware;
+// clang ext_vector_type works across address spaces with no ctors
+typedef u32 U4 __attribute__((ext_vector_type(4)));
+typedef float F4 __attribute__((ext_vector_type(4)));
+
 // pure same-dtype copy, 16B vectorized when aligned. 8-deep software
 // pipeline: 8 loads in flight per lane (8 KiB per wave) so HBM (~1 us over
 // xGMI) latency is covered by MLP, not occupancy alone.
@@ -149,13 +160,13 @@ __device__ void tile_copy(const MoveDesc& m, u64 lo, u64 hi, int lane) {
   const char* s = (const char*)m.src[0] + lo * esz;
   char* d = (char*)m.dst + lo * esz;
   if (aligned16(s) && aligned16(d) && (bytes & 15) == 0) {
-    const uint4* s4 = (const uint4*)s;
-    uint4* d4 = (uint4*)d;
+    GAS const U4* s4 = (GAS const U4*)s;
+    GAS U4* d4 = (GAS U4*)d;
     u64 n = bytes / 16;
     u64 i = lane;
     for (; i + 7 * 64 < n; i += 8 * 64) {
-      uint4 v0 = s4[i], v1 = s4[i + 64], v2 = s4[i + 2 * 64], v3 = s4[i + 3 * 64];
-      uint4 v4 = s4[i + 4 * 64], v5 = s4[i + 5 * 64], v6 = s4[i + 6 * 64],
+      U4 v0 = s4[i], v1 = s4[i + 64], v2 = s4[i + 2 * 64], v3 = s4[i + 3 * 64];
+      U4 v4 = s4[i + 4 * 64], v5 = s4[i + 5 * 64], v6 = s4[i + 6 * 64],
             v7 = s4[i + 7 * 64];
       d4[i] = v0; d4[i + 64] = v1; d4[i + 2 * 64] = v2; d4[i + 3 * 64] = v3;
       d4[i + 4 * 64] = v4; d4[i + 5 * 64] = v5; d4[i + 6 * 64] = v6;
@@ -169,8 +180,8 @@ __device__ void tile_copy(const MoveDesc& m, u64 lo, u64 hi, int lane) {
       u64 i1 = i + 64 < n ? i + 64 : last, i2 = i + 128 < n ? i + 128 : last;
       u64 i3 = i + 192 < n ? i + 192 : last, i4 = i + 256 < n ? i + 256 : last;
       u64 i5 = i + 320 < n ? i + 320 : last, i6 = i + 384 < n ? i + 384 : last;
-      uint4 v0 = s4[i], v1 = s4[i1], v2 = s4[i2], v3 = s4[i3];
-      uint4 v4 = s4[i4], v5 = s4[i5], v6 = s4[i6];
+      U4 v0 = s4[i], v1 = s4[i1], v2 = s4[i2], v3 = s4[i3];
+      U4 v4 = s4[i4], v5 = s4[i5], v6 = s4[i6];
       d4[i] = v0;
       if (i + 64 < n) d4[i1] = v1;
       if (i + 128 < n) d4[i2] = v2;
@@ -180,8 +191,8 @@ __device__ void tile_copy(const MoveDesc& m, u64 lo, u64 hi, int lane) {
       if (i + 384 < n) d4[i6] = v6;
     }
   } else if ((u64(s) & 3) == 0 && (u64(d) & 3) == 0 && (bytes & 3) == 0) {
-    const u32* s1 = (const u32*)s;
-    u32* d1 = (u32*)d;
+    GAS const u32* s1 = (GAS const u32*)s;
+    GAS u32* d1 = (GAS u32*)d;
     u64 n = bytes / 4;
     u64 i = lane;
     for (; i + 7 * 64 < n; i += 8 * 64) {
@@ -194,7 +205,9 @@ __device__ void tile_copy(const MoveDesc& m, u64 lo, u64 hi, int lane) {
     }
     for (; i < n; i += 64) d1[i] = s1[i];
   } else {
-    for (u64 i = lane; i < bytes; i += 64) d[i] = s[i];
+    GAS const char* sc = (GAS const char*)s;
+    GAS char* dc = (GAS char*)d;
+    for (u64 i = lane; i < bytes; i += 64) dc[i] = sc[i];
   }
 }
 
@@ -209,18 +222,18 @@ __device__ void tile_reduce_f32(const MoveDesc& m, u64 lo, u64 hi, int lane) {
   bool v16 = aligned16(d) && aligned16(s0) && aligned16(s1) && (n & 3) == 0 &&
              m.nsrc == 2;
   if (v16) {
-    const float4* a = (const float4*)s0;
-    const float4* b = (const float4*)s1;
-    float4* o = (float4*)d;
+    GAS const F4* a = (GAS const F4*)s0;
+    GAS const F4* b = (GAS const F4*)s1;
+    GAS F4* o = (GAS F4*)d;
     u64 n4 = n / 4;
     u64 i = lane;
     // 4-deep pipeline x 2 operand streams = 8 loads in flight per lane
     for (; i + 3 * 64 < n4; i += 4 * 64) {
-      float4 x0 = a[i], x1 = a[i + 64], x2 = a[i + 2 * 64], x3 = a[i + 3 * 64];
-      float4 y0 = b[i], y1 = b[i + 64], y2 = b[i + 2 * 64], y3 = b[i + 3 * 64];
-#define ACCL_R4(a_, b_) make_float4(OP<float>::apply(a_.x, b_.x), \
+      F4 x0 = a[i], x1 = a[i + 64], x2 = a[i + 2 * 64], x3 = a[i + 3 * 64];
+      F4 y0 = b[i], y1 = b[i + 64], y2 = b[i + 2 * 64], y3 = b[i + 3 * 64];
+#define ACCL_R4(a_, b_) (F4){OP<float>::apply(a_.x, b_.x), \
     OP<float>::apply(a_.y, b_.y), OP<float>::apply(a_.z, b_.z), \
-    OP<float>::apply(a_.w, b_.w))
+    OP<float>::apply(a_.w, b_.w)}
       o[i] = ACCL_R4(x0, y0); o[i + 64] = ACCL_R4(x1, y1);
       o[i + 2 * 64] = ACCL_R4(x2, y2); o[i + 3 * 64] = ACCL_R4(x3, y3);
     }
@@ -228,8 +241,8 @@ __device__ void tile_reduce_f32(const MoveDesc& m, u64 lo, u64 hi, int lane) {
       u64 last = n4 - 1;
       u64 i1 = i + 64 < n4 ? i + 64 : last, i2 = i + 128 < n4 ? i + 128 : last;
       u64 i3 = i + 192 < n4 ? i + 192 : last;
-      float4 x0 = a[i], x1 = a[i1], x2 = a[i2], x3 = a[i3];
-      float4 y0 = b[i], y1 = b[i1], y2 = b[i2], y3 = b[i3];
+      F4 x0 = a[i], x1 = a[i1], x2 = a[i2], x3 = a[i3];
+      F4 y0 = b[i], y1 = b[i1], y2 = b[i2], y3 = b[i3];
       o[i] = ACCL_R4(x0, y0);
       if (i + 64 < n4) o[i1] = ACCL_R4(x1, y1);
       if (i + 128 < n4) o[i2] = ACCL_R4(x2, y2);
@@ -239,10 +252,10 @@ __device__ void tile_reduce_f32(const MoveDesc& m, u64 lo, u64 hi, int lane) {
     return;
   }
   for (u64 i = lane; i < n; i += 64) {
-    float acc = ((const float*)m.src[0])[lo + i];
+    float acc = ((GAS const float*)m.src[0])[lo + i];
     for (u32 k = 1; k < m.nsrc; ++k)
-      acc = OP<float>::apply(acc, ((const float*)m.src[k])[lo + i]);
-    d[i] = acc;
+      acc = OP<float>::apply(acc, ((GAS const float*)m.src[k])[lo + i]);
+    ((GAS float*)m.dst)[lo + i] = acc;
   }
 }
 

@@ -21,8 +21,6 @@ namespace accl {
 
 #define AGENT __HIP_MEMORY_SCOPE_AGENT
 
-__device__ __forceinline__ u32 move_waves(const MoveDesc& m, u32 nwaves);
-
 // ------------------------------------------------------------- GpuMover
 __device__ u32 GpuMover::submit(const MoveDesc& m) {
   u64 h = head_cache;
@@ -68,7 +66,7 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
   d = m;
   d.inline_done = 0;
   d.epoch = h + 1;
-  s.tiles_total = move_waves(m, nwaves);  // completion = one add per wave
+  s.tiles_total = move_tiles(m);
   __hip_atomic_store(&s.tiles_claimed, 0u, __ATOMIC_RELAXED, AGENT);
   __hip_atomic_store(&s.tiles_done, 0u, __ATOMIC_RELAXED, AGENT);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -153,120 +151,119 @@ __device__ __forceinline__ bool aligned16(const void* p) {
 typedef u32 U4 __attribute__((ext_vector_type(4)));
 typedef float F4 __attribute__((ext_vector_type(4)));
 
-// ---------------------------------------------------------------- spans
-// The fleet executes each move as ONE grid-stride loop over 64-lane "lines"
-// (1KB lines on the 16B-vector path): at any instant the active addresses of
-// all W participating waves are CONTIGUOUS, sweeping memory front-to-back so
-// every HBM channel is hit evenly. (The previous tile-per-wave layout put
-// concurrent accesses 128KB apart — a handful of channels per phase — and
-// capped the fleet at ~1.6 of 8 TB/s.)
-
-// Must agree between submit() (completion accounting) and the execution
-// path dispatch below.
-__device__ __forceinline__ bool move_vec16(const MoveDesc& m) {
-  if (m.nsrc == 1 && m.src_dt[0] == m.dst_dt) {
-    u64 bytes = move_bytes(m);
-    return ((m.src[0] | m.dst) & 15) == 0 && (bytes & 15) == 0;
-  }
-  if (m.nsrc == 2 && m.dst_dt == u8(DataType::float32) &&
-      m.src_dt[0] == u8(DataType::float32) &&
-      m.src_dt[1] == u8(DataType::float32))
-    return ((m.src[0] | m.src[1] | m.dst) & 15) == 0 && (m.count & 3) == 0;
-  return false;
-}
-
-__device__ __forceinline__ u64 move_lines(const MoveDesc& m) {
-  u64 units = move_vec16(m)
-      ? (m.nsrc == 1 ? move_bytes(m) / 16 : m.count / 4)
-      : m.count;
-  return (units + 63) / 64;
-}
-
-__device__ __forceinline__ u32 move_waves(const MoveDesc& m, u32 nwaves) {
-  u64 l = move_lines(m);
-  return l < nwaves ? u32(l) : nwaves;
-}
-
-// 16B-vector copy span: wave wi of W, 8-deep software pipeline.
-__device__ void span_copy16(const MoveDesc& m, u32 wi, u32 W, int lane) {
-  GAS const U4* s4 = (GAS const U4*)m.src[0];
-  GAS U4* d4 = (GAS U4*)m.dst;
-  const u64 n = move_bytes(m) / 16;
-  const u64 S = u64(W) * 64;
-  u64 i = u64(wi) * 64 + lane;
-  for (; i + 7 * S < n; i += 8 * S) {
-    U4 v0 = s4[i], v1 = s4[i + S], v2 = s4[i + 2 * S], v3 = s4[i + 3 * S];
-    U4 v4 = s4[i + 4 * S], v5 = s4[i + 5 * S], v6 = s4[i + 6 * S],
-       v7 = s4[i + 7 * S];
-    d4[i] = v0; d4[i + S] = v1; d4[i + 2 * S] = v2; d4[i + 3 * S] = v3;
-    d4[i + 4 * S] = v4; d4[i + 5 * S] = v5; d4[i + 6 * S] = v6;
-    d4[i + 7 * S] = v7;
-  }
-  if (i < n) {
-    // clamped tail: duplicate loads are free, stores predicated, so the
-    // remaining loads all issue together (a 4KB move IS this tail)
-    u64 last = n - 1;
-    u64 i1 = i + S < n ? i + S : last, i2 = i + 2 * S < n ? i + 2 * S : last;
-    u64 i3 = i + 3 * S < n ? i + 3 * S : last,
-        i4 = i + 4 * S < n ? i + 4 * S : last;
-    u64 i5 = i + 5 * S < n ? i + 5 * S : last,
-        i6 = i + 6 * S < n ? i + 6 * S : last;
-    U4 v0 = s4[i], v1 = s4[i1], v2 = s4[i2], v3 = s4[i3];
-    U4 v4 = s4[i4], v5 = s4[i5], v6 = s4[i6];
-    d4[i] = v0;
-    if (i + S < n) d4[i1] = v1;
-    if (i + 2 * S < n) d4[i2] = v2;
-    if (i + 3 * S < n) d4[i3] = v3;
-    if (i + 4 * S < n) d4[i4] = v4;
-    if (i + 5 * S < n) d4[i5] = v5;
-    if (i + 6 * S < n) d4[i6] = v6;
+// pure same-dtype copy, 16B vectorized when aligned. 8-deep software
+// pipeline: 8 loads in flight per lane (8 KiB per wave) so HBM (~1 us over
+// xGMI) latency is covered by MLP, not occupancy alone.
+__device__ void tile_copy(const MoveDesc& m, u64 lo, u64 hi, int lane) {
+  u32 esz = dtype_size(DataType(m.dst_dt));
+  u64 bytes = (hi - lo) * esz;
+  const char* s = (const char*)m.src[0] + lo * esz;
+  char* d = (char*)m.dst + lo * esz;
+  if (aligned16(s) && aligned16(d) && (bytes & 15) == 0) {
+    GAS const U4* s4 = (GAS const U4*)s;
+    GAS U4* d4 = (GAS U4*)d;
+    u64 n = bytes / 16;
+    u64 i = lane;
+    for (; i + 7 * 64 < n; i += 8 * 64) {
+      U4 v0 = s4[i], v1 = s4[i + 64], v2 = s4[i + 2 * 64], v3 = s4[i + 3 * 64];
+      U4 v4 = s4[i + 4 * 64], v5 = s4[i + 5 * 64], v6 = s4[i + 6 * 64],
+            v7 = s4[i + 7 * 64];
+      d4[i] = v0; d4[i + 64] = v1; d4[i + 2 * 64] = v2; d4[i + 3 * 64] = v3;
+      d4[i + 4 * 64] = v4; d4[i + 5 * 64] = v5; d4[i + 6 * 64] = v6;
+      d4[i + 7 * 64] = v7;
+    }
+    if (i < n) {
+      // tail: clamp load indices (duplicate loads are free) and predicate
+      // stores, so all remaining loads issue together instead of a
+      // one-load-one-store dependent chain (a 4KB move IS this tail)
+      u64 last = n - 1;
+      u64 i1 = i + 64 < n ? i + 64 : last, i2 = i + 128 < n ? i + 128 : last;
+      u64 i3 = i + 192 < n ? i + 192 : last, i4 = i + 256 < n ? i + 256 : last;
+      u64 i5 = i + 320 < n ? i + 320 : last, i6 = i + 384 < n ? i + 384 : last;
+      U4 v0 = s4[i], v1 = s4[i1], v2 = s4[i2], v3 = s4[i3];
+      U4 v4 = s4[i4], v5 = s4[i5], v6 = s4[i6];
+      d4[i] = v0;
+      if (i + 64 < n) d4[i1] = v1;
+      if (i + 128 < n) d4[i2] = v2;
+      if (i + 192 < n) d4[i3] = v3;
+      if (i + 256 < n) d4[i4] = v4;
+      if (i + 320 < n) d4[i5] = v5;
+      if (i + 384 < n) d4[i6] = v6;
+    }
+  } else if ((u64(s) & 3) == 0 && (u64(d) & 3) == 0 && (bytes & 3) == 0) {
+    GAS const u32* s1 = (GAS const u32*)s;
+    GAS u32* d1 = (GAS u32*)d;
+    u64 n = bytes / 4;
+    u64 i = lane;
+    for (; i + 7 * 64 < n; i += 8 * 64) {
+      u32 v0 = s1[i], v1 = s1[i + 64], v2 = s1[i + 2 * 64], v3 = s1[i + 3 * 64];
+      u32 v4 = s1[i + 4 * 64], v5 = s1[i + 5 * 64], v6 = s1[i + 6 * 64],
+          v7 = s1[i + 7 * 64];
+      d1[i] = v0; d1[i + 64] = v1; d1[i + 2 * 64] = v2; d1[i + 3 * 64] = v3;
+      d1[i + 4 * 64] = v4; d1[i + 5 * 64] = v5; d1[i + 6 * 64] = v6;
+      d1[i + 7 * 64] = v7;
+    }
+    for (; i < n; i += 64) d1[i] = s1[i];
+  } else {
+    GAS const char* sc = (GAS const char*)s;
+    GAS char* dc = (GAS char*)d;
+    for (u64 i = lane; i < bytes; i += 64) dc[i] = sc[i];
   }
 }
 
-// f32 2-source reduce span (the reduce_ops hot path: reference
-// kernels/plugins/reduce_ops/reduce_ops.cpp:83-106), 4-deep x 2 streams.
+// f32 n-ary reduce, 16B vectorized (the reduce_ops hot path:
+// reference kernels/plugins/reduce_ops/reduce_ops.cpp:83-106)
 template <template <class> class OP>
-__device__ void span_reduce_f32(const MoveDesc& m, u32 wi, u32 W, int lane) {
+__device__ void tile_reduce_f32(const MoveDesc& m, u64 lo, u64 hi, int lane) {
+  float* d = (float*)m.dst + lo;
+  const float* s0 = (const float*)m.src[0] + lo;
+  const float* s1 = (const float*)m.src[1] + lo;
+  u64 n = hi - lo;
+  bool v16 = aligned16(d) && aligned16(s0) && aligned16(s1) && (n & 3) == 0 &&
+             m.nsrc == 2;
+  if (v16) {
+    GAS const F4* a = (GAS const F4*)s0;
+    GAS const F4* b = (GAS const F4*)s1;
+    GAS F4* o = (GAS F4*)d;
+    u64 n4 = n / 4;
+    u64 i = lane;
+    // 4-deep pipeline x 2 operand streams = 8 loads in flight per lane
+    for (; i + 3 * 64 < n4; i += 4 * 64) {
+      F4 x0 = a[i], x1 = a[i + 64], x2 = a[i + 2 * 64], x3 = a[i + 3 * 64];
+      F4 y0 = b[i], y1 = b[i + 64], y2 = b[i + 2 * 64], y3 = b[i + 3 * 64];
 #define ACCL_R4(a_, b_) (F4){OP<float>::apply(a_.x, b_.x), \
     OP<float>::apply(a_.y, b_.y), OP<float>::apply(a_.z, b_.z), \
     OP<float>::apply(a_.w, b_.w)}
-  GAS const F4* a = (GAS const F4*)m.src[0];
-  GAS const F4* b = (GAS const F4*)m.src[1];
-  GAS F4* o = (GAS F4*)m.dst;
-  const u64 n = m.count / 4;
-  const u64 S = u64(W) * 64;
-  u64 i = u64(wi) * 64 + lane;
-  for (; i + 3 * S < n; i += 4 * S) {
-    F4 x0 = a[i], x1 = a[i + S], x2 = a[i + 2 * S], x3 = a[i + 3 * S];
-    F4 y0 = b[i], y1 = b[i + S], y2 = b[i + 2 * S], y3 = b[i + 3 * S];
-    o[i] = ACCL_R4(x0, y0); o[i + S] = ACCL_R4(x1, y1);
-    o[i + 2 * S] = ACCL_R4(x2, y2); o[i + 3 * S] = ACCL_R4(x3, y3);
-  }
-  if (i < n) {
-    u64 last = n - 1;
-    u64 i1 = i + S < n ? i + S : last, i2 = i + 2 * S < n ? i + 2 * S : last;
-    u64 i3 = i + 3 * S < n ? i + 3 * S : last;
-    F4 x0 = a[i], x1 = a[i1], x2 = a[i2], x3 = a[i3];
-    F4 y0 = b[i], y1 = b[i1], y2 = b[i2], y3 = b[i3];
-    o[i] = ACCL_R4(x0, y0);
-    if (i + S < n) o[i1] = ACCL_R4(x1, y1);
-    if (i + 2 * S < n) o[i2] = ACCL_R4(x2, y2);
-    if (i + 3 * S < n) o[i3] = ACCL_R4(x3, y3);
-  }
+      o[i] = ACCL_R4(x0, y0); o[i + 64] = ACCL_R4(x1, y1);
+      o[i + 2 * 64] = ACCL_R4(x2, y2); o[i + 3 * 64] = ACCL_R4(x3, y3);
+    }
+    if (i < n4) {  // clamped-tail (see tile_copy): all loads in flight
+      u64 last = n4 - 1;
+      u64 i1 = i + 64 < n4 ? i + 64 : last, i2 = i + 128 < n4 ? i + 128 : last;
+      u64 i3 = i + 192 < n4 ? i + 192 : last;
+      F4 x0 = a[i], x1 = a[i1], x2 = a[i2], x3 = a[i3];
+      F4 y0 = b[i], y1 = b[i1], y2 = b[i2], y3 = b[i3];
+      o[i] = ACCL_R4(x0, y0);
+      if (i + 64 < n4) o[i1] = ACCL_R4(x1, y1);
+      if (i + 128 < n4) o[i2] = ACCL_R4(x2, y2);
+      if (i + 192 < n4) o[i3] = ACCL_R4(x3, y3);
+    }
 #undef ACCL_R4
+    return;
+  }
+  for (u64 i = lane; i < n; i += 64) {
+    float acc = ((GAS const float*)m.src[0])[lo + i];
+    for (u32 k = 1; k < m.nsrc; ++k)
+      acc = OP<float>::apply(acc, ((GAS const float*)m.src[k])[lo + i]);
+    ((GAS float*)m.dst)[lo + i] = acc;
+  }
 }
 
-__device__ bool dtype_is_floatish(DataType d) {
-  return d == DataType::float32 || d == DataType::float16 ||
-         d == DataType::bfloat16;
-}
-
-// float-domain span for any f32/f16/bf16 mix (cast + reduce fused — the
+// float-domain path for any f32/f16/bf16 mix (cast + reduce fused — the
 // hp_compression + reduce_ops lanes in one pass)
-__device__ void span_float_generic(const MoveDesc& m, u32 wi, u32 W, int lane) {
+__device__ void tile_float_generic(const MoveDesc& m, u64 lo, u64 hi, int lane) {
   const bool is_max = m.nsrc >= 2 && ReduceFunction(m.func) == ReduceFunction::MAX;
-  const u64 S = u64(W) * 64;
-  for (u64 i = u64(wi) * 64 + lane; i < m.count; i += S) {
+  for (u64 i = lo + lane; i < hi; i += 64) {
     float acc = ld_f32((const void*)m.src[0], i, DataType(m.src_dt[0]));
     for (u32 k = 1; k < m.nsrc; ++k) {
       float v = ld_f32((const void*)m.src[k], i, DataType(m.src_dt[k]));
@@ -276,34 +273,49 @@ __device__ void span_float_generic(const MoveDesc& m, u32 wi, u32 W, int lane) {
   }
 }
 
-// Execute wave wi's share (of W waves) of a move.
-__device__ void run_span(const MoveDesc& m, u32 wi, u32 W, int lane) {
-  if (move_vec16(m)) {
-    if (m.nsrc == 1) { span_copy16(m, wi, W, lane); return; }
-    if (ReduceFunction(m.func) == ReduceFunction::SUM)
-      span_reduce_f32<SumOp>(m, wi, W, lane);
-    else
-      span_reduce_f32<MaxOp>(m, wi, W, lane);
+__device__ bool dtype_is_floatish(DataType d) {
+  return d == DataType::float32 || d == DataType::float16 ||
+         d == DataType::bfloat16;
+}
+
+__device__ void run_tile(const MoveDesc& m, u32 t, int lane) {
+  u64 te = move_tile_elems(m);
+  u64 lo = u64(t) * te;
+  u64 hi = lo + te;
+  if (hi > m.count) hi = m.count;
+  if (lo >= hi) return;
+  // dispatch
+  if (m.nsrc == 1 && m.src_dt[0] == m.dst_dt) {
+    tile_copy(m, lo, hi, lane);
     return;
   }
+  bool all_f32 = m.dst_dt == u8(DataType::float32);
   bool floatish = dtype_is_floatish(DataType(m.dst_dt));
-  for (u32 k = 0; k < m.nsrc; ++k)
+  for (u32 k = 0; k < m.nsrc; ++k) {
+    all_f32 = all_f32 && m.src_dt[k] == u8(DataType::float32);
     floatish = floatish && dtype_is_floatish(DataType(m.src_dt[k]));
-  if (floatish && !(m.nsrc == 1 && m.src_dt[0] == m.dst_dt)) {
-    span_float_generic(m, wi, W, lane);
+  }
+  if (all_f32 && m.nsrc >= 2) {
+    if (ReduceFunction(m.func) == ReduceFunction::SUM)
+      tile_reduce_f32<SumOp>(m, lo, hi, lane);
+    else
+      tile_reduce_f32<MaxOp>(m, lo, hi, lane);
     return;
   }
-  // exact scalar fallback (f64 / int dtypes / mixed / unaligned same-dtype)
-  const u64 S = u64(W) * 64;
-  for (u64 i = u64(wi) * 64 + lane; i < m.count; i += S)
-    execute_move_range(m, i, i + 1);
+  if (floatish) {
+    tile_float_generic(m, lo, hi, lane);
+    return;
+  }
+  // exact scalar fallback (f64 / int dtypes / mixed)
+  for (u64 i = lo + lane; i < hi; i += 64) execute_move_range(m, i, i + 1);
 }
 
 // ------------------------------------------------------------- mover main
-// Wave wi (a per-move rotation of gw) executes a grid-stride span of each
-// move; completion = ONE tiles_done add per participating wave. The doorbell
-// is replicated (DOORBELL_REPS lines, ~nwaves/64 pollers each) and packs the
-// latest move's wave count so non-participants never touch the descriptor.
+// Static tile partitioning: move mi's tile t belongs to wave
+// (t + rot(mi)) % nwaves == gw — no claim atomics, no shared hot line.
+// Completion = ONE tiles_done add per participating wave. The doorbell is
+// replicated (DOORBELL_REPS lines, ~nwaves/64 pollers each) so noticing a
+// new move costs ~L2-hit latency, not a fleet-wide serialized cacheline.
 __device__ void mover_main(GpuEngineState* S) {
   const int lane = int(threadIdx.x) & 63;
   const u32 gw = blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
@@ -316,7 +328,7 @@ __device__ void mover_main(GpuEngineState* S) {
   for (;;) {
     u64 packed = __hip_atomic_load((const u64*)&my_rep[0], __ATOMIC_RELAXED, AGENT);
     u64 h = packed >> 24;
-    u32 latest_waves = u32(packed & 0xFFFFFFu);
+    u32 latest_total = u32(packed & 0xFFFFFFu);
     if (cursor == h) {
       if (__hip_atomic_load((const u64*)&my_rep[1], __ATOMIC_RELAXED, AGENT))
         return;
@@ -333,25 +345,26 @@ __device__ void mover_main(GpuEngineState* S) {
     while (cursor < h) {
       u32 slot = u32(cursor % MOVE_RING);
       u32 rot = u32(cursor * 37) % nwaves;
-      u32 wi = (gw + nwaves - rot) % nwaves;
+      u32 first = (gw + nwaves - rot) % nwaves;
       // when caught up, the doorbell word tells us our share without a
       // descriptor read (the common small-move fast path)
-      if (cursor == h - 1 && wi >= latest_waves) { cursor++; continue; }
+      if (cursor == h - 1 && first >= latest_total) { cursor++; continue; }
       const MoveDesc& m = ring[slot];
       // desc publish happens-before the doorbell store; sanity-check epoch
       if (u32(m.epoch) != u32(cursor + 1)) break;  // not visible yet: retry
-      u32 W = m.inline_done ? 0 : move_waves(m, nwaves);
-      if (wi < W) {
-        if (lane == 0 && wi == 0) S->dbg[1] = wallclock();
-        run_span(m, wi, W, lane);
+      u32 total = m.inline_done ? 0 : move_tiles(m);
+      if (first < total) {
+        if (lane == 0 && first == 0) S->dbg[1] = wallclock();
+        u32 cnt = 0;
+        for (u32 t = first; t < total; t += nwaves, ++cnt) run_tile(m, t, lane);
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         if (lane == 0) {
-          S->wave_tiles[gw & 4095] += 1;
-          __hip_atomic_fetch_add(&st[slot].tiles_done, 1u, __ATOMIC_RELEASE,
+          S->wave_tiles[gw & 4095] += cnt;
+          __hip_atomic_fetch_add(&st[slot].tiles_done, cnt, __ATOMIC_RELEASE,
                                  AGENT);
-          if (wi == 0) S->dbg[2] = wallclock();
+          if (first == 0) S->dbg[2] = wallclock();
         }
       }
       cursor++;
@@ -427,7 +440,7 @@ __device__ void small_mover(SmallMb* mb, int lane) {
       __builtin_amdgcn_s_sleep(1);
       continue;
     }
-    run_span(mb->d, 0, 1, lane);  // one wave executes the whole small move
+    run_tile(mb->d, 0, lane);  // a small move always fits one tile
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");

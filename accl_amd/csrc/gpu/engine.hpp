@@ -39,7 +39,6 @@ struct GpuMover {
   u64 (*rep)[8];
   void* small_mb;          // LDS SmallMb (set by the scheduler kernel)
   u64 small_seq;           // scheduler-private inline-move counter
-  u32 nwaves;              // mover fleet size (host sets: engine_wgs * 4)
 
 #if defined(__HIPCC__)
   __device__ u32 submit(const MoveDesc& m);

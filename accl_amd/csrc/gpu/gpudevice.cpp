@@ -139,7 +139,6 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
   st->mover.stop = &dstate->stop;
   st->mover.dbg = dstate->dbg;
   st->mover.rep = dstate->head_rep;
-  st->mover.nwaves = u32(engine_wgs_) * 4;  // 256-thread WGs = 4 waves
   // pinned pointers as seen by the device
   RingPage* rp = (RingPage*)ring_pinned_;
   void* dev_ptr = nullptr;

@@ -164,13 +164,25 @@ __device__ void tile_copy(const MoveDesc& m, u64 lo, u64 hi, int lane) {
     GAS U4* d4 = (GAS U4*)d;
     u64 n = bytes / 16;
     u64 i = lane;
+    // non-temporal: payload streams bypass L1/L2 (no dirty lines -> the
+    // mover needs no L2-writeback release fence, and no cache pollution)
     for (; i + 7 * 64 < n; i += 8 * 64) {
-      U4 v0 = s4[i], v1 = s4[i + 64], v2 = s4[i + 2 * 64], v3 = s4[i + 3 * 64];
-      U4 v4 = s4[i + 4 * 64], v5 = s4[i + 5 * 64], v6 = s4[i + 6 * 64],
-            v7 = s4[i + 7 * 64];
-      d4[i] = v0; d4[i + 64] = v1; d4[i + 2 * 64] = v2; d4[i + 3 * 64] = v3;
-      d4[i + 4 * 64] = v4; d4[i + 5 * 64] = v5; d4[i + 6 * 64] = v6;
-      d4[i + 7 * 64] = v7;
+      U4 v0 = __builtin_nontemporal_load(&s4[i]);
+      U4 v1 = __builtin_nontemporal_load(&s4[i + 64]);
+      U4 v2 = __builtin_nontemporal_load(&s4[i + 2 * 64]);
+      U4 v3 = __builtin_nontemporal_load(&s4[i + 3 * 64]);
+      U4 v4 = __builtin_nontemporal_load(&s4[i + 4 * 64]);
+      U4 v5 = __builtin_nontemporal_load(&s4[i + 5 * 64]);
+      U4 v6 = __builtin_nontemporal_load(&s4[i + 6 * 64]);
+      U4 v7 = __builtin_nontemporal_load(&s4[i + 7 * 64]);
+      __builtin_nontemporal_store(v0, &d4[i]);
+      __builtin_nontemporal_store(v1, &d4[i + 64]);
+      __builtin_nontemporal_store(v2, &d4[i + 2 * 64]);
+      __builtin_nontemporal_store(v3, &d4[i + 3 * 64]);
+      __builtin_nontemporal_store(v4, &d4[i + 4 * 64]);
+      __builtin_nontemporal_store(v5, &d4[i + 5 * 64]);
+      __builtin_nontemporal_store(v6, &d4[i + 6 * 64]);
+      __builtin_nontemporal_store(v7, &d4[i + 7 * 64]);
     }
     if (i < n) {
       // tail: clamp load indices (duplicate loads are free) and predicate
@@ -180,15 +192,20 @@ __device__ void tile_copy(const MoveDesc& m, u64 lo, u64 hi, int lane) {
       u64 i1 = i + 64 < n ? i + 64 : last, i2 = i + 128 < n ? i + 128 : last;
       u64 i3 = i + 192 < n ? i + 192 : last, i4 = i + 256 < n ? i + 256 : last;
       u64 i5 = i + 320 < n ? i + 320 : last, i6 = i + 384 < n ? i + 384 : last;
-      U4 v0 = s4[i], v1 = s4[i1], v2 = s4[i2], v3 = s4[i3];
-      U4 v4 = s4[i4], v5 = s4[i5], v6 = s4[i6];
-      d4[i] = v0;
-      if (i + 64 < n) d4[i1] = v1;
-      if (i + 128 < n) d4[i2] = v2;
-      if (i + 192 < n) d4[i3] = v3;
-      if (i + 256 < n) d4[i4] = v4;
-      if (i + 320 < n) d4[i5] = v5;
-      if (i + 384 < n) d4[i6] = v6;
+      U4 v0 = __builtin_nontemporal_load(&s4[i]);
+      U4 v1 = __builtin_nontemporal_load(&s4[i1]);
+      U4 v2 = __builtin_nontemporal_load(&s4[i2]);
+      U4 v3 = __builtin_nontemporal_load(&s4[i3]);
+      U4 v4 = __builtin_nontemporal_load(&s4[i4]);
+      U4 v5 = __builtin_nontemporal_load(&s4[i5]);
+      U4 v6 = __builtin_nontemporal_load(&s4[i6]);
+      __builtin_nontemporal_store(v0, &d4[i]);
+      if (i + 64 < n) __builtin_nontemporal_store(v1, &d4[i1]);
+      if (i + 128 < n) __builtin_nontemporal_store(v2, &d4[i2]);
+      if (i + 192 < n) __builtin_nontemporal_store(v3, &d4[i3]);
+      if (i + 256 < n) __builtin_nontemporal_store(v4, &d4[i4]);
+      if (i + 320 < n) __builtin_nontemporal_store(v5, &d4[i5]);
+      if (i + 384 < n) __builtin_nontemporal_store(v6, &d4[i6]);
     }
   } else if ((u64(s) & 3) == 0 && (u64(d) & 3) == 0 && (bytes & 3) == 0) {
     GAS const u32* s1 = (GAS const u32*)s;
@@ -278,16 +295,22 @@ __device__ bool dtype_is_floatish(DataType d) {
          d == DataType::bfloat16;
 }
 
-__device__ void run_tile(const MoveDesc& m, u32 t, int lane) {
+// Returns true if the executed path used only non-temporal accesses (no
+// dirty L2 lines -> the caller may skip the L2-writeback release fence).
+__device__ bool run_tile(const MoveDesc& m, u32 t, int lane) {
   u64 te = move_tile_elems(m);
   u64 lo = u64(t) * te;
   u64 hi = lo + te;
   if (hi > m.count) hi = m.count;
-  if (lo >= hi) return;
+  if (lo >= hi) return true;
   // dispatch
   if (m.nsrc == 1 && m.src_dt[0] == m.dst_dt) {
+    u32 esz = dtype_size(DataType(m.dst_dt));
+    u64 bytes = (hi - lo) * esz;
+    bool v16 = aligned16((const void*)(m.src[0] + lo * esz)) &&
+               aligned16((const void*)(m.dst + lo * esz)) && (bytes & 15) == 0;
     tile_copy(m, lo, hi, lane);
-    return;
+    return v16;
   }
   bool all_f32 = m.dst_dt == u8(DataType::float32);
   bool floatish = dtype_is_floatish(DataType(m.dst_dt));
@@ -296,18 +319,23 @@ __device__ void run_tile(const MoveDesc& m, u32 t, int lane) {
     floatish = floatish && dtype_is_floatish(DataType(m.src_dt[k]));
   }
   if (all_f32 && m.nsrc >= 2) {
+    bool v16 = aligned16((const void*)(m.dst + lo * 4)) &&
+               aligned16((const void*)(m.src[0] + lo * 4)) &&
+               aligned16((const void*)(m.src[1] + lo * 4)) &&
+               ((hi - lo) & 3) == 0 && m.nsrc == 2;
     if (ReduceFunction(m.func) == ReduceFunction::SUM)
       tile_reduce_f32<SumOp>(m, lo, hi, lane);
     else
       tile_reduce_f32<MaxOp>(m, lo, hi, lane);
-    return;
+    return v16;
   }
   if (floatish) {
     tile_float_generic(m, lo, hi, lane);
-    return;
+    return false;
   }
   // exact scalar fallback (f64 / int dtypes / mixed)
   for (u64 i = lo + lane; i < hi; i += 64) execute_move_range(m, i, i + 1);
+  return false;
 }
 
 // ------------------------------------------------------------- mover main
@@ -440,10 +468,12 @@ __device__ void small_mover(SmallMb* mb, int lane) {
       __builtin_amdgcn_s_sleep(1);
       continue;
     }
-    run_tile(mb->d, 0, lane);  // a small move always fits one tile
+    bool nt = run_tile(mb->d, 0, lane);  // a small move fits one tile
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    if (!nt) {
+      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
     if (lane == 0)
       __hip_atomic_store(&mb->done, sq, __ATOMIC_RELEASE,
                          __HIP_MEMORY_SCOPE_WORKGROUP);

@@ -158,6 +158,7 @@ class ACCL:
         if name in ("copy", "combine", "send", "recv", "bcast", "scatter",
                     "gather", "allgather", "reduce", "allreduce",
                     "reduce_scatter", "alltoall", "barrier", "nop",
+                    "stream_put", "pop_stream", "stream_ready",
                     "create_communicator", "split_communicator",
                     "free_request", "deinit"):
             return getattr(self._a, name)

@@ -240,6 +240,26 @@ PYBIND11_MODULE(_core, m) {
            py::arg("from_device") = false, py::arg("to_device") = false,
            py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
            py::call_guard<py::gil_scoped_release>())
+      .def("stream_put", &ACCL::stream_put, py::arg("src"), py::arg("count"),
+           py::arg("dst"), py::arg("tag") = 0, py::arg("comm") = GLOBAL_COMM,
+           py::arg("from_device") = false,
+           py::arg("compress") = DataType::none, py::arg("run_async") = false,
+           py::return_value_policy::reference, py::keep_alive<0, 1>(),
+           py::call_guard<py::gil_scoped_release>())
+      .def("pop_stream",
+           [](ACCL& a, u32 src, py::buffer out, u64 timeout_ms) {
+             py::buffer_info info = out.request(true);
+             u64 maxb = u64(info.size) * u64(info.itemsize);
+             u32 tag = 0;
+             u64 n;
+             {
+               py::gil_scoped_release rel;
+               n = a.pop_stream(src, info.ptr, maxb, &tag, timeout_ms);
+             }
+             return py::make_tuple(n, tag);
+           },
+           py::arg("src"), py::arg("out"), py::arg("timeout_ms") = 10000)
+      .def("stream_ready", &ACCL::stream_ready, py::arg("src"))
       .def("barrier", &ACCL::barrier, py::arg("comm") = GLOBAL_COMM,
            py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
            py::call_guard<py::gil_scoped_release>())
@@ -253,6 +273,9 @@ PYBIND11_MODULE(_core, m) {
     if (opts.contains("slot_bytes")) c.slot_bytes = opts["slot_bytes"].cast<u32>();
     if (opts.contains("max_eager")) c.max_eager = opts["max_eager"].cast<u64>();
     if (opts.contains("timeout_us")) c.timeout_us = opts["timeout_us"].cast<u64>();
+    if (opts.contains("n_stream")) c.n_stream = opts["n_stream"].cast<u32>();
+    if (opts.contains("stream_bytes")) c.stream_bytes = opts["stream_bytes"].cast<u32>();
+    if (opts.contains("n_rndzv")) c.n_rndzv = opts["n_rndzv"].cast<u32>();
     return c;
   };
 

@@ -125,7 +125,12 @@ inline ProtoConfig default_proto_config(u32 nranks, u32 rank) {
 //   [0]                ArenaHdr
 //   hdr.eager_off      per src-peer EagerChan control (headers + credit)
 //   hdr.rndzv_off      per peer RndzvRing pair (addr ring + done ring)
-//   hdr.stream_off     stream rings (device-initiated collectives, stream_put)
+//   hdr.stream_off     per src-peer stream rings (stream_put / device-
+//                      initiated collectives): ctl (credit) + SlotHdr[n] +
+//                      payload, mirror of the eager channel but consumed by
+//                      the APPLICATION (host or device kernel), not by a
+//                      posted recv — the depacketizer-bypass path of the
+//                      reference (udp_depacketizer.cpp:135-148 strm TDEST)
 //   hdr.slots_off      eager payload slots [src_peer][slot]
 //   hdr.heap_off       buffer heap until hdr.arena_bytes
 
@@ -203,8 +208,11 @@ inline ArenaLayout arena_layout(const ProtoConfig& c, u64 spare_bytes = 32u << 2
   off += u64(c.nranks) * c.n_rndzv * sizeof(RndzvRec);
   L.rndzv_done_off = off = align_up(off, 256);
   off += u64(c.nranks) * c.n_rndzv * sizeof(RndzvRec);
-  L.stream_off = off = align_up(off, 256);
-  off += u64(c.n_stream) * (sizeof(SlotHdr) + c.stream_bytes);
+  L.stream_off = off = align_up(off, 4096);
+  // per-src-peer: EagerChanCtl (credit) + SlotHdr[n_stream] + payload slots
+  off += u64(c.nranks) *
+         (sizeof(EagerChanCtl) + u64(c.n_stream) * sizeof(SlotHdr) +
+          u64(c.n_stream) * c.stream_bytes);
   L.barrier_off = off = align_up(off, 256);
   off += u64(c.nranks) * sizeof(u64);
   L.direct_off = off = align_up(off, 256);
@@ -260,6 +268,26 @@ struct TransportView {
     return (volatile u64*)(arena[r] + hdr(r)->direct_off) + s;
   }
   ACCL_HD char* heap_ptr(u32 r, u64 off) const { return arena[r] + off; }
+
+  // ---- stream channel (s -> r): ctl+hdrs+payload in r's arena lane [s];
+  // credit word in s's arena lane [r] (advanced by the consumer at r).
+  ACCL_HD u64 stream_lane_bytes(u32 r) const {
+    const ArenaHdr* h = hdr(r);
+    return sizeof(EagerChanCtl) + u64(h->n_stream) * sizeof(SlotHdr) +
+           u64(h->n_stream) * h->stream_bytes;
+  }
+  ACCL_HD EagerChanCtl* stream_ctl(u32 r, u32 s) const {
+    return (EagerChanCtl*)(arena[r] + hdr(r)->stream_off +
+                           u64(s) * stream_lane_bytes(r));
+  }
+  ACCL_HD SlotHdr* stream_hdr(u32 r, u32 s, u32 slot) const {
+    return (SlotHdr*)((char*)stream_ctl(r, s) + sizeof(EagerChanCtl)) + slot;
+  }
+  ACCL_HD char* stream_payload(u32 r, u32 s, u32 slot) const {
+    const ArenaHdr* h = hdr(r);
+    return (char*)stream_hdr(r, s, 0) + u64(h->n_stream) * sizeof(SlotHdr) +
+           u64(slot) * h->stream_bytes;
+  }
 };
 
 // Per-pair sequence state, PRIVATE to one engine (not shared): lives in the
@@ -273,6 +301,7 @@ struct PairSeq {
   u64 rndzv_addr_rx[MAX_RANKS];
   u64 rndzv_done_tx[MAX_RANKS];
   u64 rndzv_done_rx[MAX_RANKS];
+  u64 stream_tx[MAX_RANKS];    // stream segments sent to peer
   u64 direct_tx[MAX_RANKS];    // cumulative bytes direct-written to peer
   u64 direct_rx[MAX_RANKS];    // cumulative bytes direct-received from peer
   u64 barrier_epoch[MAX_RANKS];// per-PAIR barrier epoch (must match both ends)

@@ -49,6 +49,7 @@ struct PendSeg {
 
 struct Flow {
   u8 kind;
+  u8 to_stream;   // FLOW_TX: target the peer's STREAM ring, not eager slots
   u8 func;        // ReduceFunction+1; 0 = plain copy
   u8 sdt, wdt, ddt, bdt;  // src, wire, dst, reduce-operand dtypes
   u32 gpeer;
@@ -148,7 +149,8 @@ struct Cclo {
         case FLOW_TX: {
           // payload complete -> publish the slot header (release chain:
           // mover released on completion, poll() acquired).
-          SlotHdr* h = tv.slot_hdr(f.gpeer, me(), p.slot);
+          SlotHdr* h = f.to_stream ? tv.stream_hdr(f.gpeer, me(), p.slot)
+                                   : tv.slot_hdr(f.gpeer, me(), p.slot);
           h->tag = f.tag;
           h->bytes = u32(p.elems * dtype_size(DataType(f.wdt)));
           h->msg_count = f.msg_count_hdr;
@@ -213,13 +215,19 @@ struct Cclo {
           break;
         }
         case FLOW_TX: {
-          u64 next = sq.eager_tx[f.gpeer];          // segments sent so far
-          const u32 n_slots = cfg.n_slots;
-          if (next - tx_credit(f.gpeer) >= n_slots) return any;  // no credit
-          const u64 seg_cap = u64(cfg.slot_bytes) / dtype_size(DataType(f.wdt));
+          u64* txs = f.to_stream ? sq.stream_tx : sq.eager_tx;
+          u64 next = txs[f.gpeer];                  // segments sent so far
+          const u32 n_slots = f.to_stream ? cfg.n_stream : cfg.n_slots;
+          u64 credit = f.to_stream
+              ? ld_sys(&tv.stream_ctl(me(), f.gpeer)->credit)
+              : tx_credit(f.gpeer);
+          if (next - credit >= n_slots) return any;  // no credit
+          const u32 seg_bytes = f.to_stream ? cfg.stream_bytes : cfg.slot_bytes;
+          const u64 seg_cap = u64(seg_bytes) / dtype_size(DataType(f.wdt));
           u64 n = min64(avail - f.submitted, seg_cap);
           u32 slot = u32(next % n_slots);
-          m.dst = (u64)tv.slot_payload(f.gpeer, me(), slot);
+          m.dst = f.to_stream ? (u64)tv.stream_payload(f.gpeer, me(), slot)
+                              : (u64)tv.slot_payload(f.gpeer, me(), slot);
           m.dst_dt = f.wdt;
           m.src[0] = (u64)(f.src + f.submitted * dtype_size(DataType(f.sdt)));
           m.src_dt[0] = f.sdt;
@@ -230,7 +238,7 @@ struct Cclo {
                    (f.submitted + n >= f.count ? SEG_LAST : 0);
           f.pend[f.pt % FLOW_INFLIGHT] = {tok, slot, n, next + 1, fl};
           f.pt++; f.submitted += n;
-          sq.eager_tx[f.gpeer] = next + 1;
+          txs[f.gpeer] = next + 1;
           any = true;
           break;
         }
@@ -339,11 +347,11 @@ struct Cclo {
   }
   ACCL_HD void mk_tx(u32 i, u32 gpeer, const char* src, DataType sdt,
                      DataType wdt, u64 count, u32 tag,
-                     const u64* gate = nullptr) {
+                     const u64* gate = nullptr, bool to_stream = false) {
     Flow& f = fl(i);
     f.kind = FLOW_TX; f.gpeer = gpeer; f.src = src;
     f.sdt = u8(sdt); f.wdt = u8(wdt); f.count = count; f.tag = tag;
-    f.gate = gate; f.msg_count_hdr = count;
+    f.gate = gate; f.msg_count_hdr = count; f.to_stream = to_stream ? 1 : 0;
   }
   ACCL_HD void mk_rx(u32 i, u32 gpeer, char* dst, DataType ddt, DataType wdt,
                      u64 count, u32 tag, const char* red = nullptr,
@@ -833,6 +841,19 @@ struct Cclo {
     return E_OK;
   }
 
+  // stream_put: push a buffer into the peer's STREAM ring — consumed by the
+  // application at the peer (host pop_stream or a device kernel), never by a
+  // posted recv. reference: stream_put (accl.hpp:204-238), depacketizer
+  // strm-TDEST bypass (udp_depacketizer.cpp:135-148).
+  ACCL_HD u32 op_stream_put(const CallDesc& d, const CommView& c) {
+    u64 n = desc_count(d);
+    u32 peer = c.global(d.root_src_dst);
+    DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    mk_tx(0, peer, src, dt, wdt, n, d.tag, nullptr, /*to_stream=*/true);
+    return run_flows(1);
+  }
+
   // ---------------- dispatch ----------------
   // reference: run() scenario switch (ccl_offload_control.c:2375-2459)
   ACCL_HD u32 run_call(const CallDesc& d) {
@@ -857,6 +878,7 @@ struct Cclo {
       case Op::reduce_scatter: return op_reduce_scatter(d, c);
       case Op::alltoall: return op_alltoall(d, c);
       case Op::barrier: return op_barrier(d, c);
+      case Op::stream_put: return op_stream_put(d, c);
       case Op::config: return run_config(d);
       default: return E_INVALID_OP;
     }

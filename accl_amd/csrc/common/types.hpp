@@ -46,6 +46,9 @@ enum class Op : u32 {
   reduce_scatter = 11,
   alltoall = 12,
   barrier = 13,
+  stream_put = 14,  // send into the peer's stream ring (reference:
+                    // stream_put scenario, accl.hpp:204-238; remote side is
+                    // consumed by the application, not a posted recv)
   halt = 254,  // engine shutdown (no reference analogue; replaces kernel exit)
   nop = 255,
 };

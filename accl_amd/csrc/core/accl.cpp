@@ -179,6 +179,75 @@ Request* ACCL::send(BaseBuffer& src, u64 count, u32 dst, u32 tag, u32 comm,
   return finish(d, run_async, nullptr, 0, from_device ? nullptr : &src, count);
 }
 
+Request* ACCL::stream_put(BaseBuffer& src, u64 count, u32 dst, u32 tag,
+                          u32 comm, bool from_device, DataType compress,
+                          bool run_async) {
+  if (tag > MAX_USER_TAG) throw accl_error("stream_put: tag out of range");
+  CallDesc d = make_desc(Op::stream_put, count, src.dtype(), compress);
+  d.addr0 = src.arena_offset();
+  d.root_src_dst = dst;
+  d.tag = tag;
+  d.comm_id = comm;
+  d.flags = F_SRC_ARENA;
+  return finish(d, run_async, nullptr, 0, from_device ? nullptr : &src, count);
+}
+
+// ---- host-side stream consumption (the application end of the ring) ----
+namespace {
+struct StreamLane {
+  u64 ctl_off, hdr_off, payload_off;
+};
+StreamLane stream_lane(const ProtoConfig& c, u32 src_lane) {
+  ArenaLayout L = arena_layout(c);
+  u64 lane_bytes = sizeof(EagerChanCtl) +
+                   u64(c.n_stream) * sizeof(SlotHdr) +
+                   u64(c.n_stream) * c.stream_bytes;
+  StreamLane s{};
+  s.ctl_off = L.stream_off + u64(src_lane) * lane_bytes;
+  s.hdr_off = s.ctl_off + sizeof(EagerChanCtl);
+  s.payload_off = s.hdr_off + u64(c.n_stream) * sizeof(SlotHdr);
+  return s;
+}
+}  // namespace
+
+bool ACCL::stream_ready(u32 src) {
+  const ProtoConfig& c = be_->cfg();
+  StreamLane L = stream_lane(c, src);
+  u64 seq = stream_rx_seq_[src] + 1;
+  u32 slot = u32((seq - 1) % c.n_stream);
+  SlotHdr h{};
+  be_->read_arena(L.hdr_off + slot * sizeof(SlotHdr), &h, sizeof(h));
+  return h.seq == seq;
+}
+
+u64 ACCL::pop_stream(u32 src, void* out, u64 max_bytes, u32* tag,
+                     u64 timeout_ms) {
+  const ProtoConfig& c = be_->cfg();
+  if (src >= c.nranks) throw accl_error("pop_stream: bad src");
+  StreamLane L = stream_lane(c, src);
+  u64 seq = stream_rx_seq_[src] + 1;
+  u32 slot = u32((seq - 1) % c.n_stream);
+  SlotHdr h{};
+  u64 t0 = wallclock_host_ns();
+  for (;;) {
+    be_->read_arena(L.hdr_off + slot * sizeof(SlotHdr), &h, sizeof(h));
+    if (h.seq == seq) break;
+    if (wallclock_host_ns() - t0 > timeout_ms * 1000000ull) return 0;
+    cpu_pause();
+  }
+  u64 n = h.bytes;
+  if (n > max_bytes)
+    throw accl_error("pop_stream: segment larger than out buffer");
+  be_->read_arena(L.payload_off + u64(slot) * c.stream_bytes, out, n);
+  if (tag) *tag = h.tag;
+  stream_rx_seq_[src] = seq;
+  // return credit: the word lives in the SENDER's arena, lane [me]
+  StreamLane mine = stream_lane(c, c.rank);
+  be_->write_peer(src, mine.ctl_off + offsetof(EagerChanCtl, credit), &seq,
+                  sizeof(seq));
+  return n;
+}
+
 Request* ACCL::recv(BaseBuffer& dst, u64 count, u32 src, u32 tag, u32 comm,
                     bool to_device, DataType compress, bool run_async) {
   CallDesc d = make_desc(Op::recv, count, dst.dtype(), compress);

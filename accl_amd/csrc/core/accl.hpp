@@ -157,6 +157,20 @@ class ACCL {
   Request* barrier(u32 comm = GLOBAL_COMM, bool run_async = false);
   Request* nop(bool run_async = false);
 
+  // --- streaming surface (reference: stream_put accl.hpp:204-238, remote
+  // side consumed by the application via the depacketizer strm bypass) ---
+  // Push `count` elements of src into dst's stream ring (engine-segmented).
+  Request* stream_put(BaseBuffer& src, u64 count, u32 dst, u32 tag = 0,
+                      u32 comm = GLOBAL_COMM, bool from_device = false,
+                      DataType compress = DataType::none,
+                      bool run_async = false);
+  // Consume the next stream segment from global rank `src`: copies payload
+  // into out (<= max_bytes), returns bytes (0 on timeout), fills *tag.
+  u64 pop_stream(u32 src, void* out, u64 max_bytes, u32* tag = nullptr,
+                 u64 timeout_ms = 10000);
+  // Non-destructive check: is a stream segment from `src` pending?
+  bool stream_ready(u32 src);
+
   void free_request(Request* r);
   u32 comm_size(u32 comm) const { return comm_sizes_.at(comm); }
   u32 comm_rank(u32 comm) const { return comm_ranks_.at(comm); }
@@ -171,6 +185,7 @@ class ACCL {
   std::unique_ptr<Backend> be_;
   std::vector<Request*> reqs_;
   std::vector<u32> comm_sizes_, comm_ranks_;
+  u64 stream_rx_seq_[MAX_RANKS] = {};  // host-consumed stream segments
 };
 
 }  // namespace accl

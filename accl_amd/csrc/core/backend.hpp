@@ -91,6 +91,9 @@ class Backend {
   void free_block(u64 off) { heap_.free_block(off); }
   virtual void write_arena(u64 off, const void* src, u64 bytes) = 0;
   virtual void read_arena(u64 off, void* dst, u64 bytes) = 0;
+  // write into a PEER rank's arena (credit returns for stream consumption;
+  // xGMI/IPC on GPU, shm on the emulator)
+  virtual void write_peer(u32 rank, u64 off, const void* src, u64 bytes) = 0;
 
   // --- communicators (quiesce, rewrite mirror, bump generation) ---
   // reference: Communicator rank-table write, driver/xrt/src/

@@ -177,6 +177,13 @@ void EmuDevice::write_arena(u64 off, const void* src, u64 bytes) {
   std::memcpy(arena_base_ + off, src, bytes);
   __atomic_thread_fence(__ATOMIC_RELEASE);
 }
+
+void EmuDevice::write_peer(u32 rank, u64 off, const void* src, u64 bytes) {
+  if (rank >= cfg_.nranks || !peer_base_[rank])
+    throw accl_error("emu: write_peer to unmapped rank");
+  std::memcpy(peer_base_[rank] + off, src, bytes);
+  __atomic_thread_fence(__ATOMIC_RELEASE);
+}
 void EmuDevice::read_arena(u64 off, void* dst, u64 bytes) {
   __atomic_thread_fence(__ATOMIC_ACQUIRE);
   std::memcpy(dst, arena_base_ + off, bytes);

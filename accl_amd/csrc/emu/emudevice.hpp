@@ -40,6 +40,7 @@ class EmuDevice : public Backend {
 
   void write_arena(u64 off, const void* src, u64 bytes) override;
   void read_arena(u64 off, void* dst, u64 bytes) override;
+  void write_peer(u32 rank, u64 off, const void* src, u64 bytes) override;
 
  private:
   void engine_main();

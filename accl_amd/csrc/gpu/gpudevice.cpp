@@ -207,6 +207,13 @@ void GpuDevice::write_arena(u64 off, const void* src, u64 bytes) {
   hip_check(hipMemcpy(arena_base_ + off, src, bytes, hipMemcpyHostToDevice),
             "write_arena");
 }
+void GpuDevice::write_peer(u32 rank, u64 off, const void* src, u64 bytes) {
+  if (rank >= cfg_.nranks || !peer_base_[rank])
+    throw accl_error("gpu: write_peer to unmapped rank");
+  hip_check(hipMemcpy(peer_base_[rank] + off, src, bytes,
+                      hipMemcpyHostToDevice), "write_peer");
+}
+
 void GpuDevice::read_arena(u64 off, void* dst, u64 bytes) {
   hip_check(hipMemcpy(dst, arena_base_ + off, bytes, hipMemcpyDeviceToHost),
             "read_arena");

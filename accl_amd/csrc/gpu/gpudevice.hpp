@@ -25,6 +25,7 @@ class GpuDevice : public Backend {
 
   void write_arena(u64 off, const void* src, u64 bytes) override;
   void read_arena(u64 off, void* dst, u64 bytes) override;
+  void write_peer(u32 rank, u64 off, const void* src, u64 bytes) override;
 
   int device_index() const { return dev_; }
   u64 arena_bytes() const { return arena_bytes_; }

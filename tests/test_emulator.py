@@ -318,3 +318,88 @@ def _async_and_duration(a, rank, n):
 
 def test_async():
     run_ranks(_async_and_duration, 2)
+
+
+# ------------------------------------------------------------- streaming
+# reference: stream_put (accl.hpp:204-238) + depacketizer strm-TDEST bypass
+# (udp_depacketizer.cpp:135-148): the remote side is consumed by the
+# APPLICATION (pop_stream / a device kernel), never by a posted recv.
+def _stream_put(a, rank, n):
+    cnt = 3000
+    s = a.create_buffer(cnt, DT.float32)
+    if rank == 0:
+        s.write(pattern(cnt, 7))
+        a.stream_put(s, cnt, dst=1, tag=42)
+    elif rank == 1:
+        out = np.zeros(cnt, np.float32)
+        nb, tag = a.pop_stream(0, out)
+        assert nb == cnt * 4 and tag == 42
+        assert np.array_equal(out, pattern(cnt, 7))
+    a.barrier()
+
+
+def _stream_put_segmented(a, rank, n):
+    # larger than one stream slot: engine segments, consumer drains in order
+    total = 5 * 1024 + 131  # elements; stream_bytes default 1 MiB -> use opts
+    if rank == 0:
+        s = a.create_buffer(total, DT.float32)
+        s.write(pattern(total, 3))
+        a.stream_put(s, total, dst=1, tag=9)
+    elif rank == 1:
+        got = np.zeros(0, np.float32)
+        buf = np.zeros(total, np.float32)
+        while got.size < total:
+            nb, tag = a.pop_stream(0, buf)
+            assert tag == 9 and nb > 0 and nb % 4 == 0
+            got = np.concatenate([got, buf[:nb // 4]])
+        assert np.array_equal(got, pattern(total, 3))
+    a.barrier()
+
+
+def _stream_credit_wrap(a, rank, n):
+    # more segments than ring slots: sender must block on credit, not corrupt
+    cnt = 256
+    s = a.create_buffer(cnt, DT.float32)
+    rounds = 11  # > n_stream
+    if rank == 0:
+        for i in range(rounds):
+            s.write(pattern(cnt, i))
+            a.stream_put(s, cnt, dst=1, tag=i)
+    elif rank == 1:
+        out = np.zeros(cnt, np.float32)
+        for i in range(rounds):
+            nb, tag = a.pop_stream(0, out)
+            assert nb == cnt * 4 and tag == i
+            assert np.array_equal(out, pattern(cnt, i))
+    a.barrier()
+
+
+def _stream_self(a, rank, n):
+    # loopback: stream_put to self lands in own ring
+    cnt = 100
+    s = a.create_buffer(cnt, DT.float32)
+    s.write(pattern(cnt, 5))
+    a.stream_put(s, cnt, dst=rank, tag=1)
+    out = np.zeros(cnt, np.float32)
+    nb, tag = a.pop_stream(rank, out)
+    assert nb == cnt * 4 and tag == 1 and np.array_equal(out, pattern(cnt, 5))
+
+
+STREAM_SMALL = {"n_stream": 8, "stream_bytes": 4096,
+                "timeout_us": 20_000_000}
+
+
+def test_stream_put():
+    run_ranks(_stream_put, 2)
+
+
+def test_stream_put_segmented():
+    run_ranks(_stream_put_segmented, 2, opts=STREAM_SMALL)
+
+
+def test_stream_credit_wrap():
+    run_ranks(_stream_credit_wrap, 2, opts=STREAM_SMALL)
+
+
+def test_stream_self():
+    run_ranks(_stream_self, 1)

@@ -22,6 +22,7 @@ SOURCES = [
     "emu/emudevice.cpp",
     "gpu/gpudevice.cpp",
     "gpu/engine.hip",
+    "gpu/plugins.hip",
     "bindings/module.cpp",
 ]
 

@@ -7,6 +7,7 @@
 #include <cstring>
 #include "../core/accl.hpp"
 #include "../emu/emudevice.hpp"
+#include <hip/hip_runtime.h>
 #include "../gpu/gpudevice.hpp"
 
 namespace py = pybind11;
@@ -303,4 +304,23 @@ PYBIND11_MODULE(_core, m) {
         py::arg("opts") = py::dict());
 
   m.def("error_to_string", &error_to_string);
+
+  // demo plugin: device-initiated stream_put (reference vadd_put analogue)
+  m.def("demo_vadd_put",
+        [](ACCL& a, BaseBuffer& src, u64 count, u32 dst, u32 tag, float addv) {
+          auto* g = dynamic_cast<GpuDevice*>(a.backend());
+          if (!g) throw accl_error("demo_vadd_put: gpu backend only");
+          u32 seg = g->cfg().stream_bytes < (32u << 10) ? g->cfg().stream_bytes
+                                                        : (32u << 10);
+          launch_vadd_put(g->arena_local() + src.arena_offset(), count, tag,
+                          g->arena_local(), g->peer_base(dst), g->cfg().rank,
+                          dst, seg, addv, g->op_stream());
+          hipError_t e = hipStreamSynchronize((hipStream_t)g->op_stream());
+          if (e != hipSuccess)
+            throw accl_error(std::string("demo_vadd_put: ") +
+                             hipGetErrorString(e));
+        },
+        py::arg("a"), py::arg("src"), py::arg("count"), py::arg("dst"),
+        py::arg("tag") = 0, py::arg("addv") = 1.0f,
+        py::call_guard<py::gil_scoped_release>());
 }

@@ -48,6 +48,13 @@ ACCL_HD inline void st_sys(volatile u64* p, u64 v) {
   __atomic_store_n((u64*)p, v, __ATOMIC_RELAXED);
 #endif
 }
+ACCL_HD inline u64 afadd_sys(volatile u64* p, u64 v) {
+#if defined(__HIP_DEVICE_COMPILE__)
+  return __hip_atomic_fetch_add((u64*)p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
+#else
+  return __atomic_fetch_add((u64*)p, v, __ATOMIC_RELAXED);
+#endif
+}
 ACCL_HD inline void st_sys32(volatile u32* p, u32 v) {
 #if defined(__HIP_DEVICE_COMPILE__)
   __hip_atomic_store((u32*)p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
@@ -165,7 +172,9 @@ struct alignas(64) EagerChanCtl {
   // written by the receiver of the OPPOSITE direction (remote), polled locally
   // by this rank acting as sender towards `peer`:
   u64 credit;     // number of slots this rank's messages to `peer`... see note
-  u64 _pad[7];
+  u64 tx_ctr;     // STREAM lanes only: shared tx-seq allocator so the engine,
+                  // the host and device kernels can produce into one channel
+  u64 _pad[6];
 };
 // NOTE on credit placement: for channel (s -> r), slot headers+payload live in
 // r's arena at index [s]; the credit word lives in s's arena at index [r]

@@ -62,6 +62,8 @@ struct Flow {
   u64 submitted;          // elements handed to the mover
   u64 done;               // elements retired IN ORDER (gates dependents)
   const u64* gate;        // if set: submitted may not pass *gate
+  u64 claimed;            // stream tx: seq claimed from the shared allocator
+                          // but not yet submitted (0 = none)
   // direct mode:
   u64 peer_off;           // destination offset in peer arena (tx_direct)
   u64 prog_base;          // cumulative-bytes baseline (both direct kinds)
@@ -215,13 +217,20 @@ struct Cclo {
           break;
         }
         case FLOW_TX: {
-          u64* txs = f.to_stream ? sq.stream_tx : sq.eager_tx;
-          u64 next = txs[f.gpeer];                  // segments sent so far
+          u64 next;
           const u32 n_slots = f.to_stream ? cfg.n_stream : cfg.n_slots;
-          u64 credit = f.to_stream
-              ? ld_sys(&tv.stream_ctl(me(), f.gpeer)->credit)
-              : tx_credit(f.gpeer);
-          if (next - credit >= n_slots) return any;  // no credit
+          if (f.to_stream) {
+            // seq from the channel's shared allocator (engine + host + device
+            // producers interoperate); claim once, hold across credit stalls
+            EagerChanCtl* ctl = tv.stream_ctl(me(), f.gpeer);
+            if (!f.claimed) f.claimed = afadd_sys(&ctl->tx_ctr, 1) + 1;
+            next = f.claimed - 1;
+            u64 credit = ld_sys(&ctl->credit);
+            if (credit + n_slots < f.claimed) return any;  // slot not yet free
+          } else {
+            next = sq.eager_tx[f.gpeer];            // segments sent so far
+            if (next - tx_credit(f.gpeer) >= n_slots) return any;  // no credit
+          }
           const u32 seg_bytes = f.to_stream ? cfg.stream_bytes : cfg.slot_bytes;
           const u64 seg_cap = u64(seg_bytes) / dtype_size(DataType(f.wdt));
           u64 n = min64(avail - f.submitted, seg_cap);
@@ -238,7 +247,8 @@ struct Cclo {
                    (f.submitted + n >= f.count ? SEG_LAST : 0);
           f.pend[f.pt % FLOW_INFLIGHT] = {tok, slot, n, next + 1, fl};
           f.pt++; f.submitted += n;
-          txs[f.gpeer] = next + 1;
+          if (f.to_stream) f.claimed = 0;
+          else sq.eager_tx[f.gpeer] = next + 1;
           any = true;
           break;
         }

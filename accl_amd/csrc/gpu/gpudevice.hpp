@@ -11,6 +11,14 @@
 
 namespace accl {
 
+// plugin launchers (gpu/plugins.hip)
+void launch_vadd_put(const void* in, u64 count, u32 tag, void* my_arena,
+                     void* peer_arena, u32 me, u32 peer, u32 seg_bytes,
+                     float addv, void* stream);
+void launch_stream_drain(void* out, u64 max_elems, u32 nseg, void* my_arena,
+                         void* peer_arena, u32 me, u32 peer, u64 start_seq,
+                         void* stream);
+
 class GpuDevice : public Backend {
  public:
   GpuDevice(u32 nranks, u32 rank, int device_index,
@@ -32,6 +40,8 @@ class GpuDevice : public Backend {
   // debug: per-wave executed-tile counters from the engine state
   std::vector<u32> debug_wave_tiles();
   std::vector<u64> debug_timeline();
+  char* peer_base(u32 r) const { return r < MAX_RANKS ? peer_base_[r] : nullptr; }
+  void* op_stream() const { return stream_; }  // for plugin launches
 
  private:
   int dev_ = 0;

@@ -52,9 +52,23 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
     d.epoch = h + 1;
     s.tiles_total = 0;
     __hip_atomic_store(&s.tiles_done, 0u, __ATOMIC_RELAXED, AGENT);
-    while (__hip_atomic_load(&mb->done, __ATOMIC_ACQUIRE,
-                             __HIP_MEMORY_SCOPE_WORKGROUP) != sq)
+    // bounded wait: if the sibling wave is wedged/slow, execute the move
+    // here (scalar) — double execution of a deterministic elementwise move
+    // is idempotent, and an unbounded spin here would deadlock the engine
+    u64 dl = wallclock() + 100000;  // 1 ms
+    for (;;) {
+      if (__hip_atomic_load(&mb->done, __ATOMIC_ACQUIRE,
+                            __HIP_MEMORY_SCOPE_WORKGROUP) == sq)
+        break;
+      if (wallclock() > dl) {
+        execute_move_range(m, 0, m.count);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        break;
+      }
       __builtin_amdgcn_s_sleep(1);
+    }
     if (dbg) { dbg[6] = wallclock(); dbg[7]++; }
     // no fleet doorbell: movers discover this slot (and skip it) when the
     // next fleet move advances the packed head past it; the ring/desc

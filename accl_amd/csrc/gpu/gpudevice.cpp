@@ -178,8 +178,20 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
 void GpuDevice::shutdown() {
   if (!launched_) return;
   __atomic_store_n((u64*)&ring_->ctrl.shutdown, 1, __ATOMIC_RELEASE);
-  hipError_t e = hipStreamSynchronize((hipStream_t)stream_);
-  hipError_t e2 = hipStreamSynchronize((hipStream_t)mover_stream_);
+  // bounded: a wedged engine must not hang the process forever (process
+  // teardown destroys the HIP context and with it the kernels)
+  u64 t0 = wallclock_host_ns();
+  hipError_t e = hipErrorUnknown, e2 = hipErrorUnknown;
+  for (;;) {
+    e = hipStreamQuery((hipStream_t)stream_);
+    e2 = hipStreamQuery((hipStream_t)mover_stream_);
+    if (e != hipErrorNotReady && e2 != hipErrorNotReady) break;
+    if (wallclock_host_ns() - t0 > 15ull * 1000000000) {
+      launched_ = false;
+      throw accl_error("gpu: engine did not acknowledge shutdown (wedged)");
+    }
+    usleep(200);
+  }
   launched_ = false;
   hip_check(e, "engine shutdown");
   hip_check(e2, "mover shutdown");

@@ -5,6 +5,7 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 #include <cstring>
+#include <unistd.h>
 #include "../core/accl.hpp"
 #include "../emu/emudevice.hpp"
 #include <hip/hip_runtime.h>
@@ -312,10 +313,26 @@ PYBIND11_MODULE(_core, m) {
           if (!g) throw accl_error("demo_vadd_put: gpu backend only");
           u32 seg = g->cfg().stream_bytes < (32u << 10) ? g->cfg().stream_bytes
                                                         : (32u << 10);
+          // fresh stream: the engine's streams hold never-ending persistent
+          // kernels, so anything queued behind them would never run
+          hipStream_t st{};
+          if (hipStreamCreateWithFlags(&st, hipStreamNonBlocking) != hipSuccess)
+            throw accl_error("demo_vadd_put: stream create failed");
           launch_vadd_put(g->arena_local() + src.arena_offset(), count, tag,
                           g->arena_local(), g->peer_base(dst), g->cfg().rank,
-                          dst, seg, addv, g->op_stream());
-          hipError_t e = hipStreamSynchronize((hipStream_t)g->op_stream());
+                          dst, seg, addv, st);
+          hipError_t e = hipGetLastError();
+          u64 t0 = wallclock_host_ns();
+          while (e == hipSuccess) {  // bounded sync (30 s)
+            hipError_t q = hipStreamQuery(st);
+            if (q != hipErrorNotReady) { e = q; break; }
+            if (wallclock_host_ns() - t0 > 30ull * 1000000000) {
+              e = hipErrorUnknown;
+              break;
+            }
+            usleep(100);
+          }
+          (void)hipStreamDestroy(st);
           if (e != hipSuccess)
             throw accl_error(std::string("demo_vadd_put: ") +
                              hipGetErrorString(e));

@@ -33,6 +33,16 @@ def parse_args():
     p.add_argument("--bytes", type=int, default=256 << 20)
     p.add_argument("--sweep", action="store_true",
                    help="also print the 4KB..1GB busbw curve (rank 0)")
+    p.add_argument("--collective", default="allreduce",
+                   choices=["allreduce", "bcast", "scatter", "gather",
+                            "allgather", "reduce_scatter", "alltoall",
+                            "reduce", "sendrecv"],
+                   help="BASELINE configs 2-4: which collective to time")
+    p.add_argument("--dtype", default="fp32", choices=["fp32", "bf16", "fp16"],
+                   help="BASELINE config 4: bf16 grad/param patterns")
+    p.add_argument("--overlap-gemm", action="store_true",
+                   help="BASELINE config 5: run a co-resident torch GEMM "
+                        "stream while timing the collective")
     p.add_argument("--backend", default="auto", choices=["auto", "gpu", "emu"])
     p.add_argument("--out-csv", default=None)
     return p.parse_args()
@@ -62,25 +72,96 @@ def main():
                heap_bytes=heap,
                opts={"slot_bytes": slot_mb << 20, "n_slots": 8})
 
-    count = args.bytes // 4
-    src = a.create_buffer(count, A.DataType.float32, device_only=True)
-    dst = a.create_buffer(count, A.DataType.float32, device_only=True)
+    DT = {"fp32": A.DataType.float32, "bf16": A.DataType.bfloat16,
+          "fp16": A.DataType.float16}[args.dtype]
+    esz = 4 if args.dtype == "fp32" else 2
+    count = args.bytes // esz
+    # collectives with per-rank chunking need count divisible by world
+    count -= count % max(world, 1)
+    src = a.create_buffer(count, DT, device_only=True)
+    dst = a.create_buffer(count, DT, device_only=True)
     if backend == "gpu":
         import torch
         t = a.tensor(src)
-        t.copy_(torch.randn(count, device=t.device))
+        t.copy_(torch.randn(count, device=t.device, dtype=torch.float32)
+                .to(t.dtype))
         # NOT torch.cuda.synchronize(): hipDeviceSynchronize would wait on
         # the persistent engine kernel's stream (which never ends).
         torch.cuda.current_stream().synchronize()
     else:
-        tmp = a.create_buffer(min(count, 1 << 20), A.DataType.float32)
+        tmp = a.create_buffer(min(count, 1 << 20), DT)
         tmp.write(np.random.default_rng(0).standard_normal(
-            min(count, 1 << 20), dtype=np.float32))
+            min(count, 1 << 20), dtype=np.float32).astype(
+                np.float32 if esz == 4 else np.float16).view(np.int8))
         a.copy(tmp, src, min(count, 1 << 20))
 
+    RF = A.ReduceFunction.SUM
+    per = count // max(world, 1)
+
     def one_allreduce(s=src, d=dst, n=count):
-        a.allreduce(s, d, n, A.ReduceFunction.SUM,
-                    from_device=True, to_device=True)
+        a.allreduce(s, d, n, RF, from_device=True, to_device=True)
+
+    def one_bcast():
+        a.bcast(src, count, 0, from_device=True, to_device=True)
+
+    def one_scatter():
+        a.scatter(src, dst, per, 0, from_device=True, to_device=True)
+
+    def one_gather():
+        a.gather(src, dst, per, 0, from_device=True, to_device=True)
+
+    def one_allgather():
+        a.allgather(src, dst, per, from_device=True, to_device=True)
+
+    def one_reduce_scatter():
+        a.reduce_scatter(src, dst, per, RF, from_device=True, to_device=True)
+
+    def one_reduce():
+        a.reduce(src, dst, count, 0, RF, from_device=True, to_device=True)
+
+    def one_alltoall():
+        a.alltoall(src, dst, per, from_device=True, to_device=True)
+
+    def one_sendrecv():
+        if world == 1:
+            a.copy(src, dst, count, from_device=True, to_device=True)
+        elif rank % 2 == 0:
+            peer = (rank + 1) % world
+            a.send(src, count, dst=peer, tag=1, from_device=True)
+            a.recv(dst, count, src=peer, tag=2, to_device=True)
+        else:
+            peer = (rank - 1) % world
+            a.recv(dst, count, src=peer, tag=1, to_device=True)
+            a.send(src, count, dst=peer, tag=2, from_device=True)
+
+    ONE = {"allreduce": one_allreduce, "bcast": one_bcast,
+           "scatter": one_scatter, "gather": one_gather,
+           "allgather": one_allgather, "reduce_scatter": one_reduce_scatter,
+           "reduce": one_reduce, "alltoall": one_alltoall,
+           "sendrecv": one_sendrecv}
+    one_op = ONE[args.collective]
+
+    overlap_stop = None
+    if args.overlap_gemm and backend == "gpu":
+        # BASELINE config 5: a co-resident GEMM stream keeps MFMA busy while
+        # the engine's mover fleet (10 of 16 wave slots per CU) moves data
+        import threading
+        import torch
+        overlap_stop = threading.Event()
+        gemm_stats = {"n": 0}
+
+        def gemm_loop():
+            st = torch.cuda.Stream()
+            with torch.cuda.stream(st):
+                m = torch.randn(4096, 4096, device="cuda", dtype=torch.bfloat16)
+                while not overlap_stop.is_set():
+                    m = m @ m
+                    m = m / m.norm()
+                    st.synchronize()
+                    gemm_stats["n"] += 1
+
+        gemm_thread = threading.Thread(target=gemm_loop, daemon=True)
+        gemm_thread.start()
 
     def timed(fn, steps, warmup):
         for _ in range(warmup):
@@ -103,8 +184,23 @@ def main():
         factor = 2.0 * (P - 1) / P if P > 1 else 1.0
         return factor * nbytes / sec / 1e9
 
-    el = timed(one_allreduce, args.steps, args.warmup)
-    value = busbw(args.bytes, el, world)
+    el = timed(one_op, args.steps, args.warmup)
+    if overlap_stop is not None:
+        overlap_stop.set()
+        gemm_thread.join(timeout=10)
+    # bus-bandwidth accounting per collective (NCCL-tests conventions; the
+    # reference computes Gbit/s the same way, test/host/Coyote/test.cpp:523)
+    nbytes = count * esz
+    P = world
+    factor = {
+        "allreduce": 2.0 * (P - 1) / P if P > 1 else 1.0,
+        "reduce_scatter": (P - 1) / P if P > 1 else 1.0,
+        "allgather": (P - 1) / P if P > 1 else 1.0,
+        "alltoall": (P - 1) / P if P > 1 else 1.0,
+        "bcast": 1.0, "scatter": 1.0 / P, "gather": 1.0 / P,
+        "reduce": 1.0, "sendrecv": 1.0,
+    }[args.collective]
+    value = factor * nbytes / el / 1e9
 
     sweep_rows = []
     if args.sweep:
@@ -129,7 +225,7 @@ def main():
 
     if rank == 0:
         out = {
-            "metric": "allreduce_busbw_GBps",
+            "metric": f"{args.collective}_busbw_GBps",
             "value": round(value, 3),
             "unit": "GB/s",
             "n_gpus": world,
@@ -139,19 +235,21 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": round(value / BASELINE_GBPS, 3),
-            "dtype": "fp32",
+            "dtype": args.dtype,
             "data": "synthetic",
             "config": {
-                "model": "allreduce",
-                "collective": "allreduce",
-                "message_bytes": args.bytes,
+                "model": args.collective,
+                "collective": args.collective,
+                "message_bytes": count * esz,
+                "overlap_gemm": bool(args.overlap_gemm),
                 "global_batch": 1,
                 "seq_len": args.bytes // 4,
                 "parallelism": f"fullmesh_p{world}",
                 "backend": backend,
-                "metric_desc": ("all-reduce bus bandwidth (GB/s), fp32, "
-                                "per BASELINE.json; vs_baseline anchors the "
-                                "reference's 100 Gbps (12.5 GB/s) line rate"),
+                "metric_desc": (f"{args.collective} bus bandwidth (GB/s), "
+                                f"{args.dtype}, per BASELINE.json; "
+                                "vs_baseline anchors the reference's "
+                                "100 Gbps (12.5 GB/s) line rate"),
             },
         }
         print(json.dumps(out))

@@ -159,6 +159,8 @@ class ACCL:
                     "gather", "allgather", "reduce", "allreduce",
                     "reduce_scatter", "alltoall", "barrier", "nop",
                     "stream_put", "pop_stream", "stream_ready",
+                    "dump_communicator", "dump_eager_rx_buffers",
+                    "dump_streams", "dump_engine_status",
                     "create_communicator", "split_communicator",
                     "free_request", "deinit"):
             return getattr(self._a, name)

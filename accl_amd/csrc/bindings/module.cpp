@@ -262,6 +262,12 @@ PYBIND11_MODULE(_core, m) {
            },
            py::arg("src"), py::arg("out"), py::arg("timeout_ms") = 10000)
       .def("stream_ready", &ACCL::stream_ready, py::arg("src"))
+      .def("dump_communicator", &ACCL::dump_communicator,
+           py::arg("comm") = GLOBAL_COMM)
+      .def("dump_eager_rx_buffers", &ACCL::dump_eager_rx_buffers,
+           py::arg("verbose") = false)
+      .def("dump_streams", &ACCL::dump_streams)
+      .def("dump_engine_status", &ACCL::dump_engine_status)
       .def("barrier", &ACCL::barrier, py::arg("comm") = GLOBAL_COMM,
            py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
            py::call_guard<py::gil_scoped_release>())

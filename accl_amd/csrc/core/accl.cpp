@@ -1,7 +1,9 @@
 #include "accl.hpp"
 #include <algorithm>
 #include <cstdlib>
+#include <cstdio>
 #include <cstring>
+#include <string>
 
 namespace accl {
 
@@ -246,6 +248,100 @@ u64 ACCL::pop_stream(u32 src, void* out, u64 max_bytes, u32* tag,
   be_->write_peer(src, mine.ctl_off + offsetof(EagerChanCtl, credit), &seq,
                   sizeof(seq));
   return n;
+}
+
+// ---------------- debug dumps ----------------
+// reference: dump_communicator (accl.cpp:1429-1439), dump_rx_buffers
+// (accl.cpp:964-1048: status/occupancy/tag/seqn per rx buffer), here read
+// from the live arena (slot headers + credit words are the ground truth).
+std::string ACCL::dump_communicator(u32 comm) {
+  const CommView& c = be_->comm_view(comm);
+  std::string out = "communicator " + std::to_string(comm) + ": rank " +
+                    std::to_string(c.rank) + " of " + std::to_string(c.size) +
+                    ", members [";
+  for (u32 i = 0; i < c.size; ++i)
+    out += (i ? " " : "") + std::to_string(c.members[i]);
+  out += "]\n";
+  return out;
+}
+
+std::string ACCL::dump_engine_status() {
+  auto v = be_->ctrl_view();
+  char buf[256];
+  std::snprintf(buf, sizeof(buf),
+                "engine: up=%llu doorbell=%llu submitted=%llu retired=%llu "
+                "heartbeat=%llu ncomms=%llu\n",
+                (unsigned long long)v.engine_up,
+                (unsigned long long)v.doorbell,
+                (unsigned long long)v.submitted,
+                (unsigned long long)v.retired,
+                (unsigned long long)v.heartbeat,
+                (unsigned long long)v.ncomms);
+  return buf;
+}
+
+std::string ACCL::dump_eager_rx_buffers(bool verbose) {
+  const ProtoConfig& c = be_->cfg();
+  ArenaLayout L = arena_layout(c);
+  std::string out = "eager rx channels (" + std::to_string(c.n_slots) +
+                    " slots x " + std::to_string(c.slot_bytes) + " B):\n";
+  u64 lane_bytes = sizeof(EagerChanCtl) + u64(c.n_slots) * sizeof(SlotHdr);
+  for (u32 s = 0; s < c.nranks; ++s) {
+    u64 ctl_off = L.eager_off + u64(s) * lane_bytes;
+    EagerChanCtl ctl{};
+    be_->read_arena(ctl_off, &ctl, sizeof(ctl));
+    u64 newest = 0, occupied = 0;
+    std::string slots;
+    for (u32 i = 0; i < c.n_slots; ++i) {
+      SlotHdr h{};
+      be_->read_arena(ctl_off + sizeof(EagerChanCtl) + i * sizeof(SlotHdr),
+                      &h, sizeof(h));
+      if (h.seq > newest) newest = h.seq;
+      if (h.seq) occupied++;
+      if (verbose) {
+        char b[128];
+        std::snprintf(b, sizeof(b), "    slot %u: seq=%llu tag=%u bytes=%u\n",
+                      i, (unsigned long long)h.seq, h.tag, h.bytes);
+        slots += b;
+      }
+    }
+    char b[160];
+    std::snprintf(b, sizeof(b),
+                  "  from rank %u: newest_seq=%llu written_slots=%llu "
+                  "credit_to_peer=%llu\n",
+                  s, (unsigned long long)newest, (unsigned long long)occupied,
+                  (unsigned long long)ctl.credit);
+    out += b;
+    out += slots;
+  }
+  return out;
+}
+
+std::string ACCL::dump_streams() {
+  const ProtoConfig& c = be_->cfg();
+  std::string out = "stream rings (" + std::to_string(c.n_stream) +
+                    " slots x " + std::to_string(c.stream_bytes) + " B):\n";
+  for (u32 s = 0; s < c.nranks; ++s) {
+    StreamLane L = stream_lane(c, s);
+    EagerChanCtl ctl{};
+    be_->read_arena(L.ctl_off, &ctl, sizeof(ctl));
+    u64 newest = 0;
+    for (u32 i = 0; i < c.n_stream; ++i) {
+      SlotHdr h{};
+      be_->read_arena(L.hdr_off + i * sizeof(SlotHdr), &h, sizeof(h));
+      if (h.seq > newest) newest = h.seq;
+    }
+    char b[200];
+    std::snprintf(b, sizeof(b),
+                  "  lane %u: newest_rx_seq=%llu consumed=%llu "
+                  "tx_ctr(as sender to %u)=%llu credit=%llu\n",
+                  s, (unsigned long long)newest,
+                  (unsigned long long)stream_rx_seq_[s], s,
+                  (unsigned long long)ctl.tx_ctr,
+                  (unsigned long long)ctl.credit);
+    out += b;
+  }
+  return out;
 }
 
 Request* ACCL::recv(BaseBuffer& dst, u64 count, u32 src, u32 tag, u32 comm,

@@ -13,6 +13,7 @@
 //    backed by the descriptor ring sequence number.
 #pragma once
 #include <memory>
+#include <string>
 #include <vector>
 #include "backend.hpp"
 
@@ -172,6 +173,13 @@ class ACCL {
   bool stream_ready(u32 src);
 
   void free_request(Request* r);
+
+  // --- debug dumps (reference: ACCL::dump_rx_buffers / dump_communicator /
+  // dump_exchange_memory, driver/xrt/src/accl.cpp:964-1048) ---
+  std::string dump_communicator(u32 comm = GLOBAL_COMM);
+  std::string dump_eager_rx_buffers(bool verbose = false);
+  std::string dump_streams();
+  std::string dump_engine_status();
   u32 comm_size(u32 comm) const { return comm_sizes_.at(comm); }
   u32 comm_rank(u32 comm) const { return comm_ranks_.at(comm); }
 

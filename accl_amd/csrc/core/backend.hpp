@@ -115,6 +115,27 @@ class Backend {
     if (head_) wait(head_ - 1);
   }
 
+  // liveness/debug view of the control page (reference: exchange-memory
+  // debug dumps, accl.cpp:964-1048; heartbeat has no direct analogue — the
+  // persistent engine replaces the MicroBlaze's implicit liveness)
+  struct CtrlView {
+    u64 doorbell, shutdown, comm_gen, ncomms, heartbeat, engine_up;
+    u64 submitted, retired;
+  };
+  CtrlView ctrl_view() {
+    CtrlView v{};
+    v.doorbell = __atomic_load_n((u64*)&ring_->ctrl.doorbell, __ATOMIC_RELAXED);
+    v.shutdown = ring_->ctrl.shutdown;
+    v.comm_gen = ring_->ctrl.comm_gen;
+    v.ncomms = ring_->ctrl.ncomms;
+    v.heartbeat = ring_->ctrl.heartbeat;
+    v.engine_up = ring_->ctrl.engine_up;
+    v.submitted = head_;
+    v.retired = tail_retired();
+    return v;
+  }
+  const CommView& comm_view(u32 id) const { return ring_->comm_mirror[id]; }
+
  protected:
   u64 tail_retired() {
     // ring slots free once their RetEntry is published

@@ -403,3 +403,35 @@ def test_stream_credit_wrap():
 
 def test_stream_self():
     run_ranks(_stream_self, 1)
+
+
+# ---------------------------------------------------------- debug dumps
+# reference: ACCL::dump_communicator / dump_rx_buffers / exchange-memory
+# dumps (driver/xrt/src/accl.cpp:964-1048, 1429-1439)
+def _dumps(a, rank, n):
+    s = a.dump_communicator(0)
+    assert f"rank {rank} of {n}" in s
+    st = a.dump_engine_status()
+    assert "up=1" in st and "doorbell=" in st
+    cnt = 2000
+    b = a.create_buffer(cnt, DT.float32)
+    b.write(pattern(cnt, rank))
+    if rank == 0:
+        a.send(b, cnt, dst=1, tag=1)
+    elif rank == 1:
+        d = a.create_buffer(cnt, DT.float32)
+        a.recv(d, cnt, src=0, tag=1)
+        rx = a.dump_eager_rx_buffers()
+        assert "from rank 0: newest_seq=" in rx
+    a.barrier()
+    s2 = a.create_buffer(100, DT.float32)
+    s2.write(pattern(100, 1))
+    a.stream_put(s2, 100, dst=rank, tag=2)
+    out = np.zeros(100, np.float32)
+    a.pop_stream(rank, out)
+    ds = a.dump_streams()
+    assert "newest_rx_seq=1" in ds and "consumed=1" in ds
+
+
+def test_debug_dumps():
+    run_ranks(_dumps, 2)

@@ -262,6 +262,10 @@ PYBIND11_MODULE(_core, m) {
            },
            py::arg("src"), py::arg("out"), py::arg("timeout_ms") = 10000)
       .def("stream_ready", &ACCL::stream_ready, py::arg("src"))
+      .def("set_timeout_ms", &ACCL::set_timeout_ms,
+           py::call_guard<py::gil_scoped_release>())
+      .def("set_max_eager_size", &ACCL::set_max_eager_size,
+           py::call_guard<py::gil_scoped_release>())
       .def("dump_communicator", &ACCL::dump_communicator,
            py::arg("comm") = GLOBAL_COMM)
       .def("dump_eager_rx_buffers", &ACCL::dump_eager_rx_buffers,

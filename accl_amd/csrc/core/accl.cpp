@@ -250,6 +250,17 @@ u64 ACCL::pop_stream(u32 src, void* out, u64 max_bytes, u32* tag,
   return n;
 }
 
+void ACCL::set_timeout_ms(u64 ms) {
+  CallDesc d = make_desc(Op::config, ms, DataType::none, DataType::none);
+  d.function = u32(CfgFunc::set_timeout);
+  be_->call(d);
+}
+void ACCL::set_max_eager_size(u64 bytes) {
+  CallDesc d = make_desc(Op::config, bytes, DataType::none, DataType::none);
+  d.function = u32(CfgFunc::set_max_eager_size);
+  be_->call(d);
+}
+
 // ---------------- debug dumps ----------------
 // reference: dump_communicator (accl.cpp:1429-1439), dump_rx_buffers
 // (accl.cpp:964-1048: status/occupancy/tag/seqn per rx buffer), here read

@@ -174,6 +174,11 @@ class ACCL {
 
   void free_request(Request* r);
 
+  // --- runtime config (reference: set_timeout accl.cpp:1096, max eager /
+  // rendezvous sizes accl.hpp:103-104, cfgFunc calls) ---
+  void set_timeout_ms(u64 ms);
+  void set_max_eager_size(u64 bytes);
+
   // --- debug dumps (reference: ACCL::dump_rx_buffers / dump_communicator /
   // dump_exchange_memory, driver/xrt/src/accl.cpp:964-1048) ---
   std::string dump_communicator(u32 comm = GLOBAL_COMM);

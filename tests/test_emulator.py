@@ -435,3 +435,58 @@ def _dumps(a, rank, n):
 
 def test_debug_dumps():
     run_ranks(_dumps, 2)
+
+
+# ------------------------------------------------------ stress + config
+def _stress(a, rank, n):
+    # reference: test/host/xrt/src/stress.cpp — randomized repeated
+    # send/recv + collectives, fixed seed on every rank
+    rng = np.random.default_rng(123)
+    big = a.create_buffer(20000, DT.float32)
+    out = a.create_buffer(20000 * n, DT.float32)
+    for it in range(30):
+        op = rng.integers(0, 4)
+        cnt = int(rng.integers(1, 20000))
+        if op == 0:  # ping-pong
+            x = pattern(cnt, rank, seed=it)
+            big.write(x)
+            if rank == 0:
+                a.send(big, cnt, dst=1, tag=it)
+            elif rank == 1:
+                d = a.create_buffer(cnt, DT.float32)
+                a.recv(d, cnt, src=0, tag=it)
+                assert np.array_equal(rd(d, cnt), pattern(cnt, 0, seed=it))
+        elif op == 1:
+            big.write(pattern(cnt, rank, seed=it))
+            a.allreduce(big, out, cnt, RF.SUM)
+            exp = np.stack([pattern(cnt, r, seed=it) for r in range(n)]).sum(0)
+            assert np.allclose(rd(out, cnt), exp)
+        elif op == 2:
+            big.write(pattern(cnt, rank, seed=it))
+            a.allgather(big, out, cnt)
+            exp = np.concatenate([pattern(cnt, r, seed=it) for r in range(n)])
+            assert np.array_equal(rd(out, cnt * n), exp)
+        else:
+            a.barrier()
+    a.barrier()
+
+
+def test_stress():
+    run_ranks(_stress, 2, opts=SMALL, timeout=240)
+
+
+def _cfg_and_perf(a, rank, n):
+    # reference: set_timeout / max-eager cfg calls + PERFCNT duration
+    a.set_timeout_ms(30000)
+    a.set_max_eager_size(1 << 20)
+    cnt = 4096
+    s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt, DT.float32)
+    s.write(pattern(cnt, rank))
+    r = a.allreduce(s, d, cnt, RF.SUM, run_async=True)
+    r.wait()
+    assert r.retcode() == 0
+    assert r.duration_us() > 0.0  # engine perf counter (PERFCNT analogue)
+
+
+def test_config_and_perfcounter():
+    run_ranks(_cfg_and_perf, 2)

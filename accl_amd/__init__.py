@@ -35,9 +35,9 @@ TAG_ANY = _core.TAG_ANY
 GLOBAL_COMM = _core.GLOBAL_COMM
 error_to_string = _core.error_to_string
 
-from .accl import ACCL, emu_job_name  # noqa: E402,F401
+from .accl import ACCL, emu_job_name, generate_ranks  # noqa: E402,F401
 
 __all__ = [
     "ACCL", "DataType", "ReduceFunction", "TAG_ANY", "GLOBAL_COMM",
-    "error_to_string", "emu_job_name",
+    "error_to_string", "emu_job_name", "generate_ranks",
 ]

@@ -6,12 +6,32 @@ test/host/xrt/include/fixture.hpp:127); here the bootstrap allgather of the
 arena IPC handles runs over torch.distributed when it is initialized, and is
 computed locally for the emulator (shm names are deterministic).
 """
+import json
 import os
 
 from . import _core
 
 DataType = _core.DataType
 ReduceFunction = _core.ReduceFunction
+
+
+def generate_ranks(config_file=None, local=True, world_size=1,
+                   start_port=5500, rxbuf_size=1 << 20):
+    """Reference-compatible rank map (reference: accl_network_utils
+    generate_ranks, driver/utils/accl_network_utils/include/
+    accl_network_utils.hpp:41-53; config JSON = {"ips": [...]}).
+
+    On MI355X the transport is xGMI peer memory, so ip/port are carried for
+    config-format compatibility but do not open sockets; max_segment_size
+    maps to the eager slot size (the reference's rxbuf_size)."""
+    if config_file is not None:
+        with open(config_file) as f:
+            ips = json.load(f)["ips"]
+    else:
+        ips = ["127.0.0.1" if local else f"10.10.10.{i + 1}"
+               for i in range(world_size)]
+    return [{"ip": ip, "port": start_port + i, "session_id": i,
+             "max_segment_size": rxbuf_size} for i, ip in enumerate(ips)]
 
 
 def emu_job_name(seed=None):
@@ -47,7 +67,15 @@ class ACCL:
     """
 
     def __init__(self, nranks=None, rank=None, backend="auto", job=None,
-                 device=None, heap_bytes=None, bootstrap="auto", **opts):
+                 device=None, heap_bytes=None, bootstrap="auto", ranks=None,
+                 **opts):
+        if ranks is not None:
+            # reference-style rank map (generate_ranks): world size + eager
+            # slot size come from it (reference: ACCL ctor takes the rank
+            # vector, driver/xrt/include/accl.hpp:57)
+            nranks = len(ranks)
+            opts.setdefault("opts", {}).setdefault(
+                "slot_bytes", int(ranks[0].get("max_segment_size", 1 << 20)))
         if nranks is None:
             nranks = int(os.environ.get("WORLD_SIZE", "1"))
         if rank is None:

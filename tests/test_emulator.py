@@ -490,3 +490,42 @@ def _cfg_and_perf(a, rank, n):
 
 def test_config_and_perfcounter():
     run_ranks(_cfg_and_perf, 2)
+
+
+def test_generate_ranks(tmp_path):
+    # reference config format: {"ips": [...]} (accl_network_utils get_ips)
+    cfg = tmp_path / "ranks.json"
+    cfg.write_text('{"ips": ["10.0.0.1", "10.0.0.2"]}')
+    rk = A.generate_ranks(config_file=str(cfg), rxbuf_size=8192)
+    assert len(rk) == 2 and rk[1]["port"] == 5501
+    assert rk[0]["max_segment_size"] == 8192
+    rk = A.generate_ranks(local=True, world_size=3)
+    assert [r["ip"] for r in rk] == ["127.0.0.1"] * 3
+
+
+def _ranks_ctor(a, rank, n):
+    cnt = 500
+    s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt, DT.float32)
+    s.write(pattern(cnt, rank))
+    a.allreduce(s, d, cnt, RF.SUM)
+    exp = np.stack([pattern(cnt, r) for r in range(n)]).sum(0)
+    assert np.allclose(rd(d, cnt), exp)
+
+
+def test_rank_map_ctor():
+    # construct through the reference-style rank map
+    def worker(fn, r, n, job, opts, q, backend):
+        pass
+    import emu_util
+
+    def fn(a, rank, n):
+        _ranks_ctor(a, rank, n)
+    # run via run_ranks but passing ranks through opts is the harness's job;
+    # direct 1-rank construction check here:
+    rk = A.generate_ranks(local=True, world_size=1, rxbuf_size=16384)
+    a = A.ACCL(rank=0, backend="emu", job="rkmap", ranks=rk)
+    try:
+        assert a.nranks == 1
+        _ranks_ctor(a, 0, 1)
+    finally:
+        a.close()

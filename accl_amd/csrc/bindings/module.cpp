@@ -262,6 +262,22 @@ PYBIND11_MODULE(_core, m) {
            },
            py::arg("src"), py::arg("out"), py::arg("timeout_ms") = 10000)
       .def("stream_ready", &ACCL::stream_ready, py::arg("src"))
+      .def("info",
+           [](ACCL& a) {
+             // reference: parse_hwid capability decode (accl.cpp:1050-1064)
+             const ProtoConfig& c = a.backend()->cfg();
+             py::dict d;
+             d["backend"] = a.backend()->is_gpu() ? "gpu" : "emu";
+             d["rank"] = c.rank;
+             d["nranks"] = c.nranks;
+             d["n_eager_slots"] = c.n_slots;
+             d["eager_slot_bytes"] = c.slot_bytes;
+             d["n_stream_slots"] = c.n_stream;
+             d["stream_slot_bytes"] = c.stream_bytes;
+             d["max_eager_bytes"] = c.max_eager;
+             d["timeout_us"] = c.timeout_us;
+             return d;
+           })
       .def("set_timeout_ms", &ACCL::set_timeout_ms,
            py::call_guard<py::gil_scoped_release>())
       .def("set_max_eager_size", &ACCL::set_max_eager_size,
@@ -315,6 +331,27 @@ PYBIND11_MODULE(_core, m) {
         py::arg("opts") = py::dict());
 
   m.def("error_to_string", &error_to_string);
+
+  // deployment introspection (reference: xclbin_scan enumerating kernels /
+  // memory banks, driver/utils/xclbin_scan)
+  m.def("device_info", []() {
+    py::list out;
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess) return out;
+    for (int i = 0; i < n; ++i) {
+      hipDeviceProp_t p{};
+      if (hipGetDeviceProperties(&p, i) != hipSuccess) continue;
+      py::dict d;
+      d["index"] = i;
+      d["name"] = std::string(p.name);
+      d["gcn_arch"] = std::string(p.gcnArchName);
+      d["total_mem_gb"] = double(p.totalGlobalMem) / (1u << 30);
+      d["multiprocessors"] = p.multiProcessorCount;
+      d["xgmi_capable"] = bool(p.isLargeBar);
+      out.append(d);
+    }
+    return out;
+  });
 
   // demo plugin: device-initiated stream_put (reference vadd_put analogue)
   m.def("demo_vadd_put",

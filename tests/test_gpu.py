@@ -209,3 +209,33 @@ def _stream2(a, rank, n):
 
 def test_two_ranks_stream():
     run_ranks(_stream2, 2, backend="gpu", timeout=180)
+
+
+def _subcomm2(a, rank, n):
+    """Subgroup communicator on the GPU engine (reference multicomm tests,
+    test/host/xrt/src/test.cpp:756-833)."""
+    sub = a.split_communicator([0, 1])
+    cnt = 10000
+    s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt, DT.float32)
+    s.write(pattern(cnt, rank, seed=11))
+    a.allreduce(s, d, cnt, RF.SUM, comm=sub)
+    exp = np.stack([pattern(cnt, r, seed=11) for r in range(n)]).sum(0)
+    assert np.allclose(rd(d, cnt), exp)
+    a.barrier(comm=sub)
+
+
+def _compressed2(a, rank, n):
+    """f32 data, f16 wire dtype (reference compression lane tests)."""
+    cnt = 4096
+    s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt, DT.float32)
+    x = pattern(cnt, rank, seed=13) / 64.0
+    s.write(x)
+    a.allreduce(s, d, cnt, RF.SUM, compress_dtype=DT.float16)
+    exp = np.stack([pattern(cnt, r, seed=13) / 64.0 for r in range(n)]).sum(0)
+    assert np.allclose(rd(d, cnt), exp, atol=0.05), "f16-wire allreduce"
+
+
+@pytest.mark.parametrize("fn", [_subcomm2, _compressed2],
+                         ids=["subcomm", "compressed"])
+def test_two_ranks_extra(fn):
+    run_ranks(fn, 2, backend="gpu", timeout=180)

@@ -529,3 +529,27 @@ def test_rank_map_ctor():
         _ranks_ctor(a, 0, 1)
     finally:
         a.close()
+
+
+def test_cxx_api_example():
+    """The C++ facade is first-class: compile and run the standalone example
+    against the already-built objects (reference: the C++ gtest suite is the
+    reference's primary surface, test/host/xrt/src/test.cpp)."""
+    import pathlib
+    import subprocess
+    root = pathlib.Path(__file__).resolve().parent.parent
+    b = root / "accl_amd" / ".build"
+    objs = [b / "core_util.cpp.o", b / "core_accl.cpp.o",
+            b / "emu_emudevice.cpp.o"]
+    if not all(o.exists() for o in objs):
+        pytest.skip("build objects missing (run python -m accl_amd.build)")
+    exe = "/tmp/accl_cxx_example"
+    r = subprocess.run(
+        ["g++", "-std=c++17", "-O2",
+         str(root / "examples/cxx/allreduce_emu.cpp"),
+         *[str(o) for o in objs], "-o", exe, "-lpthread"],
+        capture_output=True, text=True, timeout=240)
+    assert r.returncode == 0, r.stderr[-2000:]
+    r = subprocess.run([exe], capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr[-1000:]
+    assert "cxx example OK" in r.stdout

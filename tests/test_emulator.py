@@ -553,3 +553,79 @@ def test_cxx_api_example():
     r = subprocess.run([exe], capture_output=True, text=True, timeout=120)
     assert r.returncode == 0, r.stderr[-1000:]
     assert "cxx example OK" in r.stdout
+
+
+# ---------------------------------------------- reduce roots x functions
+# reference: reduce parameterized over (root, func) (test.cpp reduce matrix)
+def _reduce_roots_funcs(a, rank, n):
+    cnt = 700
+    for root in range(n):
+        for f, npf in ((RF.SUM, np.sum), (RF.MAX, np.max)):
+            s, d = _mk(a, cnt), _mk(a, cnt)
+            s.write(pattern(cnt, rank, seed=root * 7))
+            a.reduce(s, d, cnt, root, f)
+            if rank == root:
+                stk = np.stack([pattern(cnt, r, seed=root * 7)
+                                for r in range(n)])
+                exp = stk.sum(0) if f == RF.SUM else stk.max(0)
+                assert np.allclose(rd(d, cnt), exp), f"root={root} f={f}"
+    a.barrier()
+
+
+def test_reduce_roots_funcs():
+    run_ranks(_reduce_roots_funcs, 3, opts=SMALL, timeout=240)
+
+
+# --------------------------------------------- subgroup allgather (comms)
+# reference: allgather_comms (test.cpp allgather over split communicator)
+def _allgather_subcomm(a, rank, n):
+    half = [r for r in range(n) if r >= n // 2]
+    if rank in half:
+        cid = a.split_communicator(half)
+        cnt = 300
+        s = _mk(a, cnt)
+        d = _mk(a, cnt * len(half))
+        s.write(pattern(cnt, rank, seed=21))
+        a.allgather(s, d, cnt, comm=cid)
+        exp = np.concatenate([pattern(cnt, r, seed=21) for r in half])
+        assert np.array_equal(rd(d, cnt * len(half)), exp)
+    a.barrier()
+
+
+def test_allgather_subcomm():
+    run_ranks(_allgather_subcomm, 4, opts=SMALL, timeout=240)
+
+
+# --------------------------------------------------------- error paths
+# reference: 27-bit error word decoded host-side (check_return_value,
+# accl.cpp:1210-1234) — mismatches must surface as named errors, not hangs
+def _err_timeout(a, rank, n):
+    cnt = 64
+    if rank == 0:
+        a.set_timeout_ms(300)
+        try:
+            d = _mk(a, cnt)
+            a.recv(d, cnt, src=1, tag=999)  # never sent
+            raise AssertionError("recv of unsent message must fail")
+        except RuntimeError as e:
+            assert "TIMEOUT" in str(e)
+        a.set_timeout_ms(10000)  # restore before the closing barrier
+    a.barrier()
+
+
+def test_error_timeout():
+    run_ranks(_err_timeout, 2)
+
+
+def _err_badcomm(a, rank, n):
+    cnt = 16
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    try:
+        a.allreduce(s, d, cnt, RF.SUM, comm=9)
+        raise AssertionError("bad communicator must fail")
+    except RuntimeError as e:
+        assert "COMM" in str(e)
+
+
+def test_error_badcomm():
+    run_ranks(_err_badcomm, 1)

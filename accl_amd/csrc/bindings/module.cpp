@@ -353,6 +353,91 @@ PYBIND11_MODULE(_core, m) {
     return out;
   });
 
+  // host-side injector into the device-call ring: exercises the
+  // client_arbiter path on the emulator (and doubles as a raw call API)
+  m.def("inject_device_call",
+        [](ACCL& a, u32 scenario, u64 count, u32 root, u32 tag, u64 addr0,
+           u64 addr2, u32 flags, u32 function) {
+          Backend* be = a.backend();
+          const ProtoConfig& c = be->cfg();
+          ArenaLayout L = arena_layout(c);
+          DevCallRing ring{};
+          be->read_arena(L.devcall_off, &ring, sizeof(ring));
+          u64 idx = ring.head;
+          ring.head = idx + 1;
+          be->write_arena(L.devcall_off, &ring, sizeof(u64));
+          DevCallSlot slot{};
+          slot.d.scenario = scenario;
+          slot.d.count_lo = u32(count);
+          slot.d.count_hi = u32(count >> 32);
+          slot.d.root_src_dst = root;
+          slot.d.tag = tag;
+          slot.d.addr0 = addr0;
+          slot.d.addr2 = addr2;
+          slot.d.flags = flags;
+          slot.d.function = function;
+          slot.d.arith = u32(DataType::float32) | (u32(DataType::float32) << 8);
+          u64 off = L.devcall_off + sizeof(DevCallRing) +
+                    (idx % DEVCALL_RING) * sizeof(DevCallSlot);
+          be->write_arena(off + offsetof(DevCallSlot, d), &slot.d,
+                          sizeof(CallDesc));
+          u64 seq = idx + 1;
+          be->write_arena(off + offsetof(DevCallSlot, seq), &seq, sizeof(seq));
+          return idx;
+        })
+  ;
+  m.def("wait_device_call",
+        [](ACCL& a, u64 token, u64 timeout_ms) {
+          Backend* be = a.backend();
+          ArenaLayout L = arena_layout(be->cfg());
+          u64 off = L.devcall_off + sizeof(DevCallRing) +
+                    u64(DEVCALL_RING) * sizeof(DevCallSlot) +
+                    (token % DEVCALL_RING) * sizeof(DevCallRet);
+          DevCallRet r{};
+          u64 t0 = wallclock_host_ns();
+          for (;;) {
+            be->read_arena(off, &r, sizeof(r));
+            if (r.seq >= token + 1) return u32(r.errcode);
+            if (wallclock_host_ns() - t0 > timeout_ms * 1000000ull)
+              throw accl_error("wait_device_call: timeout");
+            cpu_pause();
+          }
+        },
+        py::arg("a"), py::arg("token"), py::arg("timeout_ms") = 10000);
+
+  m.def("demo_vadd_devicecall",
+        [](ACCL& a, BaseBuffer& src, BaseBuffer& scratch, u64 count, u32 dst,
+           u32 tag, float addv) {
+          auto* g = dynamic_cast<GpuDevice*>(a.backend());
+          if (!g) throw accl_error("demo_vadd_devicecall: gpu backend only");
+          hipStream_t st{};
+          if (hipStreamCreateWithFlags(&st, hipStreamNonBlocking) != hipSuccess)
+            throw accl_error("demo: stream create failed");
+          launch_vadd_devicecall(
+              g->arena_local() + src.arena_offset(),
+              g->arena_local() + scratch.arena_offset(),
+              scratch.arena_offset(), count, tag, g->arena_local(), dst, addv,
+              st);
+          hipError_t e = hipGetLastError();
+          u64 t0 = wallclock_host_ns();
+          while (e == hipSuccess) {
+            hipError_t q = hipStreamQuery(st);
+            if (q != hipErrorNotReady) { e = q; break; }
+            if (wallclock_host_ns() - t0 > 30ull * 1000000000) {
+              e = hipErrorUnknown;
+              break;
+            }
+            usleep(100);
+          }
+          (void)hipStreamDestroy(st);
+          if (e != hipSuccess)
+            throw accl_error(std::string("demo_vadd_devicecall: ") +
+                             hipGetErrorString(e));
+        },
+        py::arg("a"), py::arg("src"), py::arg("scratch"), py::arg("count"),
+        py::arg("dst"), py::arg("tag") = 0, py::arg("addv") = 1.0f,
+        py::call_guard<py::gil_scoped_release>());
+
   // demo plugin: device-initiated stream_put (reference vadd_put analogue)
   m.def("demo_vadd_put",
         [](ACCL& a, BaseBuffer& src, u64 count, u32 dst, u32 tag, float addv) {

@@ -181,6 +181,29 @@ struct alignas(64) EagerChanCtl {
 // (field `credit` of s's EagerChanCtl[r]) and is advanced by r. Each side
 // polls only its own HBM.
 
+// Device-initiated call ring, resident in the OWNER's arena: device kernels
+// (or test harnesses) allocate a slot via atomic head, publish a CallDesc by
+// seq, and the engine consumes it alongside the host ring — the analogue of
+// the reference's client_arbiter merging PL-kernel and host command streams
+// (kernels/plugins/client_arbiter/client_arbiter.cpp:21-51).
+constexpr u32 DEVCALL_RING = 32;
+struct alignas(64) DevCallRing {
+  u64 head;                 // atomic allocator (producers fetch_add)
+  u64 _pad[7];
+  // CallDesc[DEVCALL_RING] follows, then RetEntry-style u64 seq+err pairs
+};
+struct alignas(64) DevCallSlot {
+  u64 seq;                  // published last by the producer (idx+1)
+  u64 _pad0[7];
+  CallDesc d;
+};
+struct alignas(64) DevCallRet {
+  u64 seq;                  // published by the engine when done (idx+1)
+  u64 errcode;
+  u64 t_start, t_end;
+  u64 _pad[4];
+};
+
 struct alignas(64) ArenaHdr {
   u32 magic;            // 'ACCL'
   u32 version;
@@ -198,12 +221,13 @@ struct alignas(64) ArenaHdr {
   u64 direct_off;       // per-peer cumulative direct-write progress u64[nranks]
   u64 spare_off;        // staging region for non-arena rendezvous targets
   u64 spare_bytes;      //   (reference: spare buffers, accl.cpp:1174-1196)
+  u64 devcall_off;      // DevCallRing + slots + rets (device-initiated calls)
 };
 constexpr u32 ARENA_MAGIC = 0x4143434Cu;  // "ACCL"
 
 struct ArenaLayout {
   u64 eager_off, rndzv_addr_off, rndzv_done_off, stream_off, slots_off,
-      barrier_off, direct_off, spare_off, spare_bytes, heap_off,
+      barrier_off, direct_off, spare_off, spare_bytes, devcall_off, heap_off,
       total_ctl_bytes;
 };
 
@@ -226,6 +250,9 @@ inline ArenaLayout arena_layout(const ProtoConfig& c, u64 spare_bytes = 32u << 2
   off += u64(c.nranks) * sizeof(u64);
   L.direct_off = off = align_up(off, 256);
   off += u64(c.nranks) * sizeof(u64);
+  L.devcall_off = off = align_up(off, 256);
+  off += sizeof(DevCallRing) + u64(DEVCALL_RING) * sizeof(DevCallSlot) +
+         u64(DEVCALL_RING) * sizeof(DevCallRet);
   L.slots_off = off = align_up(off, 4096);
   off += u64(c.nranks) * c.n_slots * u64(c.slot_bytes);
   L.spare_off = off = align_up(off, 4096);
@@ -277,6 +304,18 @@ struct TransportView {
     return (volatile u64*)(arena[r] + hdr(r)->direct_off) + s;
   }
   ACCL_HD char* heap_ptr(u32 r, u64 off) const { return arena[r] + off; }
+
+  // ---- device-call ring (own arena only) ----
+  ACCL_HD DevCallRing* devcall_ring(u32 r) const {
+    return (DevCallRing*)(arena[r] + hdr(r)->devcall_off);
+  }
+  ACCL_HD DevCallSlot* devcall_slot(u32 r, u32 i) const {
+    return (DevCallSlot*)((char*)devcall_ring(r) + sizeof(DevCallRing)) + i;
+  }
+  ACCL_HD DevCallRet* devcall_ret(u32 r, u32 i) const {
+    return (DevCallRet*)((char*)devcall_slot(r, 0) +
+                         u64(DEVCALL_RING) * sizeof(DevCallSlot)) + i;
+  }
 
   // ---- stream channel (s -> r): ctl+hdrs+payload in r's arena lane [s];
   // credit word in s's arena lane [r] (advanced by the consumer at r).

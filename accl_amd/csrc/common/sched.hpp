@@ -864,6 +864,31 @@ struct Cclo {
     return run_flows(1);
   }
 
+  // Consume pending device-initiated calls (client_arbiter analogue):
+  // returns number executed. Called from both engine main loops between
+  // host-ring batches.
+  ACCL_HD u32 poll_device_calls(u64& consumed) {
+    DevCallRing* ring = tv.devcall_ring(me());
+    u32 did = 0;
+    for (;;) {
+      u32 i = u32(consumed % DEVCALL_RING);
+      DevCallSlot* s = tv.devcall_slot(me(), i);
+      if (ld_sys(&s->seq) != consumed + 1) break;
+      fence_acquire_sys();
+      DevCallRet* r = tv.devcall_ret(me(), i);
+      r->t_start = wallclock();
+      u32 e = run_call(s->d);
+      r->t_end = wallclock();
+      r->errcode = e;
+      fence_release_sys();
+      st_sys(&r->seq, consumed + 1);
+      consumed++;
+      did++;
+    }
+    (void)ring;
+    return did;
+  }
+
   // ---------------- dispatch ----------------
   // reference: run() scenario switch (ccl_offload_control.c:2375-2459)
   ACCL_HD u32 run_call(const CallDesc& d) {

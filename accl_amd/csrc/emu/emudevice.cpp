@@ -52,6 +52,7 @@ EmuDevice::EmuDevice(u32 nranks, u32 rank, const std::string& job,
   h->direct_off = layout_.direct_off;
   h->spare_off = layout_.spare_off;
   h->spare_bytes = layout_.spare_bytes;
+  h->devcall_off = layout_.devcall_off;
   __atomic_store_n(&h->magic, ARENA_MAGIC, __ATOMIC_RELEASE);
 
   heap_.init(layout_.heap_off, arena_bytes_ - layout_.heap_off);
@@ -136,22 +137,24 @@ void EmuDevice::engine_main() {
   auto& C = eng_->cclo;
   CtrlPage& ctrl = ring_->ctrl;
   __atomic_store_n((u64*)&ctrl.engine_up, 1, __ATOMIC_RELEASE);
-  u64 consumed = 0;
+  u64 consumed = 0, dev_consumed = 0;
   for (;;) {
-    u64 db = __atomic_load_n((u64*)&ctrl.doorbell, __ATOMIC_ACQUIRE);
-    if (consumed == db) {
-      if (__atomic_load_n((u64*)&ctrl.shutdown, __ATOMIC_RELAXED)) return;
-      __atomic_fetch_add((u64*)&ctrl.heartbeat, 1, __ATOMIC_RELAXED);
-      usleep(20);
-      continue;
-    }
     // refresh communicator cache (reference: run() re-caches the
-    // communicator per call, ccl_offload_control.c:2308-2360)
+    // communicator per call, ccl_offload_control.c:2308-2360); must precede
+    // device-ring calls too, which name communicators
     u64 gen = __atomic_load_n((u64*)&ctrl.comm_gen, __ATOMIC_ACQUIRE);
     if (gen != eng_->cached_comm_gen) {
       C.ncomms = u32(__atomic_load_n((u64*)&ctrl.ncomms, __ATOMIC_ACQUIRE));
       for (u32 i = 0; i < C.ncomms; ++i) C.comms[i] = ring_->comm_mirror[i];
       eng_->cached_comm_gen = gen;
+    }
+    u64 db = __atomic_load_n((u64*)&ctrl.doorbell, __ATOMIC_ACQUIRE);
+    if (consumed == db) {
+      if (C.poll_device_calls(dev_consumed)) continue;
+      if (__atomic_load_n((u64*)&ctrl.shutdown, __ATOMIC_RELAXED)) return;
+      __atomic_fetch_add((u64*)&ctrl.heartbeat, 1, __ATOMIC_RELAXED);
+      usleep(20);
+      continue;
     }
     while (consumed < db) {
       const CallDesc d = ring_->descs[consumed % RING_CAP];

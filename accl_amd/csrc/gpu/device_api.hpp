@@ -156,6 +156,49 @@ __device__ inline u32 stream_pop(const StreamRx& r, u64 seq, void* dst,
   return n;
 }
 
+// ---------------- device-initiated collective calls ----------------
+// A kernel enqueues a full collective on its OWN engine (the reference's
+// ACCLCommand::start_call path, driver/hls/accl_hls.h:134-188, arbitrated
+// with host calls like client_arbiter.cpp:21-51). Lane-0-only (call from
+// one lane or guard with lane==0); returns a token for device_call_wait.
+__device__ inline u64 device_call(char* my_arena, const CallDesc& d) {
+  const ArenaHdr* h = (const ArenaHdr*)my_arena;
+  DevCallRing* ring = (DevCallRing*)(my_arena + h->devcall_off);
+  u64 idx = __hip_atomic_fetch_add(&ring->head, 1ull, __ATOMIC_RELAXED,
+                                   ACCL_DEV_SYS);
+  DevCallSlot* s = (DevCallSlot*)((char*)ring + sizeof(DevCallRing)) +
+                   (idx % DEVCALL_RING);
+  // slot reuse guard: wait until the previous occupant's ret was published
+  if (idx >= DEVCALL_RING) {
+    DevCallRet* prev =
+        (DevCallRet*)((char*)ring + sizeof(DevCallRing) +
+                      u64(DEVCALL_RING) * sizeof(DevCallSlot)) +
+        (idx % DEVCALL_RING);
+    while (__hip_atomic_load(&prev->seq, __ATOMIC_RELAXED, ACCL_DEV_SYS) <
+           idx + 1 - DEVCALL_RING)
+      __builtin_amdgcn_s_sleep(16);
+  }
+  s->d = d;
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __hip_atomic_store(&s->seq, idx + 1, __ATOMIC_RELAXED, ACCL_DEV_SYS);
+  return idx;
+}
+
+// Poll completion of a device_call; returns errcode (0 = OK). Blocking.
+__device__ inline u32 device_call_wait(char* my_arena, u64 token) {
+  const ArenaHdr* h = (const ArenaHdr*)my_arena;
+  DevCallRing* ring = (DevCallRing*)(my_arena + h->devcall_off);
+  DevCallRet* r = (DevCallRet*)((char*)ring + sizeof(DevCallRing) +
+                                u64(DEVCALL_RING) * sizeof(DevCallSlot)) +
+                  (token % DEVCALL_RING);
+  while (__hip_atomic_load(&r->seq, __ATOMIC_RELAXED, ACCL_DEV_SYS) < token + 1)
+    __builtin_amdgcn_s_sleep(16);
+  __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+  return u32(r->errcode);
+}
+
 #undef ACCL_DEV_SYS
 
 }  // namespace device_api

@@ -18,6 +18,9 @@ void launch_vadd_put(const void* in, u64 count, u32 tag, void* my_arena,
 void launch_stream_drain(void* out, u64 max_elems, u32 nseg, void* my_arena,
                          void* peer_arena, u32 me, u32 peer, u64 start_seq,
                          void* stream);
+void launch_vadd_devicecall(const void* in, void* scratch, u64 scratch_off,
+                            u64 count, u32 tag, void* my_arena, u32 dst_rank,
+                            float addv, void* stream);
 
 class GpuDevice : public Backend {
  public:

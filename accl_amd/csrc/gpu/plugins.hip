@@ -40,6 +40,44 @@ __global__ void __launch_bounds__(64, 1) stream_drain_kernel(
   }
 }
 
+// Full reference-flow demo: the kernel computes into an ARENA buffer and
+// then issues the engine's own stream_put through the device-call ring —
+// kernel -> device_call -> scheduler -> movers (vadd_put.cpp's
+// ACCLCommand::stream_put flow, arbitrated with host calls).
+__global__ void __launch_bounds__(64, 1) vadd_devicecall_kernel(
+    const float* in, float* scratch_arena_buf, u64 scratch_off, u64 count,
+    u32 tag, char* my_arena, u32 dst_rank, float addv) {
+  const int lane = int(threadIdx.x) & 63;
+  for (u64 i = lane; i < count; i += 64)
+    scratch_arena_buf[i] = in[i] + addv;
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  if (lane == 0) {
+    CallDesc d{};
+    d.scenario = u32(Op::stream_put);
+    d.count_lo = u32(count & 0xFFFFFFFFu);
+    d.count_hi = u32(count >> 32);
+    d.comm_id = 0;
+    d.root_src_dst = dst_rank;
+    d.tag = tag;
+    d.arith = u32(DataType::float32) | (u32(DataType::float32) << 8);
+    d.addr0 = scratch_off;
+    d.flags = F_SRC_ARENA;
+    u64 tok = device_api::device_call(my_arena, d);
+    u32 e = device_api::device_call_wait(my_arena, tok);
+    (void)e;
+  }
+}
+
+void launch_vadd_devicecall(const void* in, void* scratch, u64 scratch_off,
+                            u64 count, u32 tag, void* my_arena, u32 dst_rank,
+                            float addv, void* stream) {
+  hipLaunchKernelGGL(vadd_devicecall_kernel, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, (const float*)in, (float*)scratch,
+                     scratch_off, count, tag, (char*)my_arena, dst_rank, addv);
+}
+
 void launch_vadd_put(const void* in, u64 count, u32 tag, void* my_arena,
                      void* peer_arena, u32 me, u32 peer, u32 seg_bytes,
                      float addv, void* stream) {

@@ -629,3 +629,34 @@ def _err_badcomm(a, rank, n):
 
 def test_error_badcomm():
     run_ranks(_err_badcomm, 1)
+
+
+# ----------------------------------------- device-call ring (client_arbiter)
+# reference: PL kernels issue collectives through client_arbiter merged with
+# host calls (client_arbiter.cpp:21-51, accl_hls.h:134-188). The emulator
+# exercises the same ring the GPU device_api::device_call uses.
+def _devcall(a, rank, n):
+    import accl_amd._core as core
+    cnt = 1024
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    s.write(pattern(cnt, 9))
+    # Op::copy = 1, F_SRC_ARENA|F_DST_ARENA = (1<<4)|(1<<5)
+    tok = core.inject_device_call(a._a, 1, cnt, 0, 0,
+                                  s.arena_offset, d.arena_offset, 48, 0)
+    assert core.wait_device_call(a._a, tok) == 0
+    assert np.array_equal(rd(d, cnt), pattern(cnt, 9))
+    # interleave with a host-ring call: both paths share one engine
+    a.barrier() if n > 1 else None
+    # stream_put via device ring (scenario 14), consumed by pop_stream
+    tok = core.inject_device_call(a._a, 14, cnt, rank, 5,
+                                  s.arena_offset, 0, 16, 0)
+    assert core.wait_device_call(a._a, tok) == 0
+    out = np.zeros(cnt, np.float32)
+    nb, tag = a.pop_stream(rank, out)
+    assert nb == cnt * 4 and tag == 5
+    assert np.array_equal(out, pattern(cnt, 9))
+
+
+def test_device_call_ring():
+    run_ranks(_devcall, 1)
+    run_ranks(_devcall, 2)

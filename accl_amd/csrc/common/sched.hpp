@@ -864,6 +864,11 @@ struct Cclo {
     return run_flows(1);
   }
 
+  ACCL_HD bool device_call_pending(u64 consumed) const {
+    const DevCallSlot* s = tv.devcall_slot(cfg.rank, u32(consumed % DEVCALL_RING));
+    return ld_sys(&s->seq) == consumed + 1;
+  }
+
   // Consume pending device-initiated calls (client_arbiter analogue):
   // returns number executed. Called from both engine main loops between
   // host-ring batches.

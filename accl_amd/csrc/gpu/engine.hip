@@ -429,24 +429,27 @@ __device__ void scheduler_main(GpuEngineState* S, Cclo<GpuMover>& C) {
   st_sys(&ctrl->engine_up, 1);
   u64 consumed = 0, dev_consumed = 0, cached_gen = 0, beat = 0;
   for (;;) {
-    // communicator cache refresh precedes BOTH rings (device calls name
-    // communicators too). PCIe read cost is amortized by the gen check.
-    u64 gen = ld_sys(&ctrl->comm_gen);
-    if (gen != cached_gen) {
-      fence_acquire_sys();
-      C.ncomms = u32(ld_sys(&ctrl->ncomms));
-      for (u32 i = 0; i < C.ncomms; ++i) C.comms[i] = S->comm_mirror[i];
-      cached_gen = gen;
-    }
     u64 db = ld_sys(&ctrl->doorbell);
-    if (consumed == db) {
-      if (C.poll_device_calls(dev_consumed)) continue;
+    bool dev_pending = C.device_call_pending(dev_consumed);
+    if (consumed == db && !dev_pending) {
       if (ld_sys(&ctrl->shutdown)) break;
       if ((++beat & 0x3FF) == 0) st_sys(&ctrl->heartbeat, beat);
       __builtin_amdgcn_s_sleep(16);
       continue;
     }
     fence_acquire_sys();
+    // communicator cache refresh precedes BOTH rings (device calls name
+    // communicators too); only paid when there is work (PCIe read)
+    u64 gen = ld_sys(&ctrl->comm_gen);
+    if (gen != cached_gen) {
+      C.ncomms = u32(ld_sys(&ctrl->ncomms));
+      for (u32 i = 0; i < C.ncomms; ++i) C.comms[i] = S->comm_mirror[i];
+      cached_gen = gen;
+    }
+    if (dev_pending) {
+      C.poll_device_calls(dev_consumed);
+      continue;
+    }
     bool halted = false;
     while (consumed < db) {
       CallDesc d = S->descs[consumed % RING_CAP];

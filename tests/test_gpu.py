@@ -239,3 +239,21 @@ def _compressed2(a, rank, n):
                          ids=["subcomm", "compressed"])
 def test_two_ranks_extra(fn):
     run_ranks(fn, 2, backend="gpu", timeout=180)
+
+
+def test_device_call_collective(acc1):
+    """A HIP kernel issues the engine's own stream_put through the
+    device-call ring (reference: ACCLCommand via client_arbiter,
+    accl_hls.h:134-188 + client_arbiter.cpp:21-51)."""
+    import accl_amd._core as core
+    a = acc1
+    cnt = 3000
+    s = a.create_buffer(cnt, DT.float32)
+    scratch = a.create_buffer(cnt, DT.float32, device_only=True)
+    x = np.random.default_rng(8).standard_normal(cnt, dtype=np.float32)
+    s.write(x)
+    core.demo_vadd_devicecall(a._a, s, scratch, cnt, 0, 31, 2.5)
+    out = np.zeros(cnt, np.float32)
+    nb, tag = a.pop_stream(0, out)
+    assert nb == cnt * 4 and tag == 31
+    assert np.allclose(out, x + 2.5)

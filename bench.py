@@ -123,14 +123,15 @@ def main():
         a.alltoall(src, dst, per, from_device=True, to_device=True)
 
     def one_sendrecv():
-        if world == 1:
+        # disjoint pairs (0,1)(2,3)...; odd world: last rank self-copies
+        if world == 1 or (world % 2 == 1 and rank == world - 1):
             a.copy(src, dst, count, from_device=True, to_device=True)
         elif rank % 2 == 0:
-            peer = (rank + 1) % world
+            peer = rank + 1
             a.send(src, count, dst=peer, tag=1, from_device=True)
             a.recv(dst, count, src=peer, tag=2, to_device=True)
         else:
-            peer = (rank - 1) % world
+            peer = rank - 1
             a.recv(dst, count, src=peer, tag=1, to_device=True)
             a.send(src, count, dst=peer, tag=2, from_device=True)
 

@@ -187,6 +187,7 @@ class ACCL:
                     "gather", "allgather", "reduce", "allreduce",
                     "reduce_scatter", "alltoall", "barrier", "nop",
                     "stream_put", "pop_stream", "stream_ready",
+                    "copy_from_stream", "send_from_stream", "alive",
                     "info", "set_timeout_ms", "set_max_eager_size",
                     "dump_communicator", "dump_eager_rx_buffers",
                     "dump_streams", "dump_engine_status",

@@ -262,6 +262,17 @@ PYBIND11_MODULE(_core, m) {
            },
            py::arg("src"), py::arg("out"), py::arg("timeout_ms") = 10000)
       .def("stream_ready", &ACCL::stream_ready, py::arg("src"))
+      .def("copy_from_stream", &ACCL::copy_from_stream, py::arg("lane"),
+           py::arg("dst"), py::arg("count"), py::arg("to_device") = false,
+           py::arg("run_async") = false, py::return_value_policy::reference,
+           py::keep_alive<0, 1>(), py::call_guard<py::gil_scoped_release>())
+      .def("send_from_stream", &ACCL::send_from_stream, py::arg("lane"),
+           py::arg("count"), py::arg("dst"), py::arg("tag") = u32(TAG_ANY),
+           py::arg("comm") = GLOBAL_COMM,
+           py::arg("compress_dtype") = DataType::none,
+           py::arg("run_async") = false, py::return_value_policy::reference,
+           py::keep_alive<0, 1>(), py::call_guard<py::gil_scoped_release>())
+      .def("alive", &ACCL::alive)
       .def("info",
            [](ACCL& a) {
              // reference: parse_hwid capability decode (accl.cpp:1050-1064)

@@ -350,6 +350,8 @@ struct PairSeq {
   u64 rndzv_done_tx[MAX_RANKS];
   u64 rndzv_done_rx[MAX_RANKS];
   u64 stream_tx[MAX_RANKS];    // stream segments sent to peer
+  u64 stream_fed_rx[MAX_RANKS];// segments the ENGINE consumed as a stream-fed
+                               // op source (one consumer per lane)
   u64 direct_tx[MAX_RANKS];    // cumulative bytes direct-written to peer
   u64 direct_rx[MAX_RANKS];    // cumulative bytes direct-received from peer
   u64 barrier_epoch[MAX_RANKS];// per-PAIR barrier epoch (must match both ends)

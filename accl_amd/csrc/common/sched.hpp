@@ -437,6 +437,9 @@ struct Cclo {
   ACCL_HD u32 op_copy(const CallDesc& d) {
     stamp(8);
     u64 n = desc_count(d);
+    if (d.flags & F_SRC_STREAM)  // stream2mem: drain ring lane addr0
+      return stream_fed(u32(d.addr0), local_ptr(d.addr2, d.flags & F_DST_ARENA),
+                        desc_dtype(d), n, -1, desc_dtype(d), 0);
     mk_local(0, local_ptr(d.addr0, d.flags & F_SRC_ARENA), desc_dtype(d),
              local_ptr(d.addr2, d.flags & F_DST_ARENA), desc_dtype(d), n);
     stamp(9);
@@ -465,6 +468,8 @@ struct Cclo {
     u64 n = desc_count(d);
     u32 peer = c.global(d.root_src_dst);
     DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    if (d.flags & F_SRC_STREAM)  // send-from-stream (OP0_STREAM analogue)
+      return stream_fed(u32(d.addr0), nullptr, dt, n, i64(peer), wdt, d.tag);
     const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
     if (peer == me()) {  // self-send: defer to matching recv via loopback slot
       mk_tx(0, peer, src, dt, wdt, n, d.tag);
@@ -848,6 +853,44 @@ struct Cclo {
         if (!wait_pred_tick(deadline)) return err;
     }
     fence_acquire_sys();
+    return E_OK;
+  }
+
+  // Drain `count` elements from my stream ring lane [lane] into either a
+  // local buffer (stream2mem; dst != null) or a forward-to-peer eager tx
+  // (send-from-stream; fwd_peer >= 0). One segment at a time: wait for the
+  // slot, move its payload, return credit. reference: OP0_STREAM operand
+  // routing (dma_mover.cpp:497, router :92-98) and the stream2mem /
+  // mem2stream reduce+send test matrix (test.cpp).
+  ACCL_HD u32 stream_fed(u32 lane, char* dst, DataType ddt, u64 count,
+                         i64 fwd_peer, DataType wdt, u32 tag) {
+    u64 drained = 0;
+    const ArenaHdr* h = tv.hdr(me());
+    while (drained < count) {
+      u64 seq = sq.stream_fed_rx[lane] + 1;
+      u32 slot = u32((seq - 1) % h->n_stream);
+      SlotHdr* sh = tv.stream_hdr(me(), lane, slot);
+      u64 deadline = deadline_now();
+      while (ld_sys(&sh->seq) != seq)
+        if (!wait_pred_tick(deadline)) return err;
+      fence_acquire_sys();
+      u32 esz = dtype_size(ddt);
+      u64 n = sh->bytes / esz;
+      if (n > count - drained) { err |= E_SEGMENT; return err; }
+      const char* pay = tv.stream_payload(me(), lane, slot);
+      if (dst) {
+        mk_local(0, pay, ddt, dst + drained * esz, ddt, n);
+      } else {
+        mk_tx(0, u32(fwd_peer), pay, ddt, wdt, n, tag);
+      }
+      u32 e = run_flows(1);
+      if (e) return e;
+      sq.stream_fed_rx[lane] = seq;
+      // credit back to the producer's arena so its ring keeps flowing
+      fence_release_sys();
+      st_sys(&tv.stream_ctl(lane, me())->credit, seq);
+      drained += n;
+    }
     return E_OK;
   }
 

@@ -102,6 +102,10 @@ enum CallFlags : u32 {
   F_SRC_ARENA = 1u << 4,      // addresses are arena offsets (rendezvous-capable)
   F_DST_ARENA = 1u << 5,
   F_OP1_ARENA = 1u << 6,
+  F_SRC_STREAM = 1u << 7,     // addr0 = stream lane id: the engine consumes
+                              // segments from its own stream ring (reference:
+                              // OP0_STREAM send/reduce-from-krnl-stream,
+                              // dma_mover.cpp:497)
 };
 
 // -------------------------------------------------------------------- errors

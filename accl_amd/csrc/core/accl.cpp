@@ -261,6 +261,31 @@ void ACCL::set_max_eager_size(u64 bytes) {
   be_->call(d);
 }
 
+Request* ACCL::copy_from_stream(u32 lane, BaseBuffer& dst, u64 count,
+                                bool to_device, bool run_async) {
+  CallDesc d = make_desc(Op::copy, count, dst.dtype(), DataType::none);
+  d.addr0 = lane;
+  d.addr2 = dst.arena_offset();
+  d.flags = F_SRC_STREAM | F_DST_ARENA;
+  return finish(d, run_async, to_device ? nullptr : &dst, count);
+}
+
+Request* ACCL::send_from_stream(u32 lane, u64 count, u32 dst, u32 tag,
+                                u32 comm, DataType compress, bool run_async) {
+  CallDesc d = make_desc(Op::send, count, DataType::float32, compress);
+  d.addr0 = lane;
+  d.root_src_dst = dst;
+  d.tag = tag;
+  d.comm_id = comm;
+  d.flags = F_SRC_STREAM;
+  return finish(d, run_async, nullptr, 0);
+}
+
+bool ACCL::alive() {
+  auto v = be_->ctrl_view();
+  return v.engine_up != 0;
+}
+
 // ---------------- debug dumps ----------------
 // reference: dump_communicator (accl.cpp:1429-1439), dump_rx_buffers
 // (accl.cpp:964-1048: status/occupancy/tag/seqn per rx buffer), here read

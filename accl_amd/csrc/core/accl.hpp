@@ -171,6 +171,17 @@ class ACCL {
                  u64 timeout_ms = 10000);
   // Non-destructive check: is a stream segment from `src` pending?
   bool stream_ready(u32 src);
+  // Stream-fed ops: the ENGINE consumes ring lane [lane] (one consumer per
+  // lane — do not mix with pop_stream on the same lane). reference:
+  // OP0_STREAM operand routing (dma_mover.cpp:497; stream2mem tests).
+  Request* copy_from_stream(u32 lane, BaseBuffer& dst, u64 count,
+                            bool to_device = false, bool run_async = false);
+  Request* send_from_stream(u32 lane, u64 count, u32 dst, u32 tag = TAG_ANY,
+                            u32 comm = GLOBAL_COMM,
+                            DataType compress = DataType::none,
+                            bool run_async = false);
+  // liveness: engine heartbeat advancing / engine_up
+  bool alive();
 
   void free_request(Request* r);
 

@@ -660,3 +660,36 @@ def _devcall(a, rank, n):
 def test_device_call_ring():
     run_ranks(_devcall, 1)
     run_ranks(_devcall, 2)
+
+
+# ----------------------------------------------- stream-fed ops (OP0_STREAM)
+# reference: send/reduce with the op0 operand sourced from the kernel stream
+# (dma_mover.cpp:497 OP0_STREAM; stream2mem / mem2stream test matrix)
+def _stream_fed(a, rank, n):
+    cnt = 2000
+    if rank == 0:
+        # produce into rank 1's ring, rank 1's ENGINE forwards it onward
+        s = a.create_buffer(cnt, DT.float32)
+        s.write(pattern(cnt, 4))
+        a.stream_put(s, cnt, dst=1, tag=8)
+    elif rank == 1:
+        # stream2mem: engine drains lane 0 into a buffer
+        d = a.create_buffer(cnt, DT.float32)
+        a.copy_from_stream(0, d, cnt)
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 4))
+    a.barrier()
+    # self-produced lane -> engine forwards to peer (send-from-stream)
+    s2 = a.create_buffer(cnt, DT.float32)
+    s2.write(pattern(cnt, 40 + rank))
+    a.stream_put(s2, cnt, dst=rank, tag=9)  # own loopback lane
+    peer = (rank + 1) % n
+    a.send_from_stream(rank, cnt, dst=peer, tag=10)
+    d2 = a.create_buffer(cnt, DT.float32)
+    src_peer = (rank - 1) % n
+    a.recv(d2, cnt, src=src_peer, tag=10)
+    assert np.array_equal(rd(d2, cnt), pattern(cnt, 40 + src_peer))
+    a.barrier()
+
+
+def test_stream_fed_ops():
+    run_ranks(_stream_fed, 2, opts=STREAM_SMALL)

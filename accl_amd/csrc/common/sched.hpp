@@ -87,6 +87,19 @@ struct Cclo {
   u32 err;                 // error bits of the current call
   Flow flows[MAX_FLOWS];
 
+  // ---- unexpected-message queue (the rxbuf_seek match-engine analogue:
+  // reference kernels/cclo/hls/rxbuf_offload/rxbuf_seek.cpp:53-72 searches
+  // pending rx buffers by (tag, src, seqn) in any order). Segments whose tag
+  // does not match the posted recv are spilled to the spare region's upper
+  // half so later recvs can match them out of order.
+  static constexpr u32 UQ_DEPTH = 8;
+  struct Unexpected {
+    u32 tag; u32 arith; u64 bytes; u64 msg_count; u32 spare_slot; u32 flags;
+  };
+  Unexpected uq[MAX_RANKS][UQ_DEPTH];
+  u32 uq_h[MAX_RANKS], uq_t[MAX_RANKS];
+  u64 spill_busy;          // bitmap over spill slots (<= 64)
+
   ACCL_HD u32 me() const { return cfg.rank; }
 
   // device-only micro-timeline (GPU: GpuMover::dbg; emulator mover: no-op)
@@ -496,6 +509,107 @@ struct Cclo {
     return run_flows(1);
   }
 
+  ACCL_HD u32 spill_slot_count() {
+    u64 ns = (tv.hdr(me())->spare_bytes / 2) / cfg.slot_bytes;
+    return ns > 64 ? 64 : u32(ns);
+  }
+  ACCL_HD char* spill_ptr(u32 sp) {
+    const ArenaHdr* h = tv.hdr(me());
+    return tv.arena[me()] + h->spare_off + h->spare_bytes / 2 +
+           u64(sp) * cfg.slot_bytes;
+  }
+
+  // move the head segment of channel (peer -> me) into the spill pool and
+  // release the rx slot; false when queue/pool is full (caller keeps waiting)
+  ACCL_HD bool spill_head(u32 peer, const SlotHdr* hd, u32 sl, u64 seq) {
+    if (uq_t[peer] - uq_h[peer] >= UQ_DEPTH) return false;
+    u32 ns = spill_slot_count();
+    u32 sp = 0;
+    while (sp < ns && ((spill_busy >> sp) & 1)) ++sp;
+    if (sp >= ns || hd->bytes > cfg.slot_bytes) return false;
+    mk_local(0, tv.slot_payload(me(), peer, sl), DataType::int8,
+             spill_ptr(sp), DataType::int8, hd->bytes);
+    u32 e = run_flows(1);
+    if (e) return false;
+    Unexpected& u = uq[peer][uq_t[peer] % UQ_DEPTH];
+    u.tag = hd->tag; u.arith = hd->arith; u.bytes = hd->bytes;
+    u.msg_count = hd->msg_count; u.flags = hd->flags; u.spare_slot = sp;
+    spill_busy |= 1ull << sp;
+    uq_t[peer]++;
+    sq.eager_rx[peer] = seq;
+    sq.credit_ret[peer] = seq;
+    st_sys(&tv.chan_ctl(peer, me())->credit, seq);
+    return true;
+  }
+
+  // eager receive with out-of-order tag matching: drain matching spilled
+  // segments first, then the live channel; mismatched head segments are
+  // spilled so a different-tag recv can run ahead (MPI matching semantics,
+  // reference rxbuf_seek + pending queue).
+  ACCL_HD u32 op_recv_eager(u32 peer, char* dst, DataType ddt, DataType wdt,
+                            u64 n, u32 want_tag) {
+    u64 got = 0;
+    u32 mtag = want_tag;
+    const u32 wsz = dtype_size(wdt), dsz = dtype_size(ddt);
+    u64 deadline = deadline_now();
+    while (got < n) {
+      bool progressed = false;
+      // 1) spill queue, in arrival order (per-tag FIFO preserved)
+      for (u32 qi = uq_h[peer]; qi != uq_t[peer]; ++qi) {
+        Unexpected& u = uq[peer][qi % UQ_DEPTH];
+        if (u.bytes == 0) continue;  // consumed hole
+        if (mtag != TAG_ANY && u.tag != mtag) continue;
+        if (u.arith != u32(wdt)) { err |= E_COMPRESSION; return err; }
+        if (got == 0) mtag = u.tag;
+        else if (u.tag != mtag) continue;
+        u64 nseg = u.bytes / wsz;
+        if (nseg > n - got) { err |= E_SEGMENT; return err; }
+        mk_local(0, spill_ptr(u.spare_slot), wdt, dst + got * dsz, ddt, nseg);
+        u32 e = run_flows(1);
+        if (e) return e;
+        spill_busy &= ~(1ull << u.spare_slot);
+        u.bytes = 0;
+        while (uq_h[peer] != uq_t[peer] &&
+               uq[peer][uq_h[peer] % UQ_DEPTH].bytes == 0)
+          uq_h[peer]++;
+        got += nseg;
+        progressed = true;
+        deadline = deadline_now();
+        break;
+      }
+      if (progressed) continue;
+      if (got >= n) break;
+      // 2) live channel head
+      u64 seq = sq.eager_rx[peer] + 1;
+      u32 sl = u32((seq - 1) % cfg.n_slots);
+      SlotHdr* hd = tv.slot_hdr(me(), peer, sl);
+      if (ld_sys(&hd->seq) == seq) {
+        fence_acquire_sys();
+        bool match = (mtag == TAG_ANY) || (hd->tag == mtag);
+        if (match) {
+          if (hd->arith != u32(wdt)) { err |= E_COMPRESSION; return err; }
+          if (got == 0) mtag = hd->tag;
+          u64 nseg = u64(hd->bytes) / wsz;
+          if (nseg > n - got) { err |= E_SEGMENT; return err; }
+          mk_local(0, tv.slot_payload(me(), peer, sl), wdt, dst + got * dsz,
+                   ddt, nseg);
+          u32 e = run_flows(1);
+          if (e) return e;
+          sq.eager_rx[peer] = seq;
+          sq.credit_ret[peer] = seq;
+          st_sys(&tv.chan_ctl(peer, me())->credit, seq);
+          got += nseg;
+          deadline = deadline_now();
+          continue;
+        }
+        if (spill_head(peer, hd, sl, seq)) { deadline = deadline_now(); continue; }
+        if (err) return err;
+      }
+      if (!wait_pred_tick(deadline)) return err;
+    }
+    return E_OK;
+  }
+
   ACCL_HD u32 op_recv(const CallDesc& d, const CommView& c) {
     u64 n = desc_count(d);
     u32 peer = c.global(d.root_src_dst);
@@ -509,9 +623,10 @@ struct Cclo {
         if (e) return e;
         return wait_done(peer) ? E_OK : err;
       }
-      // stage through the spare region in windows (double-buffered)
+      // stage through the spare region's LOWER half in windows (double-
+      // buffered); the upper half is the unexpected-message spill pool
       ArenaHdr* h = tv.hdr(me());
-      u64 half = (h->spare_bytes / 2) / dtype_size(dt);
+      u64 half = (h->spare_bytes / 4) / dtype_size(dt);
       if (!half) { err |= E_INVALID_ARG; return err; }
       u64 got = 0; int cur = 0; u64 pending = 0; u64 pend_off = 0;
       u64 posted = 0;
@@ -542,8 +657,7 @@ struct Cclo {
       }
       return wait_done(peer) ? E_OK : err;
     }
-    mk_rx(0, peer, dst, dt, wdt, n, d.tag);
-    return run_flows(1);
+    return op_recv_eager(peer, dst, dt, wdt, n, d.tag);
   }
 
   // fullmesh bcast: root pushes to every peer; direct when dst offsets can

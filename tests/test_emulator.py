@@ -693,3 +693,43 @@ def _stream_fed(a, rank, n):
 
 def test_stream_fed_ops():
     run_ranks(_stream_fed, 2, opts=STREAM_SMALL)
+
+
+# ---------------------------------------- out-of-order tag matching (seek)
+# reference: rxbuf_seek matches pending rx buffers by (tag, src, seqn) in
+# ANY order (rxbuf_seek.cpp:53-72) — a recv may match a later-tagged message
+# while an earlier one waits for its own recv.
+def _ooo_tags(a, rank, n):
+    cnt = 1500
+    if rank == 0:
+        for tag in (11, 22, 33):
+            s = _mk(a, cnt)
+            s.write(pattern(cnt, tag))
+            a.send(s, cnt, dst=1, tag=tag)
+    elif rank == 1:
+        for tag in (33, 11, 22):  # reversed / shuffled consumption
+            d = _mk(a, cnt)
+            a.recv(d, cnt, src=0, tag=tag)
+            assert np.array_equal(rd(d, cnt), pattern(cnt, tag)), tag
+    a.barrier()
+
+
+def _ooo_tags_segmented(a, rank, n):
+    # multi-segment messages (count >> slot_bytes) spilled mid-message
+    cnt = 3500  # ~3.4 segments at 4KB slots
+    if rank == 0:
+        for tag in (1, 2):
+            s = _mk(a, cnt)
+            s.write(pattern(cnt, tag, seed=3))
+            a.send(s, cnt, dst=1, tag=tag)
+    elif rank == 1:
+        for tag in (2, 1):
+            d = _mk(a, cnt)
+            a.recv(d, cnt, src=0, tag=tag)
+            assert np.array_equal(rd(d, cnt), pattern(cnt, tag, seed=3)), tag
+    a.barrier()
+
+
+def test_out_of_order_tags():
+    run_ranks(_ooo_tags, 2, opts=SMALL)
+    run_ranks(_ooo_tags_segmented, 2, opts=SMALL)

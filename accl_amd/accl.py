@@ -188,6 +188,7 @@ class ACCL:
                     "reduce_scatter", "alltoall", "barrier", "nop",
                     "stream_put", "pop_stream", "stream_ready",
                     "copy_from_stream", "send_from_stream", "alive",
+                    "soft_reset",
                     "info", "set_timeout_ms", "set_max_eager_size",
                     "dump_communicator", "dump_eager_rx_buffers",
                     "dump_streams", "dump_engine_status",

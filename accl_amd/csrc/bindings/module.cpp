@@ -273,6 +273,8 @@ PYBIND11_MODULE(_core, m) {
            py::arg("run_async") = false, py::return_value_policy::reference,
            py::keep_alive<0, 1>(), py::call_guard<py::gil_scoped_release>())
       .def("alive", &ACCL::alive)
+      .def("soft_reset", &ACCL::soft_reset,
+           py::call_guard<py::gil_scoped_release>())
       .def("info",
            [](ACCL& a) {
              // reference: parse_hwid capability decode (accl.cpp:1050-1064)

@@ -1089,7 +1089,18 @@ struct Cclo {
       case CfgFunc::set_max_eager_size:
         max_eager_bytes = desc_count(d);
         return E_OK;
-      case CfgFunc::reset:
+      case CfgFunc::reset: {
+        // soft reset (reference: encore_soft_reset drains retry queue +
+        // resets peripherals, ccl_offload_control.c:2249-2261). Local-only:
+        // clears the flow table, unexpected-message queue and spill pool;
+        // pair sequence counters are PROTOCOL state shared with peers and
+        // survive (desynced pairs need a reset on both ends).
+        for (u32 i = 0; i < MAX_FLOWS; ++i) flows[i] = Flow{};
+        for (u32 r = 0; r < MAX_RANKS; ++r) { uq_h[r] = uq_t[r] = 0; }
+        spill_busy = 0;
+        err = 0;
+        return E_OK;
+      }
       case CfgFunc::enable_pkt:
         return E_OK;
       default: return E_OK;

@@ -281,6 +281,12 @@ Request* ACCL::send_from_stream(u32 lane, u64 count, u32 dst, u32 tag,
   return finish(d, run_async, nullptr, 0);
 }
 
+void ACCL::soft_reset() {
+  CallDesc d = make_desc(Op::config, 0, DataType::none, DataType::none);
+  d.function = u32(CfgFunc::reset);
+  be_->call(d);
+}
+
 bool ACCL::alive() {
   auto v = be_->ctrl_view();
   return v.engine_up != 0;

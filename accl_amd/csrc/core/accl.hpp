@@ -189,6 +189,8 @@ class ACCL {
   // rendezvous sizes accl.hpp:103-104, cfgFunc calls) ---
   void set_timeout_ms(u64 ms);
   void set_max_eager_size(u64 bytes);
+  // local engine soft reset (reference: ACCL soft_reset, accl.cpp:57-69)
+  void soft_reset();
 
   // --- debug dumps (reference: ACCL::dump_rx_buffers / dump_communicator /
   // dump_exchange_memory, driver/xrt/src/accl.cpp:964-1048) ---

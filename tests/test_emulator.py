@@ -733,3 +733,28 @@ def _ooo_tags_segmented(a, rank, n):
 def test_out_of_order_tags():
     run_ranks(_ooo_tags, 2, opts=SMALL)
     run_ranks(_ooo_tags_segmented, 2, opts=SMALL)
+
+
+def _soft_reset(a, rank, n):
+    # an errored recv followed by reset; the engine keeps serving calls
+    # (reference: soft_reset drains + re-arms, accl.cpp:57-69)
+    cnt = 256
+    if rank == 0:
+        a.set_timeout_ms(300)
+        d = _mk(a, cnt)
+        try:
+            a.recv(d, cnt, src=1, tag=77)
+        except RuntimeError as e:
+            assert "TIMEOUT" in str(e)
+        a.set_timeout_ms(10000)
+        a.soft_reset()
+    a.barrier()
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    s.write(pattern(cnt, rank))
+    a.allreduce(s, d, cnt, RF.SUM)
+    exp = np.stack([pattern(cnt, r) for r in range(n)]).sum(0)
+    assert np.allclose(rd(d, cnt), exp)
+
+
+def test_soft_reset():
+    run_ranks(_soft_reset, 2)

@@ -758,3 +758,42 @@ def _soft_reset(a, rank, n):
 
 def test_soft_reset():
     run_ranks(_soft_reset, 2)
+
+
+# ----------------------------------- rendezvous/direct collective paths
+# Force the direct (address-exchange + peer-write) schedules by lowering
+# max_eager (reference: rendezvous selection, ccl_offload_control.c:587-610;
+# direct allgather/bcast are the RDMA-design analogues)
+DIRECT = {"max_eager": 4096, "n_slots": 4, "slot_bytes": 4096,
+          "timeout_us": 20_000_000}
+
+
+def _direct_paths(a, rank, n):
+    cnt = 5000  # 20 KB > max_eager -> rendezvous/direct
+    s, d = _mk(a, cnt), _mk(a, cnt * n)
+    s.write(pattern(cnt, rank, seed=31))
+    a.allgather(s, d, cnt)
+    exp = np.concatenate([pattern(cnt, r, seed=31) for r in range(n)])
+    assert np.array_equal(rd(d, cnt * n), exp), "direct allgather"
+    b = _mk(a, cnt)
+    if rank == 0:
+        b.write(pattern(cnt, 77))
+    a.bcast(b, cnt, 0)
+    assert np.array_equal(rd(b, cnt), pattern(cnt, 77)), "direct bcast"
+    if rank == 0:
+        a.send(s, cnt, dst=1, tag=4)
+    elif rank == 1:
+        d2 = _mk(a, cnt)
+        a.recv(d2, cnt, src=0, tag=4)
+        assert np.array_equal(rd(d2, cnt), pattern(cnt, 0, seed=31)), "rndzv sr"
+    a.barrier()
+    # large allreduce stays eager-correct under the tiny threshold too
+    d3 = _mk(a, cnt)
+    a.allreduce(s, d3, cnt, RF.SUM)
+    exp = np.stack([pattern(cnt, r, seed=31) for r in range(n)]).sum(0)
+    assert np.allclose(rd(d3, cnt), exp)
+
+
+def test_direct_rendezvous_paths():
+    run_ranks(_direct_paths, 2, opts=DIRECT)
+    run_ranks(_direct_paths, 3, opts=DIRECT)

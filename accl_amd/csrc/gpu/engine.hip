@@ -379,7 +379,8 @@ __device__ void mover_main(GpuEngineState* S) {
       idle = idle < 240 ? idle + 8 : 240;
       if (idle < 64) __builtin_amdgcn_s_sleep(8);
       else if (idle < 160) __builtin_amdgcn_s_sleep(32);
-      else __builtin_amdgcn_s_sleep(64);
+      else if (idle < 240) __builtin_amdgcn_s_sleep(64);
+      else __builtin_amdgcn_s_sleep(127);  // deep idle: ~3 us wake worst-case
       continue;
     }
     idle = 0;

@@ -797,3 +797,19 @@ def _direct_paths(a, rank, n):
 def test_direct_rendezvous_paths():
     run_ranks(_direct_paths, 2, opts=DIRECT)
     run_ranks(_direct_paths, 3, opts=DIRECT)
+
+
+def _ring_ar(a, rank, n):
+    # P > 9 falls back from fullmesh to the segmented ring schedule
+    # (reference shape: ccl_offload_control.c:1888-2071)
+    cnt = 2200
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    s.write(pattern(cnt, rank, seed=17))
+    a.allreduce(s, d, cnt, RF.SUM)
+    exp = np.stack([pattern(cnt, r, seed=17) for r in range(n)]).sum(0)
+    assert np.allclose(rd(d, cnt), exp)
+    a.barrier()
+
+
+def test_ring_allreduce_p10():
+    run_ranks(_ring_ar, 10, opts=SMALL, timeout=300)

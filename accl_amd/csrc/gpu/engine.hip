@@ -258,28 +258,37 @@ __device__ void tile_reduce_f32(const MoveDesc& m, u64 lo, u64 hi, int lane) {
     GAS F4* o = (GAS F4*)d;
     u64 n4 = n / 4;
     u64 i = lane;
-    // 4-deep pipeline x 2 operand streams = 8 loads in flight per lane
-    for (; i + 3 * 64 < n4; i += 4 * 64) {
-      F4 x0 = a[i], x1 = a[i + 64], x2 = a[i + 2 * 64], x3 = a[i + 3 * 64];
-      F4 y0 = b[i], y1 = b[i + 64], y2 = b[i + 2 * 64], y3 = b[i + 3 * 64];
+#define ACCL_NTL(p_) __builtin_nontemporal_load(p_)
 #define ACCL_R4(a_, b_) (F4){OP<float>::apply(a_.x, b_.x), \
     OP<float>::apply(a_.y, b_.y), OP<float>::apply(a_.z, b_.z), \
     OP<float>::apply(a_.w, b_.w)}
-      o[i] = ACCL_R4(x0, y0); o[i + 64] = ACCL_R4(x1, y1);
-      o[i + 2 * 64] = ACCL_R4(x2, y2); o[i + 3 * 64] = ACCL_R4(x3, y3);
+    // 4-deep pipeline x 2 non-temporal operand streams (8 loads in flight);
+    // nt stores leave no dirty L2 so the mover can skip the writeback fence
+    for (; i + 3 * 64 < n4; i += 4 * 64) {
+      F4 x0 = ACCL_NTL(&a[i]), x1 = ACCL_NTL(&a[i + 64]);
+      F4 x2 = ACCL_NTL(&a[i + 2 * 64]), x3 = ACCL_NTL(&a[i + 3 * 64]);
+      F4 y0 = ACCL_NTL(&b[i]), y1 = ACCL_NTL(&b[i + 64]);
+      F4 y2 = ACCL_NTL(&b[i + 2 * 64]), y3 = ACCL_NTL(&b[i + 3 * 64]);
+      __builtin_nontemporal_store(ACCL_R4(x0, y0), &o[i]);
+      __builtin_nontemporal_store(ACCL_R4(x1, y1), &o[i + 64]);
+      __builtin_nontemporal_store(ACCL_R4(x2, y2), &o[i + 2 * 64]);
+      __builtin_nontemporal_store(ACCL_R4(x3, y3), &o[i + 3 * 64]);
     }
     if (i < n4) {  // clamped-tail (see tile_copy): all loads in flight
       u64 last = n4 - 1;
       u64 i1 = i + 64 < n4 ? i + 64 : last, i2 = i + 128 < n4 ? i + 128 : last;
       u64 i3 = i + 192 < n4 ? i + 192 : last;
-      F4 x0 = a[i], x1 = a[i1], x2 = a[i2], x3 = a[i3];
-      F4 y0 = b[i], y1 = b[i1], y2 = b[i2], y3 = b[i3];
-      o[i] = ACCL_R4(x0, y0);
-      if (i + 64 < n4) o[i1] = ACCL_R4(x1, y1);
-      if (i + 128 < n4) o[i2] = ACCL_R4(x2, y2);
-      if (i + 192 < n4) o[i3] = ACCL_R4(x3, y3);
+      F4 x0 = ACCL_NTL(&a[i]), x1 = ACCL_NTL(&a[i1]);
+      F4 x2 = ACCL_NTL(&a[i2]), x3 = ACCL_NTL(&a[i3]);
+      F4 y0 = ACCL_NTL(&b[i]), y1 = ACCL_NTL(&b[i1]);
+      F4 y2 = ACCL_NTL(&b[i2]), y3 = ACCL_NTL(&b[i3]);
+      __builtin_nontemporal_store(ACCL_R4(x0, y0), &o[i]);
+      if (i + 64 < n4) __builtin_nontemporal_store(ACCL_R4(x1, y1), &o[i1]);
+      if (i + 128 < n4) __builtin_nontemporal_store(ACCL_R4(x2, y2), &o[i2]);
+      if (i + 192 < n4) __builtin_nontemporal_store(ACCL_R4(x3, y3), &o[i3]);
     }
 #undef ACCL_R4
+#undef ACCL_NTL
     return;
   }
   for (u64 i = lane; i < n; i += 64) {
@@ -288,6 +297,81 @@ __device__ void tile_reduce_f32(const MoveDesc& m, u64 lo, u64 hi, int lane) {
       acc = OP<float>::apply(acc, ((GAS const float*)m.src[k])[lo + i]);
     ((GAS float*)m.dst)[lo + i] = acc;
   }
+}
+
+// Packed 16-bit reduce (the bf16/f16 grad hot path, BASELINE config 4):
+// 16B vectors = 8 elements per load; f16 pairs use v_pk (__half2) math,
+// bf16 pairs convert through f32 with RNE repack (no native bf16 pk-add on
+// CDNA4's VALU path we rely on). Non-temporal streams like the f32 path.
+template <bool MAX_, bool BF16>
+__device__ __forceinline__ u32 h2_op(u32 pa, u32 pb) {
+  if (BF16) {
+    float a0 = __uint_as_float(pa << 16), a1 = __uint_as_float(pa & 0xFFFF0000u);
+    float b0 = __uint_as_float(pb << 16), b1 = __uint_as_float(pb & 0xFFFF0000u);
+    float r0 = MAX_ ? (a0 > b0 ? a0 : b0) : a0 + b0;
+    float r1 = MAX_ ? (a1 > b1 ? a1 : b1) : a1 + b1;
+    auto pack = [](float v) -> u32 {
+      u32 x = __float_as_uint(v);
+      if ((x & 0x7F800000u) == 0x7F800000u && (x & 0x7FFFFFu))
+        return (x >> 16) | 0x40;
+      x += 0x7FFFu + ((x >> 16) & 1);
+      return x >> 16;
+    };
+    return pack(r0) | (pack(r1) << 16);
+  }
+  __half2 a = *(__half2*)&pa, b = *(__half2*)&pb;
+  __half2 r;
+  if (MAX_) {
+    r.x = __hgt(a.x, b.x) ? a.x : b.x;
+    r.y = __hgt(a.y, b.y) ? a.y : b.y;
+  } else {
+    r = __hadd2(a, b);  // v_pk_add_f16
+  }
+  return *(u32*)&r;
+}
+
+template <bool MAX_, bool BF16>
+__device__ void tile_reduce_h16(const MoveDesc& m, u64 lo, u64 hi, int lane) {
+  GAS const U4* a = (GAS const U4*)((const u16*)m.src[0] + lo);
+  GAS const U4* b = (GAS const U4*)((const u16*)m.src[1] + lo);
+  GAS U4* o = (GAS U4*)((u16*)m.dst + lo);
+  u64 n8 = (hi - lo) / 8;  // 8 halves per 16B vector
+#define ACCL_H8(x_, y_) (U4){h2_op<MAX_, BF16>(x_.x, y_.x), \
+    h2_op<MAX_, BF16>(x_.y, y_.y), h2_op<MAX_, BF16>(x_.z, y_.z), \
+    h2_op<MAX_, BF16>(x_.w, y_.w)}
+  u64 i = lane;
+  for (; i + 3 * 64 < n8; i += 4 * 64) {
+    U4 x0 = __builtin_nontemporal_load(&a[i]);
+    U4 x1 = __builtin_nontemporal_load(&a[i + 64]);
+    U4 x2 = __builtin_nontemporal_load(&a[i + 2 * 64]);
+    U4 x3 = __builtin_nontemporal_load(&a[i + 3 * 64]);
+    U4 y0 = __builtin_nontemporal_load(&b[i]);
+    U4 y1 = __builtin_nontemporal_load(&b[i + 64]);
+    U4 y2 = __builtin_nontemporal_load(&b[i + 2 * 64]);
+    U4 y3 = __builtin_nontemporal_load(&b[i + 3 * 64]);
+    __builtin_nontemporal_store(ACCL_H8(x0, y0), &o[i]);
+    __builtin_nontemporal_store(ACCL_H8(x1, y1), &o[i + 64]);
+    __builtin_nontemporal_store(ACCL_H8(x2, y2), &o[i + 2 * 64]);
+    __builtin_nontemporal_store(ACCL_H8(x3, y3), &o[i + 3 * 64]);
+  }
+  if (i < n8) {
+    u64 last = n8 - 1;
+    u64 i1 = i + 64 < n8 ? i + 64 : last, i2 = i + 128 < n8 ? i + 128 : last;
+    u64 i3 = i + 192 < n8 ? i + 192 : last;
+    U4 x0 = __builtin_nontemporal_load(&a[i]);
+    U4 x1 = __builtin_nontemporal_load(&a[i1]);
+    U4 x2 = __builtin_nontemporal_load(&a[i2]);
+    U4 x3 = __builtin_nontemporal_load(&a[i3]);
+    U4 y0 = __builtin_nontemporal_load(&b[i]);
+    U4 y1 = __builtin_nontemporal_load(&b[i1]);
+    U4 y2 = __builtin_nontemporal_load(&b[i2]);
+    U4 y3 = __builtin_nontemporal_load(&b[i3]);
+    __builtin_nontemporal_store(ACCL_H8(x0, y0), &o[i]);
+    if (i + 64 < n8) __builtin_nontemporal_store(ACCL_H8(x1, y1), &o[i1]);
+    if (i + 128 < n8) __builtin_nontemporal_store(ACCL_H8(x2, y2), &o[i2]);
+    if (i + 192 < n8) __builtin_nontemporal_store(ACCL_H8(x3, y3), &o[i3]);
+  }
+#undef ACCL_H8
 }
 
 // float-domain path for any f32/f16/bf16 mix (cast + reduce fused — the
@@ -342,6 +426,24 @@ __device__ bool run_tile(const MoveDesc& m, u32 t, int lane) {
     else
       tile_reduce_f32<MaxOp>(m, lo, hi, lane);
     return v16;
+  }
+  // packed same-dtype f16/bf16 two-source reduce (config-4 hot path)
+  if (m.nsrc == 2 && m.src_dt[0] == m.dst_dt && m.src_dt[1] == m.dst_dt &&
+      (m.dst_dt == u8(DataType::float16) || m.dst_dt == u8(DataType::bfloat16)) &&
+      ((hi - lo) & 7) == 0 &&
+      aligned16((const void*)(m.dst + lo * 2)) &&
+      aligned16((const void*)(m.src[0] + lo * 2)) &&
+      aligned16((const void*)(m.src[1] + lo * 2))) {
+    bool mx = ReduceFunction(m.func) == ReduceFunction::MAX;
+    bool bf = m.dst_dt == u8(DataType::bfloat16);
+    if (bf) {
+      if (mx) tile_reduce_h16<true, true>(m, lo, hi, lane);
+      else tile_reduce_h16<false, true>(m, lo, hi, lane);
+    } else {
+      if (mx) tile_reduce_h16<true, false>(m, lo, hi, lane);
+      else tile_reduce_h16<false, false>(m, lo, hi, lane);
+    }
+    return true;
   }
   if (floatish) {
     tile_float_generic(m, lo, hi, lane);

@@ -25,7 +25,9 @@ struct alignas(128) MoveDesc {
   u8 dst_dt;
   u8 inline_done;         // GPU: executed inline by the scheduler WG's small
                           // mover wave; fleet must skip (bookkeeping only)
-  u8 _pad[6];
+  u8 tile_log2;           // per-move tile size override (0 = default 128KB);
+                          // lets perf sweeps tune tiling without rebuilds
+  u8 _pad[5];
   u64 epoch;              // published last (GPU queue); emulator ignores
 };
 static_assert(sizeof(MoveDesc) == 128, "");
@@ -45,16 +47,21 @@ constexpr u64 MOVE_TILE_BYTES = 1u << 17;  // 128 KiB per mover tile
 ACCL_HD inline u64 move_bytes(const MoveDesc& m) {
   return m.count * dtype_size(DataType(m.dst_dt));
 }
+ACCL_HD inline u64 move_tile_bytes(const MoveDesc& m) {
+  return m.tile_log2 ? (1ull << m.tile_log2) : MOVE_TILE_BYTES;
+}
+
 ACCL_HD inline u32 move_tiles(const MoveDesc& m) {
   // tile over DST elements; sources are index-aligned
+  u64 tb = move_tile_bytes(m);
   u64 bytes = move_bytes(m);
-  u64 t = (bytes + MOVE_TILE_BYTES - 1) / MOVE_TILE_BYTES;
+  u64 t = (bytes + tb - 1) / tb;
   return t ? u32(t) : 1;
 }
 
 ACCL_HD inline u64 move_tile_elems(const MoveDesc& m) {
   u32 dsz = dtype_size(DataType(m.dst_dt));
-  return MOVE_TILE_BYTES / (dsz ? dsz : 1);
+  return move_tile_bytes(m) / (dsz ? dsz : 1);
 }
 
 }  // namespace accl

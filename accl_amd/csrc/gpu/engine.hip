@@ -37,6 +37,7 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
   bool small = small_mb && move_bytes(m) <= SMALL_INLINE_MAX;
   MoveDesc& d = ring[slot];
   MoveState& s = st[slot];
+  const_cast<MoveDesc&>(m).tile_log2 = u8(tile_log2);
   if (small) {
     // kick the sibling wave over LDS FIRST, overlap the ring bookkeeping
     // with its copy, then block until its system-release completes

@@ -39,6 +39,7 @@ struct GpuMover {
   u64 (*rep)[8];
   void* small_mb;          // LDS SmallMb (set by the scheduler kernel)
   u64 small_seq;           // scheduler-private inline-move counter
+  u32 tile_log2;           // 0 = default tile; ACCL_TILE_KB env override
 
 #if defined(__HIPCC__)
   __device__ u32 submit(const MoveDesc& m);

@@ -139,6 +139,12 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
   st->mover.head = &dstate->mq_head;
   st->mover.stop = &dstate->stop;
   st->mover.dbg = dstate->dbg;
+  if (const char* t = std::getenv("ACCL_TILE_KB")) {
+    u64 kb = strtoull(t, nullptr, 10);
+    u32 lg = 0;
+    while ((1ull << (lg + 1)) <= kb * 1024) ++lg;
+    st->mover.tile_log2 = lg;
+  }
   st->mover.rep = dstate->head_rep;
   // pinned pointers as seen by the device
   RingPage* rp = (RingPage*)ring_pinned_;

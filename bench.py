@@ -205,19 +205,27 @@ def main():
 
     sweep_rows = []
     if args.sweep:
+        # reference protocol: count sweep per collective with device-side
+        # duration recorded to CSV (bench.cpp:25-61 + fixture CSV)
         sz = 4096
         while sz <= (1 << 30):
-            c = sz // 4
-            if c <= count:
+            c = (sz // esz) - ((sz // esz) % max(world, 1))
+            if 0 < c <= count:
                 s2, d2 = src.slice(0, c), dst.slice(0, c)
-                e = timed(lambda s=s2, d=d2, c=c: a.allreduce(
-                    s, d, c, A.ReduceFunction.SUM, from_device=True,
-                    to_device=True), max(3, min(20, (1 << 26) // sz)), 2)
-                sweep_rows.append((sz, e * 1e6, busbw(sz, e, world)))
+                dev_us = [0.0]
+
+                def one(s=s2, d=d2, c=c):
+                    r = a.allreduce(s, d, c, A.ReduceFunction.SUM,
+                                    from_device=True, to_device=True)
+                    dev_us[0] = r.duration_us()
+
+                e = timed(one, max(3, min(20, (1 << 26) // sz)), 2)
+                sweep_rows.append((sz, e * 1e6, dev_us[0],
+                                   busbw(sz, e, world)))
             sz *= 4
         if rank == 0:
-            lines = ["bytes,usec,busbw_GBps"] + [
-                f"{b},{u:.2f},{g:.2f}" for b, u, g in sweep_rows]
+            lines = ["bytes,usec,engine_usec,busbw_GBps"] + [
+                f"{b},{u:.2f},{du:.2f},{g:.2f}" for b, u, du, g in sweep_rows]
             csv = "\n".join(lines)
             print(csv, file=sys.stderr)
             if args.out_csv:

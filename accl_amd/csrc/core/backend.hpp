@@ -8,6 +8,7 @@
 #pragma once
 #include <cstring>
 #include <memory>
+#include <mutex>
 #include <string>
 #include <vector>
 #include "../common/proto.hpp"
@@ -50,10 +51,12 @@ class Backend {
   char* arena_local() const { return arena_base_; }
 
   // --- calls ---
+  // thread-safe in-order ring (reference: FPGAQueue serializes calls from
+  // any host thread to the single engine, acclrequest.hpp:153-211)
   u64 submit(CallDesc d) {
-    // in-order ring; throttle on ring occupancy
+    std::lock_guard<std::mutex> lk(ring_mu_);
     u64 seq = head_;
-    while (seq - tail_retired() >= RING_CAP) cpu_pause();
+    while (seq - tail_retired_locked() >= RING_CAP) cpu_pause();
     d.seq = u32(seq);
     ring_->descs[seq % RING_CAP] = d;
     head_ = seq + 1;
@@ -138,6 +141,10 @@ class Backend {
 
  protected:
   u64 tail_retired() {
+    std::lock_guard<std::mutex> lk(ring_mu_);
+    return tail_retired_locked();
+  }
+  u64 tail_retired_locked() {
     // ring slots free once their RetEntry is published
     while (retired_ < head_ && test(retired_, nullptr)) retired_++;
     return retired_;
@@ -148,6 +155,7 @@ class Backend {
   RingPage* ring_ = nullptr;   // host-visible (pinned on GPU)
   HeapAlloc heap_;
   u64 head_ = 0, retired_ = 0;
+  std::mutex ring_mu_;
 };
 
 }  // namespace accl

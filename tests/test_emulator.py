@@ -813,3 +813,32 @@ def _ring_ar(a, rank, n):
 
 def test_ring_allreduce_p10():
     run_ranks(_ring_ar, 10, opts=SMALL, timeout=300)
+
+
+def _threaded_submit(a, rank, n):
+    # reference FPGAQueue semantics: concurrent host threads may submit
+    import threading
+    cnt = 512
+    errs = []
+
+    def worker(seed):
+        try:
+            for i in range(10):
+                s, d = _mk(a, cnt), _mk(a, cnt)
+                x = pattern(cnt, seed * 100 + i)
+                s.write(x)
+                a.copy(s, d, cnt)
+                assert np.array_equal(rd(d, cnt), x)
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    ts = [threading.Thread(target=worker, args=(t,)) for t in range(4)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    assert not errs, errs
+
+
+def test_threaded_submit():
+    run_ranks(_threaded_submit, 1)

@@ -4,6 +4,7 @@
 // ACCL_FPGA_ALIGNMENT, driver/xrt/include/accl/common.hpp:29).
 #pragma once
 #include <map>
+#include <mutex>
 #include <stdexcept>
 #include "../common/types.hpp"
 
@@ -20,6 +21,7 @@ class HeapAlloc {
     if (base_ < end_) free_[base_] = end_ - base_;
   }
   u64 alloc(u64 bytes) {
+    std::lock_guard<std::mutex> lk(mu_);
     bytes = (bytes + ALIGN - 1) & ~(ALIGN - 1);
     if (!bytes) bytes = ALIGN;
     for (auto it = free_.begin(); it != free_.end(); ++it) {
@@ -34,6 +36,7 @@ class HeapAlloc {
     throw std::runtime_error("accl: arena heap exhausted");
   }
   void free_block(u64 off) {
+    std::lock_guard<std::mutex> lk(mu_);
     auto it = used_.find(off);
     if (it == used_.end()) return;
     u64 sz = it->second;
@@ -54,6 +57,7 @@ class HeapAlloc {
     free_[off] = sz;
   }
   u64 bytes_free() const {
+    std::lock_guard<std::mutex> lk(mu_);
     u64 t = 0;
     for (auto& kv : free_) t += kv.second;
     return t;
@@ -61,6 +65,7 @@ class HeapAlloc {
 
  private:
   u64 base_ = 0, end_ = 0;
+  mutable std::mutex mu_;
   std::map<u64, u64> free_;   // offset -> size
   std::map<u64, u64> used_;
 };

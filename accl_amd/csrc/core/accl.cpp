@@ -113,6 +113,27 @@ CallDesc ACCL::make_desc(Op op, u64 count, DataType dt, DataType wire) {
   return d;
 }
 
+static const char* op_name(u32 sc) {
+  switch (Op(sc)) {
+    case Op::copy: return "copy";
+    case Op::combine: return "combine";
+    case Op::send: return "send";
+    case Op::recv: return "recv";
+    case Op::bcast: return "bcast";
+    case Op::scatter: return "scatter";
+    case Op::gather: return "gather";
+    case Op::allgather: return "allgather";
+    case Op::reduce: return "reduce";
+    case Op::allreduce: return "allreduce";
+    case Op::reduce_scatter: return "reduce_scatter";
+    case Op::alltoall: return "alltoall";
+    case Op::barrier: return "barrier";
+    case Op::stream_put: return "stream_put";
+    case Op::config: return "config";
+    default: return "op";
+  }
+}
+
 Request* ACCL::finish(CallDesc d, bool run_async, BaseBuffer* sync_out,
                       u64 out_count, BaseBuffer* sync_in0, u64 in0_count,
                       BaseBuffer* sync_in1, u64 in1_count) {
@@ -123,11 +144,21 @@ Request* ACCL::finish(CallDesc d, bool run_async, BaseBuffer* sync_out,
     be_->write_arena(sync_in1->arena_offset(), sync_in1->host_ptr(),
                      in1_count * dtype_size(sync_in1->dtype()));
   u64 seq = be_->submit(d);
+  if (debug_enabled())
+    debug_log(std::string(op_name(d.scenario)) + " seq=" +
+              std::to_string(seq) + " count=" +
+              std::to_string((u64(d.count_hi) << 32) | d.count_lo) +
+              " tag=" + std::to_string(d.tag) +
+              (run_async ? " async" : ""));
   Request* r = new Request(be_.get(), seq);
   reqs_.push_back(r);
   if (!run_async) {
     u32 e = r->wait();
-    if (e) throw accl_error("accl op failed: " + error_to_string(e), e);
+    if (e) {
+      debug_log(std::string(op_name(d.scenario)) + " seq=" +
+                std::to_string(seq) + " FAILED: " + error_to_string(e));
+      throw accl_error("accl op failed: " + error_to_string(e), e);
+    }
     if (sync_out && sync_out->host_ptr())
       be_->read_arena(sync_out->arena_offset(), sync_out->host_ptr(),
                       out_count * dtype_size(sync_out->dtype()));

@@ -1,5 +1,7 @@
 #include "util.hpp"
 #include <chrono>
+#include <cstdio>
+#include <cstdlib>
 #include <sstream>
 
 namespace accl {
@@ -8,6 +10,15 @@ u64 wallclock_host_ns() {
   return u64(std::chrono::duration_cast<std::chrono::nanoseconds>(
                  std::chrono::steady_clock::now().time_since_epoch())
                  .count());
+}
+
+bool debug_enabled() {
+  static const bool on = std::getenv("ACCL_DEBUG") != nullptr;
+  return on;
+}
+
+void debug_log(const std::string& msg) {
+  if (debug_enabled()) std::fprintf(stderr, "[accl] %s\n", msg.c_str());
 }
 
 std::string error_to_string(u32 bits) {

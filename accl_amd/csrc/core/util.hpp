@@ -20,4 +20,9 @@ class accl_error : public std::runtime_error {
 // error_code_to_string, driver/xrt/src/accl.cpp:1210-1234)
 std::string error_to_string(u32 bits);
 
+// ACCL_DEBUG-gated host-side debug log (reference: debug() stderr logging
+// under ACCL_DEBUG, driver/xrt/include/accl/common.hpp:38-58)
+bool debug_enabled();
+void debug_log(const std::string& msg);
+
 }  // namespace accl

@@ -842,3 +842,16 @@ def _threaded_submit(a, rank, n):
 
 def test_threaded_submit():
     run_ranks(_threaded_submit, 1)
+
+
+def test_sanitized_protocol():
+    """ASan+UBSan over the 2-rank protocol surface (scripts/sanitize.sh):
+    the emulator and GPU engines share the scheduler/transport source, so a
+    sanitizer-clean emulator run vouches for the shared protocol code."""
+    import pathlib
+    import subprocess
+    root = pathlib.Path(__file__).resolve().parent.parent
+    r = subprocess.run(["bash", str(root / "scripts/sanitize.sh")],
+                       capture_output=True, text=True, timeout=420)
+    assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-1500:])
+    assert "asan multirank OK" in r.stdout

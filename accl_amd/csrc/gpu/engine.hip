@@ -54,14 +54,24 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
     s.tiles_total = 0;
     __hip_atomic_store(&s.tiles_done, 0u, __ATOMIC_RELAXED, AGENT);
     // bounded wait: if the sibling wave is wedged/slow, execute the move
-    // here (scalar) — double execution of a deterministic elementwise move
-    // is idempotent, and an unbounded spin here would deadlock the engine
+    // here (scalar). Double execution of a NON-aliasing elementwise move is
+    // idempotent (same inputs, same outputs); an ALIASING fused reduce
+    // (dst is also an operand, e.g. dst += slot) is NOT — re-execution or a
+    // racing partial sibling pass would double-accumulate, so for those we
+    // only ever wait (the sibling shares our workgroup: if it is dead the
+    // kernel is gone anyway, and the host's bounded waits still fire).
+    bool aliased = false;
+    u64 dbytes = m.count * dtype_size(DataType(m.dst_dt));
+    for (u32 k = 0; k < m.nsrc; ++k) {
+      u64 sb = m.count * dtype_size(DataType(m.src_dt[k]));
+      if (m.src[k] < m.dst + dbytes && m.dst < m.src[k] + sb) aliased = true;
+    }
     u64 dl = wallclock() + 100000;  // 1 ms
     for (;;) {
       if (__hip_atomic_load(&mb->done, __ATOMIC_ACQUIRE,
                             __HIP_MEMORY_SCOPE_WORKGROUP) == sq)
         break;
-      if (wallclock() > dl) {
+      if (!aliased && wallclock() > dl) {
         execute_move_range(m, 0, m.count);
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
         __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");

@@ -45,6 +45,9 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
     if (dbg) dbg[5] = wallclock();
     SmallMb* mb = (SmallMb*)small_mb;
     mb->d = m;
+    mb->d.tile_log2 = 0;  // sibling executes tile 0 only: one default-size
+                          // tile always covers a <=32KB move, a small
+                          // ACCL_TILE_KB override would not
     u64 sq = ++small_seq;
     __hip_atomic_store(&mb->seq, sq, __ATOMIC_RELEASE,
                        __HIP_MEMORY_SCOPE_WORKGROUP);

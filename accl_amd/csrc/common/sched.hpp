@@ -749,6 +749,7 @@ struct Cclo {
     const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
     char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
     u32 r = c.rank;
+    if (2 * u64(c.size) - 1 > MAX_FLOWS) return allgather_batched(d, c);
     u32 nf = 0;
     mk_local(nf++, src, dt, dst + u64(r) * n * dtype_size(dt), dt, n);
     bool direct = use_rndzv(n, dt, wdt) && (d.flags & F_DST_ARENA) &&
@@ -772,6 +773,32 @@ struct Cclo {
       mk_rx(nf++, c.global(p), dst + u64(p) * n * dtype_size(dt), dt, wdt, n, tag);
     }
     return run_flows(nf);
+  }
+
+  // large-P fallback: (tx,rx) pairs in batches that fit the flow table
+  // (P > MAX_FLOWS/2 would overflow flows[] — LDS corruption otherwise)
+  ACCL_HD u32 allgather_batched(const CallDesc& d, const CommView& c) {
+    u64 n = desc_count(d);
+    DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    u32 tag = TAG_COLL | (u32(Op::allgather) << 16) | d.comm_id;
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    u32 r = c.rank;
+    mk_local(0, src, dt, dst + u64(r) * n * dtype_size(dt), dt, n);
+    u32 e = run_flows(1);
+    if (e) return e;
+    const u32 B = (MAX_FLOWS / 2) - 1;
+    for (u32 base = 0; base < c.size; base += B) {
+      u32 nf = 0;
+      for (u32 j = base; j < c.size && j < base + B; ++j) {
+        if (j == r) continue;
+        mk_tx(nf++, c.global(j), src, dt, wdt, n, tag);
+        mk_rx(nf++, c.global(j), dst + u64(j) * n * dtype_size(dt), dt, wdt,
+              n, tag);
+      }
+      if (nf && (e = run_flows(nf))) return e;
+    }
+    return E_OK;
   }
 
   // reduce at root: fan-in with a serialized reduce chain per segment
@@ -817,6 +844,7 @@ struct Cclo {
       mk_local(0, src, dt, dst, dt, n);
       return run_flows(1);
     }
+    if (2 * u64(c.size) - 1 > MAX_FLOWS) return reduce_scatter_batched(d, c);
     u32 nf = 0;
     // outbound: my chunk p -> rank p, all links at once
     for (u32 p = 0; p < c.size; ++p)
@@ -833,6 +861,34 @@ struct Cclo {
       nf++;
     }
     return run_flows(nf);
+  }
+
+  // large-P fallback: batched outbound, then one fused recv-reduce at a time
+  ACCL_HD u32 reduce_scatter_batched(const CallDesc& d, const CommView& c) {
+    u64 n = desc_count(d);
+    DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
+    u32 tag = TAG_COLL | (u32(Op::reduce_scatter) << 16) | d.comm_id;
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    u32 r = c.rank;
+    const u32 B = MAX_FLOWS - 1;
+    u32 e;
+    for (u32 base = 0; base < c.size; base += B) {
+      u32 nf = 0;
+      for (u32 p = base; p < c.size && p < base + B; ++p)
+        if (p != r)
+          mk_tx(nf++, c.global(p), src + u64(p) * n * dtype_size(dt), dt, wdt,
+                n, tag);
+      if (nf && (e = run_flows(nf))) return e;
+    }
+    mk_local(0, src + u64(r) * n * dtype_size(dt), dt, dst, dt, n);
+    if ((e = run_flows(1))) return e;
+    for (u32 p = 0; p < c.size; ++p) {
+      if (p == r) continue;
+      mk_rx(0, c.global(p), dst, dt, wdt, n, tag, dst, dt, int(d.function));
+      if ((e = run_flows(1))) return e;
+    }
+    return E_OK;
   }
 
   // allreduce = fullmesh reduce_scatter + fullmesh allgather, all phases as
@@ -937,6 +993,23 @@ struct Cclo {
     u32 nf = 0;
     mk_local(nf++, src + u64(r) * n * dtype_size(dt), dt,
              dst + u64(r) * n * dtype_size(dt), dt, n);
+    if (2 * u64(c.size) - 1 > MAX_FLOWS) {
+      u32 e = run_flows(1);  // self copy first
+      if (e) return e;
+      const u32 B = (MAX_FLOWS / 2) - 1;
+      for (u32 base = 0; base < c.size; base += B) {
+        nf = 0;
+        for (u32 j = base; j < c.size && j < base + B; ++j) {
+          if (j == r) continue;
+          mk_tx(nf++, c.global(j), src + u64(j) * n * dtype_size(dt), dt, wdt,
+                n, tag);
+          mk_rx(nf++, c.global(j), dst + u64(j) * n * dtype_size(dt), dt, wdt,
+                n, tag);
+        }
+        if (nf && (e = run_flows(nf))) return e;
+      }
+      return E_OK;
+    }
     for (u32 p = 0; p < c.size; ++p) {
       if (p == r) continue;
       mk_tx(nf++, c.global(p), src + u64(p) * n * dtype_size(dt), dt, wdt, n, tag);

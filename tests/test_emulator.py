@@ -855,3 +855,33 @@ def test_sanitized_protocol():
                        capture_output=True, text=True, timeout=420)
     assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-1500:])
     assert "asan multirank OK" in r.stdout
+
+
+def _large_p(a, rank, n):
+    # P > MAX_FLOWS/2: exercises the batched allgather/reduce_scatter/
+    # alltoall fallbacks (flow-table bound)
+    cnt = 64
+    s, d = _mk(a, cnt), _mk(a, cnt * n)
+    s.write(pattern(cnt, rank, seed=41))
+    a.allgather(s, d, cnt)
+    exp = np.concatenate([pattern(cnt, r, seed=41) for r in range(n)])
+    assert np.array_equal(rd(d, cnt * n), exp)
+    s2, d2 = _mk(a, cnt * n), _mk(a, cnt)
+    s2.write(np.concatenate([pattern(cnt, rank + 3 * j, seed=42)
+                             for j in range(n)]))
+    a.reduce_scatter(s2, d2, cnt, RF.SUM)
+    exp = np.stack([pattern(cnt, r + 3 * rank, seed=42)
+                    for r in range(n)]).sum(0)
+    assert np.allclose(rd(d2, cnt), exp)
+    a.alltoall(s2, d, cnt)
+    exp = np.concatenate([pattern(cnt, r + 3 * rank, seed=42)
+                          for r in range(n)])
+    assert np.array_equal(rd(d, cnt * n), exp)
+    a.barrier()
+
+
+def test_large_p_batched():
+    run_ranks(_large_p, 38,
+              opts={"n_slots": 4, "slot_bytes": 4096, "n_stream": 2,
+                    "stream_bytes": 2048, "timeout_us": 60_000_000},
+              timeout=280)

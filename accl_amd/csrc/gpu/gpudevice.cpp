@@ -36,17 +36,18 @@ GpuDevice::GpuDevice(u32 nranks, u32 rank, int device_index,
   layout_ = arena_layout(cfg_);
   arena_bytes_ = layout_.total_ctl_bytes + heap_bytes;
 
-  // Fine-grained HBM so peer stores + system-scope flags are coherent over
-  // xGMI without kernel boundaries. ACCL_COARSE_ARENA=1 falls back to
-  // coarse (measurement escape hatch).
-  bool coarse = std::getenv("ACCL_COARSE_ARENA") != nullptr;
+  // Coarse-grained HBM by default: the standard hipMalloc+IPC path (what
+  // RCCL uses) — cross-agent visibility comes from system-scope atomics and
+  // release fences, and measured bandwidth is identical to fine-grained.
+  // ACCL_FINE_ARENA=1 opts into fine-grained allocation.
+  bool fine = std::getenv("ACCL_FINE_ARENA") != nullptr;
   hipError_t e = hipErrorUnknown;
-  if (!coarse) {
+  if (fine) {
     e = hipExtMallocWithFlags((void**)&arena_base_, arena_bytes_,
                               hipDeviceMallocFinegrained);
     fine_grained_ = (e == hipSuccess);
   }
-  if (coarse || e != hipSuccess) {
+  if (!fine || e != hipSuccess) {
     hip_check(hipMalloc((void**)&arena_base_, arena_bytes_), "hipMalloc arena");
     fine_grained_ = false;
   }

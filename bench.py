@@ -261,7 +261,7 @@ def main():
                                 "100 Gbps (12.5 GB/s) line rate"),
             },
         }
-        print(json.dumps(out))
+        print(json.dumps(out), flush=True)
     a.close()
     if dist is not None:
         dist.barrier()
@@ -269,4 +269,12 @@ def main():
 
 
 if __name__ == "__main__":
-    main()
+    try:
+        main()
+    except Exception:
+        # post-mortem detail for multi-rank failures (driver captures stderr)
+        import traceback
+        rank = os.environ.get("RANK", "?")
+        print(f"[bench rank {rank}] FAILED:", file=sys.stderr, flush=True)
+        traceback.print_exc()
+        raise

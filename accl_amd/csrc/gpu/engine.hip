@@ -1,17 +1,21 @@
-// The persistent CCLO engine kernel for MI355X (gfx950).
+// The persistent CCLO engine for MI355X (gfx950): TWO kernels on separate
+// non-blocking streams (docs/DESIGN.md "Persistent engine").
 //
-// One cooperative-by-construction grid per GPU (grid <= 1 workgroup per CU,
-// so residency is guaranteed by size — cdna_hip_programming.md §1): lane 0 of
-// workgroup 0 runs the collective scheduler (common/sched.hpp, the microcode
-// analogue of reference ccl_offload_control.c); every other wave is a mover
-// executing MoveDescs — vectorized copy / cast / n-ary reduce between local
-// HBM and peer HBM over xGMI (the data plane: reference dma_mover +
-// reduce_ops + hp_compression, kernels/cclo/hls, kernels/plugins).
+//  * accl_scheduler_kernel — one 128-thread WG: wave 0 lane 0 runs the
+//    collective scheduler (common/sched.hpp, the microcode analogue of
+//    reference ccl_offload_control.c) with the Cclo state staged in LDS;
+//    wave 1 is the inline small-mover (sub-32KB moves over an LDS mailbox).
+//  * accl_mover_kernel — the mover fleet (~10 of 16 wave slots per CU):
+//    static rotated tile partitioning, replicated packed doorbells,
+//    non-temporal software-pipelined copy / cast / n-ary reduce tiles
+//    between local HBM and peer HBM over xGMI (the data plane: reference
+//    dma_mover + reduce_ops + hp_compression, kernels/cclo/hls, plugins).
 //
 // Memory-ordering discipline (MI355X_MICROARCH.md §Workgroup dispatch):
-//  * movers: payload stores -> s_waitcnt vmcnt(0) (asm) -> system release
-//    fence -> tiles_done release-add (agent) — the release chain the
-//    scheduler extends to peers when it publishes slot headers.
+//  * movers: payload stores -> s_waitcnt vmcnt(0) (asm) -> release fence
+//    (skippable when the tile used only non-temporal accesses) ->
+//    tiles_done release-add (agent) — the release chain the scheduler
+//    extends to peers when it publishes slot headers (system release).
 //  * all polls are relaxed + s_sleep; one acquire fence after a match.
 #include <hip/hip_runtime.h>
 #include <hip/hip_fp16.h>

@@ -109,6 +109,11 @@ void EmuDevice::connect(const std::vector<std::vector<char>>& blobs) {
         throw accl_error("emu: peer arena " + name + " never became ready");
       usleep(1000);
     }
+    // ranks must agree on the arena geometry (a size mismatch would SIGBUS
+    // on access past the peer's shm object, not error)
+    if (((ArenaHdr*)p)->arena_bytes != arena_bytes_)
+      throw accl_error("emu: rank " + std::to_string(r) +
+                       " arena size mismatch (configs must agree)");
     peer_base_[r] = p;
   }
 

@@ -118,6 +118,9 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
     if (r == cfg_.rank) { peer_base_[r] = arena_base_; continue; }
     GpuBlob b{};
     std::memcpy(&b, blobs[r].data(), sizeof(b));
+    if (b.arena_bytes != arena_bytes_)
+      throw accl_error("gpu: rank " + std::to_string(r) +
+                       " arena size mismatch (configs must agree)");
     void* p = nullptr;
     hip_check(hipIpcOpenMemHandle(&p, b.handle, hipIpcMemLazyEnablePeerAccess),
               "hipIpcOpenMemHandle");

@@ -111,6 +111,10 @@ class ACCL:
         if nranks == 1:
             self._a.connect([blob])
             return
+        if callable(bootstrap):
+            # caller-supplied allgather: fn(my_blob, nranks, rank) -> blobs
+            self._a.connect(list(bootstrap(blob, nranks, rank)))
+            return
         if self.backend_name == "emu":
             # shm names are deterministic — compute peers' blobs locally
             blobs = [f"/{emu_job_name(job)}_r{r}".encode() for r in range(nranks)]

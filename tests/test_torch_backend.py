@@ -90,3 +90,59 @@ def test_torch_backend_collectives_and_ddp():
                 p.terminate()
                 errs.append("rank hung")
     assert not errs, "\n".join(errs)
+
+
+# ---------------- GPU engine variant (driver runs with -m gpu) ----------
+def _gpu_rank(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        os.environ.setdefault("ACCL_ENGINE_WGS", "64")  # 2 engines, 1 GPU
+        import torch.distributed as dist
+
+        import accl_amd.torch_backend  # noqa: F401
+        dist.init_process_group("accl", rank=rank, world_size=world)
+        t = torch.full((4096,), float(rank + 1), device="cuda")
+        dist.all_reduce(t)
+        assert torch.allclose(t.cpu(), torch.full((4096,), 3.0))
+        b = torch.full((256,), float(rank), device="cuda")
+        dist.broadcast(b, src=0)
+        assert b.cpu().eq(0.0).all()
+        out = torch.zeros(2 * 128, device="cuda")
+        dist.all_gather_into_tensor(out, torch.full((128,), float(rank + 5),
+                                                    device="cuda"))
+        assert out[:128].cpu().eq(5.0).all() and out[128:].cpu().eq(6.0).all()
+        dist.barrier()
+        dist.destroy_process_group()
+        q.put((rank, None))
+    except Exception:
+        import traceback
+        q.put((rank, traceback.format_exc()))
+
+
+@pytest.mark.gpu
+def test_torch_backend_gpu():
+    try:
+        if not torch.cuda.is_available():
+            pytest.skip("no HIP GPU")
+    except Exception:
+        pytest.skip("no HIP GPU")
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29700 + os.getpid() % 200
+    ps = [ctx.Process(target=_gpu_rank, args=(r, 2, port, q)) for r in range(2)]
+    for p in ps:
+        p.start()
+    errs = []
+    try:
+        for _ in range(2):
+            rank, err = q.get(timeout=180)
+            if err:
+                errs.append(f"rank {rank}:\n{err}")
+    finally:
+        for p in ps:
+            p.join(timeout=15)
+            if p.is_alive():
+                p.terminate()
+                errs.append("rank hung")
+    assert not errs, "\n".join(errs)

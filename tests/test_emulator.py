@@ -885,3 +885,54 @@ def test_large_p_batched():
               opts={"n_slots": 4, "slot_bytes": 4096, "n_stream": 2,
                     "stream_bytes": 2048, "timeout_us": 60_000_000},
               timeout=280)
+
+
+# --------------------------------------------------- fault surface
+def test_peer_death_surfaces_timeout():
+    """A rank dying mid-collective must surface E_TIMEOUT on survivors —
+    never a hang (reference's failure surface is the per-call error word;
+    ours adds deadline-bounded spins everywhere)."""
+    import multiprocessing as mp
+    import os
+    from emu_util import fresh_job
+
+    def victim(job):
+        import accl_amd as A
+        a = A.ACCL(nranks=2, rank=1, backend="emu", job=job,
+                   opts={"timeout_us": 2_000_000})
+        # connect, then die without participating
+        del a._a  # hard teardown path
+        os._exit(0)
+
+    def survivor(job, q):
+        try:
+            import accl_amd as A
+            a = A.ACCL(nranks=2, rank=0, backend="emu", job=job,
+                       opts={"timeout_us": 2_000_000})
+            s = a.create_buffer(1000, DT.float32)
+            d = a.create_buffer(1000, DT.float32)
+            s.write(pattern(1000, 0))
+            try:
+                a.allreduce(s, d, 1000, RF.SUM)
+                q.put("no-error")
+                return
+            except RuntimeError as e:
+                assert "TIMEOUT" in str(e), str(e)
+            # engine stays serviceable for local work after the failure
+            a.soft_reset()
+            a.copy(s, d, 1000)
+            assert np.array_equal(rd(d, 1000), pattern(1000, 0))
+            q.put("ok")
+        except Exception:
+            import traceback
+            q.put(traceback.format_exc())
+
+    ctx = mp.get_context("fork")
+    job = fresh_job()
+    q = ctx.Queue()
+    pv = ctx.Process(target=victim, args=(job,))
+    ps = ctx.Process(target=survivor, args=(job, q))
+    pv.start(); ps.start()
+    res = q.get(timeout=120)
+    pv.join(10); ps.join(10)
+    assert res == "ok", res

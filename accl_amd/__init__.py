@@ -35,9 +35,10 @@ TAG_ANY = _core.TAG_ANY
 GLOBAL_COMM = _core.GLOBAL_COMM
 error_to_string = _core.error_to_string
 
-from .accl import ACCL, emu_job_name, generate_ranks  # noqa: E402,F401
+from .accl import (ACCL, emu_job_name, generate_ranks,  # noqa: E402,F401
+                   load_tuning)
 
 __all__ = [
     "ACCL", "DataType", "ReduceFunction", "TAG_ANY", "GLOBAL_COMM",
-    "error_to_string", "emu_job_name", "generate_ranks",
+    "error_to_string", "emu_job_name", "generate_ranks", "load_tuning",
 ]

@@ -34,6 +34,21 @@ def generate_ranks(config_file=None, local=True, world_size=1,
              "max_segment_size": rxbuf_size} for i, ip in enumerate(ips)]
 
 
+def load_tuning(path, message_bytes=None):
+    """Read a tools/autotune.py table; returns the opts dict for the size
+    bracket closest to message_bytes (largest entry if None)."""
+    with open(path) as f:
+        meta = json.load(f)
+    table = meta["table"]
+    keys = sorted(int(k) for k in table)
+    if message_bytes is None:
+        pick = keys[-1]
+    else:
+        pick = min(keys, key=lambda k: abs(k - message_bytes))
+    e = table[str(pick)]
+    return {"slot_bytes": int(e["slot_bytes"]), "n_slots": int(e["n_slots"])}
+
+
 def emu_job_name(seed=None):
     """A job name every rank can derive identically (pass a shared seed)."""
     seed = seed if seed is not None else os.environ.get("ACCL_EMU_JOB", "accljob")

@@ -936,3 +936,22 @@ def test_peer_death_surfaces_timeout():
     res = q.get(timeout=120)
     pv.join(10); ps.join(10)
     assert res == "ok", res
+
+
+def test_load_tuning(tmp_path):
+    import json
+    t = {"backend": "emu", "ranks": 2, "collective": "allreduce",
+         "table": {"16384": {"slot_bytes": 65536, "n_slots": 8, "usec": 1},
+                   "4194304": {"slot_bytes": 1 << 20, "n_slots": 8,
+                               "usec": 2}}}
+    p = tmp_path / "t.json"
+    p.write_text(json.dumps(t))
+    o = A.load_tuning(str(p), message_bytes=20000)
+    assert o == {"slot_bytes": 65536, "n_slots": 8}
+    o = A.load_tuning(str(p))
+    assert o["slot_bytes"] == 1 << 20
+    a = A.ACCL(nranks=1, rank=0, backend="emu", job="tload", opts=o)
+    try:
+        assert a.info()["eager_slot_bytes"] == 1 << 20
+    finally:
+        a.close()

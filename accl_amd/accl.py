@@ -187,6 +187,22 @@ class ACCL:
         cap = self._a.buffer_dlpack(buf, self.device_index)
         return torch.from_dlpack(cap)
 
+    def reduce_from_stream(self, lane, count, root, func, comm=0,
+                           dst_stream=None, dtype=DataType.float32):
+        """Reduce where this rank's operand comes from stream ring `lane`
+        (reference: reduce stream2mem/stream2stream, test.cpp matrix) —
+        composition of copy_from_stream + reduce (+ stream_put of the root
+        result when dst_stream is given). Returns the result buffer on the
+        root, else None."""
+        staged = self.create_buffer(count, dtype)
+        self.copy_from_stream(lane, staged, count)
+        out = self.create_buffer(count, dtype)
+        self.reduce(staged, out, count, root, func, comm=comm)
+        if self._a.rank == root and dst_stream is not None:
+            self.stream_put(out, count, dst=dst_stream[0],
+                            tag=dst_stream[1], comm=comm)
+        return out if self._a.rank == root else None
+
     def buffer_like(self, tensor, device_only=True):
         dt = {v: k for k, v in _torch_dtype_map().items()}[tensor.dtype]
         return self.create_buffer(tensor.numel(), dt, device_only)

@@ -955,3 +955,27 @@ def test_load_tuning(tmp_path):
         assert a.info()["eager_slot_bytes"] == 1 << 20
     finally:
         a.close()
+
+
+def _reduce_stream2stream(a, rank, n):
+    # reference: test_reduce_stream2mem / stream2stream — operand from the
+    # kernel stream, result to a stream (composition helper)
+    cnt = 800
+    src = a.create_buffer(cnt, DT.float32)
+    src.write(pattern(cnt, rank, seed=51))
+    a.stream_put(src, cnt, dst=rank, tag=1)       # own lane = "krnl stream"
+    out = a.reduce_from_stream(rank, cnt, root=0, func=RF.SUM,
+                               dst_stream=(1, 7))
+    if rank == 0:
+        exp = np.stack([pattern(cnt, r, seed=51) for r in range(n)]).sum(0)
+        assert np.allclose(rd(out, cnt), exp)
+    elif rank == 1:
+        buf = np.zeros(cnt, np.float32)
+        nb, tag = a.pop_stream(0, buf)
+        exp = np.stack([pattern(cnt, r, seed=51) for r in range(n)]).sum(0)
+        assert nb == cnt * 4 and tag == 7 and np.allclose(buf, exp)
+    a.barrier()
+
+
+def test_reduce_stream2stream():
+    run_ranks(_reduce_stream2stream, 2, opts=STREAM_SMALL)

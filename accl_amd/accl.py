@@ -89,7 +89,8 @@ class ACCL:
             # slot size come from it (reference: ACCL ctor takes the rank
             # vector, driver/xrt/include/accl.hpp:57)
             nranks = len(ranks)
-            opts.setdefault("opts", {}).setdefault(
+            # opts itself is the kwargs dict make_config reads (top-level key)
+            opts.setdefault(
                 "slot_bytes", int(ranks[0].get("max_segment_size", 1 << 20)))
         if nranks is None:
             nranks = int(os.environ.get("WORLD_SIZE", "1"))
@@ -225,6 +226,7 @@ class ACCL:
                     "copy_from_stream", "send_from_stream", "alive",
                     "soft_reset",
                     "info", "set_timeout_ms", "set_max_eager_size",
+                    "set_max_rendezvous_size",
                     "dump_communicator", "dump_eager_rx_buffers",
                     "dump_streams", "dump_engine_status",
                     "create_communicator", "split_communicator",

@@ -293,6 +293,8 @@ PYBIND11_MODULE(_core, m) {
            })
       .def("set_timeout_ms", &ACCL::set_timeout_ms,
            py::call_guard<py::gil_scoped_release>())
+      .def("set_max_rendezvous_size", &ACCL::set_max_rendezvous_size,
+           py::call_guard<py::gil_scoped_release>())
       .def("set_max_eager_size", &ACCL::set_max_eager_size,
            py::call_guard<py::gil_scoped_release>())
       .def("dump_communicator", &ACCL::dump_communicator,
@@ -368,6 +370,10 @@ PYBIND11_MODULE(_core, m) {
 
   // host-side injector into the device-call ring: exercises the
   // client_arbiter path on the emulator (and doubles as a raw call API)
+  // TEST-ONLY single-producer injection: emulates device_call() from the
+  // host via read/modify/write of the ring head, which is NOT atomic
+  // against concurrent device-side device_call() fetch_adds. Use only in
+  // harnesses where no kernel produces device calls at the same time.
   m.def("inject_device_call",
         [](ACCL& a, u32 scenario, u64 count, u32 root, u32 tag, u64 addr0,
            u64 addr2, u32 flags, u32 function) {
@@ -379,6 +385,21 @@ PYBIND11_MODULE(_core, m) {
           u64 idx = ring.head;
           ring.head = idx + 1;
           be->write_arena(L.devcall_off, &ring, sizeof(u64));
+          if (idx >= DEVCALL_RING) {
+            // slot-reuse guard (mirrors device_call): wait until the
+            // previous occupant's ret was published by the engine
+            u64 ret_off = L.devcall_off + sizeof(DevCallRing) +
+                          u64(DEVCALL_RING) * sizeof(DevCallSlot) +
+                          (idx % DEVCALL_RING) * sizeof(DevCallRet);
+            u64 t0 = wallclock_host_ns(), prev_seq = 0;
+            do {
+              be->read_arena(ret_off, &prev_seq, sizeof(prev_seq));
+              if (prev_seq >= idx + 1 - DEVCALL_RING) break;
+              if (wallclock_host_ns() - t0 > 30ull * 1000000000)
+                throw accl_error("inject_device_call: ring slot never freed");
+              usleep(50);
+            } while (true);
+          }
           DevCallSlot slot{};
           slot.d.scenario = scenario;
           slot.d.count_lo = u32(count);

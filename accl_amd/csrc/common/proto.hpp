@@ -222,14 +222,17 @@ struct alignas(64) ArenaHdr {
   u64 spare_off;        // staging region for non-arena rendezvous targets
   u64 spare_bytes;      //   (reference: spare buffers, accl.cpp:1174-1196)
   u64 devcall_off;      // DevCallRing + slots + rets (device-initiated calls)
+  u64 dbg_off;          // engine flow-state dump on timeout (debug_dump)
 };
 constexpr u32 ARENA_MAGIC = 0x4143434Cu;  // "ACCL"
 
 struct ArenaLayout {
   u64 eager_off, rndzv_addr_off, rndzv_done_off, stream_off, slots_off,
-      barrier_off, direct_off, spare_off, spare_bytes, devcall_off, heap_off,
-      total_ctl_bytes;
+      barrier_off, direct_off, spare_off, spare_bytes, devcall_off, dbg_off,
+      heap_off, total_ctl_bytes;
 };
+
+constexpr u64 DBG_DUMP_BYTES = 8192;  // flow-state dump region size
 
 inline ArenaLayout arena_layout(const ProtoConfig& c, u64 spare_bytes = 32u << 20) {
   auto align_up = [](u64 x, u64 a) { return (x + a - 1) & ~(a - 1); };
@@ -253,6 +256,8 @@ inline ArenaLayout arena_layout(const ProtoConfig& c, u64 spare_bytes = 32u << 2
   L.devcall_off = off = align_up(off, 256);
   off += sizeof(DevCallRing) + u64(DEVCALL_RING) * sizeof(DevCallSlot) +
          u64(DEVCALL_RING) * sizeof(DevCallRet);
+  L.dbg_off = off = align_up(off, 256);
+  off += DBG_DUMP_BYTES;
   L.slots_off = off = align_up(off, 4096);
   off += u64(c.nranks) * c.n_slots * u64(c.slot_bytes);
   L.spare_off = off = align_up(off, 4096);

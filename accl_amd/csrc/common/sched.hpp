@@ -84,6 +84,9 @@ struct Cclo {
   u32 ncomms;
   u64 timeout_ticks;
   u64 max_eager_bytes;     // above this, arena<->arena transfers go direct
+  u64 max_rndzv_bytes;     // window cap for a single posted rendezvous
+                           // window (0 = unlimited); reference:
+                           // set_max_rendezvous_size (accl.hpp:103-104)
   u32 err;                 // error bits of the current call
   Flow flows[MAX_FLOWS];
 
@@ -283,6 +286,12 @@ struct Cclo {
           if (h->arith != u32(f.wdt)) { err |= E_COMPRESSION; return any; }
           u64 n = u64(h->bytes) / wsz;
           if (n > f.count - f.submitted) { err |= E_SEGMENT; return any; }
+          // a segment is consumed whole: a gated flow (fused reduce chained
+          // on its predecessor's done counter) must wait until the gate
+          // covers the ENTIRE segment, or the reduce would read dst elements
+          // the predecessor has not produced yet (race on the GPU's async
+          // movers; the synchronous emulator mover never exposes it)
+          if (f.gate && f.submitted + n > avail) return any;
           m.dst = (u64)(f.dst + f.submitted * dtype_size(DataType(f.ddt)));
           m.dst_dt = f.ddt;
           m.src[0] = (u64)tv.slot_payload(me(), f.gpeer, slot);
@@ -617,10 +626,25 @@ struct Cclo {
     char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
     if (use_rndzv(n, dt, wdt) && peer != me()) {
       if (d.flags & F_DST_ARENA) {
-        post_addr(peer, d.addr2, n, d.tag, u32(dt));
-        mk_rx_direct(0, peer, n, dt);
-        u32 e = run_flows(1);
-        if (e) return e;
+        // window the posting by max_rndzv_bytes (reference:
+        // set_max_rendezvous_size caps a single rendezvous transfer);
+        // sender follows each posted window (op_send's wait_addr loop)
+        u64 wmax = max_rndzv_bytes ? max_rndzv_bytes / dtype_size(dt) : n;
+        if (!wmax) wmax = 1;
+        u64 posted = 0, got = 0;
+        while (got < n) {
+          while (posted < n && posted - got < 2 * wmax) {
+            u64 w = min64(n - posted, wmax);
+            post_addr(peer, d.addr2 + posted * dtype_size(dt), w, d.tag,
+                      u32(dt));
+            posted += w;
+          }
+          u64 w = min64(n - got, wmax);
+          mk_rx_direct(0, peer, w, dt);
+          u32 e = run_flows(1);
+          if (e) return e;
+          got += w;
+        }
         return wait_done(peer) ? E_OK : err;
       }
       // stage through the spare region's LOWER half in windows (double-
@@ -863,7 +887,11 @@ struct Cclo {
     return run_flows(nf);
   }
 
-  // large-P fallback: batched outbound, then one fused recv-reduce at a time
+  // large-P fallback: (tx, fused recv-reduce) PAIRS batched to the flow
+  // table, reduce chain serialized by gates. Running tx and rx in the same
+  // flow set is what keeps credit flowing: an all-tx phase would stall once
+  // the per-peer message exceeds the eager window, with every rank waiting
+  // for credit only a posted recv returns (deadlock).
   ACCL_HD u32 reduce_scatter_batched(const CallDesc& d, const CommView& c) {
     u64 n = desc_count(d);
     DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
@@ -871,22 +899,23 @@ struct Cclo {
     const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
     char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
     u32 r = c.rank;
-    const u32 B = MAX_FLOWS - 1;
     u32 e;
-    for (u32 base = 0; base < c.size; base += B) {
-      u32 nf = 0;
-      for (u32 p = base; p < c.size && p < base + B; ++p)
-        if (p != r)
-          mk_tx(nf++, c.global(p), src + u64(p) * n * dtype_size(dt), dt, wdt,
-                n, tag);
-      if (nf && (e = run_flows(nf))) return e;
-    }
     mk_local(0, src + u64(r) * n * dtype_size(dt), dt, dst, dt, n);
     if ((e = run_flows(1))) return e;
-    for (u32 p = 0; p < c.size; ++p) {
-      if (p == r) continue;
-      mk_rx(0, c.global(p), dst, dt, wdt, n, tag, dst, dt, int(d.function));
-      if ((e = run_flows(1))) return e;
+    const u32 B = MAX_FLOWS / 2;
+    for (u32 base = 0; base < c.size; base += B) {
+      u32 nf = 0;
+      const u64* gate = nullptr;  // predecessor finished in an earlier batch
+      for (u32 p = base; p < c.size && p < base + B; ++p) {
+        if (p == r) continue;
+        mk_tx(nf++, c.global(p), src + u64(p) * n * dtype_size(dt), dt, wdt,
+              n, tag);
+        mk_rx(nf, c.global(p), dst, dt, wdt, n, tag, dst, dt, int(d.function),
+              gate);
+        gate = &flows[nf].done;
+        nf++;
+      }
+      if (nf && (e = run_flows(nf))) return e;
     }
     return E_OK;
   }
@@ -1094,6 +1123,57 @@ struct Cclo {
     return run_flows(1);
   }
 
+  // ---- timeout forensics ----
+  // On E_TIMEOUT, snapshot every live flow (plus what it is waiting on:
+  // awaited slot seq vs observed header, credit, direct progress, mover poll
+  // state of the head pending segment) into the arena's dbg region. The
+  // host appends the formatted dump to the thrown error
+  // (Backend::timeout_dump_str), so a wedged GPU run is root-causeable from
+  // the pytest log alone.
+  ACCL_HD void dump_timeout(u32 scen) {
+    volatile u64* w = (volatile u64*)(tv.arena[me()] + tv.hdr(me())->dbg_off);
+    u32 nd = 0;
+    for (u32 i = 0; i < MAX_FLOWS && nd < 24; ++i) {
+      Flow& f = flows[i];
+      if (f.kind == FLOW_IDLE || flow_done(f)) continue;
+      volatile u64* fw = w + 8 + u64(nd) * 16;
+      for (int k = 0; k < 16; ++k) fw[k] = 0;
+      fw[0] = u64(f.kind) | (u64(f.to_stream) << 8) | (u64(f.func) << 16) |
+              (u64(f.gpeer) << 32);
+      fw[1] = u64(f.tag) | (u64(f.matched_tag) << 32);
+      fw[2] = f.count; fw[3] = f.submitted; fw[4] = f.done;
+      fw[5] = u64(f.ph) | (u64(f.pt) << 32);
+      fw[6] = f.gate ? ld_sys((const volatile u64*)f.gate) : ~0ull;
+      if (f.kind == FLOW_RX) {
+        u64 seq = sq.eager_rx[f.gpeer] + 1;
+        u32 sl = u32((seq - 1) % cfg.n_slots);
+        SlotHdr* h = tv.slot_hdr(me(), f.gpeer, sl);
+        fw[7] = seq;
+        fw[8] = ld_sys(&h->seq);
+        fw[9] = u64(h->tag) | (u64(h->bytes) << 32);
+      } else if (f.kind == FLOW_TX) {
+        fw[7] = sq.eager_tx[f.gpeer];
+        fw[8] = tx_credit(f.gpeer);
+      } else if (f.kind == FLOW_RX_DIRECT) {
+        fw[7] = f.prog_base;
+        fw[8] = ld_sys(tv.direct_word(me(), f.gpeer));
+      }
+      if (f.ph != f.pt) {
+        PendSeg& p = f.pend[f.ph % FLOW_INFLIGHT];
+        fw[11] = u64(p.token) | (u64(mv->poll(p.token) ? 1 : 0) << 32);
+        fw[12] = p.elems;
+        fw[13] = p.seq;
+      }
+      nd++;
+    }
+    w[1] = u64(scen) | (u64(err) << 32);
+    w[2] = nd;
+    w[3] = wallclock();
+    w[4] = u64(cfg.rank) | (u64(cfg.nranks) << 32);
+    fence_release_sys();
+    st_sys((volatile u64*)&w[0], ld_sys((const volatile u64*)&w[0]) + 1);
+  }
+
   ACCL_HD bool device_call_pending(u64 consumed) const {
     const DevCallSlot* s = tv.devcall_slot(cfg.rank, u32(consumed % DEVCALL_RING));
     return ld_sys(&s->seq) == consumed + 1;
@@ -1127,6 +1207,12 @@ struct Cclo {
   // ---------------- dispatch ----------------
   // reference: run() scenario switch (ccl_offload_control.c:2375-2459)
   ACCL_HD u32 run_call(const CallDesc& d) {
+    u32 e = run_call_inner(d);
+    if (e & E_TIMEOUT) dump_timeout(d.scenario);
+    return e;
+  }
+
+  ACCL_HD u32 run_call_inner(const CallDesc& d) {
     err = 0;
     if (d.comm_id >= ncomms && Op(d.scenario) != Op::copy &&
         Op(d.scenario) != Op::combine && Op(d.scenario) != Op::config &&
@@ -1161,6 +1247,9 @@ struct Cclo {
         return E_OK;
       case CfgFunc::set_max_eager_size:
         max_eager_bytes = desc_count(d);
+        return E_OK;
+      case CfgFunc::set_max_rendezvous_size:
+        max_rndzv_bytes = desc_count(d);
         return E_OK;
       case CfgFunc::reset: {
         // soft reset (reference: encore_soft_reset drains retry queue +

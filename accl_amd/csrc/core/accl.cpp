@@ -291,6 +291,11 @@ void ACCL::set_max_eager_size(u64 bytes) {
   d.function = u32(CfgFunc::set_max_eager_size);
   be_->call(d);
 }
+void ACCL::set_max_rendezvous_size(u64 bytes) {
+  CallDesc d = make_desc(Op::config, bytes, DataType::none, DataType::none);
+  d.function = u32(CfgFunc::set_max_rendezvous_size);
+  be_->call(d);
+}
 
 Request* ACCL::copy_from_stream(u32 lane, BaseBuffer& dst, u64 count,
                                 bool to_device, bool run_async) {

@@ -189,6 +189,8 @@ class ACCL {
   // rendezvous sizes accl.hpp:103-104, cfgFunc calls) ---
   void set_timeout_ms(u64 ms);
   void set_max_eager_size(u64 bytes);
+  void set_max_rendezvous_size(u64 bytes);  // window cap for one posted
+                                            // rendezvous transfer
   // local engine soft reset (reference: ACCL soft_reset, accl.cpp:57-69)
   void soft_reset();
 

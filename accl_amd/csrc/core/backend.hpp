@@ -6,6 +6,7 @@
 // buffers + heap). Concrete backends: EmuDevice (CPU engine thread over shm)
 // and GpuDevice (persistent HIP kernel over HBM).
 #pragma once
+#include <cstdio>
 #include <cstring>
 #include <memory>
 #include <mutex>
@@ -85,8 +86,70 @@ class Backend {
   u64 call(const CallDesc& d, RetEntry* out = nullptr) {
     u64 s = submit(d);
     u32 e = wait(s, out);
-    if (e) throw accl_error("accl call failed: " + error_to_string(e), e);
+    if (e) {
+      std::string msg = "accl call failed: " + error_to_string(e);
+      if (e & 1u /*E_TIMEOUT*/) msg += "\n" + timeout_dump_str();
+      throw accl_error(msg, e);
+    }
     return s;
+  }
+
+  // Formatted engine flow-state snapshot written by Cclo::dump_timeout —
+  // what each live flow was waiting on when the deadline fired.
+  std::string timeout_dump_str() {
+    ArenaLayout L = arena_layout(cfg_);
+    std::vector<u64> w(8 + 24 * 16);
+    read_arena(L.dbg_off, w.data(), w.size() * sizeof(u64));
+    if (!w[0]) return "(no engine flow dump)";
+    static const char* kinds[] = {"IDLE", "LOCAL", "TX", "RX", "TX_DIRECT",
+                                  "RX_DIRECT"};
+    char buf[256];
+    snprintf(buf, sizeof(buf),
+             "engine flow dump #%llu: scenario=%llu err=0x%llx rank=%llu/%llu "
+             "flows=%llu",
+             (unsigned long long)w[0], (unsigned long long)(w[1] & 0xFFFFFFFF),
+             (unsigned long long)(w[1] >> 32),
+             (unsigned long long)(w[4] & 0xFFFFFFFF),
+             (unsigned long long)(w[4] >> 32), (unsigned long long)w[2]);
+    std::string out = buf;
+    for (u64 i = 0; i < w[2] && i < 24; ++i) {
+      const u64* f = &w[8 + i * 16];
+      u32 kind = u32(f[0] & 0xFF);
+      snprintf(buf, sizeof(buf),
+               "\n  [%llu] %s peer=%llu func=%llu tag=0x%llx cnt=%llu "
+               "sub=%llu done=%llu ph/pt=%llu/%llu gate=%lld",
+               (unsigned long long)i, kind < 6 ? kinds[kind] : "?",
+               (unsigned long long)(f[0] >> 32),
+               (unsigned long long)((f[0] >> 16) & 0xFF),
+               (unsigned long long)(f[1] & 0xFFFFFFFF),
+               (unsigned long long)f[2], (unsigned long long)f[3],
+               (unsigned long long)f[4], (unsigned long long)(f[5] & 0xFFFFFFFF),
+               (unsigned long long)(f[5] >> 32), (long long)f[6]);
+      out += buf;
+      if (kind == 3) {  // RX: awaited seq vs observed slot header
+        snprintf(buf, sizeof(buf),
+                 " await_seq=%llu hdr_seq=%llu hdr_tag=0x%llx hdr_bytes=%llu",
+                 (unsigned long long)f[7], (unsigned long long)f[8],
+                 (unsigned long long)(f[9] & 0xFFFFFFFF),
+                 (unsigned long long)(f[9] >> 32));
+        out += buf;
+      } else if (kind == 2) {  // TX: sent vs credit
+        snprintf(buf, sizeof(buf), " sent=%llu credit=%llu",
+                 (unsigned long long)f[7], (unsigned long long)f[8]);
+        out += buf;
+      } else if (kind == 5) {  // RX_DIRECT: progress word
+        snprintf(buf, sizeof(buf), " prog_base=%llu prog=%llu",
+                 (unsigned long long)f[7], (unsigned long long)f[8]);
+        out += buf;
+      }
+      if ((f[5] & 0xFFFFFFFF) != (f[5] >> 32)) {
+        snprintf(buf, sizeof(buf), " head_pend{tok=%llu done=%llu elems=%llu}",
+                 (unsigned long long)(f[11] & 0xFFFFFFFF),
+                 (unsigned long long)(f[11] >> 32), (unsigned long long)f[12]);
+        out += buf;
+      }
+    }
+    return out;
   }
 
   // --- arena memory ---

@@ -53,6 +53,7 @@ EmuDevice::EmuDevice(u32 nranks, u32 rank, const std::string& job,
   h->spare_off = layout_.spare_off;
   h->spare_bytes = layout_.spare_bytes;
   h->devcall_off = layout_.devcall_off;
+  h->dbg_off = layout_.dbg_off;
   __atomic_store_n(&h->magic, ARENA_MAGIC, __ATOMIC_RELEASE);
 
   heap_.init(layout_.heap_off, arena_bytes_ - layout_.heap_off);

@@ -133,10 +133,11 @@ __device__ inline u32 stream_pop(const StreamRx& r, u64 seq, void* dst,
   const int lane = int(threadIdx.x) & 63;
   u32 slot = u32((seq - 1) % r.n_stream);
   SlotHdr* h = (SlotHdr*)(r.my_arena + r.hdr_off_mine) + slot;
-  if (lane == 0)
-    while (__hip_atomic_load(&h->seq, __ATOMIC_RELAXED, ACCL_DEV_SYS) != seq)
-      __builtin_amdgcn_s_sleep(8);
-  __builtin_amdgcn_s_barrier();  // all lanes of wave see arrival via lane 0
+  // every lane spins convergently under the exec mask (no workgroup barrier:
+  // this function is wave-collective and may be called from ONE wave of a
+  // multi-wave workgroup — an s_barrier here would hang the sibling waves)
+  while (__hip_atomic_load(&h->seq, __ATOMIC_RELAXED, ACCL_DEV_SYS) != seq)
+    __builtin_amdgcn_s_sleep(8);
   __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
   u32 bytes = h->bytes;
   if (tag && lane == 0) *tag = h->tag;

@@ -109,12 +109,14 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
   // doorbell word packs (head, latest move's tile count): a wave that is
   // caught up decides ownership without touching the descriptor line, so a
   // 1-tile move wakes 1 wave's worth of desc traffic, not the whole fleet's.
-  // PLAIN volatile stores: they pipeline (atomic stores issue serially at
-  // ~0.4us each — 64 of them cost 26us, measured); the fence above ordered
-  // the desc, and readers use atomic loads that hit the same L2.
+  // ATOMIC relaxed stores (-> global_store sc0 sc1, write-through): a plain
+  // store stays dirty in THIS XCD's L2, which the other XCDs' L2-served
+  // polls never see until a later fence writes it back — the final move of
+  // a flow set then never executes and the engine times out (guideline-16
+  // "plain flag" invalid form; root cause of the round-1 fresh-box hang).
   u64 packed = ((h + 1) << 24) | (s.tiles_total & 0xFFFFFFu);
   for (u32 i = 0; i < DOORBELL_REPS; ++i)
-    *(volatile u64*)&rep[i][0] = packed;
+    __hip_atomic_store(&rep[i][0], packed, __ATOMIC_RELAXED, AGENT);
   head_cache = h + 1;
   if (dbg) dbg[13] = wallclock();
   return u32(h);
@@ -500,9 +502,16 @@ __device__ void mover_main(GpuEngineState* S) {
       if (idle < 64) __builtin_amdgcn_s_sleep(8);
       else if (idle < 160) __builtin_amdgcn_s_sleep(32);
       else if (idle < 240) __builtin_amdgcn_s_sleep(64);
-      else __builtin_amdgcn_s_sleep(127);  // deep idle: ~3 us wake worst-case
+      else {
+        __builtin_amdgcn_s_sleep(127);  // deep idle: ~3 us wake worst-case
+        // liveness net: consult the true head (atomic, always visible) in
+        // case this wave's doorbell replica was ever missed
+        u64 hv = __hip_atomic_load(S->mover.head, __ATOMIC_RELAXED, AGENT);
+        if (hv > cursor) { h = hv; latest_total = 0xFFFFFFu; goto have_work; }
+      }
       continue;
     }
+  have_work:
     idle = 0;
     __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     while (cursor < h) {
@@ -513,11 +522,24 @@ __device__ void mover_main(GpuEngineState* S) {
       // descriptor read (the common small-move fast path)
       if (cursor == h - 1 && first >= latest_total) { cursor++; continue; }
       const MoveDesc& m = ring[slot];
-      // desc publish happens-before the doorbell store; sanity-check epoch
-      if (u32(m.epoch) != u32(cursor + 1)) break;  // not visible yet: retry
+      // desc publish happens-before the doorbell store; sanity-check epoch.
+      // Epoch AHEAD of cursor+1 = the slot was recycled past us by a burst
+      // of >MOVE_RING inline submits (which never ring the fleet doorbell):
+      // the occupant we missed completed inline, so step over it. Only an
+      // OLDER epoch means the desc is not visible yet.
+      if (u32(m.epoch) != u32(cursor + 1)) {
+        if (i32(u32(m.epoch) - u32(cursor + 1)) > 0) { cursor++; continue; }
+        break;  // not published yet: retry via doorbell
+      }
       u32 total = m.inline_done ? 0 : move_tiles(m);
       if (first < total) {
         if (lane == 0 && first == 0) S->dbg[1] = wallclock();
+        // System acquire before touching payload: slot/spare sources were
+        // written by a PEER (xGMI or another process on this GPU) and this
+        // XCD's L2 may hold stale clean lines from a previous occupancy of
+        // the same addresses; a persistent kernel never gets the implicit
+        // launch-boundary invalidate ordinary kernels rely on.
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
         u32 cnt = 0;
         for (u32 t = first; t < total; t += nwaves, ++cnt) run_tile(m, t, lane);
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");

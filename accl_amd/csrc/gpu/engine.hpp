@@ -47,7 +47,14 @@ struct GpuMover {
 #endif
 };
 
-constexpr u32 DOORBELL_REPS = 64;
+// Doorbell replicas are published with ATOMIC relaxed agent stores (they
+// lower to global_store sc0 sc1 — write-through, line dropped from the
+// producer XCD's L2). A PLAIN store would sit dirty in the scheduler's XCD
+// L2 (per-XCD L2s are not coherent) and only become visible to the other 7
+// XCDs' movers when a LATER release fence wrote it back — so the last move
+// of a flow set was invisible to ~7/8 of the fleet and the engine hung when
+// its owner waves were cross-XCD (round-1 fresh-box allgather_rs timeout).
+constexpr u32 DOORBELL_REPS = 16;
 
 struct GpuEngineState {
   Cclo<GpuMover> cclo;     // trivially-copyable; host fills, device runs

@@ -71,6 +71,7 @@ GpuDevice::GpuDevice(u32 nranks, u32 rank, int device_index,
   h.spare_off = layout_.spare_off;
   h.spare_bytes = layout_.spare_bytes;
   h.devcall_off = layout_.devcall_off;
+  h.dbg_off = layout_.dbg_off;
   h.magic = ARENA_MAGIC;
   hip_check(hipMemcpy(arena_base_, &h, sizeof(h), hipMemcpyHostToDevice),
             "write ArenaHdr");

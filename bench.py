@@ -68,6 +68,8 @@ def main():
         backend = "gpu" if A.ACCL._has_gpu() else "emu"
     slot_mb = 4 if backend == "gpu" else 1
     heap = max(4 * args.bytes + (64 << 20), 1 << 30) if backend == "gpu" else None
+    if heap is not None and args.sweep:
+        heap += 2 << 30  # dedicated 1 GiB sweep src+dst (full-range sweep)
     a = A.ACCL(nranks=world, rank=rank, backend=backend,
                heap_bytes=heap,
                opts={"slot_bytes": slot_mb << 20, "n_slots": 8})
@@ -206,12 +208,28 @@ def main():
     sweep_rows = []
     if args.sweep:
         # reference protocol: count sweep per collective with device-side
-        # duration recorded to CSV (bench.cpp:25-61 + fixture CSV)
+        # duration recorded to CSV (bench.cpp:25-61 + fixture CSV). The
+        # sweep always covers the full 4 KB..1 GiB range: allocate dedicated
+        # sweep buffers when --bytes is smaller than the 1 GiB endpoint.
+        sweep_count = (1 << 30) // esz
+        sweep_count -= sweep_count % max(world, 1)
+        if sweep_count > count:
+            sw_src = a.create_buffer(sweep_count, DT, device_only=True)
+            sw_dst = a.create_buffer(sweep_count, DT, device_only=True)
+            if backend == "gpu":
+                import torch
+                t = a.tensor(sw_src)
+                t.copy_(torch.randn(sweep_count, device=t.device,
+                                    dtype=torch.float32).to(t.dtype))
+                torch.cuda.current_stream().synchronize()
+        else:
+            sw_src, sw_dst = src, dst
+            sweep_count = count
         sz = 4096
         while sz <= (1 << 30):
             c = (sz // esz) - ((sz // esz) % max(world, 1))
-            if 0 < c <= count:
-                s2, d2 = src.slice(0, c), dst.slice(0, c)
+            if 0 < c <= sweep_count:
+                s2, d2 = sw_src.slice(0, c), sw_dst.slice(0, c)
                 dev_us = [0.0]
 
                 def one(s=s2, d=d2, c=c):

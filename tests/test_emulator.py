@@ -979,3 +979,80 @@ def _reduce_stream2stream(a, rank, n):
 
 def test_reduce_stream2stream():
     run_ranks(_reduce_stream2stream, 2, opts=STREAM_SMALL)
+
+
+# ------------------------------------------- round-2 regression coverage
+def _rs_batched_window(a, rank, n):
+    # per-peer message (32 KB) exceeds the eager window (4x4 KB): the
+    # batched fallback must run tx and recv-reduce in the SAME flow set or
+    # every rank stalls waiting for credit only a posted recv returns
+    cnt = 8192
+    s, d = _mk(a, cnt * n), _mk(a, cnt)
+    s.write(np.concatenate([pattern(cnt, rank + 7 * j, seed=51)
+                            for j in range(n)]))
+    a.reduce_scatter(s, d, cnt, RF.SUM)
+    exp = np.stack([pattern(cnt, r + 7 * rank, seed=51)
+                    for r in range(n)]).sum(0)
+    assert np.allclose(rd(d, cnt), exp)
+    a.barrier()
+
+
+def test_reduce_scatter_batched_window():
+    run_ranks(_rs_batched_window, 38,
+              opts={"n_slots": 4, "slot_bytes": 4096, "n_stream": 2,
+                    "stream_bytes": 2048, "timeout_us": 120_000_000},
+              timeout=420)
+
+
+def _rndzv_window(a, rank, n):
+    # set_max_rendezvous_size caps one posted window: a 40 KB rendezvous
+    # message must flow as multiple 8 KB windows (reference:
+    # set_max_rendezvous_size, driver/xrt/include/accl.hpp:103-104)
+    a.set_max_rendezvous_size(8192)
+    cnt = 10_000  # 40 KB > max_eager(4 KB) -> rendezvous, 5 windows
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    s.write(pattern(cnt, rank, seed=61))
+    if rank == 0:
+        a.send(s, cnt, dst=1, tag=9)
+        a.recv(d, cnt, src=1, tag=10)
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 1, seed=61))
+    else:
+        a.recv(d, cnt, src=0, tag=9)
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 0, seed=61))
+        a.send(s, cnt, dst=0, tag=10)
+    a.barrier()
+
+
+def test_max_rendezvous_window():
+    run_ranks(_rndzv_window, 2, opts=DIRECT)
+
+
+def _back_to_back(a, rank, n):
+    # barrier-free chains of different collectives: a rank running ahead
+    # into the next collective must not head-of-line-block its peer
+    # (round-1 fresh-box failure shape: allgather -> reduce_scatter)
+    cnt = 5000
+    for it in range(4):
+        s, d = _mk(a, cnt), _mk(a, cnt * n)
+        s.write(pattern(cnt, rank, seed=70 + it))
+        a.allgather(s, d, cnt)
+        exp = np.concatenate([pattern(cnt, r, seed=70 + it)
+                              for r in range(n)])
+        assert np.array_equal(rd(d, cnt * n), exp)
+        s2, d2 = _mk(a, cnt * n), _mk(a, cnt)
+        s2.write(np.concatenate([pattern(cnt, 100 * rank + j, seed=80 + it)
+                                 for j in range(n)]))
+        a.reduce_scatter(s2, d2, cnt, RF.SUM)
+        exp = np.stack([pattern(cnt, 100 * r + rank, seed=80 + it)
+                        for r in range(n)]).sum(0)
+        assert np.allclose(rd(d2, cnt), exp)
+        d3 = _mk(a, cnt)
+        a.allreduce(s, d3, cnt, RF.SUM)
+        exp = np.stack([pattern(cnt, r, seed=70 + it) for r in range(n)]).sum(0)
+        assert np.allclose(rd(d3, cnt), exp)
+    a.barrier()
+
+
+def test_back_to_back_no_barrier():
+    run_ranks(_back_to_back, 2, opts=SMALL)
+    run_ranks(_back_to_back, 3, opts=SMALL)

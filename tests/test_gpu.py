@@ -257,3 +257,94 @@ def test_device_call_collective(acc1):
     nb, tag = a.pop_stream(0, out)
     assert nb == cnt * 4 and tag == 31
     assert np.allclose(out, x + 2.5)
+
+
+# ---------------- round-2: barrier-free sequences + stress ----------------
+def _b2b2(a, rank, n):
+    """Barrier-free multi-collective chains (round-1 fresh-box failure
+    shape) — a rank running ahead must not wedge its peer."""
+    for it in range(3):
+        cnt = 50_000 if it == 0 else 7000
+        s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt * n, DT.float32)
+        s.write(pattern(cnt, rank, seed=90 + it))
+        a.allgather(s, d, cnt)
+        exp = np.concatenate([pattern(cnt, r, seed=90 + it) for r in range(n)])
+        assert np.array_equal(rd(d, cnt * n), exp)
+        s2 = a.create_buffer(cnt * n, DT.float32)
+        d2 = a.create_buffer(cnt, DT.float32)
+        s2.write(np.concatenate([pattern(cnt, 100 * rank + j, seed=95 + it)
+                                 for j in range(n)]))
+        a.reduce_scatter(s2, d2, cnt, RF.SUM)
+        exp = np.stack([pattern(cnt, 100 * r + rank, seed=95 + it)
+                        for r in range(n)]).sum(0)
+        assert np.allclose(rd(d2, cnt), exp)
+        d3 = a.create_buffer(cnt, DT.float32)
+        a.allreduce(s, d3, cnt, RF.SUM)
+        exp = np.stack([pattern(cnt, r, seed=90 + it) for r in range(n)]).sum(0)
+        assert np.allclose(rd(d3, cnt), exp)
+
+
+def test_two_ranks_back_to_back():
+    run_ranks(_b2b2, 2, backend="gpu", timeout=240)
+
+
+def _stress2(a, rank, n):
+    """Randomized op mix: sizes spanning inline (<32 KB) and fleet moves,
+    bursts of small ops followed by large ones (the mover-ring recycle
+    path), mixed tags."""
+    rng = np.random.default_rng(123)  # same sequence on all ranks
+    for it in range(30):
+        op = rng.integers(0, 4)
+        cnt = int(rng.integers(64, 300_000))
+        if op == 0:
+            s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt, DT.float32)
+            s.write(pattern(cnt, rank, seed=it))
+            a.allreduce(s, d, cnt, RF.SUM)
+            exp = np.stack([pattern(cnt, r, seed=it) for r in range(n)]).sum(0)
+            assert np.allclose(rd(d, cnt), exp), f"it={it} allreduce cnt={cnt}"
+        elif op == 1:
+            s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt * n, DT.float32)
+            s.write(pattern(cnt, rank, seed=it))
+            a.allgather(s, d, cnt)
+            exp = np.concatenate([pattern(cnt, r, seed=it) for r in range(n)])
+            assert np.array_equal(rd(d, cnt * n), exp), f"it={it} allgather"
+        elif op == 2:
+            tag = int(rng.integers(1, 1000))
+            s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt, DT.float32)
+            s.write(pattern(cnt, rank, seed=it))
+            if rank == 0:
+                a.send(s, cnt, dst=1, tag=tag)
+            elif rank == 1:
+                a.recv(d, cnt, src=0, tag=tag)
+                assert np.array_equal(rd(d, cnt), pattern(cnt, 0, seed=it))
+        else:
+            b = a.create_buffer(cnt, DT.float32)
+            if rank == 0:
+                b.write(pattern(cnt, 55, seed=it))
+            a.bcast(b, cnt, 0)
+            assert np.array_equal(rd(b, cnt), pattern(cnt, 55, seed=it))
+    a.barrier()
+
+
+def test_two_ranks_gpu_stress():
+    run_ranks(_stress2, 2, backend="gpu", timeout=420)
+
+
+def test_inline_burst_then_fleet(acc1):
+    """>MOVE_RING consecutive sub-32KB moves (inline path, no fleet
+    doorbell) followed by a large fleet move: the mover cursors must step
+    over the recycled slots instead of wedging (round-1 advisor finding)."""
+    a = acc1
+    small = a.create_buffer(256, DT.float32)
+    sd = a.create_buffer(256, DT.float32)
+    x = np.arange(256, dtype=np.float32)
+    small.write(x)
+    for _ in range(80):  # > MOVE_RING=64 inline submits
+        a.copy(small, sd, 256)
+    cnt = 1 << 22  # 16 MB fleet move
+    s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt, DT.float32)
+    y = np.random.default_rng(9).standard_normal(cnt, dtype=np.float32)
+    s.write(y)
+    a.copy(s, d, cnt)
+    assert np.array_equal(rd(d, cnt), y)
+    assert np.array_equal(rd(sd, 256), x)

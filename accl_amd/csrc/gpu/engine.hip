@@ -513,7 +513,17 @@ __device__ void mover_main(GpuEngineState* S) {
     }
   have_work:
     idle = 0;
-    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    // ONE SYSTEM acquire per wake batch: invalidates this CU's L1 and the
+    // XCD L2's stale lines. Covers the descs AND every payload of the
+    // moves <= h — peer payload writes happen-before the scheduler's
+    // doorbell store (peer release -> scheduler acquire at hdr match ->
+    // desc release -> doorbell), so acquiring after the doorbell read is
+    // transitively sufficient. Persistent kernels never get the implicit
+    // launch-boundary invalidate ordinary kernels rely on, and slot
+    // payloads are peer-written: without this, an XCD L2 line cached from
+    // a slot's previous occupancy reads stale.
+    if (!S->no_acq) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+    else __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
     while (cursor < h) {
       u32 slot = u32(cursor % MOVE_RING);
       u32 rot = u32(cursor * 37) % nwaves;
@@ -534,12 +544,6 @@ __device__ void mover_main(GpuEngineState* S) {
       u32 total = m.inline_done ? 0 : move_tiles(m);
       if (first < total) {
         if (lane == 0 && first == 0) S->dbg[1] = wallclock();
-        // System acquire before touching payload: slot/spare sources were
-        // written by a PEER (xGMI or another process on this GPU) and this
-        // XCD's L2 may hold stale clean lines from a previous occupancy of
-        // the same addresses; a persistent kernel never gets the implicit
-        // launch-boundary invalidate ordinary kernels rely on.
-        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
         u32 cnt = 0;
         for (u32 t = first; t < total; t += nwaves, ++cnt) run_tile(m, t, lane);
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");

@@ -58,6 +58,10 @@ constexpr u32 DOORBELL_REPS = 16;
 
 struct GpuEngineState {
   Cclo<GpuMover> cclo;     // trivially-copyable; host fills, device runs
+  u32 no_acq;              // ACCL_NO_ACQ=1: skip the mover wake-batch system
+                           // acquire (measurement only — UNSOUND for peer-
+                           // written payload, see mover_main)
+  u32 _pad0;
   GpuMover mover;
   MoveDesc mq[MOVE_RING];
   MoveState mst[MOVE_RING];

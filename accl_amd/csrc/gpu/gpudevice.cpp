@@ -144,6 +144,7 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
   st->mover.head = &dstate->mq_head;
   st->mover.stop = &dstate->stop;
   st->mover.dbg = dstate->dbg;
+  st->no_acq = std::getenv("ACCL_NO_ACQ") ? 1u : 0u;
   if (const char* t = std::getenv("ACCL_TILE_KB")) {
     u64 kb = strtoull(t, nullptr, 10);
     u32 lg = 0;

@@ -252,7 +252,7 @@ inline ArenaLayout arena_layout(const ProtoConfig& c, u64 spare_bytes = 32u << 2
   L.barrier_off = off = align_up(off, 256);
   off += u64(c.nranks) * sizeof(u64);
   L.direct_off = off = align_up(off, 256);
-  off += u64(c.nranks) * sizeof(u64);
+  off += u64(c.nranks) * c.n_rndzv * sizeof(u64);
   L.devcall_off = off = align_up(off, 256);
   off += sizeof(DevCallRing) + u64(DEVCALL_RING) * sizeof(DevCallSlot) +
          u64(DEVCALL_RING) * sizeof(DevCallRet);
@@ -304,9 +304,14 @@ struct TransportView {
   ACCL_HD u64* barrier_word(u32 r, u32 peer) const {
     return (u64*)(arena[r] + hdr(r)->barrier_off) + peer;
   }
-  // cumulative direct-write progress: in r's arena, written by peer s
-  ACCL_HD volatile u64* direct_word(u32 r, u32 s) const {
-    return (volatile u64*)(arena[r] + hdr(r)->direct_off) + s;
+  // per-WINDOW direct-write progress words: in r's arena, lane [s] slot [i]
+  // (i = the addr-ring slot of the posted window). The sender of a window
+  // writes cumulative bytes WITHIN the window; the receiver zeroes the word
+  // when posting and polls it locally. Window-scoped (not pair-cumulative)
+  // so out-of-order matched rendezvous ops never corrupt each other.
+  ACCL_HD volatile u64* direct_word(u32 r, u32 s, u32 i) const {
+    return (volatile u64*)(arena[r] + hdr(r)->direct_off) +
+           u64(s) * hdr(r)->n_rndzv + i;
   }
   ACCL_HD char* heap_ptr(u32 r, u64 off) const { return arena[r] + off; }
 
@@ -357,8 +362,6 @@ struct PairSeq {
   u64 stream_tx[MAX_RANKS];    // stream segments sent to peer
   u64 stream_fed_rx[MAX_RANKS];// segments the ENGINE consumed as a stream-fed
                                // op source (one consumer per lane)
-  u64 direct_tx[MAX_RANKS];    // cumulative bytes direct-written to peer
-  u64 direct_rx[MAX_RANKS];    // cumulative bytes direct-received from peer
   u64 barrier_epoch[MAX_RANKS];// per-PAIR barrier epoch (must match both ends)
 };
 

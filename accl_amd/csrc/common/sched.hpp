@@ -66,7 +66,9 @@ struct Flow {
                           // but not yet submitted (0 = none)
   // direct mode:
   u64 peer_off;           // destination offset in peer arena (tx_direct)
-  u64 prog_base;          // cumulative-bytes baseline (both direct kinds)
+  u64 prog_addr;          // window progress word: tx_direct = in PEER's
+                          // arena (we store cumulative window bytes),
+                          // rx_direct = in OWN arena (we poll)
   // pending segment fifo:
   PendSeg pend[FLOW_INFLIGHT];
   u32 ph, pt;             // head/tail (pt-ph = in flight)
@@ -102,6 +104,15 @@ struct Cclo {
   Unexpected uq[MAX_RANKS][UQ_DEPTH];
   u32 uq_h[MAX_RANKS], uq_t[MAX_RANKS];
   u64 spill_busy;          // bitmap over spill slots (<= 64)
+
+  // ---- rendezvous pending sets (the RNDZV_PENDING spill queue analogue,
+  // reference ccl_offload_control.c:154-408): in-seq records whose tag does
+  // not match the current wait are parked here for later out-of-order match.
+  static constexpr u32 RNDZV_PEND = 4;
+  struct PendRndzv { u64 seq; u64 offset; u64 count; u32 tag; u32 arith;
+                     u32 valid; };
+  PendRndzv pa[MAX_RANKS][RNDZV_PEND];  // addr records
+  PendRndzv pd[MAX_RANKS][RNDZV_PEND];  // done records
 
   ACCL_HD u32 me() const { return cfg.rank; }
 
@@ -185,10 +196,11 @@ struct Cclo {
           break;
         }
         case FLOW_TX_DIRECT: {
-          // advance the cumulative progress word in the peer's arena
+          // advance the WINDOW's progress word in the peer's arena
+          // (cumulative bytes within this posted window)
           fence_release_sys();
-          st_sys(tv.direct_word(f.gpeer, me()),
-                 f.prog_base + (f.done + p.elems) * dtype_size(DataType(f.ddt)));
+          st_sys((volatile u64*)f.prog_addr,
+                 (f.done + p.elems) * dtype_size(DataType(f.ddt)));
           break;
         }
         default: break;
@@ -327,9 +339,9 @@ struct Cclo {
           break;
         }
         case FLOW_RX_DIRECT: {
-          // pure wait: done advances with the peer's progress word
-          u64 w = ld_sys(tv.direct_word(me(), f.gpeer));
-          u64 avail_elems = min64(f.count, (w - f.prog_base) / dtype_size(DataType(f.ddt)));
+          // pure wait: done advances with the window's progress word
+          u64 w = ld_sys((const volatile u64*)f.prog_addr);
+          u64 avail_elems = min64(f.count, w / dtype_size(DataType(f.ddt)));
           if (avail_elems > f.done) {
             fence_acquire_sys();
             f.done = avail_elems;
@@ -394,46 +406,94 @@ struct Cclo {
     f.ddt = u8(ddt); f.wdt = u8(wdt); f.bdt = u8(bdt);
     f.func = u8(func + 1); f.count = count; f.tag = tag; f.gate = gate;
   }
+  // slot = the addr-ring slot of the matched/posted window (determines the
+  // progress word both sides use)
   ACCL_HD void mk_tx_direct(u32 i, u32 gpeer, const char* src, DataType sdt,
-                            DataType ddt, u64 count, u64 peer_off,
+                            DataType ddt, u64 count, u64 peer_off, u32 slot,
                             const u64* gate = nullptr) {
     Flow& f = fl(i);
     f.kind = FLOW_TX_DIRECT; f.gpeer = gpeer; f.src = src;
     f.sdt = u8(sdt); f.ddt = u8(ddt); f.count = count;
     f.peer_off = peer_off; f.gate = gate;
-    f.prog_base = sq.direct_tx[gpeer];
-    sq.direct_tx[gpeer] += count * dtype_size(ddt);
+    f.prog_addr = (u64)tv.direct_word(gpeer, me(), slot);
   }
-  ACCL_HD void mk_rx_direct(u32 i, u32 gpeer, u64 count, DataType ddt) {
+  ACCL_HD void mk_rx_direct(u32 i, u32 gpeer, u64 count, DataType ddt,
+                            u32 slot) {
     Flow& f = fl(i);
     f.kind = FLOW_RX_DIRECT; f.gpeer = gpeer; f.ddt = u8(ddt); f.count = count;
-    f.prog_base = sq.direct_rx[gpeer];
-    sq.direct_rx[gpeer] += count * dtype_size(ddt);
+    f.prog_addr = (u64)tv.direct_word(me(), gpeer, slot);
   }
 
   // ---------------- rendezvous record rings ----------------
   // post {offset,count,tag} into PEER's addr ring (peer = the sender that
-  // will write to us).  reference: rendezvous_send_addr
+  // will write to us); zero the window's progress word first. Returns the
+  // ring slot. reference: rendezvous_send_addr
   // (ccl_offload_control.c:142-150).
-  ACCL_HD void post_addr(u32 gpeer, u64 offset, u64 count, u32 tag, u32 arith) {
+  ACCL_HD u32 post_addr(u32 gpeer, u64 offset, u64 count, u32 tag, u32 arith) {
     u64 seq = ++sq.rndzv_addr_tx[gpeer];
-    RndzvRec* r = tv.rndzv_addr(gpeer, me(), u32((seq - 1) % cfg.n_rndzv));
+    u32 slot = u32((seq - 1) % cfg.n_rndzv);
+    st_sys(tv.direct_word(me(), gpeer, slot), 0);  // reset window progress
+    RndzvRec* r = tv.rndzv_addr(gpeer, me(), slot);
     u64 val[8] = {seq, (u64(arith) << 32) | tag, offset, count, 0, 0, 0, 0};
     publish_rec((volatile u64*)r, val);
+    return slot;
   }
-  // wait for the next addr record from gpeer (the receiver), local poll.
-  // reference: rendezvous_get_addr (ccl_offload_control.c:154-212).
-  ACCL_HD bool wait_addr(u32 gpeer, RndzvRec& out) {
+  ACCL_HD static u32 rec_slot(const RndzvRec& rec, u32 n_rndzv) {
+    return u32((rec.seq - 1) % n_rndzv);
+  }
+
+  // Wait for an addr record from gpeer MATCHING want_tag; non-matching
+  // in-seq records spill to a pending set so overlapping rendezvous ops
+  // match out of order (reference: the RNDZV_PENDING spill queue,
+  // ccl_offload_control.c:154-212 + rxbuf_seek-style any-order matching).
+  ACCL_HD bool wait_addr(u32 gpeer, u32 want_tag, RndzvRec& out) {
+    u64 deadline = deadline_now();
+    for (;;) {
+      int best = -1;
+      for (u32 k = 0; k < RNDZV_PEND; ++k) {
+        PendRndzv& p = pa[gpeer][k];
+        if (!p.valid) continue;
+        if (want_tag != TAG_ANY && p.tag != want_tag && p.tag != TAG_ANY)
+          continue;
+        if (best < 0 || p.seq < pa[gpeer][best].seq) best = int(k);
+      }
+      if (best >= 0) {
+        PendRndzv& p = pa[gpeer][best];
+        out.seq = p.seq; out.tag = p.tag; out.arith = p.arith;
+        out.offset = p.offset; out.count = p.count;
+        p.valid = 0;
+        return true;
+      }
+      u64 seq = sq.rndzv_addr_rx[gpeer] + 1;
+      RndzvRec* r = tv.rndzv_addr(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
+      if (ld_sys(&r->seq) == seq) {
+        fence_acquire_sys();
+        sq.rndzv_addr_rx[gpeer] = seq;
+        if (want_tag == TAG_ANY || r->tag == want_tag || r->tag == TAG_ANY) {
+          out.seq = seq; out.tag = r->tag; out.arith = r->arith;
+          out.offset = r->offset; out.count = r->count;
+          return true;
+        }
+        u32 k = 0;
+        while (k < RNDZV_PEND && pa[gpeer][k].valid) ++k;
+        if (k >= RNDZV_PEND) { err |= E_RNDZV; return false; }
+        pa[gpeer][k] = PendRndzv{seq, r->offset, r->count, r->tag, r->arith, 1};
+        continue;
+      }
+      if (!wait_pred_tick(deadline)) return false;
+    }
+  }
+  // non-consuming probe: is a matching addr record available right now?
+  ACCL_HD bool addr_ready(u32 gpeer, u32 want_tag) {
+    for (u32 k = 0; k < RNDZV_PEND; ++k) {
+      PendRndzv& p = pa[gpeer][k];
+      if (p.valid && (want_tag == TAG_ANY || p.tag == want_tag ||
+                      p.tag == TAG_ANY))
+        return true;
+    }
     u64 seq = sq.rndzv_addr_rx[gpeer] + 1;
     RndzvRec* r = tv.rndzv_addr(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
-    u64 deadline = deadline_now();
-    while (ld_sys(&r->seq) != seq)
-      if (!wait_pred_tick(deadline)) return false;
-    fence_acquire_sys();
-    out.seq = seq; out.tag = r->tag; out.arith = r->arith;
-    out.offset = r->offset; out.count = r->count;
-    sq.rndzv_addr_rx[gpeer] = seq;
-    return true;
+    return ld_sys(&r->seq) == seq;
   }
   ACCL_HD void post_done(u32 gpeer, u32 tag) {
     u64 seq = ++sq.rndzv_done_tx[gpeer];
@@ -441,15 +501,33 @@ struct Cclo {
     u64 val[8] = {seq, tag, 0, 0, 0, 0, 0, 0};
     publish_rec((volatile u64*)r, val);
   }
-  ACCL_HD bool wait_done(u32 gpeer) {
-    u64 seq = sq.rndzv_done_rx[gpeer] + 1;
-    RndzvRec* r = tv.rndzv_done(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
+  ACCL_HD bool wait_done(u32 gpeer, u32 want_tag) {
     u64 deadline = deadline_now();
-    while (ld_sys(&r->seq) != seq)
+    for (;;) {
+      int best = -1;
+      for (u32 k = 0; k < RNDZV_PEND; ++k) {
+        PendRndzv& p = pd[gpeer][k];
+        if (!p.valid) continue;
+        if (want_tag != TAG_ANY && p.tag != want_tag && p.tag != TAG_ANY)
+          continue;
+        if (best < 0 || p.seq < pd[gpeer][best].seq) best = int(k);
+      }
+      if (best >= 0) { pd[gpeer][best].valid = 0; return true; }
+      u64 seq = sq.rndzv_done_rx[gpeer] + 1;
+      RndzvRec* r = tv.rndzv_done(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
+      if (ld_sys(&r->seq) == seq) {
+        fence_acquire_sys();
+        sq.rndzv_done_rx[gpeer] = seq;
+        if (want_tag == TAG_ANY || r->tag == want_tag || r->tag == TAG_ANY)
+          return true;
+        u32 k = 0;
+        while (k < RNDZV_PEND && pd[gpeer][k].valid) ++k;
+        if (k >= RNDZV_PEND) { err |= E_RNDZV; return false; }
+        pd[gpeer][k] = PendRndzv{seq, 0, 0, r->tag, 0, 1};
+        continue;
+      }
       if (!wait_pred_tick(deadline)) return false;
-    fence_acquire_sys();
-    sq.rndzv_done_rx[gpeer] = seq;
-    return true;
+    }
   }
 
   // ================= collectives =================
@@ -498,15 +576,16 @@ struct Cclo {
       return run_flows(1);
     }
     if (use_rndzv(n, dt, wdt)) {
-      // follow the receiver's posted windows
+      // follow the receiver's posted windows (tag-matched, out of order
+      // w.r.t. other rendezvous ops on this pair)
       u64 sent = 0;
       while (sent < n) {
         RndzvRec rec{};
-        if (!wait_addr(peer, rec)) return err;
-        if (rec.tag != d.tag && rec.tag != TAG_ANY) { err |= E_MATCH; return err; }
+        if (!wait_addr(peer, d.tag, rec)) return err;
         u64 w = min64(n - sent, rec.count);
         mk_tx_direct(0, peer, src + sent * dtype_size(dt), dt,
-                     DataType(rec.arith ? rec.arith : u32(dt)), w, rec.offset);
+                     DataType(rec.arith ? rec.arith : u32(dt)), w, rec.offset,
+                     rec_slot(rec, cfg.n_rndzv));
         u32 e = run_flows(1);
         if (e) return e;
         sent += w;
@@ -625,61 +704,58 @@ struct Cclo {
     DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
     char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
     if (use_rndzv(n, dt, wdt) && peer != me()) {
+      const u32 esz = dtype_size(dt);
       if (d.flags & F_DST_ARENA) {
         // window the posting by max_rndzv_bytes (reference:
         // set_max_rendezvous_size caps a single rendezvous transfer);
         // sender follows each posted window (op_send's wait_addr loop)
-        u64 wmax = max_rndzv_bytes ? max_rndzv_bytes / dtype_size(dt) : n;
+        u64 wmax = max_rndzv_bytes ? max_rndzv_bytes / esz : n;
         if (!wmax) wmax = 1;
         u64 posted = 0, got = 0;
+        u32 wslot[2]; u64 wcnt[2]; u32 wi = 0, nw = 0;
         while (got < n) {
-          while (posted < n && posted - got < 2 * wmax) {
+          while (posted < n && nw < 2) {
             u64 w = min64(n - posted, wmax);
-            post_addr(peer, d.addr2 + posted * dtype_size(dt), w, d.tag,
-                      u32(dt));
-            posted += w;
+            wslot[(wi + nw) % 2] =
+                post_addr(peer, d.addr2 + posted * esz, w, d.tag, u32(dt));
+            wcnt[(wi + nw) % 2] = w;
+            posted += w; nw++;
           }
-          u64 w = min64(n - got, wmax);
-          mk_rx_direct(0, peer, w, dt);
+          mk_rx_direct(0, peer, wcnt[wi], dt, wslot[wi]);
           u32 e = run_flows(1);
           if (e) return e;
-          got += w;
+          got += wcnt[wi];
+          wi ^= 1; nw--;
         }
-        return wait_done(peer) ? E_OK : err;
+        return wait_done(peer, d.tag) ? E_OK : err;
       }
       // stage through the spare region's LOWER half in windows (double-
       // buffered); the upper half is the unexpected-message spill pool
       ArenaHdr* h = tv.hdr(me());
-      u64 half = (h->spare_bytes / 4) / dtype_size(dt);
-      if (!half) { err |= E_INVALID_ARG; return err; }
-      u64 got = 0; int cur = 0; u64 pending = 0; u64 pend_off = 0;
-      u64 posted = 0;
-      // pipeline: post window k+1 while copying window k out
-      while (got < n || pending) {
-        if (posted < n && (posted - got) < 2 * half) {
-          u64 w = min64(n - posted, half);
-          post_addr(peer, h->spare_off + u64(cur) * half * dtype_size(dt), w,
-                    d.tag, u32(dt));
-          posted += w; cur ^= 1;
+      u64 bank = (h->spare_bytes / 4) / esz;  // elems per staging bank
+      if (max_rndzv_bytes) bank = min64(bank, max_rndzv_bytes / esz);
+      if (!bank) { err |= E_INVALID_ARG; return err; }
+      u64 posted = 0, got = 0;
+      u32 wslot[2]; u64 wcnt[2], woff[2]; u32 wi = 0, nw = 0; int cur = 0;
+      while (got < n) {
+        while (posted < n && nw < 2) {
+          u64 w = min64(n - posted, bank);
+          u64 off = h->spare_off + u64(cur) * (h->spare_bytes / 4);
+          wslot[(wi + nw) % 2] = post_addr(peer, off, w, d.tag, u32(dt));
+          wcnt[(wi + nw) % 2] = w;
+          woff[(wi + nw) % 2] = off;
+          posted += w; cur ^= 1; nw++;
         }
-        if (pending) {
-          mk_local(0, tv.arena[me()] + pend_off, dt,
-                   dst + got * dtype_size(dt), dt, pending);
-          u32 e = run_flows(1);
-          if (e) return e;
-          got += pending; pending = 0;
-        }
-        if (got < posted && !pending) {
-          // wait for the oldest posted window to land
-          u64 w = min64(min64(n, got + half) - got, half);
-          mk_rx_direct(0, peer, w, dt);
-          u32 e = run_flows(1);
-          if (e) return e;
-          pending = w;
-          pend_off = h->spare_off + u64((got / half) % 2) * half * dtype_size(dt);
-        }
+        mk_rx_direct(0, peer, wcnt[wi], dt, wslot[wi]);
+        u32 e = run_flows(1);
+        if (e) return e;
+        mk_local(0, tv.arena[me()] + woff[wi], dt, dst + got * esz, dt,
+                 wcnt[wi]);
+        if ((e = run_flows(1))) return e;
+        got += wcnt[wi];
+        wi ^= 1; nw--;
       }
-      return wait_done(peer) ? E_OK : err;
+      return wait_done(peer, d.tag) ? E_OK : err;
     }
     return op_recv_eager(peer, dst, dt, wdt, n, d.tag);
   }
@@ -702,8 +778,9 @@ struct Cclo {
         if (p == root) continue;
         if (rndzv) {
           RndzvRec rec{};
-          if (!wait_addr(c.global(p), rec)) return err;
-          mk_tx_direct(nf++, c.global(p), src, dt, dt, n, rec.offset);
+          if (!wait_addr(c.global(p), tag, rec)) return err;
+          mk_tx_direct(nf++, c.global(p), src, dt, dt, n, rec.offset,
+                       rec_slot(rec, cfg.n_rndzv));
         } else {
           mk_tx(nf++, c.global(p), src, dt, wdt, n, tag);
         }
@@ -712,52 +789,97 @@ struct Cclo {
     }
     char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
     if (rndzv) {
-      post_addr(c.global(root), d.addr2, n, tag, u32(dt));
-      mk_rx_direct(0, c.global(root), n, dt);
+      u32 s = post_addr(c.global(root), d.addr2, n, tag, u32(dt));
+      mk_rx_direct(0, c.global(root), n, dt, s);
       return run_flows(1);
     }
     mk_rx(0, c.global(root), dst, dt, wdt, n, tag);
     return run_flows(1);
   }
 
-  // reference: scatter (ccl_offload_control.c:992-1123)
+  // both src and dst arena-resident, large, same wire dtype -> the
+  // address-exchange direct path is legal on every rank (flags must agree
+  // across ranks, same contract as the existing bcast/allgather direct)
+  ACCL_HD bool coll_direct_ok(const CallDesc& d, u64 n, DataType dt,
+                              DataType wdt) const {
+    return dt == wdt && n * dtype_size(dt) > max_eager_bytes &&
+           (d.flags & F_SRC_ARENA) && (d.flags & F_DST_ARENA);
+  }
+
+  // reference: scatter (ccl_offload_control.c:992-1123 — the rendezvous
+  // variant there is root self-copy + out-of-order addr-driven writes,
+  // :1010-1070; ours: every leaf posts its dst window, root direct-writes
+  // all P-1 links concurrently)
   ACCL_HD u32 op_scatter(const CallDesc& d, const CommView& c) {
     u64 n = desc_count(d);  // per-rank count
     u32 root = d.root_src_dst;
     DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
     u32 tag = TAG_COLL | (u32(Op::scatter) << 16) | d.comm_id;
     char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    bool direct = coll_direct_ok(d, n, dt, wdt) && c.size <= MAX_FLOWS;
     if (c.rank == root) {
       const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
       u32 nf = 0;
+      mk_local(nf++, src + u64(root) * n * dtype_size(dt), dt, dst, dt, n);
       for (u32 p = 0; p < c.size; ++p) {
+        if (p == root) continue;
         const char* s = src + u64(p) * n * dtype_size(dt);
-        if (p == root) mk_local(nf++, s, dt, dst, dt, n);
-        else mk_tx(nf++, c.global(p), s, dt, wdt, n, tag);
+        if (direct) {
+          RndzvRec rec{};
+          if (!wait_addr(c.global(p), tag, rec)) return err;
+          mk_tx_direct(nf++, c.global(p), s, dt, dt, n, rec.offset,
+                       rec_slot(rec, cfg.n_rndzv));
+        } else {
+          mk_tx(nf++, c.global(p), s, dt, wdt, n, tag);
+        }
       }
       return run_flows(nf);
+    }
+    if (direct) {
+      u32 s = post_addr(c.global(root), d.addr2, n, tag, u32(dt));
+      mk_rx_direct(0, c.global(root), n, dt, s);
+      return run_flows(1);
     }
     mk_rx(0, c.global(root), dst, dt, wdt, n, tag);
     return run_flows(1);
   }
 
-  // reference: gather (ccl_offload_control.c:1128-1294; ring relay there —
-  // here: direct fan-in, every inbound link concurrently)
+  // reference: gather (ccl_offload_control.c:1128-1294; rendezvous flat
+  // tree :1146-1184 — leaves write into the root's posted dst slots; ours:
+  // root posts per-leaf windows, every inbound link lands concurrently)
   ACCL_HD u32 op_gather(const CallDesc& d, const CommView& c) {
     u64 n = desc_count(d);
     u32 root = d.root_src_dst;
     DataType dt = desc_dtype(d), wdt = desc_wire_dtype(d);
     u32 tag = TAG_COLL | (u32(Op::gather) << 16) | d.comm_id;
     const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    bool direct = coll_direct_ok(d, n, dt, wdt) && c.size <= MAX_FLOWS;
     if (c.rank == root) {
       char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
       u32 nf = 0;
+      mk_local(nf++, src, dt, dst + u64(root) * n * dtype_size(dt), dt, n);
+      if (direct) {
+        for (u32 p = 0; p < c.size; ++p) {
+          if (p == root) continue;
+          u32 s = post_addr(c.global(p), d.addr2 + u64(p) * n * dtype_size(dt),
+                            n, tag, u32(dt));
+          mk_rx_direct(nf++, c.global(p), n, dt, s);
+        }
+        return run_flows(nf);
+      }
       for (u32 p = 0; p < c.size; ++p) {
-        char* dp = dst + u64(p) * n * dtype_size(dt);
-        if (p == root) mk_local(nf++, src, dt, dp, dt, n);
-        else mk_rx(nf++, c.global(p), dp, dt, wdt, n, tag);
+        if (p == root) continue;
+        mk_rx(nf++, c.global(p), dst + u64(p) * n * dtype_size(dt), dt, wdt, n,
+              tag);
       }
       return run_flows(nf);
+    }
+    if (direct) {
+      RndzvRec rec{};
+      if (!wait_addr(c.global(root), tag, rec)) return err;
+      mk_tx_direct(0, c.global(root), src, dt, dt, n, rec.offset,
+                   rec_slot(rec, cfg.n_rndzv));
+      return run_flows(1);
     }
     mk_tx(0, c.global(root), src, dt, wdt, n, tag);
     return run_flows(1);
@@ -779,15 +901,19 @@ struct Cclo {
     bool direct = use_rndzv(n, dt, wdt) && (d.flags & F_DST_ARENA) &&
                   (d.flags & F_SRC_ARENA) && c.size <= 9;
     if (direct) {
+      u32 myslot[MAX_RANKS];
       for (u32 p = 0; p < c.size; ++p)
-        if (p != r) post_addr(c.global(p), d.addr2 + u64(r) * n * dtype_size(dt),
-                              n, tag, u32(dt));
+        if (p != r)
+          myslot[p] = post_addr(c.global(p),
+                                d.addr2 + u64(r) * n * dtype_size(dt), n, tag,
+                                u32(dt));
       for (u32 p = 0; p < c.size; ++p) {
         if (p == r) continue;
         RndzvRec rec{};
-        if (!wait_addr(c.global(p), rec)) return err;
-        mk_tx_direct(nf++, c.global(p), src, dt, dt, n, rec.offset);
-        mk_rx_direct(nf++, c.global(p), n, dt);
+        if (!wait_addr(c.global(p), tag, rec)) return err;
+        mk_tx_direct(nf++, c.global(p), src, dt, dt, n, rec.offset,
+                     rec_slot(rec, cfg.n_rndzv));
+        mk_rx_direct(nf++, c.global(p), n, dt, myslot[p]);
       }
       return run_flows(nf);
     }
@@ -825,6 +951,192 @@ struct Cclo {
     return E_OK;
   }
 
+  // ---- windowed n-ary direct fan-in (the reduce family's large-message
+  // path; reference: flat-tree reduce with spare-buffer scratchpads,
+  // ccl_offload_control.c:1531-1602, redesigned MI355X-first: P-1 peers
+  // direct-write window stages over their own xGMI links concurrently,
+  // then ONE n-ary mover pass folds stage slots + own chunk into dst —
+  // root HBM traffic is O(P*read + write) per window, depth O(1)) ----
+  struct FanGeom { u64 bank_off[2]; u64 slot_bytes; u64 W; };
+  ACCL_HD bool fan_geom(u32 P, u32 esz, FanGeom& g) {
+    const ArenaHdr* h = tv.hdr(me());
+    u64 bank_bytes = (h->spare_bytes / 4) & ~4095ull;
+    g.slot_bytes = (bank_bytes / (P - 1)) & ~255ull;
+    g.W = g.slot_bytes / esz;
+    if (max_rndzv_bytes) g.W = min64(g.W, max_rndzv_bytes / esz);
+    g.bank_off[0] = h->spare_off;
+    g.bank_off[1] = h->spare_off + bank_bytes;
+    return g.W != 0;
+  }
+
+  // dst[0..count) = f(first, extra[0], ..., extra[nx-1]), chained in groups
+  // of MOVE_MAX_SRC sources (one mover pass per group; later groups
+  // accumulate into dst)
+  ACCL_HD u32 nary_reduce(char* dst, const char* first, DataType dt, u64 count,
+                          const char* const* extra, u32 nx, int func) {
+    u32 k = 0;
+    const char* head = first;
+    for (;;) {
+      MoveDesc m{};
+      m.dst = (u64)dst;
+      m.dst_dt = u8(dt);
+      m.count = count;
+      m.func = u32(func);
+      m.src[0] = (u64)head;
+      m.src_dt[0] = u8(dt);
+      m.nsrc = 1;
+      while (m.nsrc < MOVE_MAX_SRC && k < nx) {
+        m.src[m.nsrc] = (u64)extra[k];
+        m.src_dt[m.nsrc] = u8(dt);
+        m.nsrc++;
+        k++;
+      }
+      u32 tok = mv->submit(m);
+      u64 deadline = deadline_now();
+      while (!mv->poll(tok))
+        if (!wait_pred_tick(deadline)) return err;
+      if (k >= nx) return E_OK;
+      head = dst;
+    }
+  }
+
+  // non-root side of the direct reduce family: follow the root's posted
+  // stage windows (tag-matched)
+  ACCL_HD u32 reduce_direct_leaf(const CommView& c, u32 root, const char* src,
+                                 DataType dt, u64 n, u32 tag) {
+    const u32 esz = dtype_size(dt);
+    u64 sent = 0;
+    while (sent < n) {
+      RndzvRec rec{};
+      if (!wait_addr(c.global(root), tag, rec)) return err;
+      u64 w = min64(n - sent, rec.count);
+      mk_tx_direct(0, c.global(root), src + sent * esz, dt, dt, w, rec.offset,
+                   rec_slot(rec, cfg.n_rndzv));
+      u32 e = run_flows(1);
+      if (e) return e;
+      sent += w;
+    }
+    return E_OK;
+  }
+
+  ACCL_HD u32 reduce_direct_root(const CallDesc& d, const CommView& c,
+                                 u32 tag) {
+    u64 n = desc_count(d);
+    DataType dt = desc_dtype(d);
+    const u32 esz = dtype_size(dt);
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    const u32 P = c.size, r = c.rank;
+    FanGeom g{};
+    if (!fan_geom(P, esz, g)) { err |= E_INVALID_ARG; return err; }
+    u64 nwin = (n + g.W - 1) / g.W;
+    u32 sp[2][MAX_RANKS];
+    u64 w0 = min64(n, g.W);
+    for (u32 p = 0; p < P; ++p) {
+      if (p == r) continue;
+      u32 idx = p < r ? p : p - 1;
+      sp[0][p] = post_addr(c.global(p), g.bank_off[0] + idx * g.slot_bytes,
+                           w0, tag, u32(dt));
+    }
+    for (u64 w = 0; w < nwin; ++w) {
+      int bank = int(w & 1);
+      u64 off = w * g.W;
+      u64 wc = min64(g.W, n - off);
+      // post window w+1 into the other bank FIRST: senders stream it over
+      // xGMI while we still collect/reduce window w
+      if (w + 1 < nwin) {
+        u64 nc = min64(g.W, n - (w + 1) * g.W);
+        for (u32 p = 0; p < P; ++p) {
+          if (p == r) continue;
+          u32 idx = p < r ? p : p - 1;
+          sp[bank ^ 1][p] =
+              post_addr(c.global(p), g.bank_off[bank ^ 1] + idx * g.slot_bytes,
+                        nc, tag, u32(dt));
+        }
+      }
+      u32 nf = 0;
+      for (u32 p = 0; p < P; ++p)
+        if (p != r) mk_rx_direct(nf++, c.global(p), wc, dt, sp[bank][p]);
+      u32 e = run_flows(nf);
+      if (e) return e;
+      const char* extra[MAX_RANKS];
+      u32 nx = 0;
+      for (u32 p = 0; p < P; ++p) {
+        if (p == r) continue;
+        u32 idx = p < r ? p : p - 1;
+        extra[nx++] = tv.arena[me()] + g.bank_off[bank] + idx * g.slot_bytes;
+      }
+      e = nary_reduce(dst + off * esz, src + off * esz, dt, wc, extra, nx,
+                      int(d.function));
+      if (e) return e;
+    }
+    return E_OK;
+  }
+
+  // every rank runs root-and-leaf at once: direct-write window stages for
+  // chunk p to rank p while collecting own chunk's stages from all peers
+  // (reference ring reduce_scatter ccl_offload_control.c:1748-1852; ours
+  // drives all xGMI links simultaneously)
+  ACCL_HD u32 reduce_scatter_direct(const CallDesc& d, const CommView& c,
+                                    u32 tag) {
+    u64 n = desc_count(d);  // per-rank chunk
+    DataType dt = desc_dtype(d);
+    const u32 esz = dtype_size(dt);
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    const u32 P = c.size, r = c.rank;
+    FanGeom g{};
+    if (!fan_geom(P, esz, g)) { err |= E_INVALID_ARG; return err; }
+    u64 nwin = (n + g.W - 1) / g.W;
+    u32 sp[2][MAX_RANKS];
+    u64 w0 = min64(n, g.W);
+    for (u32 p = 0; p < P; ++p) {
+      if (p == r) continue;
+      u32 idx = p < r ? p : p - 1;
+      sp[0][p] = post_addr(c.global(p), g.bank_off[0] + idx * g.slot_bytes,
+                           w0, tag, u32(dt));
+    }
+    for (u64 w = 0; w < nwin; ++w) {
+      int bank = int(w & 1);
+      u64 off = w * g.W;
+      u64 wc = min64(g.W, n - off);
+      if (w + 1 < nwin) {
+        u64 nc = min64(g.W, n - (w + 1) * g.W);
+        for (u32 p = 0; p < P; ++p) {
+          if (p == r) continue;
+          u32 idx = p < r ? p : p - 1;
+          sp[bank ^ 1][p] =
+              post_addr(c.global(p), g.bank_off[bank ^ 1] + idx * g.slot_bytes,
+                        nc, tag, u32(dt));
+        }
+      }
+      u32 nf = 0;
+      for (u32 p = 0; p < P; ++p) {
+        if (p == r) continue;
+        RndzvRec rec{};
+        if (!wait_addr(c.global(p), tag, rec)) return err;
+        mk_tx_direct(nf++, c.global(p), src + (u64(p) * n + off) * esz, dt,
+                     dt, min64(wc, rec.count), rec.offset,
+                     rec_slot(rec, cfg.n_rndzv));
+      }
+      for (u32 p = 0; p < P; ++p)
+        if (p != r) mk_rx_direct(nf++, c.global(p), wc, dt, sp[bank][p]);
+      u32 e = run_flows(nf);
+      if (e) return e;
+      const char* extra[MAX_RANKS];
+      u32 nx = 0;
+      for (u32 p = 0; p < P; ++p) {
+        if (p == r) continue;
+        u32 idx = p < r ? p : p - 1;
+        extra[nx++] = tv.arena[me()] + g.bank_off[bank] + idx * g.slot_bytes;
+      }
+      e = nary_reduce(dst + off * esz, src + (u64(r) * n + off) * esz, dt, wc,
+                      extra, nx, int(d.function));
+      if (e) return e;
+    }
+    return E_OK;
+  }
+
   // reduce at root: fan-in with a serialized reduce chain per segment
   // (reference: reduce, ccl_offload_control.c:1507-1744)
   ACCL_HD u32 op_reduce(const CallDesc& d, const CommView& c) {
@@ -836,6 +1148,10 @@ struct Cclo {
     if (c.size == 1) {
       mk_local(0, src, dt, local_ptr(d.addr2, d.flags & F_DST_ARENA), dt, n);
       return run_flows(1);
+    }
+    if (coll_direct_ok(d, n, dt, wdt)) {
+      if (c.rank != root) return reduce_direct_leaf(c, root, src, dt, n, tag);
+      return reduce_direct_root(d, c, tag);
     }
     if (c.rank != root) {
       mk_tx(0, c.global(root), src, dt, wdt, n, tag);
@@ -868,6 +1184,8 @@ struct Cclo {
       mk_local(0, src, dt, dst, dt, n);
       return run_flows(1);
     }
+    if (coll_direct_ok(d, n, dt, wdt) && 2 * u64(c.size - 1) <= MAX_FLOWS)
+      return reduce_scatter_direct(d, c, tag);
     if (2 * u64(c.size) - 1 > MAX_FLOWS) return reduce_scatter_batched(d, c);
     u32 nf = 0;
     // outbound: my chunk p -> rank p, all links at once
@@ -1019,6 +1337,28 @@ struct Cclo {
     const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
     char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
     u32 r = c.rank;
+    const u32 esz = dtype_size(dt);
+    if (coll_direct_ok(d, n, dt, wdt) && 2 * u64(c.size) - 1 <= MAX_FLOWS) {
+      // pairwise address exchange + direct peer writes, every link at once
+      // (reference rendezvous all_to_all: publish P-1 addrs, write-on-addr
+      // out of order, ccl_offload_control.c:2123-2218)
+      u32 sp[MAX_RANKS];
+      for (u32 p = 0; p < c.size; ++p)
+        if (p != r)
+          sp[p] = post_addr(c.global(p), d.addr2 + u64(p) * n * esz, n, tag,
+                            u32(dt));
+      u32 nf = 0;
+      mk_local(nf++, src + u64(r) * n * esz, dt, dst + u64(r) * n * esz, dt, n);
+      for (u32 p = 0; p < c.size; ++p) {
+        if (p == r) continue;
+        RndzvRec rec{};
+        if (!wait_addr(c.global(p), tag, rec)) return err;
+        mk_tx_direct(nf++, c.global(p), src + u64(p) * n * esz, dt, dt, n,
+                     rec.offset, rec_slot(rec, cfg.n_rndzv));
+        mk_rx_direct(nf++, c.global(p), n, dt, sp[p]);
+      }
+      return run_flows(nf);
+    }
     u32 nf = 0;
     mk_local(nf++, src + u64(r) * n * dtype_size(dt), dt,
              dst + u64(r) * n * dtype_size(dt), dt, n);
@@ -1155,8 +1495,8 @@ struct Cclo {
         fw[7] = sq.eager_tx[f.gpeer];
         fw[8] = tx_credit(f.gpeer);
       } else if (f.kind == FLOW_RX_DIRECT) {
-        fw[7] = f.prog_base;
-        fw[8] = ld_sys(tv.direct_word(me(), f.gpeer));
+        fw[7] = f.count * dtype_size(DataType(f.ddt));  // window bytes wanted
+        fw[8] = ld_sys((const volatile u64*)f.prog_addr);
       }
       if (f.ph != f.pt) {
         PendSeg& p = f.pend[f.ph % FLOW_INFLIGHT];

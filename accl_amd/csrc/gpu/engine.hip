@@ -38,10 +38,20 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
       __builtin_amdgcn_s_sleep(1);
     __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
   }
-  bool small = small_mb && move_bytes(m) <= SMALL_INLINE_MAX;
+  u64 mbytes = move_bytes(m);
+  bool small = small_mb && mbytes <= SMALL_INLINE_MAX;
   MoveDesc& d = ring[slot];
   MoveState& s = st[slot];
-  const_cast<MoveDesc&>(m).tile_log2 = u8(tile_log2);
+  // Tile-size policy (measured, profiles/r2: 256 MiB 856->1199 GB/s at
+  // 512 KiB tiles, 1 GiB 1486->2016 at 1 MiB): fewer, fatter tiles beat
+  // maximum fan-out for big moves — per-wave wake/fence/desc overhead
+  // amortizes over more streaming. ACCL_TILE_KB overrides.
+  u8 tl = u8(tile_log2);
+  if (!tl) {
+    if (mbytes >= (768ull << 20)) tl = 20;      // 1 MiB tiles
+    else if (mbytes >= (96ull << 20)) tl = 19;  // 512 KiB tiles
+  }
+  const_cast<MoveDesc&>(m).tile_log2 = tl;
   if (small) {
     // kick the sibling wave over LDS FIRST, overlap the ring bookkeeping
     // with its copy, then block until its system-release completes

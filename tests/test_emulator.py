@@ -1056,3 +1056,80 @@ def _back_to_back(a, rank, n):
 def test_back_to_back_no_barrier():
     run_ranks(_back_to_back, 2, opts=SMALL)
     run_ranks(_back_to_back, 3, opts=SMALL)
+
+
+def _direct_all_colls(a, rank, n):
+    """Every collective's address-exchange direct path (max_eager forced
+    tiny): scatter/gather/alltoall single peer-write, reduce/reduce_scatter
+    windowed n-ary fan-in through spare staging banks."""
+    cnt = 6000  # 24 KB > max_eager(4 KB) -> direct
+    s = _mk(a, cnt * n)
+    d = _mk(a, cnt)
+    s.write(np.concatenate([pattern(cnt, rank * 10 + j, seed=101)
+                            for j in range(n)]))
+    a.scatter(s, d, cnt, 1 % n)
+    exp = pattern(cnt, (1 % n) * 10 + rank, seed=101)
+    assert np.array_equal(rd(d, cnt), exp), "direct scatter"
+    g = _mk(a, cnt * n)
+    a.gather(d, g, cnt, 0)
+    if rank == 0:
+        exp = np.concatenate([pattern(cnt, (1 % n) * 10 + r, seed=101)
+                              for r in range(n)])
+        assert np.array_equal(rd(g, cnt * n), exp), "direct gather"
+    a.barrier()
+    a.alltoall(s, g, cnt)
+    exp = np.concatenate([pattern(cnt, r * 10 + rank, seed=101)
+                          for r in range(n)])
+    assert np.array_equal(rd(g, cnt * n), exp), "direct alltoall"
+    dr = _mk(a, cnt * n)
+    a.reduce(s, dr, cnt * n, 0, RF.SUM)
+    if rank == 0:
+        exp = np.stack([np.concatenate([pattern(cnt, r * 10 + j, seed=101)
+                                        for j in range(n)])
+                        for r in range(n)]).sum(0)
+        assert np.allclose(rd(dr, cnt * n), exp), "direct n-ary reduce"
+    a.barrier()
+    a.reduce_scatter(s, d, cnt, RF.SUM)
+    exp = np.stack([pattern(cnt, r * 10 + rank, seed=101)
+                    for r in range(n)]).sum(0)
+    assert np.allclose(rd(d, cnt), exp), "direct n-ary reduce_scatter"
+    # MAX through the n-ary path too
+    a.reduce(s, dr, cnt * n, 0, RF.MAX)
+    if rank == 0:
+        exp = np.stack([np.concatenate([pattern(cnt, r * 10 + j, seed=101)
+                                        for j in range(n)])
+                        for r in range(n)]).max(0)
+        assert np.array_equal(rd(dr, cnt * n), exp), "direct MAX reduce"
+    a.barrier()
+
+
+def test_direct_all_collectives():
+    run_ranks(_direct_all_colls, 2, opts=DIRECT)
+    run_ranks(_direct_all_colls, 3, opts=DIRECT)
+    run_ranks(_direct_all_colls, 4, opts=DIRECT)
+
+
+def _direct_windowed(a, rank, n):
+    """Multi-window n-ary fan-in: max_rendezvous_size forces many small
+    stage windows through the double-buffered spare banks."""
+    a.set_max_rendezvous_size(8192)  # 2K f32 elems per window
+    cnt = 10_000  # 5 windows per chunk
+    s, d = _mk(a, cnt * n), _mk(a, cnt)
+    s.write(np.concatenate([pattern(cnt, rank + 5 * j, seed=111)
+                            for j in range(n)]))
+    a.reduce_scatter(s, d, cnt, RF.SUM)
+    exp = np.stack([pattern(cnt, r + 5 * rank, seed=111)
+                    for r in range(n)]).sum(0)
+    assert np.allclose(rd(d, cnt), exp), "windowed RS"
+    dr = _mk(a, cnt * n)
+    a.reduce(s, dr, cnt * n, n - 1, RF.SUM)
+    if rank == n - 1:
+        exp = np.stack([np.concatenate([pattern(cnt, r + 5 * j, seed=111)
+                                        for j in range(n)])
+                        for r in range(n)]).sum(0)
+        assert np.allclose(rd(dr, cnt * n), exp), "windowed reduce"
+    a.barrier()
+
+
+def test_direct_windowed_fan_in():
+    run_ranks(_direct_windowed, 3, opts=DIRECT)

@@ -114,6 +114,25 @@ struct Cclo {
   PendRndzv pa[MAX_RANKS][RNDZV_PEND];  // addr records
   PendRndzv pd[MAX_RANKS][RNDZV_PEND];  // done records
 
+  // ---- multi-call interleaving (the CMD_CALL_RETRY requeue analogue,
+  // reference ccl_offload_control.c:2460-2478 + current_step resume
+  // :671-675): a send/recv that would block before making ANY progress is
+  // parked and the engine serves later descriptors; parked calls are
+  // re-probed each loop. ParkState carries the resume cursors (posted
+  // rendezvous windows) so re-entry never re-posts.
+  struct ParkState { u64 w[6]; u32 step; u32 _pad; };
+  struct ParkedCall {
+    CallDesc d; u64 ring_idx; u64 deadline; u64 t_start; ParkState ps;
+    u32 used; u32 _pad;
+  };
+  ParkedCall parked[MAX_INFLIGHT];
+  u32 nparked;
+  // retry_parked() completion report:
+  u64 done_ring_idx; u64 done_t0; u32 done_err;
+  // probe plumbing (set around run_call_inner)
+  u32 probe_;
+  ParkState* ps_;
+
   ACCL_HD u32 me() const { return cfg.rank; }
 
   // device-only micro-timeline (GPU: GpuMover::dbg; emulator mover: no-op)
@@ -578,6 +597,7 @@ struct Cclo {
     if (use_rndzv(n, dt, wdt)) {
       // follow the receiver's posted windows (tag-matched, out of order
       // w.r.t. other rendezvous ops on this pair)
+      if (probe_ && !addr_ready(peer, d.tag)) return E_NOT_READY;
       u64 sent = 0;
       while (sent < n) {
         RndzvRec rec{};
@@ -593,6 +613,10 @@ struct Cclo {
       post_done(peer, d.tag);
       return E_OK;
     }
+    // eager: park when the credit window is exhausted and nothing was sent
+    // (receiver absent or far behind)
+    if (probe_ && sq.eager_tx[peer] - tx_credit(peer) >= cfg.n_slots)
+      return E_NOT_READY;
     mk_tx(0, peer, src, dt, wdt, n, d.tag);
     return run_flows(1);
   }
@@ -693,6 +717,8 @@ struct Cclo {
         if (spill_head(peer, hd, sl, seq)) { deadline = deadline_now(); continue; }
         if (err) return err;
       }
+      // park instead of spinning while nothing has matched yet
+      if (probe_ && got == 0) return E_NOT_READY;
       if (!wait_pred_tick(deadline)) return err;
     }
     return E_OK;
@@ -708,11 +734,19 @@ struct Cclo {
       if (d.flags & F_DST_ARENA) {
         // window the posting by max_rndzv_bytes (reference:
         // set_max_rendezvous_size caps a single rendezvous transfer);
-        // sender follows each posted window (op_send's wait_addr loop)
+        // sender follows each posted window (op_send's wait_addr loop).
+        // Resumable: a parked re-entry restores the window cursors from
+        // ParkState instead of re-posting (the current_step analogue).
         u64 wmax = max_rndzv_bytes ? max_rndzv_bytes / esz : n;
         if (!wmax) wmax = 1;
         u64 posted = 0, got = 0;
-        u32 wslot[2]; u64 wcnt[2]; u32 wi = 0, nw = 0;
+        u32 wslot[2] = {0, 0}; u64 wcnt[2] = {0, 0}; u32 wi = 0, nw = 0;
+        if (ps_ && ps_->step == 1) {
+          posted = ps_->w[0]; got = ps_->w[1];
+          wcnt[0] = ps_->w[2]; wcnt[1] = ps_->w[3];
+          wslot[0] = u32(ps_->w[4]); wslot[1] = u32(ps_->w[4] >> 32);
+          wi = u32(ps_->w[5]); nw = u32(ps_->w[5] >> 32);
+        }
         while (got < n) {
           while (posted < n && nw < 2) {
             u64 w = min64(n - posted, wmax);
@@ -720,6 +754,17 @@ struct Cclo {
                 post_addr(peer, d.addr2 + posted * esz, w, d.tag, u32(dt));
             wcnt[(wi + nw) % 2] = w;
             posted += w; nw++;
+          }
+          if (probe_ && got == 0 &&
+              ld_sys(tv.direct_word(me(), peer, wslot[wi])) == 0) {
+            if (ps_) {
+              ps_->w[0] = posted; ps_->w[1] = got;
+              ps_->w[2] = wcnt[0]; ps_->w[3] = wcnt[1];
+              ps_->w[4] = u64(wslot[0]) | (u64(wslot[1]) << 32);
+              ps_->w[5] = u64(wi) | (u64(nw) << 32);
+              ps_->step = 1;
+            }
+            return E_NOT_READY;
           }
           mk_rx_direct(0, peer, wcnt[wi], dt, wslot[wi]);
           u32 e = run_flows(1);
@@ -1547,9 +1592,120 @@ struct Cclo {
   // ---------------- dispatch ----------------
   // reference: run() scenario switch (ccl_offload_control.c:2375-2459)
   ACCL_HD u32 run_call(const CallDesc& d) {
+    ParkState ps{};
+    return run_call_probe(d, false, ps);
+  }
+
+  ACCL_HD u32 run_call_probe(const CallDesc& d, bool probe, ParkState& ps) {
+    probe_ = probe ? 1 : 0;
+    ps_ = &ps;
     u32 e = run_call_inner(d);
+    probe_ = 0;
+    ps_ = nullptr;
+    if (e & E_NOT_READY) return e;
     if (e & E_TIMEOUT) dump_timeout(d.scenario);
     return e;
+  }
+
+  // would running `d` now violate per-(pair, tag) FIFO with a parked call?
+  ACCL_HD bool parked_key_match(const CallDesc& d) const {
+    if (!nparked) return false;
+    for (u32 i = 0; i < MAX_INFLIGHT; ++i) {
+      const ParkedCall& p = parked[i];
+      if (p.used && p.d.scenario == d.scenario &&
+          p.d.root_src_dst == d.root_src_dst && p.d.comm_id == d.comm_id &&
+          p.d.tag == d.tag)
+        return true;
+    }
+    return false;
+  }
+
+  ACCL_HD bool park(const CallDesc& d, u64 ring_idx, const ParkState& ps) {
+    for (u32 i = 0; i < MAX_INFLIGHT; ++i) {
+      if (parked[i].used) continue;
+      parked[i].d = d;
+      parked[i].ring_idx = ring_idx;
+      parked[i].deadline = wallclock() + timeout_ticks;
+      parked[i].t_start = wallclock();
+      parked[i].ps = ps;
+      parked[i].used = 1;
+      nparked++;
+      return true;
+    }
+    return false;
+  }
+
+  // One retry round over the parked set (FIFO per channel key). Returns a
+  // parked-entry index that COMPLETED (engine publishes its ret using
+  // done_ring_idx/done_err/done_t0), or -1 if none finished this round.
+  ACCL_HD int retry_parked() {
+    if (!nparked) return -1;
+    for (u32 i = 0; i < MAX_INFLIGHT; ++i) {
+      ParkedCall& p = parked[i];
+      if (!p.used) continue;
+      bool blocked = false;
+      for (u32 j = 0; j < MAX_INFLIGHT; ++j) {
+        const ParkedCall& q = parked[j];
+        if (!q.used || j == i) continue;
+        if (q.ring_idx < p.ring_idx && q.d.scenario == p.d.scenario &&
+            q.d.root_src_dst == p.d.root_src_dst &&
+            q.d.comm_id == p.d.comm_id && q.d.tag == p.d.tag) {
+          blocked = true;
+          break;
+        }
+      }
+      if (blocked) continue;
+      u32 e = run_call_probe(p.d, true, p.ps);
+      if (e & E_NOT_READY) {
+        if (wallclock() <= p.deadline) continue;
+        err = E_TIMEOUT;
+        dump_timeout(p.d.scenario);
+        e = E_TIMEOUT;
+      }
+      done_ring_idx = p.ring_idx;
+      done_err = e & ~E_NOT_READY;
+      done_t0 = p.t_start;
+      p.used = 0;
+      nparked--;
+      return int(i);
+    }
+    return -1;
+  }
+
+  // Serve one fresh descriptor with parking. Bit E_NOT_READY in the return
+  // means "parked — publish nothing"; otherwise the value is the errcode.
+  // Parkable ops ALWAYS probe: a not-ready send/recv parks immediately
+  // (even with an empty queue — later submissions must not block behind
+  // it), and the engine loop re-probes at its retry cadence.
+  ACCL_HD u32 serve_desc(const CallDesc& d, u64 ring_idx, bool) {
+    Op op = Op(d.scenario);
+    bool parkable = (op == Op::send || op == Op::recv) &&
+                    !(d.flags & F_SRC_STREAM);
+    if (parkable && parked_key_match(d)) {
+      if (park(d, ring_idx, ParkState{})) return E_NOT_READY;
+    }
+    ParkState ps{};
+    u32 e = run_call_probe(d, parkable, ps);
+    if (e & E_NOT_READY) {
+      if (park(d, ring_idx, ps)) return E_NOT_READY;
+      e = run_call_probe(d, false, ps);  // park table full: run blocking
+    }
+    return e & ~E_NOT_READY;
+  }
+
+  // engine shutdown with calls still parked: fail them (engine going away)
+  ACCL_HD int fail_parked() {
+    for (u32 i = 0; i < MAX_INFLIGHT; ++i) {
+      ParkedCall& p = parked[i];
+      if (!p.used) continue;
+      done_ring_idx = p.ring_idx;
+      done_err = E_ENGINE_DOWN;
+      done_t0 = p.t_start;
+      p.used = 0;
+      nparked--;
+      return int(i);
+    }
+    return -1;
   }
 
   ACCL_HD u32 run_call_inner(const CallDesc& d) {

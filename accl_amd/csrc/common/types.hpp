@@ -127,6 +127,10 @@ enum ErrorCode : u32 {
   E_INVALID_ARG = 1u << 10,       // bad count/root/addr
   E_INFLIGHT_OVERFLOW = 1u << 11, // too many outstanding ops
   E_COMM = 1u << 12,              // bad communicator id / membership
+  E_NOT_READY = 1u << 31,         // INTERNAL: op parked before any progress
+                                  // (the reference's NOT_READY_ERROR retry
+                                  // path, ccl_offload_control.c:2460-2478);
+                                  // never surfaces to the host
 };
 
 // --------------------------------------------------------------- descriptor

@@ -155,24 +155,47 @@ void EmuDevice::engine_main() {
       eng_->cached_comm_gen = gen;
     }
     u64 db = __atomic_load_n((u64*)&ctrl.doorbell, __ATOMIC_ACQUIRE);
+    auto publish = [&](u64 idx, u32 e, u64 t0) {
+      RetEntry& r = ring_->rets[idx % RING_CAP];
+      r.t_start = t0;
+      r.t_end = wallclock();
+      r.errcode = e;
+      __atomic_store_n(&r.seq, u32(idx + 1), __ATOMIC_RELEASE);
+    };
+    // retry parked calls first (CMD_CALL_RETRY analogue)
+    for (int pi; (pi = C.retry_parked()) >= 0;)
+      publish(C.done_ring_idx, C.done_err, C.done_t0);
     if (consumed == db) {
       if (C.poll_device_calls(dev_consumed)) continue;
-      if (__atomic_load_n((u64*)&ctrl.shutdown, __ATOMIC_RELAXED)) return;
-      __atomic_fetch_add((u64*)&ctrl.heartbeat, 1, __ATOMIC_RELAXED);
+      if (__atomic_load_n((u64*)&ctrl.shutdown, __ATOMIC_RELAXED)) break;
+      if (!C.nparked)
+        __atomic_fetch_add((u64*)&ctrl.heartbeat, 1, __ATOMIC_RELAXED);
       usleep(20);
       continue;
     }
     while (consumed < db) {
       const CallDesc d = ring_->descs[consumed % RING_CAP];
-      RetEntry& r = ring_->rets[consumed % RING_CAP];
-      r.t_start = wallclock();
-      u32 e = (Op(d.scenario) == Op::halt) ? E_OK : C.run_call(d);
-      r.t_end = wallclock();
-      r.errcode = e;
-      __atomic_store_n(&r.seq, u32(consumed + 1), __ATOMIC_RELEASE);
+      u64 t0 = wallclock();
+      if (Op(d.scenario) == Op::halt) {
+        publish(consumed, E_OK, t0);
+        consumed++;
+        goto out;
+      }
+      u32 e = C.serve_desc(d, consumed, consumed + 1 < db);
+      if (!(e & E_NOT_READY)) publish(consumed, e, t0);
       consumed++;
-      if (Op(d.scenario) == Op::halt) return;
+      if (C.nparked) break;  // interleave: give parked calls a retry round
     }
+  }
+out:
+  // engine exiting with calls parked: fail them so host waits return
+  for (int pi; (pi = eng_->cclo.fail_parked()) >= 0;) {
+    RetEntry& r = ring_->rets[eng_->cclo.done_ring_idx % RING_CAP];
+    r.t_start = eng_->cclo.done_t0;
+    r.t_end = wallclock();
+    r.errcode = eng_->cclo.done_err;
+    __atomic_store_n(&r.seq, u32(eng_->cclo.done_ring_idx + 1),
+                     __ATOMIC_RELEASE);
   }
 }
 

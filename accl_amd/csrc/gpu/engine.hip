@@ -585,10 +585,19 @@ __device__ void scheduler_main(GpuEngineState* S, Cclo<GpuMover>& C) {
   CtrlPage* ctrl = S->ctrl;
   st_sys(&ctrl->engine_up, 1);
   u64 consumed = 0, dev_consumed = 0, cached_gen = 0, beat = 0;
+  auto publish = [&](u64 idx, u32 e, u64 t0) {
+    RetEntry& r = S->rets[idx % RING_CAP];
+    r.errcode = e;
+    r.t_start = t0;
+    r.t_end = wallclock();
+    fence_release_sys();
+    st_sys32(&r.seq, u32(idx + 1));
+  };
+  bool halted = false;
   for (;;) {
     u64 db = ld_sys(&ctrl->doorbell);
     bool dev_pending = C.device_call_pending(dev_consumed);
-    if (consumed == db && !dev_pending) {
+    if (consumed == db && !dev_pending && !C.nparked) {
       if (ld_sys(&ctrl->shutdown)) break;
       if ((++beat & 0x3FF) == 0) st_sys(&ctrl->heartbeat, beat);
       __builtin_amdgcn_s_sleep(16);
@@ -603,28 +612,33 @@ __device__ void scheduler_main(GpuEngineState* S, Cclo<GpuMover>& C) {
       for (u32 i = 0; i < C.ncomms; ++i) C.comms[i] = S->comm_mirror[i];
       cached_gen = gen;
     }
+    // retry parked calls (the CMD_CALL_RETRY queue analogue)
+    for (int pi; (pi = C.retry_parked()) >= 0;)
+      publish(C.done_ring_idx, C.done_err, C.done_t0);
     if (dev_pending) {
       C.poll_device_calls(dev_consumed);
       continue;
     }
-    bool halted = false;
-    while (consumed < db) {
+    if (consumed < db) {
       CallDesc d = S->descs[consumed % RING_CAP];
-      RetEntry& r = S->rets[consumed % RING_CAP];
       u64 t0 = wallclock();
-      u32 e;
-      if (Op(d.scenario) == Op::halt) { e = E_OK; halted = true; }
-      else e = C.run_call(d);
-      r.errcode = e;
-      r.t_start = t0;
-      r.t_end = wallclock();
-      fence_release_sys();
-      st_sys32(&r.seq, u32(consumed + 1));
+      if (Op(d.scenario) == Op::halt) {
+        publish(consumed, E_OK, t0);
+        consumed++;
+        halted = true;
+        break;
+      }
+      u32 e = C.serve_desc(d, consumed, consumed + 1 < db);
+      if (!(e & E_NOT_READY)) publish(consumed, e, t0);
       consumed++;
-      if (halted) break;
+    } else if (C.nparked) {
+      __builtin_amdgcn_s_sleep(8);  // only parked work: light retry cadence
     }
-    if (halted) break;
   }
+  (void)halted;
+  // engine exiting: fail anything still parked so host waits return
+  for (int pi; (pi = C.fail_parked()) >= 0;)
+    publish(C.done_ring_idx, C.done_err, C.done_t0);
   // tell movers to exit, then leave
   __hip_atomic_store(S->mover.stop, 1u, __ATOMIC_RELEASE, AGENT);
   for (u32 i = 0; i < DOORBELL_REPS; ++i)

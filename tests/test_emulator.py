@@ -1133,3 +1133,96 @@ def _direct_windowed(a, rank, n):
 
 def test_direct_windowed_fan_in():
     run_ranks(_direct_windowed, 3, opts=DIRECT)
+
+
+# ------------------------------------- multi-call interleaving (parking)
+def _interleave_recv(a, rank, n):
+    """A recv posted long before the matching send must not block the
+    engine: other calls keep completing while it is parked (reference:
+    CMD_CALL_RETRY requeue, ccl_offload_control.c:2460-2478)."""
+    import time
+    cnt = 5000
+    if rank == 0:
+        d = _mk(a, cnt)
+        req = a.recv(d, cnt, src=1, tag=5, run_async=True)
+        # engine must serve these while the recv is parked
+        t0 = time.monotonic()
+        for i in range(10):
+            s1, d1 = _mk(a, 2000), _mk(a, 2000)
+            x = pattern(2000, i)
+            s1.write(x)
+            a.copy(s1, d1, 2000)
+            assert np.array_equal(rd(d1, 2000), x)
+        served = time.monotonic() - t0
+        assert served < 2.0, f"copies blocked behind parked recv: {served}s"
+        assert req.wait() == 0
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 9, seed=3))
+    else:
+        time.sleep(2.5)
+        s = _mk(a, cnt)
+        s.write(pattern(cnt, 9, seed=3))
+        a.send(s, cnt, dst=0, tag=5)
+    a.barrier()
+
+
+def test_interleave_parked_recv():
+    run_ranks(_interleave_recv, 2)
+
+
+def _interleave_tags(a, rank, n, big):
+    """Reversed-order tagged pairs: recv(tag2) first, sends arrive tag1
+    then tag2 — OOO matching + parking resolve it without timeouts."""
+    cnt = big and 300_000 or 3000  # rendezvous vs eager
+    if rank == 0:
+        d2, d1 = _mk(a, cnt), _mk(a, cnt)
+        r2 = a.recv(d2, cnt, src=1, tag=2, run_async=True)
+        r1 = a.recv(d1, cnt, src=1, tag=1, run_async=True)
+        assert r1.wait() == 0
+        assert r2.wait() == 0
+        assert np.array_equal(rd(d1, cnt), pattern(cnt, 1, seed=21))
+        assert np.array_equal(rd(d2, cnt), pattern(cnt, 2, seed=21))
+    else:
+        s1, s2 = _mk(a, cnt), _mk(a, cnt)
+        s1.write(pattern(cnt, 1, seed=21))
+        s2.write(pattern(cnt, 2, seed=21))
+        a.send(s1, cnt, dst=0, tag=1)
+        a.send(s2, cnt, dst=0, tag=2)
+    a.barrier()
+
+
+def test_interleave_reversed_tags_eager():
+    run_ranks(lambda a, r, n: _interleave_tags(a, r, n, False), 2)
+
+
+def test_interleave_reversed_tags_rndzv():
+    run_ranks(lambda a, r, n: _interleave_tags(a, r, n, True), 2,
+              opts=DIRECT)
+
+
+def _fifo_same_tag(a, rank, n):
+    """Two same-tag sends where the first parks on credit exhaustion: the
+    second must stay behind it (per-(pair,tag) FIFO preserved)."""
+    import time
+    cnt = 3000  # 12 KB > 2x4 KB slots -> first send exhausts credit
+    if rank == 0:
+        s1, s2 = _mk(a, cnt), _mk(a, cnt)
+        s1.write(pattern(cnt, 100))
+        s2.write(pattern(cnt, 200))
+        q1 = a.send(s1, cnt, dst=1, tag=7, run_async=True)
+        q2 = a.send(s2, cnt, dst=1, tag=7, run_async=True)
+        assert q1.wait() == 0
+        assert q2.wait() == 0
+    else:
+        time.sleep(1.0)
+        d1, d2 = _mk(a, cnt), _mk(a, cnt)
+        a.recv(d1, cnt, src=0, tag=7)
+        a.recv(d2, cnt, src=0, tag=7)
+        assert np.array_equal(rd(d1, cnt), pattern(cnt, 100)), "FIFO violated"
+        assert np.array_equal(rd(d2, cnt), pattern(cnt, 200))
+    a.barrier()
+
+
+def test_fifo_same_tag_parked_sends():
+    run_ranks(_fifo_same_tag, 2,
+              opts={"n_slots": 2, "slot_bytes": 4096,
+                    "timeout_us": 20_000_000})

@@ -348,3 +348,88 @@ def test_inline_burst_then_fleet(acc1):
     a.copy(s, d, cnt)
     assert np.array_equal(rd(d, cnt), y)
     assert np.array_equal(rd(sd, 256), x)
+
+
+DIRECT_GPU = {"max_eager": 4096, "n_slots": 4, "slot_bytes": 4096,
+              "timeout_us": 20_000_000}
+
+
+def _direct2(a, rank, n):
+    """All collectives through the address-exchange direct path on the GPU
+    engine (max_eager forced tiny; windowed n-ary fan-in for reduce/RS)."""
+    cnt = 6000
+    s = a.create_buffer(cnt * n, DT.float32)
+    d = a.create_buffer(cnt, DT.float32)
+    s.write(np.concatenate([pattern(cnt, rank * 10 + j, seed=131)
+                            for j in range(n)]))
+    a.scatter(s, d, cnt, 0)
+    assert np.array_equal(rd(d, cnt), pattern(cnt, rank, seed=131))
+    g = a.create_buffer(cnt * n, DT.float32)
+    a.gather(d, g, cnt, 0)
+    if rank == 0:
+        exp = np.concatenate([pattern(cnt, r, seed=131) for r in range(n)])
+        assert np.array_equal(rd(g, cnt * n), exp)
+    a.barrier()
+    a.alltoall(s, g, cnt)
+    exp = np.concatenate([pattern(cnt, r * 10 + rank, seed=131)
+                          for r in range(n)])
+    assert np.array_equal(rd(g, cnt * n), exp)
+    a.reduce_scatter(s, d, cnt, RF.SUM)
+    exp = np.stack([pattern(cnt, r * 10 + rank, seed=131)
+                    for r in range(n)]).sum(0)
+    assert np.allclose(rd(d, cnt), exp)
+    dr = a.create_buffer(cnt * n, DT.float32)
+    a.reduce(s, dr, cnt * n, 0, RF.SUM)
+    if rank == 0:
+        exp = np.stack([np.concatenate([pattern(cnt, r * 10 + j, seed=131)
+                                        for j in range(n)])
+                        for r in range(n)]).sum(0)
+        assert np.allclose(rd(dr, cnt * n), exp)
+    a.barrier()
+
+
+def test_two_ranks_direct_paths():
+    run_ranks(_direct2, 2, backend="gpu", opts=DIRECT_GPU, timeout=240)
+
+
+def _parked2(a, rank, n):
+    """Parked recv + interleaved traffic + reversed-tag matching on the GPU
+    engine (multi-call interleaving / CMD_CALL_RETRY analogue)."""
+    import time
+    cnt = 5000
+    if rank == 0:
+        d = a.create_buffer(cnt, DT.float32)
+        req = a.recv(d, cnt, src=1, tag=5, run_async=True)
+        t0 = time.monotonic()
+        for i in range(6):
+            s1 = a.create_buffer(2000, DT.float32)
+            d1 = a.create_buffer(2000, DT.float32)
+            x = pattern(2000, i)
+            s1.write(x)
+            a.copy(s1, d1, 2000)
+            assert np.array_equal(rd(d1, 2000), x)
+        assert time.monotonic() - t0 < 2.0, "copies blocked behind parked recv"
+        assert req.wait() == 0
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 9, seed=3))
+        # reversed tags
+        d2, d1b = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt, DT.float32)
+        r2 = a.recv(d2, cnt, src=1, tag=12, run_async=True)
+        r1 = a.recv(d1b, cnt, src=1, tag=11, run_async=True)
+        assert r1.wait() == 0 and r2.wait() == 0
+        assert np.array_equal(rd(d1b, cnt), pattern(cnt, 11, seed=4))
+        assert np.array_equal(rd(d2, cnt), pattern(cnt, 12, seed=4))
+    else:
+        time.sleep(2.2)
+        s = a.create_buffer(cnt, DT.float32)
+        s.write(pattern(cnt, 9, seed=3))
+        a.send(s, cnt, dst=0, tag=5)
+        s1, s2 = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt, DT.float32)
+        s1.write(pattern(cnt, 11, seed=4))
+        s2.write(pattern(cnt, 12, seed=4))
+        a.send(s1, cnt, dst=0, tag=11)
+        a.send(s2, cnt, dst=0, tag=12)
+    a.barrier()
+
+
+def test_two_ranks_parked_interleave():
+    run_ranks(_parked2, 2, backend="gpu", timeout=240)

@@ -157,7 +157,9 @@ Request* ACCL::finish(CallDesc d, bool run_async, BaseBuffer* sync_out,
     if (e) {
       debug_log(std::string(op_name(d.scenario)) + " seq=" +
                 std::to_string(seq) + " FAILED: " + error_to_string(e));
-      throw accl_error("accl op failed: " + error_to_string(e), e);
+      std::string msg = "accl op failed: " + error_to_string(e);
+      if (e & E_TIMEOUT) msg += "\n" + be_->timeout_dump_str();
+      throw accl_error(msg, e);
     }
     if (sync_out && sync_out->host_ptr())
       be_->read_arena(sync_out->arena_offset(), sync_out->host_ptr(),

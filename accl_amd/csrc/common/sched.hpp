@@ -155,6 +155,15 @@ struct Cclo {
   ACCL_HD static u64 min64(u64 a, u64 b) { return a < b ? a : b; }
 
   ACCL_HD bool wait_pred_tick(u64& deadline) {
+#if defined(__HIP_DEVICE_COMPILE__)
+    // Drop stale L1/XCD-L2 lines before the next re-read. Peer-written
+    // control words (another PROCESS over IPC, another GPU over xGMI) do
+    // NOT invalidate this XCD's L2 copies on coarse-grained memory, and
+    // sc0/sc1 polls are L2-served — a line cached by our own earlier polls
+    // would read stale forever (round-2 fresh-box wedge: peer's published
+    // slot header invisible for the whole 10 s deadline).
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+#endif
     cpu_pause();
     if (wallclock() > deadline) { err |= E_TIMEOUT; return false; }
     return true;

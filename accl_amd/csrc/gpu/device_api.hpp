@@ -69,9 +69,13 @@ __device__ inline u64 stream_push(const StreamChan& c, const void* data,
     seq = __hip_atomic_fetch_add(&ctl->tx_ctr, 1ull, __ATOMIC_RELAXED,
                                  ACCL_DEV_SYS) + 1;
     // credit gate: slot free once consumer advanced past seq - n_stream
+    // (in-loop acquire: the consumer may be another process — see
+    // stream_pop's staleness note)
     while (__hip_atomic_load(&ctl->credit, __ATOMIC_RELAXED, ACCL_DEV_SYS) +
-               c.n_stream < seq)
+               c.n_stream < seq) {
       __builtin_amdgcn_s_sleep(8);
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+    }
   }
   seq = u64(__shfl(int(seq & 0xFFFFFFFF), 0, 64)) |
         (u64(u32(__shfl(int(seq >> 32), 0, 64))) << 32);
@@ -135,9 +139,13 @@ __device__ inline u32 stream_pop(const StreamRx& r, u64 seq, void* dst,
   SlotHdr* h = (SlotHdr*)(r.my_arena + r.hdr_off_mine) + slot;
   // every lane spins convergently under the exec mask (no workgroup barrier:
   // this function is wave-collective and may be called from ONE wave of a
-  // multi-wave workgroup — an s_barrier here would hang the sibling waves)
-  while (__hip_atomic_load(&h->seq, __ATOMIC_RELAXED, ACCL_DEV_SYS) != seq)
+  // multi-wave workgroup — an s_barrier here would hang the sibling waves).
+  // The in-loop acquire drops stale L2 lines: the producer may be another
+  // PROCESS (IPC) whose stores do not invalidate this XCD's L2 copies.
+  while (__hip_atomic_load(&h->seq, __ATOMIC_RELAXED, ACCL_DEV_SYS) != seq) {
     __builtin_amdgcn_s_sleep(8);
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+  }
   __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
   u32 bytes = h->bytes;
   if (tag && lane == 0) *tag = h->tag;

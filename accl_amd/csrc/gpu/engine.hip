@@ -632,7 +632,10 @@ __device__ void scheduler_main(GpuEngineState* S, Cclo<GpuMover>& C) {
       if (!(e & E_NOT_READY)) publish(consumed, e, t0);
       consumed++;
     } else if (C.nparked) {
-      __builtin_amdgcn_s_sleep(8);  // only parked work: light retry cadence
+      // only parked work: light retry cadence; the acquire drops stale L2
+      // lines so the next probe round re-reads peer-written records fresh
+      fence_acquire_sys();
+      __builtin_amdgcn_s_sleep(8);
     }
   }
   (void)halted;

@@ -404,6 +404,182 @@ __device__ void tile_reduce_h16(const MoveDesc& m, u64 lo, u64 hi, int lane) {
 #undef ACCL_H8
 }
 
+// ---- vectorized compression lanes (reference: hp_compression 2:1 width
+// converter, kernels/plugins/hp_compression/hp_compression.cpp:72-144) ----
+__device__ __forceinline__ u16 bf16_rne(float v) {
+  u32 x = __float_as_uint(v);
+  if ((x & 0x7F800000u) == 0x7F800000u && (x & 0x7FFFFFu))
+    return u16((x >> 16) | 0x40);
+  x += 0x7FFFu + ((x >> 16) & 1);
+  return u16(x >> 16);
+}
+template <bool BF16>
+__device__ __forceinline__ void unpack8(U4 v, F4& a, F4& b) {
+  if (BF16) {
+    a = (F4){__uint_as_float(v.x << 16), __uint_as_float(v.x & 0xFFFF0000u),
+             __uint_as_float(v.y << 16), __uint_as_float(v.y & 0xFFFF0000u)};
+    b = (F4){__uint_as_float(v.z << 16), __uint_as_float(v.z & 0xFFFF0000u),
+             __uint_as_float(v.w << 16), __uint_as_float(v.w & 0xFFFF0000u)};
+  } else {
+    u32 wx = v.x, wy = v.y, wz = v.z, ww = v.w;
+    float2 f0 = __half22float2(*(__half2*)&wx);
+    float2 f1 = __half22float2(*(__half2*)&wy);
+    float2 f2 = __half22float2(*(__half2*)&wz);
+    float2 f3 = __half22float2(*(__half2*)&ww);
+    a = (F4){f0.x, f0.y, f1.x, f1.y};
+    b = (F4){f2.x, f2.y, f3.x, f3.y};
+  }
+}
+template <bool BF16>
+__device__ __forceinline__ U4 pack8(F4 a, F4 b) {
+  U4 o;
+  if (BF16) {
+    o.x = u32(bf16_rne(a.x)) | (u32(bf16_rne(a.y)) << 16);
+    o.y = u32(bf16_rne(a.z)) | (u32(bf16_rne(a.w)) << 16);
+    o.z = u32(bf16_rne(b.x)) | (u32(bf16_rne(b.y)) << 16);
+    o.w = u32(bf16_rne(b.z)) | (u32(bf16_rne(b.w)) << 16);
+  } else {
+    __half2 h0 = __floats2half2_rn(a.x, a.y);
+    __half2 h1 = __floats2half2_rn(a.z, a.w);
+    __half2 h2 = __floats2half2_rn(b.x, b.y);
+    __half2 h3 = __floats2half2_rn(b.z, b.w);
+    o.x = *(u32*)&h0; o.y = *(u32*)&h1; o.z = *(u32*)&h2; o.w = *(u32*)&h3;
+  }
+  return o;
+}
+
+// pure width conversion: UP = h16 -> f32 (decompress), else f32 -> h16.
+// 2x unrolled: 2-4 independent nt loads in flight per lane.
+template <bool UP, bool BF16>
+__device__ void tile_cast(const MoveDesc& m, u64 lo, u64 hi, int lane) {
+  u64 n8 = (hi - lo) / 8;
+  if (UP) {
+    GAS const U4* s = (GAS const U4*)((const u16*)m.src[0] + lo);
+    GAS F4* d = (GAS F4*)((float*)m.dst + lo);
+    u64 i = lane;
+    for (; i + 64 < n8; i += 2 * 64) {
+      U4 v0 = __builtin_nontemporal_load(&s[i]);
+      U4 v1 = __builtin_nontemporal_load(&s[i + 64]);
+      F4 a0, b0, a1, b1;
+      unpack8<BF16>(v0, a0, b0);
+      unpack8<BF16>(v1, a1, b1);
+      __builtin_nontemporal_store(a0, &d[2 * i]);
+      __builtin_nontemporal_store(b0, &d[2 * i + 1]);
+      __builtin_nontemporal_store(a1, &d[2 * (i + 64)]);
+      __builtin_nontemporal_store(b1, &d[2 * (i + 64) + 1]);
+    }
+    if (i < n8) {
+      U4 v = __builtin_nontemporal_load(&s[i]);
+      F4 a, b;
+      unpack8<BF16>(v, a, b);
+      __builtin_nontemporal_store(a, &d[2 * i]);
+      __builtin_nontemporal_store(b, &d[2 * i + 1]);
+    }
+  } else {
+    GAS const F4* s = (GAS const F4*)((const float*)m.src[0] + lo);
+    GAS U4* d = (GAS U4*)((u16*)m.dst + lo);
+    u64 i = lane;
+    for (; i + 64 < n8; i += 2 * 64) {
+      F4 a0 = __builtin_nontemporal_load(&s[2 * i]);
+      F4 b0 = __builtin_nontemporal_load(&s[2 * i + 1]);
+      F4 a1 = __builtin_nontemporal_load(&s[2 * (i + 64)]);
+      F4 b1 = __builtin_nontemporal_load(&s[2 * (i + 64) + 1]);
+      __builtin_nontemporal_store(pack8<BF16>(a0, b0), &d[i]);
+      __builtin_nontemporal_store(pack8<BF16>(a1, b1), &d[i + 64]);
+    }
+    if (i < n8) {
+      F4 a = __builtin_nontemporal_load(&s[2 * i]);
+      F4 b = __builtin_nontemporal_load(&s[2 * i + 1]);
+      __builtin_nontemporal_store(pack8<BF16>(a, b), &d[i]);
+    }
+  }
+}
+
+// fused decompress + reduce (rx lane of a compressed collective:
+// dst_f32 = op(cvt(wire_h16), dst_f32)); src0 = wire, src1 = f32 operand
+template <bool MAX_, bool BF16>
+__device__ void tile_cast_reduce(const MoveDesc& m, u64 lo, u64 hi, int lane) {
+  u64 n8 = (hi - lo) / 8;
+  GAS const U4* s0 = (GAS const U4*)((const u16*)m.src[0] + lo);
+  GAS const F4* s1 = (GAS const F4*)((const float*)m.src[1] + lo);
+  GAS F4* d = (GAS F4*)((float*)m.dst + lo);
+  for (u64 i = lane; i < n8; i += 64) {
+    U4 w = __builtin_nontemporal_load(&s0[i]);
+    F4 x0 = __builtin_nontemporal_load(&s1[2 * i]);
+    F4 x1 = __builtin_nontemporal_load(&s1[2 * i + 1]);
+    F4 a, b;
+    unpack8<BF16>(w, a, b);
+    F4 r0, r1;
+    if (MAX_) {
+      r0 = (F4){a.x > x0.x ? a.x : x0.x, a.y > x0.y ? a.y : x0.y,
+                a.z > x0.z ? a.z : x0.z, a.w > x0.w ? a.w : x0.w};
+      r1 = (F4){b.x > x1.x ? b.x : x1.x, b.y > x1.y ? b.y : x1.y,
+                b.z > x1.z ? b.z : x1.z, b.w > x1.w ? b.w : x1.w};
+    } else {
+      r0 = a + x0;
+      r1 = b + x1;
+    }
+    __builtin_nontemporal_store(r0, &d[2 * i]);
+    __builtin_nontemporal_store(r1, &d[2 * i + 1]);
+  }
+}
+
+// n-ary f32 reduce (3..8 sources — the direct fan-in hot path): one nt
+// load per source per 16B vector, fold, one nt store. 8 concurrent source
+// streams give the MLP; dst may alias src[0].
+template <template <class> class OP>
+__device__ void tile_reduce_f32_n(const MoveDesc& m, u64 lo, u64 hi,
+                                  int lane) {
+  u64 n4 = (hi - lo) / 4;
+  GAS F4* o = (GAS F4*)((float*)m.dst + lo);
+  GAS const F4* s[MOVE_MAX_SRC];
+  for (u32 k = 0; k < m.nsrc; ++k)
+    s[k] = (GAS const F4*)((const float*)m.src[k] + lo);
+  const u32 ns = m.nsrc;
+  for (u64 i = lane; i < n4; i += 64) {
+    F4 v[MOVE_MAX_SRC];
+#pragma unroll
+    for (u32 k = 0; k < MOVE_MAX_SRC; ++k)
+      if (k < ns) v[k] = __builtin_nontemporal_load(&s[k][i]);
+    F4 acc = v[0];
+#pragma unroll
+    for (u32 k = 1; k < MOVE_MAX_SRC; ++k)
+      if (k < ns)
+        acc = (F4){OP<float>::apply(acc.x, v[k].x),
+                   OP<float>::apply(acc.y, v[k].y),
+                   OP<float>::apply(acc.z, v[k].z),
+                   OP<float>::apply(acc.w, v[k].w)};
+    __builtin_nontemporal_store(acc, &o[i]);
+  }
+}
+
+// n-ary packed 16-bit reduce (bf16/f16 config-4 fan-in)
+template <bool MAX_, bool BF16>
+__device__ void tile_reduce_h16_n(const MoveDesc& m, u64 lo, u64 hi,
+                                  int lane) {
+  u64 n8 = (hi - lo) / 8;
+  GAS U4* o = (GAS U4*)((u16*)m.dst + lo);
+  GAS const U4* s[MOVE_MAX_SRC];
+  for (u32 k = 0; k < m.nsrc; ++k)
+    s[k] = (GAS const U4*)((const u16*)m.src[k] + lo);
+  const u32 ns = m.nsrc;
+  for (u64 i = lane; i < n8; i += 64) {
+    U4 v[MOVE_MAX_SRC];
+#pragma unroll
+    for (u32 k = 0; k < MOVE_MAX_SRC; ++k)
+      if (k < ns) v[k] = __builtin_nontemporal_load(&s[k][i]);
+    U4 acc = v[0];
+#pragma unroll
+    for (u32 k = 1; k < MOVE_MAX_SRC; ++k)
+      if (k < ns)
+        acc = (U4){h2_op<MAX_, BF16>(acc.x, v[k].x),
+                   h2_op<MAX_, BF16>(acc.y, v[k].y),
+                   h2_op<MAX_, BF16>(acc.z, v[k].z),
+                   h2_op<MAX_, BF16>(acc.w, v[k].w)};
+    __builtin_nontemporal_store(acc, &o[i]);
+  }
+}
+
 // float-domain path for any f32/f16/bf16 mix (cast + reduce fused — the
 // hp_compression + reduce_ops lanes in one pass)
 __device__ void tile_float_generic(const MoveDesc& m, u64 lo, u64 hi, int lane) {
@@ -446,34 +622,100 @@ __device__ bool run_tile(const MoveDesc& m, u32 t, int lane) {
     all_f32 = all_f32 && m.src_dt[k] == u8(DataType::float32);
     floatish = floatish && dtype_is_floatish(DataType(m.src_dt[k]));
   }
-  if (all_f32 && m.nsrc >= 2) {
+  const bool mx = ReduceFunction(m.func) == ReduceFunction::MAX;
+  if (all_f32 && m.nsrc == 2) {
     bool v16 = aligned16((const void*)(m.dst + lo * 4)) &&
                aligned16((const void*)(m.src[0] + lo * 4)) &&
                aligned16((const void*)(m.src[1] + lo * 4)) &&
-               ((hi - lo) & 3) == 0 && m.nsrc == 2;
-    if (ReduceFunction(m.func) == ReduceFunction::SUM)
-      tile_reduce_f32<SumOp>(m, lo, hi, lane);
-    else
-      tile_reduce_f32<MaxOp>(m, lo, hi, lane);
+               ((hi - lo) & 3) == 0;
+    if (!mx) tile_reduce_f32<SumOp>(m, lo, hi, lane);
+    else tile_reduce_f32<MaxOp>(m, lo, hi, lane);
     return v16;
   }
-  // packed same-dtype f16/bf16 two-source reduce (config-4 hot path)
-  if (m.nsrc == 2 && m.src_dt[0] == m.dst_dt && m.src_dt[1] == m.dst_dt &&
-      (m.dst_dt == u8(DataType::float16) || m.dst_dt == u8(DataType::bfloat16)) &&
+  if (all_f32 && m.nsrc > 2) {
+    bool v16 = aligned16((const void*)(m.dst + lo * 4)) &&
+               ((hi - lo) & 3) == 0;
+    for (u32 k = 0; k < m.nsrc; ++k)
+      v16 = v16 && aligned16((const void*)(m.src[k] + lo * 4));
+    if (v16) {
+      // n-ary fan-in (direct reduce/reduce_scatter stage fold)
+      if (!mx) tile_reduce_f32_n<SumOp>(m, lo, hi, lane);
+      else tile_reduce_f32_n<MaxOp>(m, lo, hi, lane);
+      return true;
+    }
+    tile_float_generic(m, lo, hi, lane);
+    return false;
+  }
+  // pure width conversion (compression lanes): f32 <-> f16/bf16
+  if (m.nsrc == 1 && ((hi - lo) & 7) == 0) {
+    DataType sd = DataType(m.src_dt[0]), dd = DataType(m.dst_dt);
+    bool s16 = sd == DataType::float16 || sd == DataType::bfloat16;
+    bool d16 = dd == DataType::float16 || dd == DataType::bfloat16;
+    if (s16 && dd == DataType::float32 &&
+        aligned16((const void*)(m.src[0] + lo * 2)) &&
+        aligned16((const void*)(m.dst + lo * 4))) {
+      if (sd == DataType::bfloat16) tile_cast<true, true>(m, lo, hi, lane);
+      else tile_cast<true, false>(m, lo, hi, lane);
+      return true;
+    }
+    if (d16 && sd == DataType::float32 &&
+        aligned16((const void*)(m.src[0] + lo * 4)) &&
+        aligned16((const void*)(m.dst + lo * 2))) {
+      if (dd == DataType::bfloat16) tile_cast<false, true>(m, lo, hi, lane);
+      else tile_cast<false, false>(m, lo, hi, lane);
+      return true;
+    }
+  }
+  // fused decompress + reduce: wire h16 segment folded into an f32 operand
+  if (m.nsrc == 2 && m.dst_dt == u8(DataType::float32) &&
+      m.src_dt[1] == u8(DataType::float32) &&
+      (m.src_dt[0] == u8(DataType::float16) ||
+       m.src_dt[0] == u8(DataType::bfloat16)) &&
       ((hi - lo) & 7) == 0 &&
-      aligned16((const void*)(m.dst + lo * 2)) &&
       aligned16((const void*)(m.src[0] + lo * 2)) &&
-      aligned16((const void*)(m.src[1] + lo * 2))) {
-    bool mx = ReduceFunction(m.func) == ReduceFunction::MAX;
-    bool bf = m.dst_dt == u8(DataType::bfloat16);
+      aligned16((const void*)(m.src[1] + lo * 4)) &&
+      aligned16((const void*)(m.dst + lo * 4))) {
+    bool bf = m.src_dt[0] == u8(DataType::bfloat16);
     if (bf) {
-      if (mx) tile_reduce_h16<true, true>(m, lo, hi, lane);
-      else tile_reduce_h16<false, true>(m, lo, hi, lane);
+      if (mx) tile_cast_reduce<true, true>(m, lo, hi, lane);
+      else tile_cast_reduce<false, true>(m, lo, hi, lane);
     } else {
-      if (mx) tile_reduce_h16<true, false>(m, lo, hi, lane);
-      else tile_reduce_h16<false, false>(m, lo, hi, lane);
+      if (mx) tile_cast_reduce<true, false>(m, lo, hi, lane);
+      else tile_cast_reduce<false, false>(m, lo, hi, lane);
     }
     return true;
+  }
+  // packed same-dtype f16/bf16 reduce, 2..8 sources (config-4 hot path +
+  // the bf16 direct fan-in stage fold)
+  {
+    bool h16 = (m.dst_dt == u8(DataType::float16) ||
+                m.dst_dt == u8(DataType::bfloat16)) &&
+               m.nsrc >= 2 && ((hi - lo) & 7) == 0 &&
+               aligned16((const void*)(m.dst + lo * 2));
+    for (u32 k = 0; k < m.nsrc && h16; ++k)
+      h16 = m.src_dt[k] == m.dst_dt &&
+            aligned16((const void*)(m.src[k] + lo * 2));
+    if (h16) {
+      bool bf = m.dst_dt == u8(DataType::bfloat16);
+      if (m.nsrc == 2) {
+        if (bf) {
+          if (mx) tile_reduce_h16<true, true>(m, lo, hi, lane);
+          else tile_reduce_h16<false, true>(m, lo, hi, lane);
+        } else {
+          if (mx) tile_reduce_h16<true, false>(m, lo, hi, lane);
+          else tile_reduce_h16<false, false>(m, lo, hi, lane);
+        }
+      } else {
+        if (bf) {
+          if (mx) tile_reduce_h16_n<true, true>(m, lo, hi, lane);
+          else tile_reduce_h16_n<false, true>(m, lo, hi, lane);
+        } else {
+          if (mx) tile_reduce_h16_n<true, false>(m, lo, hi, lane);
+          else tile_reduce_h16_n<false, false>(m, lo, hi, lane);
+        }
+      }
+      return true;
+    }
   }
   if (floatish) {
     tile_float_generic(m, lo, hi, lane);
@@ -703,6 +945,30 @@ __global__ void __launch_bounds__(128, 1) accl_scheduler_kernel(GpuEngineState* 
 
 __global__ void __launch_bounds__(256, 2) accl_mover_kernel(GpuEngineState* S) {
   mover_main(S);
+}
+
+// Bring-up probe: ONE thread stores a magic + ack mask into every peer
+// arena through the IPC imports — validates the SHADER write path into
+// each mapping before the persistent engine relies on it (a lazily
+// enabled peer mapping has been observed to silently drop shader writes
+// when two processes share one GPU; hipMemcpy probes would not catch it).
+__global__ void accl_probe_kernel(ProbeArgs a) {
+  if (threadIdx.x != 0 || blockIdx.x != 0) return;
+  for (u32 r = 0; r < a.nranks; ++r) {
+    if (r == a.me || !a.probe[r]) continue;
+    __hip_atomic_store((u64*)a.probe[r], PROBE_MAGIC | a.me, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_SYSTEM);
+    __hip_atomic_store((u64*)a.ack[r], a.seen_mask, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_SYSTEM);
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+}
+
+void gpu_probe_launch(const ProbeArgs& a, void* stream) {
+  hipLaunchKernelGGL(accl_probe_kernel, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, a);
 }
 
 void gpu_engine_launch(GpuEngineState* state_dev, int n_wgs, void* sched_stream,

@@ -86,4 +86,14 @@ struct GpuEngineState {
 void gpu_engine_launch(GpuEngineState* state_dev, int n_wgs, void* sched_stream,
                        void* mover_stream);
 
+// bring-up handshake probe (shader-store into every peer mapping)
+constexpr u64 PROBE_MAGIC = 0x50524F4245000000ull;  // "PROBE" | rank
+struct ProbeArgs {
+  u64* probe[MAX_RANKS];  // peer arenas: my probe word in rank r's arena
+  u64* ack[MAX_RANKS];    // peer arenas: my ack word in rank r's arena
+  u32 me, nranks;
+  u64 seen_mask;
+};
+void gpu_probe_launch(const ProbeArgs& a, void* stream);
+
 }  // namespace accl

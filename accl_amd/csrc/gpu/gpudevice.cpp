@@ -115,9 +115,10 @@ std::vector<char> GpuDevice::local_blob() {
 
 void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
   if (blobs.size() != cfg_.nranks) throw accl_error("gpu: blob count != nranks");
+  std::vector<GpuBlob> pb(cfg_.nranks);
   for (u32 r = 0; r < cfg_.nranks; ++r) {
     if (r == cfg_.rank) { peer_base_[r] = arena_base_; continue; }
-    GpuBlob b{};
+    GpuBlob& b = pb[r];
     std::memcpy(&b, blobs[r].data(), sizeof(b));
     if (b.arena_bytes != arena_bytes_)
       throw accl_error("gpu: rank " + std::to_string(r) +
@@ -126,6 +127,82 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
     hip_check(hipIpcOpenMemHandle(&p, b.handle, hipIpcMemLazyEnablePeerAccess),
               "hipIpcOpenMemHandle");
     peer_base_[r] = (char*)p;
+  }
+
+  // ---- bring-up handshake: every peer mapping must pass SHADER writes
+  // before the persistent engine depends on it. A lazily enabled IPC peer
+  // mapping has been observed (2 procs / 1 GPU) to silently drop shader
+  // stores in one direction, wedging the first collective; the handshake
+  // detects the dead direction at init and re-imports the handle.
+  if (cfg_.nranks > 1) {
+    const u64 probe_base = layout_.dbg_off + DBG_DUMP_BYTES - 2048;
+    const u64 ack_base = layout_.dbg_off + DBG_DUMP_BYTES - 1024;
+    const u32 me = cfg_.rank;
+    const u64 full = (cfg_.nranks >= 64) ? ~0ull
+                                         : ((1ull << cfg_.nranks) - 1);
+    u64 seen = 1ull << me;
+    u64 t0 = wallclock_host_ns(), last_change = t0;
+    int reimports = 0;
+    auto read_word = [&](u64 off) {
+      u64 v = 0;
+      read_arena(off, &v, sizeof(v));
+      return v;
+    };
+    auto launch_probe = [&]() {
+      ProbeArgs a{};
+      a.me = me;
+      a.nranks = cfg_.nranks;
+      a.seen_mask = seen;
+      for (u32 r = 0; r < cfg_.nranks; ++r) {
+        if (r == me) continue;
+        a.probe[r] = (u64*)(peer_base_[r] + probe_base + u64(me) * 8);
+        a.ack[r] = (u64*)(peer_base_[r] + ack_base + u64(me) * 8);
+      }
+      gpu_probe_launch(a, stream_);
+      hip_check(hipStreamSynchronize((hipStream_t)stream_), "probe sync");
+    };
+    for (;;) {
+      launch_probe();
+      usleep(2000);
+      u64 ns = seen;
+      bool all_acked = true;
+      for (u32 r = 0; r < cfg_.nranks; ++r) {
+        if (r == me) continue;
+        if (read_word(probe_base + u64(r) * 8) == (PROBE_MAGIC | r))
+          ns |= 1ull << r;
+        if (!(read_word(ack_base + u64(r) * 8) & (1ull << me)))
+          all_acked = false;
+      }
+      bool progressed = ns != seen;
+      seen = ns;
+      u64 now = wallclock_host_ns();
+      if (seen == full && all_acked) {
+        launch_probe();  // final ack broadcast with the complete seen mask
+        break;
+      }
+      if (progressed) last_change = now;
+      if (now - last_change > 1500ull * 1000 * 1000) {
+        // a direction is dead: re-import every peer handle and retry
+        for (u32 r = 0; r < cfg_.nranks; ++r) {
+          if (r == me) continue;
+          (void)hipIpcCloseMemHandle(peer_base_[r]);
+          void* p = nullptr;
+          hip_check(hipIpcOpenMemHandle(&p, pb[r].handle,
+                                        hipIpcMemLazyEnablePeerAccess),
+                    "hipIpcOpenMemHandle (reimport)");
+          peer_base_[r] = (char*)p;
+        }
+        reimports++;
+        last_change = now;
+      }
+      if (now - t0 > 60ull * 1000 * 1000 * 1000)
+        throw accl_error("gpu: peer-mapping handshake never completed "
+                         "(dead IPC direction persisted)");
+    }
+    if (reimports)
+      fprintf(stderr,
+              "accl gpu rank %u: peer mapping recovered after %d "
+              "re-import(s)\n", me, reimports);
   }
 
   // build the engine state on host, copy to device

@@ -84,6 +84,22 @@ PYBIND11_MODULE(_core, m) {
       .value("bfloat16", DataType::bfloat16)
       .value("int8", DataType::int8);
 
+  py::enum_<Op>(m, "Op")
+      .value("copy", Op::copy)
+      .value("combine", Op::combine)
+      .value("send", Op::send)
+      .value("recv", Op::recv)
+      .value("bcast", Op::bcast)
+      .value("scatter", Op::scatter)
+      .value("gather", Op::gather)
+      .value("reduce", Op::reduce)
+      .value("allgather", Op::allgather)
+      .value("allreduce", Op::allreduce)
+      .value("reduce_scatter", Op::reduce_scatter)
+      .value("alltoall", Op::alltoall)
+      .value("barrier", Op::barrier)
+      .value("stream_put", Op::stream_put);
+
   py::enum_<ReduceFunction>(m, "ReduceFunction")
       .value("SUM", ReduceFunction::SUM)
       .value("MAX", ReduceFunction::MAX);
@@ -346,6 +362,51 @@ PYBIND11_MODULE(_core, m) {
         py::arg("opts") = py::dict());
 
   m.def("error_to_string", &error_to_string);
+
+  // Raw-pointer collective call: operands are DEVICE pointers (e.g. torch
+  // tensors' data_ptr), not arena buffers — zero staging copies. Eager
+  // schedules only touch local src/dst (peers go through arena slots), so
+  // raw pointers are legal; direct paths require arena residency and the
+  // engine falls back to eager when the arena flags are absent.
+  m.def("call_raw",
+        [](ACCL& a, u32 scenario, u64 count, u32 root, u32 tag, u32 comm,
+           u32 function, u64 addr0, u64 addr1, u64 addr2, u32 dtype,
+           u32 wire_dtype, bool run_async) {
+          CallDesc d{};
+          d.scenario = scenario;
+          d.count_lo = u32(count);
+          d.count_hi = u32(count >> 32);
+          d.root_src_dst = root;
+          d.tag = tag;
+          d.comm_id = comm;
+          d.function = function;
+          d.addr0 = addr0;
+          d.addr1 = addr1;
+          d.addr2 = addr2;
+          d.flags = 0;  // raw pointers: no arena flags
+          d.arith = dtype | (wire_dtype << 8);
+          Backend* be = a.backend();
+          u64 seq = be->submit(d);
+          auto* r = new Request(be, seq);
+          if (!run_async) {
+            u32 e = r->wait();
+            if (e) {
+              delete r;
+              std::string msg = "accl call_raw failed: " + error_to_string(e);
+              if (e & E_TIMEOUT) msg += "\n" + be->timeout_dump_str();
+              throw accl_error(msg, e);
+            }
+          }
+          return r;
+        },
+        py::arg("a"), py::arg("scenario"), py::arg("count"),
+        py::arg("root") = 0, py::arg("tag") = 0, py::arg("comm") = 0,
+        py::arg("function") = 0, py::arg("addr0") = 0, py::arg("addr1") = 0,
+        py::arg("addr2") = 0, py::arg("dtype") = u32(DataType::float32),
+        py::arg("wire_dtype") = u32(DataType::float32),
+        py::arg("run_async") = false,
+        py::call_guard<py::gil_scoped_release>(),
+        py::return_value_policy::take_ownership);
 
   // deployment introspection (reference: xclbin_scan enumerating kernels /
   // memory banks, driver/utils/xclbin_scan)

@@ -20,19 +20,28 @@ from .accl import ACCL, DataType, _torch_dtype_map
 
 
 class _Work(dist.Work):
-    """Engine calls used here are blocking host-side, so work is complete."""
+    """Completion wrapper: blocking ops pass req=None (already complete —
+    is_completed() True is then truthful); async ops carry the engine
+    Request and report real completion state."""
 
-    def __init__(self, result=None):
+    def __init__(self, result=None, req=None):
         super().__init__()
         self._result = result if result is not None else []
+        self._req = req
 
     def is_completed(self):
-        return True
+        return self._req is None or self._req.test()
 
     def is_success(self):
-        return True
+        return self._req is None or not self._req.test() or \
+            self._req.retcode() == 0
 
     def wait(self, timeout=None):
+        if self._req is not None:
+            e = self._req.wait()
+            if e:
+                raise RuntimeError(f"accl work failed: {_core.error_to_string(e)}")
+            self._req = None
         return True
 
     def exception(self):
@@ -40,6 +49,7 @@ class _Work(dist.Work):
 
     def get_future(self):
         # DDP's reducer consumes the bucket tensors through this future
+        self.wait()
         fut = torch.futures.Future()
         fut.set_result(self._result)
         return fut
@@ -94,6 +104,24 @@ class AcclProcessGroup(dist.ProcessGroup):
                        job=f"tpg{size}", bootstrap=store_allgather)
         self._bufs = {}  # (count, DataType) -> (buffer, torch_view_or_None)
 
+    # ---------------- zero-copy path (GPU) ----------------
+    # Contiguous CUDA tensors go to the engine as RAW device pointers: the
+    # eager schedules only touch local src/dst through the mover fleet
+    # (peers communicate via arena slots), so no staging copy or stream
+    # round-trip is needed. The producing stream is synchronized first so
+    # the engine never reads half-written operands.
+    def _rawable(self, *tensors):
+        return self._gpu and all(
+            t.is_cuda and t.is_contiguous() for t in tensors)
+
+    def _raw(self, op, count, src, dst, func=0, root=0, tag=0):
+        torch.cuda.current_stream().synchronize()
+        _core.call_raw(self._a._a, int(op), count, root=root, tag=tag,
+                       function=int(func), addr0=src.data_ptr(),
+                       addr2=dst.data_ptr(),
+                       dtype=int(_t2dt(dst.dtype)),
+                       wire_dtype=int(_t2dt(dst.dtype)))
+
     # ---------------- staging ----------------
     def _buf(self, count, tdt, which):
         key = (which, count, tdt)
@@ -138,11 +166,14 @@ class AcclProcessGroup(dist.ProcessGroup):
         op = opts.reduceOp if opts is not None else dist.ReduceOp.SUM
         for t in tensors:
             n = t.numel()
-            s, _ = self._upload(t, n, "ar_s")
-            d, dv = self._buf(n, t.dtype, "ar_d")
-            self._a.allreduce(s, d, n, _red(op), from_device=True,
-                              to_device=True)
-            self._download(t, d, dv)
+            if self._rawable(t):
+                self._raw(_core.Op.allreduce, n, t, t, func=int(_red(op)))
+            else:
+                s, _ = self._upload(t, n, "ar_s")
+                d, dv = self._buf(n, t.dtype, "ar_d")
+                self._a.allreduce(s, d, n, _red(op), from_device=True,
+                                  to_device=True)
+                self._download(t, d, dv)
             if _is_avg(op):
                 t.div_(self.size())
         return _Work(list(tensors))
@@ -151,13 +182,19 @@ class AcclProcessGroup(dist.ProcessGroup):
         root = opts.rootRank if opts is not None else 0
         for t in tensors:
             n = t.numel()
-            b, bv = self._upload(t, n, "bc")
-            self._a.bcast(b, n, root, from_device=True, to_device=True)
-            self._download(t, b, bv)
+            if self._rawable(t):
+                self._raw(_core.Op.bcast, n, t, t, root=root)
+            else:
+                b, bv = self._upload(t, n, "bc")
+                self._a.bcast(b, n, root, from_device=True, to_device=True)
+                self._download(t, b, bv)
         return _Work()
 
     def _allgather_base(self, output, input, opts=None):
         n = input.numel()
+        if self._rawable(output, input):
+            self._raw(_core.Op.allgather, n, input, output)
+            return _Work()
         s, _ = self._upload(input, n, "ag_s")
         d, dv = self._buf(n * self.size(), input.dtype, "ag_d")
         self._a.allgather(s, d, n, from_device=True, to_device=True)
@@ -181,11 +218,15 @@ class AcclProcessGroup(dist.ProcessGroup):
     def _reduce_scatter_base(self, output, input, opts=None):
         op = opts.reduceOp if opts is not None else dist.ReduceOp.SUM
         per = output.numel()
-        s, _ = self._upload(input, per * self.size(), "rs_s")
-        d, dv = self._buf(per, input.dtype, "rs_d")
-        self._a.reduce_scatter(s, d, per, _red(op), from_device=True,
-                               to_device=True)
-        self._download(output, d, dv)
+        if self._rawable(output, input):
+            self._raw(_core.Op.reduce_scatter, per, input, output,
+                      func=int(_red(op)))
+        else:
+            s, _ = self._upload(input, per * self.size(), "rs_s")
+            d, dv = self._buf(per, input.dtype, "rs_d")
+            self._a.reduce_scatter(s, d, per, _red(op), from_device=True,
+                                   to_device=True)
+            self._download(output, d, dv)
         if _is_avg(op):
             output.div_(self.size())
         return _Work()
@@ -201,6 +242,9 @@ class AcclProcessGroup(dist.ProcessGroup):
                 (in_sizes and len(set(in_sizes)) > 1):
             raise NotImplementedError("accl backend: uneven alltoall")
         per = input.numel() // self.size()
+        if self._rawable(output, input):
+            self._raw(_core.Op.alltoall, per, input, output)
+            return _Work()
         s, _ = self._upload(input, per * self.size(), "a2a_s")
         d, dv = self._buf(per * self.size(), input.dtype, "a2a_d")
         self._a.alltoall(s, d, per, from_device=True, to_device=True)

@@ -424,9 +424,60 @@ PYBIND11_MODULE(_core, m) {
       d["total_mem_gb"] = double(p.totalGlobalMem) / (1u << 30);
       d["multiprocessors"] = p.multiProcessorCount;
       d["xgmi_capable"] = bool(p.isLargeBar);
+      // peer matrix (the xclbin_scan deployment-introspection analogue,
+      // reference driver/utils/xclbin_scan/): reachability + link perf
+      // rank per peer device
+      py::list peers;
+      for (int j = 0; j < n; ++j) {
+        if (j == i) continue;
+        int can = 0;
+        (void)hipDeviceCanAccessPeer(&can, i, j);
+        py::dict pd;
+        pd["device"] = j;
+        pd["p2p"] = bool(can);
+        int perf = 0;
+        if (hipDeviceGetP2PAttribute(&perf, hipDevP2PAttrPerformanceRank, i,
+                                     j) == hipSuccess)
+          pd["perf_rank"] = perf;
+        peers.append(pd);
+      }
+      d["peers"] = peers;
       out.append(d);
     }
     return out;
+  });
+
+  // engine capability word (the HWID/parse_hwid analogue, reference
+  // driver/xrt/src/accl.cpp:1050-1064): what this build of the engine
+  // supports, for deployment introspection
+  m.def("engine_capabilities", []() {
+    py::dict d;
+    d["version"] = 2;
+    d["max_ranks"] = MAX_RANKS;
+    d["max_comms"] = MAX_COMMS;
+    d["max_inflight_parked"] = MAX_INFLIGHT;
+    d["move_ring"] = MOVE_RING;
+    d["ring_cap"] = RING_CAP;
+    d["devcall_ring"] = DEVCALL_RING;
+    py::list ops;
+    for (const char* s :
+         {"copy", "combine", "send", "recv", "bcast", "scatter", "gather",
+          "reduce", "allgather", "allreduce", "reduce_scatter", "alltoall",
+          "barrier", "stream_put"})
+      ops.append(std::string(s));
+    d["ops"] = ops;
+    py::list dts;
+    for (const char* s :
+         {"float16", "float32", "float64", "int32", "int64", "bfloat16",
+          "int8"})
+      dts.append(std::string(s));
+    d["dtypes"] = dts;
+    d["protocols"] = py::make_tuple("eager", "rendezvous_direct");
+    d["features"] = py::make_tuple(
+        "multi_call_interleaving", "ooo_rendezvous_matching",
+        "windowed_nary_fan_in", "compression_f16_bf16_wire",
+        "device_initiated_calls", "stream_rings", "torch_backend");
+    return d;
   });
 
   // host-side injector into the device-call ring: exercises the

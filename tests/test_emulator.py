@@ -1226,3 +1226,12 @@ def test_fifo_same_tag_parked_sends():
     run_ranks(_fifo_same_tag, 2,
               opts={"n_slots": 2, "slot_bytes": 4096,
                     "timeout_us": 20_000_000})
+
+
+def test_engine_capabilities():
+    """Deployment introspection (reference: xclbin_scan + parse_hwid)."""
+    import accl_amd._core as core
+    cap = core.engine_capabilities()
+    assert cap["max_ranks"] >= 64 and "allreduce" in cap["ops"]
+    assert "ooo_rendezvous_matching" in cap["features"]
+    core.device_info()  # enumerates HIP devices (empty off-GPU)

@@ -59,9 +59,15 @@ class Backend {
     u64 seq = head_;
     while (seq - tail_retired_locked() >= RING_CAP) cpu_pause();
     d.seq = u32(seq);
-    ring_->descs[seq % RING_CAP] = d;
+    // desc/doorbell destination is normally the pinned ring; the GPU
+    // backend may point these at a DEVICE-resident ring written over the
+    // large BAR (engine then polls HBM instead of PCIe)
+    CallDesc* descs = desc_ring_ ? desc_ring_ : ring_->descs;
+    volatile u64* door = door_ ? door_ : &ring_->ctrl.doorbell;
+    std::memcpy((void*)&descs[seq % RING_CAP], &d, sizeof(d));
     head_ = seq + 1;
-    __atomic_store_n((u64*)&ring_->ctrl.doorbell, head_, __ATOMIC_RELEASE);
+    __sync_synchronize();  // drain WC/UC buffers before the doorbell
+    __atomic_store_n((u64*)door, head_, __ATOMIC_RELEASE);
     return seq;
   }
   bool test(u64 seq, RetEntry* out) {
@@ -216,6 +222,8 @@ class Backend {
   ProtoConfig cfg_{};
   char* arena_base_ = nullptr;
   RingPage* ring_ = nullptr;   // host-visible (pinned on GPU)
+  CallDesc* desc_ring_ = nullptr;   // optional device-resident desc ring
+  volatile u64* door_ = nullptr;    //   (host-writable over large BAR)
   HeapAlloc heap_;
   u64 head_ = 0, retired_ = 0;
   std::mutex ring_mu_;

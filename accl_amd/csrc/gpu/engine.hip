@@ -835,13 +835,18 @@ __device__ void scheduler_main(GpuEngineState* S, Cclo<GpuMover>& C) {
     fence_release_sys();
     st_sys32(&r.seq, u32(idx + 1));
   };
+  // dev_ring: host writes descs + doorbell into HBM over the large BAR —
+  // local poll instead of a PCIe fetch per iteration
+  volatile u64* door = S->dev_ring ? &S->ddoorbell : &ctrl->doorbell;
+  CallDesc* descs = S->dev_ring ? S->dring : S->descs;
   bool halted = false;
   for (;;) {
-    u64 db = ld_sys(&ctrl->doorbell);
+    u64 db = ld_sys(door);
     bool dev_pending = C.device_call_pending(dev_consumed);
     if (consumed == db && !dev_pending && !C.nparked) {
       if (ld_sys(&ctrl->shutdown)) break;
       if ((++beat & 0x3FF) == 0) st_sys(&ctrl->heartbeat, beat);
+      if (S->dev_ring) fence_acquire_sys();  // BAR-written doorbell/descs
       __builtin_amdgcn_s_sleep(16);
       continue;
     }
@@ -862,7 +867,7 @@ __device__ void scheduler_main(GpuEngineState* S, Cclo<GpuMover>& C) {
       continue;
     }
     if (consumed < db) {
-      CallDesc d = S->descs[consumed % RING_CAP];
+      CallDesc d = descs[consumed % RING_CAP];
       u64 t0 = wallclock();
       if (Op(d.scenario) == Op::halt) {
         publish(consumed, E_OK, t0);

@@ -61,7 +61,13 @@ struct GpuEngineState {
   u32 no_acq;              // ACCL_NO_ACQ=1: skip the mover wake-batch system
                            // acquire (measurement only — UNSOUND for peer-
                            // written payload, see mover_main)
-  u32 _pad0;
+  u32 dev_ring;            // descriptor ring lives in DEVICE memory (host
+                           // writes over the large BAR): scheduler polls
+                           // HBM instead of fetching descs over PCIe
+  // device-resident descriptor ring (dev_ring == 1)
+  alignas(64) CallDesc dring[RING_CAP];
+  alignas(64) u64 ddoorbell;
+  u64 _dpad[7];
   GpuMover mover;
   MoveDesc mq[MOVE_RING];
   MoveState mst[MOVE_RING];

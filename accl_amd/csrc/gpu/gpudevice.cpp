@@ -238,8 +238,33 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
   st->rets = rp_dev->rets;
   st->ctrl = &rp_dev->ctrl;
   st->comm_mirror = rp_dev->comm_mirror;
+  // Device-resident descriptor ring (small-op latency): with a large BAR
+  // the host can store descs + doorbell straight into HBM, so the
+  // scheduler polls local memory instead of fetching over PCIe each
+  // iteration. Verified by write+readback; ACCL_NO_DEV_RING disables.
+  bool dev_ring_ok = false;
+  if (!std::getenv("ACCL_NO_DEV_RING")) {
+    int large_bar = 0;
+    (void)hipDeviceGetAttribute(&large_bar, hipDeviceAttributeIsLargeBar,
+                                dev_);
+    if (large_bar) {
+      volatile u64* bar_door = &dstate->ddoorbell;
+      *bar_door = 0xC0FFEEull;
+      __sync_synchronize();
+      u64 chk = 0;
+      if (hipMemcpy(&chk, (void*)bar_door, 8, hipMemcpyDeviceToHost) ==
+              hipSuccess &&
+          chk == 0xC0FFEEull)
+        dev_ring_ok = true;
+    }
+  }
+  st->dev_ring = dev_ring_ok ? 1u : 0u;
   hip_check(hipMemcpy(state_dev_, st, sizeof(GpuEngineState),
                       hipMemcpyHostToDevice), "state upload");
+  if (dev_ring_ok) {  // host-side submit now targets the device ring
+    desc_ring_ = dstate->dring;
+    door_ = &dstate->ddoorbell;
+  }
   delete st;
 
   // global communicator 0 (before launch: engine reads mirror at gen bump)

@@ -101,9 +101,6 @@ struct Cclo {
   struct Unexpected {
     u32 tag; u32 arith; u64 bytes; u64 msg_count; u32 spare_slot; u32 flags;
   };
-  Unexpected uq[MAX_RANKS][UQ_DEPTH];
-  u32 uq_h[MAX_RANKS], uq_t[MAX_RANKS];
-  u64 spill_busy;          // bitmap over spill slots (<= 64)
 
   // ---- rendezvous pending sets (the RNDZV_PENDING spill queue analogue,
   // reference ccl_offload_control.c:154-408): in-seq records whose tag does
@@ -111,8 +108,6 @@ struct Cclo {
   static constexpr u32 RNDZV_PEND = 4;
   struct PendRndzv { u64 seq; u64 offset; u64 count; u32 tag; u32 arith;
                      u32 valid; };
-  PendRndzv pa[MAX_RANKS][RNDZV_PEND];  // addr records
-  PendRndzv pd[MAX_RANKS][RNDZV_PEND];  // done records
 
   // ---- multi-call interleaving (the CMD_CALL_RETRY requeue analogue,
   // reference ccl_offload_control.c:2460-2478 + current_step resume
@@ -125,7 +120,20 @@ struct Cclo {
     CallDesc d; u64 ring_idx; u64 deadline; u64 t_start; ParkState ps;
     u32 used; u32 _pad;
   };
-  ParkedCall parked[MAX_INFLIGHT];
+
+  // Cold match/park tables live OUTSIDE the Cclo (GPU: device-global
+  // via GpuEngineState; emulator: heap) — the GPU scheduler stages Cclo
+  // in LDS and the per-workgroup LDS budget is 64 KB; these tables are
+  // slow-path state that does not need LDS residency.
+  struct ColdState {
+    Unexpected uq[MAX_RANKS][UQ_DEPTH];
+    u32 uq_h[MAX_RANKS], uq_t[MAX_RANKS];
+    PendRndzv pa[MAX_RANKS][RNDZV_PEND];  // addr records
+    PendRndzv pd[MAX_RANKS][RNDZV_PEND];  // done records
+    ParkedCall parked[MAX_INFLIGHT];
+  };
+  ColdState* cold;
+  u64 spill_busy;          // bitmap over spill slots (<= 64)
   u32 nparked;
   // retry_parked() completion report:
   u64 done_ring_idx; u64 done_t0; u32 done_err;
@@ -479,14 +487,14 @@ struct Cclo {
     for (;;) {
       int best = -1;
       for (u32 k = 0; k < RNDZV_PEND; ++k) {
-        PendRndzv& p = pa[gpeer][k];
+        PendRndzv& p = cold->pa[gpeer][k];
         if (!p.valid) continue;
         if (want_tag != TAG_ANY && p.tag != want_tag && p.tag != TAG_ANY)
           continue;
-        if (best < 0 || p.seq < pa[gpeer][best].seq) best = int(k);
+        if (best < 0 || p.seq < cold->pa[gpeer][best].seq) best = int(k);
       }
       if (best >= 0) {
-        PendRndzv& p = pa[gpeer][best];
+        PendRndzv& p = cold->pa[gpeer][best];
         out.seq = p.seq; out.tag = p.tag; out.arith = p.arith;
         out.offset = p.offset; out.count = p.count;
         p.valid = 0;
@@ -503,9 +511,9 @@ struct Cclo {
           return true;
         }
         u32 k = 0;
-        while (k < RNDZV_PEND && pa[gpeer][k].valid) ++k;
+        while (k < RNDZV_PEND && cold->pa[gpeer][k].valid) ++k;
         if (k >= RNDZV_PEND) { err |= E_RNDZV; return false; }
-        pa[gpeer][k] = PendRndzv{seq, r->offset, r->count, r->tag, r->arith, 1};
+        cold->pa[gpeer][k] = PendRndzv{seq, r->offset, r->count, r->tag, r->arith, 1};
         continue;
       }
       if (!wait_pred_tick(deadline)) return false;
@@ -514,7 +522,7 @@ struct Cclo {
   // non-consuming probe: is a matching addr record available right now?
   ACCL_HD bool addr_ready(u32 gpeer, u32 want_tag) {
     for (u32 k = 0; k < RNDZV_PEND; ++k) {
-      PendRndzv& p = pa[gpeer][k];
+      PendRndzv& p = cold->pa[gpeer][k];
       if (p.valid && (want_tag == TAG_ANY || p.tag == want_tag ||
                       p.tag == TAG_ANY))
         return true;
@@ -534,13 +542,13 @@ struct Cclo {
     for (;;) {
       int best = -1;
       for (u32 k = 0; k < RNDZV_PEND; ++k) {
-        PendRndzv& p = pd[gpeer][k];
+        PendRndzv& p = cold->pd[gpeer][k];
         if (!p.valid) continue;
         if (want_tag != TAG_ANY && p.tag != want_tag && p.tag != TAG_ANY)
           continue;
-        if (best < 0 || p.seq < pd[gpeer][best].seq) best = int(k);
+        if (best < 0 || p.seq < cold->pd[gpeer][best].seq) best = int(k);
       }
-      if (best >= 0) { pd[gpeer][best].valid = 0; return true; }
+      if (best >= 0) { cold->pd[gpeer][best].valid = 0; return true; }
       u64 seq = sq.rndzv_done_rx[gpeer] + 1;
       RndzvRec* r = tv.rndzv_done(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
       if (ld_sys(&r->seq) == seq) {
@@ -549,9 +557,9 @@ struct Cclo {
         if (want_tag == TAG_ANY || r->tag == want_tag || r->tag == TAG_ANY)
           return true;
         u32 k = 0;
-        while (k < RNDZV_PEND && pd[gpeer][k].valid) ++k;
+        while (k < RNDZV_PEND && cold->pd[gpeer][k].valid) ++k;
         if (k >= RNDZV_PEND) { err |= E_RNDZV; return false; }
-        pd[gpeer][k] = PendRndzv{seq, 0, 0, r->tag, 0, 1};
+        cold->pd[gpeer][k] = PendRndzv{seq, 0, 0, r->tag, 0, 1};
         continue;
       }
       if (!wait_pred_tick(deadline)) return false;
@@ -643,7 +651,7 @@ struct Cclo {
   // move the head segment of channel (peer -> me) into the spill pool and
   // release the rx slot; false when queue/pool is full (caller keeps waiting)
   ACCL_HD bool spill_head(u32 peer, const SlotHdr* hd, u32 sl, u64 seq) {
-    if (uq_t[peer] - uq_h[peer] >= UQ_DEPTH) return false;
+    if (cold->uq_t[peer] - cold->uq_h[peer] >= UQ_DEPTH) return false;
     u32 ns = spill_slot_count();
     u32 sp = 0;
     while (sp < ns && ((spill_busy >> sp) & 1)) ++sp;
@@ -652,11 +660,11 @@ struct Cclo {
              spill_ptr(sp), DataType::int8, hd->bytes);
     u32 e = run_flows(1);
     if (e) return false;
-    Unexpected& u = uq[peer][uq_t[peer] % UQ_DEPTH];
+    Unexpected& u = cold->uq[peer][cold->uq_t[peer] % UQ_DEPTH];
     u.tag = hd->tag; u.arith = hd->arith; u.bytes = hd->bytes;
     u.msg_count = hd->msg_count; u.flags = hd->flags; u.spare_slot = sp;
     spill_busy |= 1ull << sp;
-    uq_t[peer]++;
+    cold->uq_t[peer]++;
     sq.eager_rx[peer] = seq;
     sq.credit_ret[peer] = seq;
     st_sys(&tv.chan_ctl(peer, me())->credit, seq);
@@ -676,8 +684,8 @@ struct Cclo {
     while (got < n) {
       bool progressed = false;
       // 1) spill queue, in arrival order (per-tag FIFO preserved)
-      for (u32 qi = uq_h[peer]; qi != uq_t[peer]; ++qi) {
-        Unexpected& u = uq[peer][qi % UQ_DEPTH];
+      for (u32 qi = cold->uq_h[peer]; qi != cold->uq_t[peer]; ++qi) {
+        Unexpected& u = cold->uq[peer][qi % UQ_DEPTH];
         if (u.bytes == 0) continue;  // consumed hole
         if (mtag != TAG_ANY && u.tag != mtag) continue;
         if (u.arith != u32(wdt)) { err |= E_COMPRESSION; return err; }
@@ -690,9 +698,9 @@ struct Cclo {
         if (e) return e;
         spill_busy &= ~(1ull << u.spare_slot);
         u.bytes = 0;
-        while (uq_h[peer] != uq_t[peer] &&
-               uq[peer][uq_h[peer] % UQ_DEPTH].bytes == 0)
-          uq_h[peer]++;
+        while (cold->uq_h[peer] != cold->uq_t[peer] &&
+               cold->uq[peer][cold->uq_h[peer] % UQ_DEPTH].bytes == 0)
+          cold->uq_h[peer]++;
         got += nseg;
         progressed = true;
         deadline = deadline_now();
@@ -1620,7 +1628,7 @@ struct Cclo {
   ACCL_HD bool parked_key_match(const CallDesc& d) const {
     if (!nparked) return false;
     for (u32 i = 0; i < MAX_INFLIGHT; ++i) {
-      const ParkedCall& p = parked[i];
+      const ParkedCall& p = cold->parked[i];
       if (p.used && p.d.scenario == d.scenario &&
           p.d.root_src_dst == d.root_src_dst && p.d.comm_id == d.comm_id &&
           p.d.tag == d.tag)
@@ -1631,13 +1639,13 @@ struct Cclo {
 
   ACCL_HD bool park(const CallDesc& d, u64 ring_idx, const ParkState& ps) {
     for (u32 i = 0; i < MAX_INFLIGHT; ++i) {
-      if (parked[i].used) continue;
-      parked[i].d = d;
-      parked[i].ring_idx = ring_idx;
-      parked[i].deadline = wallclock() + timeout_ticks;
-      parked[i].t_start = wallclock();
-      parked[i].ps = ps;
-      parked[i].used = 1;
+      if (cold->parked[i].used) continue;
+      cold->parked[i].d = d;
+      cold->parked[i].ring_idx = ring_idx;
+      cold->parked[i].deadline = wallclock() + timeout_ticks;
+      cold->parked[i].t_start = wallclock();
+      cold->parked[i].ps = ps;
+      cold->parked[i].used = 1;
       nparked++;
       return true;
     }
@@ -1650,11 +1658,11 @@ struct Cclo {
   ACCL_HD int retry_parked() {
     if (!nparked) return -1;
     for (u32 i = 0; i < MAX_INFLIGHT; ++i) {
-      ParkedCall& p = parked[i];
+      ParkedCall& p = cold->parked[i];
       if (!p.used) continue;
       bool blocked = false;
       for (u32 j = 0; j < MAX_INFLIGHT; ++j) {
-        const ParkedCall& q = parked[j];
+        const ParkedCall& q = cold->parked[j];
         if (!q.used || j == i) continue;
         if (q.ring_idx < p.ring_idx && q.d.scenario == p.d.scenario &&
             q.d.root_src_dst == p.d.root_src_dst &&
@@ -1705,7 +1713,7 @@ struct Cclo {
   // engine shutdown with calls still parked: fail them (engine going away)
   ACCL_HD int fail_parked() {
     for (u32 i = 0; i < MAX_INFLIGHT; ++i) {
-      ParkedCall& p = parked[i];
+      ParkedCall& p = cold->parked[i];
       if (!p.used) continue;
       done_ring_idx = p.ring_idx;
       done_err = E_ENGINE_DOWN;
@@ -1763,7 +1771,7 @@ struct Cclo {
         // pair sequence counters are PROTOCOL state shared with peers and
         // survive (desynced pairs need a reset on both ends).
         for (u32 i = 0; i < MAX_FLOWS; ++i) flows[i] = Flow{};
-        for (u32 r = 0; r < MAX_RANKS; ++r) { uq_h[r] = uq_t[r] = 0; }
+        for (u32 r = 0; r < MAX_RANKS; ++r) { cold->uq_h[r] = cold->uq_t[r] = 0; }
         spill_busy = 0;
         err = 0;
         return E_OK;

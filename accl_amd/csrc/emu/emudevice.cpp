@@ -10,6 +10,7 @@ namespace accl {
 
 struct EmuDevice::Engine {
   Cclo<CpuMover> cclo;
+  Cclo<CpuMover>::ColdState cold{};
   CpuMover mover;
   u64 cached_comm_gen = 0;
 };
@@ -126,6 +127,7 @@ void EmuDevice::connect(const std::vector<std::vector<char>>& blobs) {
   for (u32 r = 0; r < cfg_.nranks; ++r) C.tv.arena[r] = peer_base_[r];
   C.tv.cfg = cfg_;
   C.mv = &eng_->mover;
+  C.cold = &eng_->cold;
   C.timeout_ticks = cfg_.timeout_us * TICKS_PER_US;
   C.max_eager_bytes = cfg_.max_eager;
   // global communicator 0

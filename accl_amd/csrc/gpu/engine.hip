@@ -818,6 +818,7 @@ __device__ void mover_main(GpuEngineState* S) {
 // of ~0.5us per dependent global-memory load (it dominated small-op latency).
 __device__ void scheduler_main(GpuEngineState* S, Cclo<GpuMover>& C) {
   C.mv = &S->mover;
+  C.cold = &S->cold;
   S->mover.ring = S->mq;
   S->mover.st = S->mst;
   S->mover.head = &S->mq_head;
@@ -929,6 +930,11 @@ __device__ void small_mover(SmallMb* mb, int lane) {
 __global__ void __launch_bounds__(128, 1) accl_scheduler_kernel(GpuEngineState* S) {
   __shared__ SmallMb mb;
   __shared__ Cclo<GpuMover> C;
+  // the per-workgroup LDS budget on CDNA is 64 KB: exceeding it makes the
+  // kernel silently unlaunchable (engine never comes up)
+  static_assert(sizeof(Cclo<GpuMover>) + sizeof(SmallMb) <= 60 * 1024,
+                "scheduler LDS state too large — move cold tables to "
+                "ColdState (device-global)");
   // both waves cooperatively stage the host-initialized Cclo into LDS
   {
     const u64* src = (const u64*)&S->cclo;

@@ -58,6 +58,8 @@ constexpr u32 DOORBELL_REPS = 16;
 
 struct GpuEngineState {
   Cclo<GpuMover> cclo;     // trivially-copyable; host fills, device runs
+  Cclo<GpuMover>::ColdState cold;  // match/park tables (device-global, not
+                                   // LDS: per-WG LDS budget is 64 KB)
   u32 no_acq;              // ACCL_NO_ACQ=1: skip the mover wake-batch system
                            // acquire (measurement only — UNSOUND for peer-
                            // written payload, see mover_main)

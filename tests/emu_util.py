@@ -59,6 +59,21 @@ def run_ranks(fn, nranks, opts=None, timeout=120, backend="emu"):
     finally:
         for p in ps:
             p.join(timeout=10)
+        for p in ps:
+            if p.is_alive():
+                # SIGINT first: KeyboardInterrupt unwinds the worker so
+                # a.close() halts the persistent engine kernels — a hard
+                # terminate would leave them running on the GPU and poison
+                # every later test on the box
+                import os as _os
+                import signal as _signal
+                try:
+                    _os.kill(p.pid, _signal.SIGINT)
+                except OSError:
+                    pass
+        for p in ps:
+            if p.is_alive():
+                p.join(timeout=20)
             if p.is_alive():
                 p.terminate()
                 errs.append(f"rank {p.pid} hung; terminated")

@@ -22,6 +22,9 @@ struct GpuBlob {
   u32 _pad;
 };
 
+static bool dbg_connect() { return std::getenv("ACCL_DEBUG_CONNECT"); }
+#define CSTAGE(msg) do { if (dbg_connect()) { fprintf(stderr, "[accl connect r%u] %s\n", cfg_.rank, msg); fflush(stderr); } } while (0)
+
 GpuDevice::GpuDevice(u32 nranks, u32 rank, int device_index,
                      const ProtoConfig* cfg_override, u64 heap_bytes,
                      int engine_wgs) {
@@ -205,6 +208,7 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
               "re-import(s)\n", me, reimports);
   }
 
+  CSTAGE("peers mapped + handshake done");
   // build the engine state on host, copy to device
   auto* st = new GpuEngineState();
   std::memset((void*)st, 0, sizeof(GpuEngineState));
@@ -242,6 +246,7 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
   // the host can store descs + doorbell straight into HBM, so the
   // scheduler polls local memory instead of fetching over PCIe each
   // iteration. Verified by write+readback; ACCL_NO_DEV_RING disables.
+  CSTAGE("state built");
   bool dev_ring_ok = false;
   if (!std::getenv("ACCL_NO_DEV_RING")) {
     int large_bar = 0;
@@ -258,6 +263,7 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
         dev_ring_ok = true;
     }
   }
+  CSTAGE(dev_ring_ok ? "dev_ring ON" : "dev_ring off");
   st->dev_ring = dev_ring_ok ? 1u : 0u;
   hip_check(hipMemcpy(state_dev_, st, sizeof(GpuEngineState),
                       hipMemcpyHostToDevice), "state upload");
@@ -277,10 +283,12 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
   __atomic_store_n((u64*)&ring_->ctrl.ncomms, 1, __ATOMIC_RELEASE);
   __atomic_store_n((u64*)&ring_->ctrl.comm_gen, 1, __ATOMIC_RELEASE);
 
+  CSTAGE("state uploaded; launching engine");
   gpu_engine_launch((GpuEngineState*)state_dev_, engine_wgs_, stream_,
                     mover_stream_);
   hip_check(hipGetLastError(), "engine kernel launch");
   launched_ = true;
+  CSTAGE("kernels launched; waiting engine_up");
   // wait for the engine to come up
   u64 t0 = wallclock_host_ns();
   while (!__atomic_load_n((u64*)&ring_->ctrl.engine_up, __ATOMIC_ACQUIRE)) {

@@ -19,6 +19,7 @@
 //  * all polls are relaxed + s_sleep; one acquire fence after a match.
 #include <hip/hip_runtime.h>
 #include <hip/hip_fp16.h>
+#include <cstdlib>
 #include "engine.hpp"
 
 namespace accl {
@@ -955,6 +956,10 @@ __global__ void __launch_bounds__(128, 1) accl_scheduler_kernel(GpuEngineState* 
 }
 
 __global__ void __launch_bounds__(256, 2) accl_mover_kernel(GpuEngineState* S) {
+  // bring-up telemetry: count mover WGs that actually started (dbg[15])
+  if (threadIdx.x == 0)
+    __hip_atomic_fetch_add(&S->dbg[15], 1ull, __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
   mover_main(S);
 }
 
@@ -984,10 +989,20 @@ void gpu_probe_launch(const ProbeArgs& a, void* stream) {
 
 void gpu_engine_launch(GpuEngineState* state_dev, int n_wgs, void* sched_stream,
                        void* mover_stream) {
-  hipLaunchKernelGGL(accl_mover_kernel, dim3(n_wgs), dim3(256), 0,
-                     (hipStream_t)mover_stream, state_dev);
-  hipLaunchKernelGGL(accl_scheduler_kernel, dim3(1), dim3(128), 0,
-                     (hipStream_t)sched_stream, state_dev);
+  // scheduler FIRST: its single WG must never be starved behind the
+  // 640-WG fleet dispatch (ACCL_MOVER_FIRST restores the old order for
+  // comparison)
+  if (std::getenv("ACCL_MOVER_FIRST")) {
+    hipLaunchKernelGGL(accl_mover_kernel, dim3(n_wgs), dim3(256), 0,
+                       (hipStream_t)mover_stream, state_dev);
+    hipLaunchKernelGGL(accl_scheduler_kernel, dim3(1), dim3(128), 0,
+                       (hipStream_t)sched_stream, state_dev);
+  } else {
+    hipLaunchKernelGGL(accl_scheduler_kernel, dim3(1), dim3(128), 0,
+                       (hipStream_t)sched_stream, state_dev);
+    hipLaunchKernelGGL(accl_mover_kernel, dim3(n_wgs), dim3(256), 0,
+                       (hipStream_t)mover_stream, state_dev);
+  }
 }
 
 }  // namespace accl

@@ -292,8 +292,16 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
   // wait for the engine to come up
   u64 t0 = wallclock_host_ns();
   while (!__atomic_load_n((u64*)&ring_->ctrl.engine_up, __ATOMIC_ACQUIRE)) {
-    if (wallclock_host_ns() - t0 > 30ull * 1000000000)
-      throw accl_error("gpu: engine kernel never came up");
+    if (wallclock_host_ns() - t0 > 30ull * 1000000000) {
+      u64 movers_started = 0;
+      (void)hipMemcpy(&movers_started,
+                      (char*)state_dev_ + offsetof(GpuEngineState, dbg) +
+                          15 * sizeof(u64),
+                      sizeof(u64), hipMemcpyDeviceToHost);
+      throw accl_error("gpu: engine kernel never came up (mover WGs "
+                       "started: " + std::to_string(movers_started) + "/" +
+                       std::to_string(engine_wgs_) + ")");
+    }
     usleep(100);
   }
 }

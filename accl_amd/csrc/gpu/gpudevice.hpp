@@ -48,11 +48,9 @@ class GpuDevice : public Backend {
 
  private:
   int dev_ = 0;
-  int engine_wgs_ = 512;  // 256-thr WGs: uniform 2 WGs/CU (8 of 16 wave
-                          // slots + >200 VGPRs/SIMD left for co-resident
-                          // kernels — torch interop and overlapped-GEMM
-                          // need the headroom; round-2 tiles raised mover
-                          // VGPRs to ~147, capping occupancy at 3 WGs/CU)
+  int engine_wgs_ = 640;  // 256-thr WGs, ~2.5 per CU (mover VGPR use caps
+                          // occupancy at 3 WGs/CU; wave slots and >70
+                          // VGPRs/SIMD stay free for co-resident kernels)
   u64 arena_bytes_ = 0;
   ArenaLayout layout_{};
   bool fine_grained_ = true;

@@ -68,7 +68,9 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
                        __HIP_MEMORY_SCOPE_WORKGROUP);
     d = m;
     d.inline_done = 1;
-    d.epoch = h + 1;
+    // epoch published LAST with release: lagging movers validate descs
+    // seqlock-style against it (fields -> release -> epoch)
+    __hip_atomic_store(&d.epoch, h + 1, __ATOMIC_RELEASE, AGENT);
     s.tiles_total = 0;
     __hip_atomic_store(&s.tiles_done, 0u, __ATOMIC_RELAXED, AGENT);
     // bounded wait: if the sibling wave is wedged/slow, execute the move
@@ -108,7 +110,7 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
   }
   d = m;
   d.inline_done = 0;
-  d.epoch = h + 1;
+  __hip_atomic_store(&d.epoch, h + 1, __ATOMIC_RELEASE, AGENT);
   s.tiles_total = move_tiles(m);
   __hip_atomic_store(&s.tiles_claimed, 0u, __ATOMIC_RELAXED, AGENT);
   __hip_atomic_store(&s.tiles_done, 0u, __ATOMIC_RELAXED, AGENT);
@@ -796,6 +798,16 @@ __device__ void mover_main(GpuEngineState* S) {
       }
       u32 total = m.inline_done ? 0 : move_tiles(m);
       if (first < total) {
+        // seqlock re-check: the slot may have been recycled mid-read by an
+        // inline burst (>MOVE_RING moves with no fleet doorbell); a torn
+        // desc must never execute. Same epoch on both sides of the field
+        // reads => consistent (writer publishes epoch last, release).
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        if (u32(__hip_atomic_load(&ring[slot].epoch, __ATOMIC_RELAXED,
+                                  AGENT)) != u32(cursor + 1)) {
+          cursor++;
+          continue;
+        }
         if (lane == 0 && first == 0) S->dbg[1] = wallclock();
         u32 cnt = 0;
         for (u32 t = first; t < total; t += nwaves, ++cnt) run_tile(m, t, lane);

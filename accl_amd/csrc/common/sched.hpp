@@ -519,7 +519,10 @@ struct Cclo {
       if (!wait_pred_tick(deadline)) return false;
     }
   }
-  // non-consuming probe: is a matching addr record available right now?
+  // non-consuming probe: is a TAG-MATCHING addr record available right now?
+  // (a foreign-tag head record must NOT make a parked send commit — it
+  // would block the engine inside wait_addr while the matching send for
+  // that record sits queued behind it)
   ACCL_HD bool addr_ready(u32 gpeer, u32 want_tag) {
     for (u32 k = 0; k < RNDZV_PEND; ++k) {
       PendRndzv& p = cold->pa[gpeer][k];
@@ -529,7 +532,10 @@ struct Cclo {
     }
     u64 seq = sq.rndzv_addr_rx[gpeer] + 1;
     RndzvRec* r = tv.rndzv_addr(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
-    return ld_sys(&r->seq) == seq;
+    if (ld_sys(&r->seq) != seq) return false;
+    fence_acquire_sys();
+    u32 t = r->tag;
+    return want_tag == TAG_ANY || t == want_tag || t == TAG_ANY;
   }
   ACCL_HD void post_done(u32 gpeer, u32 tag) {
     u64 seq = ++sq.rndzv_done_tx[gpeer];

@@ -68,6 +68,41 @@ static void rank_main(u32 rank, const std::string& job) {
     assert(nb == 1024 * 4 && tag == 3);
   }
   a.barrier();
+
+  // round-2 surface: direct (rendezvous) collectives + windowed n-ary
+  // fan-in under the sanitizers (forced by the tiny max_eager above)
+  const u64 m = 6000;  // 24 KB > max_eager
+  auto s2 = a.create_buffer(m * 2, DataType::float32);
+  auto d2 = a.create_buffer(m, DataType::float32);
+  float* s2p = (float*)s2->host_ptr();
+  for (u64 i = 0; i < m * 2; ++i)
+    s2p[i] = float((i * 3 + rank * 11) % 53) - 26;
+  a.reduce_scatter(*s2, *d2, m, ReduceFunction::SUM);
+  float* d2p = (float*)d2->host_ptr();
+  for (u64 i = 0; i < m; ++i) {
+    u64 gi = rank * m + i;
+    float e = (float((gi * 3) % 53) - 26) + (float((gi * 3 + 11) % 53) - 26);
+    assert(d2p[i] == e);
+  }
+  a.barrier();
+
+  // parked-call interleaving: rank 1 posts the recv seconds "early" (rank 0
+  // delays its send behind other traffic), exercising park/retry + resume
+  if (rank == 1) {
+    Request* rq = a.recv(*d, n, 0, 77, 0, false, DataType::none, true);
+    for (int it = 0; it < 5; ++it) a.allreduce(*s, *d2, 100, ReduceFunction::SUM);
+    // the engine must have served the allreduces while the recv was parked
+    a.barrier();           // release rank 0's send
+    assert(rq->wait() == 0);
+    a.free_request(rq);
+  } else {
+    for (int it = 0; it < 5; ++it) a.allreduce(*s, *d2, 100, ReduceFunction::SUM);
+    a.barrier();
+    for (u64 i = 0; i < n; ++i) sp[i] = float(i % 31);
+    s->sync_to_device();
+    a.send(*s, n, 1, 77);
+  }
+  a.barrier();
   std::printf("rank %u OK\n", rank);
 }
 

@@ -45,6 +45,9 @@ struct PendSeg {
   u64 elems;
   u64 seq;        // eager: segment seq to publish / credit value to return
   u32 first_last; // SEG_FIRST/SEG_LAST for hdr publish
+  u32 spill;      // FLOW_RX: spare_slot+1 when the segment was consumed
+                  // from the unexpected-spill pool (0 = live ring; credit
+                  // was already returned at spill time)
 };
 
 struct Flow {
@@ -226,6 +229,12 @@ struct Cclo {
           break;
         }
         case FLOW_RX: {
+          if (p.spill) {
+            // segment came from the spill pool (credit returned when it
+            // was spilled); free the spare slot now that the move retired
+            spill_busy &= ~(1ull << (p.spill - 1));
+            break;
+          }
           // payload consumed -> return credit to the sender (cumulative).
           sq.credit_ret[f.gpeer] = p.seq;
           st_sys(&tv.chan_ctl(f.gpeer, me())->credit, p.seq);
@@ -317,6 +326,53 @@ struct Cclo {
           break;
         }
         case FLOW_RX: {
+          // 0) spill pool first: a parked recv's probe may have spilled
+          // this flow's segment out of the live ring (spilled entries are
+          // strictly older than the current head, so consuming them first
+          // preserves per-tag send order)
+          {
+            bool consumed_spill = false;
+            for (u32 qi = cold->uq_h[f.gpeer]; qi != cold->uq_t[f.gpeer];
+                 ++qi) {
+              Unexpected& u = cold->uq[f.gpeer][qi % UQ_DEPTH];
+              if (u.bytes == 0) continue;
+              u32 want = (f.submitted == 0) ? f.tag : f.matched_tag;
+              if (want != TAG_ANY && u.tag != want) continue;
+              if (u.arith != u32(f.wdt)) { err |= E_COMPRESSION; return any; }
+              u32 wsz0 = dtype_size(DataType(f.wdt));
+              u64 n0 = u.bytes / wsz0;
+              if (n0 > f.count - f.submitted) { err |= E_SEGMENT; return any; }
+              if (f.gate && f.submitted + n0 > gate_limit(f)) return any;
+              if (f.submitted == 0) f.matched_tag = u.tag;
+              m.dst = (u64)(f.dst + f.submitted * dtype_size(DataType(f.ddt)));
+              m.dst_dt = f.ddt;
+              m.src[0] = (u64)spill_ptr(u.spare_slot);
+              m.src_dt[0] = f.wdt;
+              m.nsrc = 1;
+              if (f.func) {
+                m.src[1] =
+                    (u64)(f.red + f.submitted * dtype_size(DataType(f.bdt)));
+                m.src_dt[1] = f.bdt;
+                m.nsrc = 2;
+                m.func = f.func - 1;
+              }
+              m.count = n0;
+              u32 tok0 = mv->submit(m);
+              f.pend[f.pt % FLOW_INFLIGHT] =
+                  PendSeg{tok0, 0, n0, 0, 0, u.spare_slot + 1};
+              f.pt++;
+              f.submitted += n0;
+              u.bytes = 0;  // entry claimed; spill_busy freed at retire
+              while (cold->uq_h[f.gpeer] != cold->uq_t[f.gpeer] &&
+                     cold->uq[f.gpeer][cold->uq_h[f.gpeer] % UQ_DEPTH].bytes ==
+                         0)
+                cold->uq_h[f.gpeer]++;
+              any = true;
+              consumed_spill = true;
+              break;
+            }
+            if (consumed_spill) continue;
+          }
           u64 next = sq.eager_rx[f.gpeer];          // segments consumed
           u32 slot = u32(next % cfg.n_slots);
           SlotHdr* h = tv.slot_hdr(me(), f.gpeer, slot);

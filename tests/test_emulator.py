@@ -1235,3 +1235,33 @@ def test_engine_capabilities():
     assert cap["max_ranks"] >= 64 and "allreduce" in cap["ops"]
     assert "ooo_rendezvous_matching" in cap["features"]
     core.device_info()  # enumerates HIP devices (empty off-GPU)
+
+
+def _parked_vs_collectives(a, rank, n):
+    """A parked recv's probes must never steal (spill) a collective's eager
+    segments in a way the collective's flows cannot recover — regression
+    for the spill-pool/flow integration (caught by the ASan harness)."""
+    cnt = 2000
+    if rank == 1:
+        d = _mk(a, cnt)
+        req = a.recv(d, cnt, src=0, tag=77, run_async=True)
+    s, dr = _mk(a, 100), _mk(a, 100)
+    s.write(pattern(100, rank, seed=3))
+    for it in range(12):
+        a.allreduce(s, dr, 100, RF.SUM)
+    exp = np.stack([pattern(100, r, seed=3) for r in range(n)]).sum(0)
+    assert np.allclose(rd(dr, 100), exp)
+    a.barrier()
+    if rank == 0:
+        sv = _mk(a, cnt)
+        sv.write(pattern(cnt, 9, seed=7))
+        a.send(sv, cnt, dst=1, tag=77)
+    else:
+        assert req.wait() == 0
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 9, seed=7))
+    a.barrier()
+
+
+def test_parked_recv_vs_collectives():
+    for _ in range(3):
+        run_ranks(_parked_vs_collectives, 2, opts=SMALL)

@@ -93,6 +93,8 @@ struct Cclo {
                            // window (0 = unlimited); reference:
                            // set_max_rendezvous_size (accl.hpp:103-104)
   u32 err;                 // error bits of the current call
+  u32 nflows_;             // size of the flow set run_flows is executing
+                           // (sibling-flow scan for cross-call spills)
   Flow flows[MAX_FLOWS];
 
   // ---- unexpected-message queue (the rxbuf_seek match-engine analogue:
@@ -380,11 +382,28 @@ struct Cclo {
           fence_acquire_sys();
           // Per-segment tag demultiplex: several rx flows may drain one
           // pair channel concurrently (e.g. allreduce phase 1 + phase 2);
-          // a head segment that is not ours is left for the sibling flow.
-          // (reference analogue: rxbuf_seek matching by (tag, src, seqn),
-          // kernels/cclo/hls/rxbuf_offload/rxbuf_seek.cpp:53-72)
+          // a head segment that belongs to a SIBLING flow in this set is
+          // left in place, but a CROSS-CALL segment (a user send whose
+          // matching recv comes later in program order) must be spilled to
+          // the unexpected pool or it head-of-line-blocks this collective
+          // forever. (reference analogue: rxbuf_seek matching by
+          // (tag, src, seqn), rxbuf_seek.cpp:53-72)
           u32 want = (f.submitted == 0) ? f.tag : f.matched_tag;
-          if (want != TAG_ANY && h->tag != want) return any;
+          if (want != TAG_ANY && h->tag != want) {
+            bool sibling = false;
+            for (u32 j = 0; j < nflows_; ++j) {
+              const Flow& g = flows[j];
+              if (g.kind != FLOW_RX || flow_done(g) || g.gpeer != f.gpeer)
+                continue;
+              u32 gw = (g.submitted == 0) ? g.tag : g.matched_tag;
+              if (gw == TAG_ANY || gw == h->tag) { sibling = true; break; }
+            }
+            if (!sibling) {
+              if (spill_head(f.gpeer, h, slot, next + 1)) { any = true; continue; }
+              if (err) return any;
+            }
+            return any;
+          }
           if (f.submitted == 0) f.matched_tag = h->tag;
           u32 wsz = dtype_size(DataType(f.wdt));
           if (h->arith != u32(f.wdt)) { err |= E_COMPRESSION; return any; }
@@ -450,6 +469,7 @@ struct Cclo {
 
   // run a set of flows to completion (the engine inner loop)
   ACCL_HD u32 run_flows(u32 n) {
+    nflows_ = n;
     stamp(10);
     u64 deadline = deadline_now();
     for (;;) {
@@ -711,17 +731,27 @@ struct Cclo {
   }
 
   // move the head segment of channel (peer -> me) into the spill pool and
-  // release the rx slot; false when queue/pool is full (caller keeps waiting)
+  // release the rx slot; false when queue/pool is full (caller keeps
+  // waiting). NON-REENTRANT w.r.t. the flow table: the copy goes straight
+  // to the mover (no flows), so it is safe to call from INSIDE
+  // flow_submit when a flow meets a foreign-tag head segment.
   ACCL_HD bool spill_head(u32 peer, const SlotHdr* hd, u32 sl, u64 seq) {
     if (cold->uq_t[peer] - cold->uq_h[peer] >= UQ_DEPTH) return false;
     u32 ns = spill_slot_count();
     u32 sp = 0;
     while (sp < ns && ((spill_busy >> sp) & 1)) ++sp;
     if (sp >= ns || hd->bytes > cfg.slot_bytes) return false;
-    mk_local(0, tv.slot_payload(me(), peer, sl), DataType::int8,
-             spill_ptr(sp), DataType::int8, hd->bytes);
-    u32 e = run_flows(1);
-    if (e) return false;
+    MoveDesc m{};
+    m.dst = (u64)spill_ptr(sp);
+    m.dst_dt = u8(DataType::int8);
+    m.src[0] = (u64)tv.slot_payload(me(), peer, sl);
+    m.src_dt[0] = u8(DataType::int8);
+    m.nsrc = 1;
+    m.count = hd->bytes;
+    u32 tok = mv->submit(m);
+    u64 deadline = deadline_now();
+    while (!mv->poll(tok))
+      if (!wait_pred_tick(deadline)) return false;
     Unexpected& u = cold->uq[peer][cold->uq_t[peer] % UQ_DEPTH];
     u.tag = hd->tag; u.arith = hd->arith; u.bytes = hd->bytes;
     u.msg_count = hd->msg_count; u.flags = hd->flags; u.spare_slot = sp;

@@ -1265,3 +1265,113 @@ def _parked_vs_collectives(a, rank, n):
 def test_parked_recv_vs_collectives():
     for _ in range(3):
         run_ranks(_parked_vs_collectives, 2, opts=SMALL)
+
+
+def _fuzz_script(seed, n, nops=40):
+    """Deterministic op script shared by all ranks: mixes collectives,
+    tagged pairwise send/recv (sync + async), sizes spanning inline/fleet/
+    rendezvous. Every rank derives the same script from the seed."""
+    rng = np.random.default_rng(seed)
+    ops = []
+    for _ in range(nops):
+        k = int(rng.integers(0, 7))
+        cnt = int(rng.integers(16, 30_000))
+        tag = int(rng.integers(1, 500))
+        if k == 0:
+            ops.append(("allreduce", cnt))
+        elif k == 1:
+            ops.append(("allgather", cnt))
+        elif k == 2:
+            ops.append(("reduce_scatter", cnt))
+        elif k == 3:
+            ops.append(("bcast", cnt, int(rng.integers(0, n))))
+        elif k == 4:
+            a_, b_ = rng.choice(n, size=2, replace=False)
+            ops.append(("sendrecv", cnt, int(a_), int(b_), tag,
+                        bool(rng.integers(0, 2))))
+        elif k == 5:
+            ops.append(("alltoall", cnt))
+        else:
+            ops.append(("barrier",))
+    return ops
+
+
+def _fuzz(a, rank, n, seed):
+    ops = _fuzz_script(seed, n)
+    pending = []
+    for i, op in enumerate(ops):
+        if op[0] == "allreduce":
+            cnt = op[1]
+            s, d = _mk(a, cnt), _mk(a, cnt)
+            s.write(pattern(cnt, rank, seed=i))
+            a.allreduce(s, d, cnt, RF.SUM)
+            exp = np.stack([pattern(cnt, r, seed=i) for r in range(n)]).sum(0)
+            assert np.allclose(rd(d, cnt), exp), f"op{i} allreduce"
+        elif op[0] == "allgather":
+            cnt = op[1]
+            s, d = _mk(a, cnt), _mk(a, cnt * n)
+            s.write(pattern(cnt, rank, seed=i))
+            a.allgather(s, d, cnt)
+            exp = np.concatenate([pattern(cnt, r, seed=i) for r in range(n)])
+            assert np.array_equal(rd(d, cnt * n), exp), f"op{i} allgather"
+        elif op[0] == "reduce_scatter":
+            cnt = op[1]
+            s, d = _mk(a, cnt * n), _mk(a, cnt)
+            s.write(np.concatenate([pattern(cnt, rank + 3 * j, seed=i)
+                                    for j in range(n)]))
+            a.reduce_scatter(s, d, cnt, RF.SUM)
+            exp = np.stack([pattern(cnt, r + 3 * rank, seed=i)
+                            for r in range(n)]).sum(0)
+            assert np.allclose(rd(d, cnt), exp), f"op{i} rs"
+        elif op[0] == "bcast":
+            cnt, root = op[1], op[2]
+            b = _mk(a, cnt)
+            if rank == root:
+                b.write(pattern(cnt, 55, seed=i))
+            a.bcast(b, cnt, root)
+            assert np.array_equal(rd(b, cnt), pattern(cnt, 55, seed=i)), \
+                f"op{i} bcast"
+        elif op[0] == "sendrecv":
+            cnt, src_r, dst_r, tag, asyn = op[1:]
+            if rank == src_r:
+                s = _mk(a, cnt)
+                s.write(pattern(cnt, src_r, seed=i))
+                r = a.send(s, cnt, dst=dst_r, tag=tag, run_async=asyn)
+                if asyn:
+                    pending.append((r, None, None, None))
+            elif rank == dst_r:
+                d = _mk(a, cnt)
+                r = a.recv(d, cnt, src=src_r, tag=tag, run_async=asyn)
+                if asyn:
+                    pending.append((r, d, cnt, pattern(cnt, src_r, seed=i)))
+                else:
+                    assert np.array_equal(rd(d, cnt),
+                                          pattern(cnt, src_r, seed=i)), \
+                        f"op{i} recv"
+        elif op[0] == "alltoall":
+            cnt = op[1]
+            s, d = _mk(a, cnt * n), _mk(a, cnt * n)
+            s.write(np.concatenate([pattern(cnt, rank * 9 + j, seed=i)
+                                    for j in range(n)]))
+            a.alltoall(s, d, cnt)
+            exp = np.concatenate([pattern(cnt, r * 9 + rank, seed=i)
+                                  for r in range(n)])
+            assert np.array_equal(rd(d, cnt * n), exp), f"op{i} alltoall"
+        else:
+            a.barrier()
+    for r, d, cnt, exp in pending:
+        assert r.wait() == 0
+        if d is not None:
+            assert np.array_equal(rd(d, cnt), exp), "async recv data"
+    a.barrier()
+
+
+@pytest.mark.parametrize("seed", [11, 23, 37])
+def test_protocol_fuzz(seed):
+    run_ranks(lambda a, r, n: _fuzz(a, r, n, seed), 2, opts=SMALL, timeout=240)
+    run_ranks(lambda a, r, n: _fuzz(a, r, n, seed), 3, opts=SMALL, timeout=240)
+
+
+def test_protocol_fuzz_direct():
+    # same scripts under forced-tiny max_eager: every op takes a direct path
+    run_ranks(lambda a, r, n: _fuzz(a, r, n, 51), 3, opts=DIRECT, timeout=240)

@@ -1053,7 +1053,7 @@ struct Cclo {
     u32 nf = 0;
     mk_local(nf++, src, dt, dst + u64(r) * n * dtype_size(dt), dt, n);
     bool direct = use_rndzv(n, dt, wdt) && (d.flags & F_DST_ARENA) &&
-                  (d.flags & F_SRC_ARENA) && c.size <= 9;
+                  (d.flags & F_SRC_ARENA) && 2 * u64(c.size) - 1 <= MAX_FLOWS;
     if (direct) {
       u32 myslot[MAX_RANKS];
       for (u32 p = 0; p < c.size; ++p)

@@ -1406,6 +1406,29 @@ struct Cclo {
       mk_local(0, src, dt, dst, dt, total);
       return run_flows(1);
     }
+    // Large-message direct path: compose direct reduce_scatter (windowed
+    // n-ary fan-in over xGMI stages) + direct allgather (single peer
+    // writes) — no eager slot staging on either phase. (reference shape:
+    // rendezvous allreduce = reduce+bcast composition,
+    // ccl_offload_control.c:1878-1887; ours keeps the RS+AG structure.)
+    if (total % P == 0 && coll_direct_ok(d, total / P, dt, wdt) &&
+        2 * u64(P - 1) <= MAX_FLOWS) {
+      u64 chunk = total / P;
+      CallDesc rs = d;
+      rs.scenario = u32(Op::reduce_scatter);
+      rs.count_lo = u32(chunk);
+      rs.count_hi = u32(chunk >> 32);
+      rs.addr2 = d.addr2 + u64(r) * chunk * dtype_size(dt);
+      u32 e = op_reduce_scatter(rs, c);
+      if (e) return e;
+      CallDesc ag = d;
+      ag.scenario = u32(Op::allgather);
+      ag.count_lo = u32(chunk);
+      ag.count_hi = u32(chunk >> 32);
+      ag.addr0 = d.addr2 + u64(r) * chunk * dtype_size(dt);
+      ag.addr2 = d.addr2;
+      return op_allgather(ag, c);
+    }
     if (P > 9) return ring_allreduce(d, c);  // flow budget: fullmesh P<=9
     const u32 dsz = dtype_size(dt);
     // chunk partition: chunk i = [off(i), off(i+1)), balanced

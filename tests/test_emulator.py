@@ -1375,3 +1375,24 @@ def test_protocol_fuzz(seed):
 def test_protocol_fuzz_direct():
     # same scripts under forced-tiny max_eager: every op takes a direct path
     run_ranks(lambda a, r, n: _fuzz(a, r, n, 51), 3, opts=DIRECT, timeout=240)
+
+
+def _ar_direct(a, rank, n):
+    """Composed direct allreduce (direct RS + direct AG, no eager
+    staging): counts divisible by P trigger the path under tiny max_eager."""
+    for cnt in (6000 * n, 12_288 * n):
+        s, d = _mk(a, cnt), _mk(a, cnt)
+        s.write(pattern(cnt, rank, seed=cnt))
+        a.allreduce(s, d, cnt, RF.SUM)
+        exp = np.stack([pattern(cnt, r, seed=cnt) for r in range(n)]).sum(0)
+        assert np.allclose(rd(d, cnt), exp), f"cnt={cnt}"
+        a.allreduce(s, d, cnt, RF.MAX)
+        exp = np.stack([pattern(cnt, r, seed=cnt) for r in range(n)]).max(0)
+        assert np.array_equal(rd(d, cnt), exp), f"MAX cnt={cnt}"
+    a.barrier()
+
+
+def test_allreduce_direct_composed():
+    run_ranks(_ar_direct, 2, opts=DIRECT)
+    run_ranks(_ar_direct, 3, opts=DIRECT)
+    run_ranks(_ar_direct, 4, opts=DIRECT)

@@ -222,7 +222,7 @@ class ACCL:
         if name in ("copy", "combine", "send", "recv", "bcast", "scatter",
                     "gather", "allgather", "reduce", "allreduce",
                     "reduce_scatter", "alltoall", "barrier", "nop",
-                    "stream_put", "pop_stream", "stream_ready",
+                    "stream_put", "pop_stream", "stream_ready", "push_stream",
                     "copy_from_stream", "send_from_stream", "alive",
                     "soft_reset",
                     "info", "set_timeout_ms", "set_max_eager_size",

@@ -278,6 +278,16 @@ PYBIND11_MODULE(_core, m) {
            },
            py::arg("src"), py::arg("out"), py::arg("timeout_ms") = 10000)
       .def("stream_ready", &ACCL::stream_ready, py::arg("src"))
+      .def("push_stream",
+           [](ACCL& a, u32 dst, py::buffer data, u32 tag, u64 timeout_ms) {
+             py::buffer_info info = data.request();
+             return a.push_stream(dst, info.ptr,
+                                  u64(info.size) * u64(info.itemsize), tag,
+                                  timeout_ms);
+           },
+           py::arg("dst"), py::arg("data"), py::arg("tag") = 0,
+           py::arg("timeout_ms") = 10000,
+           py::call_guard<py::gil_scoped_release>())
       .def("copy_from_stream", &ACCL::copy_from_stream, py::arg("lane"),
            py::arg("dst"), py::arg("count"), py::arg("to_device") = false,
            py::arg("run_async") = false, py::return_value_policy::reference,

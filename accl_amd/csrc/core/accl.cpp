@@ -245,6 +245,51 @@ StreamLane stream_lane(const ProtoConfig& c, u32 src_lane) {
 }
 }  // namespace
 
+u64 ACCL::push_stream(u32 dst, const void* data, u64 bytes, u32 tag,
+                      u64 timeout_ms) {
+  // Host-side twin of device_api::stream_push — the CCLO_BFM analogue
+  // (reference test/model/bfm/cclo_bfm.cpp:28-178): lets user "kernel"
+  // logic co-simulate its stream-producer role against the emulator
+  // without a GPU, interoperating with engine ops and real device kernels
+  // through the shared tx_ctr allocator and credit words.
+  const ProtoConfig& c = be_->cfg();
+  if (dst >= c.nranks) throw accl_error("push_stream: bad dst");
+  if (be_->is_gpu())
+    throw accl_error(
+        "push_stream is the emulator-side BFM producer; on the GPU use "
+        "device_api::stream_push from a kernel (or ACCL::stream_put)");
+  char* my = be_->arena_local();
+  // channel (me -> dst): ctl (credit + tx_ctr) lives in MY arena lane [dst]
+  StreamLane mine = stream_lane(c, dst);
+  auto* ctl = (EagerChanCtl*)(my + mine.ctl_off);
+  if (bytes > c.stream_bytes)
+    throw accl_error("push_stream: segment larger than stream slot");
+  u64 seq = __atomic_fetch_add(&ctl->tx_ctr, 1ull, __ATOMIC_RELAXED) + 1;
+  u64 t0 = wallclock_host_ns();
+  while (__atomic_load_n(&ctl->credit, __ATOMIC_ACQUIRE) + c.n_stream < seq) {
+    if (wallclock_host_ns() - t0 > timeout_ms * 1000000ull)
+      throw accl_error("push_stream: no stream credit (consumer stalled)");
+    cpu_pause();
+  }
+  // payload + header into DST's arena lane [me]
+  StreamLane theirs = stream_lane(c, c.rank);
+  u32 slot = u32((seq - 1) % c.n_stream);
+  be_->write_peer(dst, theirs.payload_off + u64(slot) * c.stream_bytes, data,
+                  bytes);
+  SlotHdr h{};
+  h.tag = tag;
+  h.bytes = u32(bytes);
+  h.msg_count = bytes;
+  h.arith = 0;
+  h.flags = SEG_FIRST | SEG_LAST;
+  h.seq = 0;  // published separately, last
+  u64 hdr_off = theirs.hdr_off + u64(slot) * sizeof(SlotHdr);
+  be_->write_peer(dst, hdr_off, &h, sizeof(h));
+  __atomic_thread_fence(__ATOMIC_RELEASE);
+  be_->write_peer(dst, hdr_off + offsetof(SlotHdr, seq), &seq, sizeof(seq));
+  return seq;
+}
+
 bool ACCL::stream_ready(u32 src) {
   const ProtoConfig& c = be_->cfg();
   StreamLane L = stream_lane(c, src);

@@ -171,6 +171,11 @@ class ACCL {
                  u64 timeout_ms = 10000);
   // Non-destructive check: is a stream segment from `src` pending?
   bool stream_ready(u32 src);
+  // Host-side BFM producer (reference CCLO_BFM): push one segment into
+  // dst's stream ring exactly like device_api::stream_push (emulator
+  // backend only — co-simulation of user-kernel stream logic).
+  u64 push_stream(u32 dst, const void* data, u64 bytes, u32 tag = 0,
+                  u64 timeout_ms = 10000);
   // Stream-fed ops: the ENGINE consumes ring lane [lane] (one consumer per
   // lane — do not mix with pop_stream on the same lane). reference:
   // OP0_STREAM operand routing (dma_mover.cpp:497; stream2mem tests).

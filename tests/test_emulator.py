@@ -1396,3 +1396,34 @@ def test_allreduce_direct_composed():
     run_ranks(_ar_direct, 2, opts=DIRECT)
     run_ranks(_ar_direct, 3, opts=DIRECT)
     run_ranks(_ar_direct, 4, opts=DIRECT)
+
+
+def _bfm(a, rank, n):
+    """CCLO_BFM analogue: host-side user-'kernel' stream producer
+    (push_stream) co-simulated against the engine — consumed by the peer's
+    pop_stream AND by an engine stream-fed op, interoperating with
+    engine-produced stream_put segments through the shared allocator."""
+    cnt = 2000
+    if rank == 0:
+        x = pattern(cnt, 4)
+        a.push_stream(1, x.view(np.int8), tag=9)       # BFM producer
+        s = _mk(a, cnt)
+        s.write(pattern(cnt, 5))
+        a.stream_put(s, cnt, dst=1, tag=10)            # engine producer
+    else:
+        out = np.zeros(cnt, np.float32)
+        nb, tag = a.pop_stream(0, out)
+        assert (nb, tag) == (cnt * 4, 9) and np.array_equal(out, pattern(cnt, 4))
+        nb, tag = a.pop_stream(0, out)
+        assert (nb, tag) == (cnt * 4, 10) and np.array_equal(out, pattern(cnt, 5))
+        # BFM self-push on lane 1 consumed by an ENGINE stream-fed op (one
+        # consumer per lane: lane 0 is the host pop_stream's)
+        a.push_stream(1, pattern(cnt, 6).view(np.int8), tag=11)
+        d = _mk(a, cnt)
+        a.copy_from_stream(1, d, cnt)
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 6))
+    a.barrier()
+
+
+def test_bfm_host_stream_producer():
+    run_ranks(_bfm, 2)

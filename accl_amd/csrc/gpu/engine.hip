@@ -583,6 +583,42 @@ __device__ void tile_reduce_h16_n(const MoveDesc& m, u64 lo, u64 hi,
   }
 }
 
+// n-ary same-dtype reduce over 16-byte vectors for the remaining reduce_ops
+// dtypes (i32: 4/vec, i64/f64: 2/vec) — reference reduce_ops.cpp:31-107
+// supports f32/f64/i32/i64/f16; ELT is the element type, EPV elems/vector.
+template <typename ELT, u32 EPV, bool MAX_>
+__device__ void tile_reduce_wide(const MoveDesc& m, u64 lo, u64 hi,
+                                 int lane) {
+  u64 nv = (hi - lo) / EPV;
+  GAS U4* o = (GAS U4*)((ELT*)m.dst + lo);
+  GAS const U4* s[MOVE_MAX_SRC];
+  for (u32 k = 0; k < m.nsrc; ++k)
+    s[k] = (GAS const U4*)((const ELT*)m.src[k] + lo);
+  const u32 ns = m.nsrc;
+  for (u64 i = lane; i < nv; i += 64) {
+    U4 v[MOVE_MAX_SRC];
+#pragma unroll
+    for (u32 k = 0; k < MOVE_MAX_SRC; ++k)
+      if (k < ns) v[k] = __builtin_nontemporal_load(&s[k][i]);
+    ELT acc[EPV];
+#pragma unroll
+    for (u32 e = 0; e < EPV; ++e) acc[e] = ((ELT*)&v[0])[e];
+#pragma unroll
+    for (u32 k = 1; k < MOVE_MAX_SRC; ++k) {
+      if (k >= ns) break;
+#pragma unroll
+      for (u32 e = 0; e < EPV; ++e) {
+        ELT b = ((ELT*)&v[k])[e];
+        acc[e] = MAX_ ? (acc[e] > b ? acc[e] : b) : ELT(acc[e] + b);
+      }
+    }
+    U4 out;
+#pragma unroll
+    for (u32 e = 0; e < EPV; ++e) ((ELT*)&out)[e] = acc[e];
+    __builtin_nontemporal_store(out, &o[i]);
+  }
+}
+
 // float-domain path for any f32/f16/bf16 mix (cast + reduce fused — the
 // hp_compression + reduce_ops lanes in one pass)
 __device__ void tile_float_generic(const MoveDesc& m, u64 lo, u64 hi, int lane) {
@@ -720,11 +756,38 @@ __device__ bool run_tile(const MoveDesc& m, u32 t, int lane) {
       return true;
     }
   }
+  // same-dtype i32/i64/f64 reduce, 2..8 sources, 16-byte vectors
+  {
+    DataType dd = DataType(m.dst_dt);
+    bool wide = (dd == DataType::int32 || dd == DataType::int64 ||
+                 dd == DataType::float64) &&
+                m.nsrc >= 2;
+    u32 esz2 = dtype_size(dd);
+    u32 epv = esz2 ? 16 / esz2 : 0;
+    wide = wide && epv && ((hi - lo) % epv) == 0 &&
+           aligned16((const void*)(m.dst + lo * esz2));
+    for (u32 k = 0; k < m.nsrc && wide; ++k)
+      wide = m.src_dt[k] == m.dst_dt &&
+             aligned16((const void*)(m.src[k] + lo * esz2));
+    if (wide) {
+      if (dd == DataType::int32) {
+        if (mx) tile_reduce_wide<i32, 4, true>(m, lo, hi, lane);
+        else tile_reduce_wide<i32, 4, false>(m, lo, hi, lane);
+      } else if (dd == DataType::int64) {
+        if (mx) tile_reduce_wide<i64, 2, true>(m, lo, hi, lane);
+        else tile_reduce_wide<i64, 2, false>(m, lo, hi, lane);
+      } else {
+        if (mx) tile_reduce_wide<double, 2, true>(m, lo, hi, lane);
+        else tile_reduce_wide<double, 2, false>(m, lo, hi, lane);
+      }
+      return true;
+    }
+  }
   if (floatish) {
     tile_float_generic(m, lo, hi, lane);
     return false;
   }
-  // exact scalar fallback (f64 / int dtypes / mixed)
+  // exact scalar fallback (int8 / mixed dtypes)
   for (u64 i = lo + lane; i < hi; i += 64) execute_move_range(m, i, i + 1);
   return false;
 }

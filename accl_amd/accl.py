@@ -226,7 +226,7 @@ class ACCL:
                     "copy_from_stream", "send_from_stream", "alive",
                     "soft_reset",
                     "info", "set_timeout_ms", "set_max_eager_size",
-                    "set_max_rendezvous_size",
+                    "set_max_rendezvous_size", "set_tuning",
                     "dump_communicator", "dump_eager_rx_buffers",
                     "dump_streams", "dump_engine_status",
                     "create_communicator", "split_communicator",

@@ -319,6 +319,8 @@ PYBIND11_MODULE(_core, m) {
            })
       .def("set_timeout_ms", &ACCL::set_timeout_ms,
            py::call_guard<py::gil_scoped_release>())
+      .def("set_tuning", &ACCL::set_tuning, py::arg("knob"),
+           py::arg("value"), py::call_guard<py::gil_scoped_release>())
       .def("set_max_rendezvous_size", &ACCL::set_max_rendezvous_size,
            py::call_guard<py::gil_scoped_release>())
       .def("set_max_eager_size", &ACCL::set_max_eager_size,

@@ -92,6 +92,9 @@ struct Cclo {
   u64 max_rndzv_bytes;     // window cap for a single posted rendezvous
                            // window (0 = unlimited); reference:
                            // set_max_rendezvous_size (accl.hpp:103-104)
+  u32 tune_fullmesh_max;   // allreduce fullmesh->ring cutoff (runtime
+                           // tuning register; reference flat-tree caps,
+                           // driver/xrt/src/accl.cpp:1198-1208); 0 = default
   u32 err;                 // error bits of the current call
   u32 nflows_;             // size of the flow set run_flows is executing
                            // (sibling-flow scan for cross-call spills)
@@ -1429,7 +1432,9 @@ struct Cclo {
       ag.addr2 = d.addr2;
       return op_allgather(ag, c);
     }
-    if (P > 9) return ring_allreduce(d, c);  // flow budget: fullmesh P<=9
+    u32 fm_max = tune_fullmesh_max ? tune_fullmesh_max : 9;
+    if (fm_max > 17) fm_max = 17;  // 4(P-1)+1 flows must fit MAX_FLOWS
+    if (P > fm_max) return ring_allreduce(d, c);
     const u32 dsz = dtype_size(dt);
     // chunk partition: chunk i = [off(i), off(i+1)), balanced
     u64 base = total / P, rem = total % P;
@@ -1878,6 +1883,11 @@ struct Cclo {
         return E_OK;
       case CfgFunc::set_max_rendezvous_size:
         max_rndzv_bytes = desc_count(d);
+        return E_OK;
+      case CfgFunc::set_tuning:
+        // tuning registers (reference configure_tuning_parameters):
+        // knob id in root_src_dst, value in count
+        if (d.root_src_dst == 0) tune_fullmesh_max = u32(desc_count(d));
         return E_OK;
       case CfgFunc::reset: {
         // soft reset (reference: encore_soft_reset drains retry queue +

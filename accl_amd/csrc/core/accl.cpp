@@ -338,6 +338,12 @@ void ACCL::set_max_eager_size(u64 bytes) {
   d.function = u32(CfgFunc::set_max_eager_size);
   be_->call(d);
 }
+void ACCL::set_tuning(u32 knob, u64 value) {
+  CallDesc d = make_desc(Op::config, value, DataType::none, DataType::none);
+  d.function = u32(CfgFunc::set_tuning);
+  d.root_src_dst = knob;
+  be_->call(d);
+}
 void ACCL::set_max_rendezvous_size(u64 bytes) {
   CallDesc d = make_desc(Op::config, bytes, DataType::none, DataType::none);
   d.function = u32(CfgFunc::set_max_rendezvous_size);

@@ -196,6 +196,9 @@ class ACCL {
   void set_max_eager_size(u64 bytes);
   void set_max_rendezvous_size(u64 bytes);  // window cap for one posted
                                             // rendezvous transfer
+  void set_tuning(u32 knob, u64 value);     // runtime tuning registers
+                                            // (knob 0: allreduce fullmesh
+                                            // -> ring cutoff)
   // local engine soft reset (reference: ACCL soft_reset, accl.cpp:57-69)
   void soft_reset();
 

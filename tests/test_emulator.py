@@ -1427,3 +1427,21 @@ def _bfm(a, rank, n):
 
 def test_bfm_host_stream_producer():
     run_ranks(_bfm, 2)
+
+
+def _tuning(a, rank, n):
+    """Runtime tuning registers (reference configure_tuning_parameters):
+    forcing the allreduce fullmesh->ring cutoff below P routes through the
+    ring schedule; results must be identical."""
+    a.set_tuning(0, 2)  # fullmesh only up to P=2 -> ring at P=3
+    cnt = 4000
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    s.write(pattern(cnt, rank, seed=8))
+    a.allreduce(s, d, cnt, RF.SUM)
+    exp = np.stack([pattern(cnt, r, seed=8) for r in range(n)]).sum(0)
+    assert np.allclose(rd(d, cnt), exp)
+    a.barrier()
+
+
+def test_tuning_registers():
+    run_ranks(_tuning, 3, opts=SMALL)

@@ -40,7 +40,7 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
     __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
   }
   u64 mbytes = move_bytes(m);
-  bool small = small_mb && mbytes <= SMALL_INLINE_MAX;
+  bool small = small_mb && mbytes <= (small_max ? small_max : SMALL_INLINE_MAX);
   MoveDesc& d = ring[slot];
   MoveState& s = st[slot];
   // Tile-size policy (measured, profiles/r2: 256 MiB 856->1199 GB/s at
@@ -60,9 +60,11 @@ __device__ u32 GpuMover::submit(const MoveDesc& m) {
     if (dbg) dbg[5] = wallclock();
     SmallMb* mb = (SmallMb*)small_mb;
     mb->d = m;
-    mb->d.tile_log2 = 0;  // sibling executes tile 0 only: one default-size
-                          // tile always covers a <=32KB move, a small
-                          // ACCL_TILE_KB override would not
+    // sibling executes tile 0 only: pick a tile size covering the WHOLE
+    // move (the default 128 KiB covers any cutoff up to ACCL_INLINE_KB=128)
+    u8 cov = 17;
+    while ((1ull << cov) < mbytes) ++cov;
+    mb->d.tile_log2 = cov;
     u64 sq = ++small_seq;
     __hip_atomic_store(&mb->seq, sq, __ATOMIC_RELEASE,
                        __HIP_MEMORY_SCOPE_WORKGROUP);

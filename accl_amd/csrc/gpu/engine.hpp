@@ -22,7 +22,9 @@ struct SmallMb {
   u64 _pad;
   MoveDesc d;
 };
-constexpr u64 SMALL_INLINE_MAX = 32u << 10;  // bytes
+constexpr u64 SMALL_INLINE_MAX = 32u << 10;  // bytes (default; the
+                                             // ACCL_INLINE_KB env knob
+                                             // overrides via small_max)
 constexpr u32 INLINE_TOKEN = 0x80000000u;    // poll(): done at submit time
 
 // Device-side mover handle. Methods (submit/poll) are device-only and live
@@ -40,6 +42,7 @@ struct GpuMover {
   void* small_mb;          // LDS SmallMb (set by the scheduler kernel)
   u64 small_seq;           // scheduler-private inline-move counter
   u32 tile_log2;           // 0 = default tile; ACCL_TILE_KB env override
+  u32 small_max;           // inline-path cutoff bytes (ACCL_INLINE_KB)
 
 #if defined(__HIPCC__)
   __device__ u32 submit(const MoveDesc& m);

@@ -226,6 +226,8 @@ void GpuDevice::connect(const std::vector<std::vector<char>>& blobs) {
   st->mover.stop = &dstate->stop;
   st->mover.dbg = dstate->dbg;
   st->no_acq = std::getenv("ACCL_NO_ACQ") ? 1u : 0u;
+  if (const char* ik = std::getenv("ACCL_INLINE_KB"))
+    st->mover.small_max = u32(strtoul(ik, nullptr, 10)) << 10;
   if (const char* t = std::getenv("ACCL_TILE_KB")) {
     u64 kb = strtoull(t, nullptr, 10);
     u32 lg = 0;

@@ -219,7 +219,7 @@ class ACCL:
 
     # ---------------- ops (delegate) ----------------
     def __getattr__(self, name):
-        if name in ("copy", "combine", "send", "recv", "bcast", "scatter",
+        if name in ("copy", "put", "combine", "send", "recv", "bcast", "scatter",
                     "gather", "allgather", "reduce", "allreduce",
                     "reduce_scatter", "alltoall", "barrier", "nop",
                     "stream_put", "pop_stream", "stream_ready", "push_stream",

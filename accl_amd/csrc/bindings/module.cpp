@@ -186,6 +186,11 @@ PYBIND11_MODULE(_core, m) {
              return make_dlpack(b, a.backend()->is_gpu(), device_id);
            },
            py::arg("buffer"), py::arg("device_id") = 0)
+      .def("put", &ACCL::put, py::arg("src"), py::arg("count"),
+           py::arg("dst_rank"), py::arg("peer_arena_offset"),
+           py::arg("from_device") = false, py::arg("run_async") = false,
+           py::call_guard<py::gil_scoped_release>(),
+           py::return_value_policy::reference)
       .def("create_communicator", &ACCL::create_communicator)
       .def("split_communicator", &ACCL::split_communicator)
       .def("copy", &ACCL::copy, py::arg("src"), py::arg("dst"), py::arg("count"),

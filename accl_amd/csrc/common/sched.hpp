@@ -661,6 +661,14 @@ struct Cclo {
     if (d.flags & F_SRC_STREAM)  // stream2mem: drain ring lane addr0
       return stream_fed(u32(d.addr0), local_ptr(d.addr2, d.flags & F_DST_ARENA),
                         desc_dtype(d), n, -1, desc_dtype(d), 0);
+    if (d.flags & F_DST_PEER) {
+      // one-sided put: write straight into the peer's mapped arena over
+      // xGMI (reference copy_p2p semantics); user-level synchronization
+      u32 peer = d.root_src_dst < cfg.nranks ? d.root_src_dst : cfg.rank;
+      mk_local(0, local_ptr(d.addr0, d.flags & F_SRC_ARENA), desc_dtype(d),
+               tv.arena[peer] + d.addr2, desc_dtype(d), n);
+      return run_flows(1);
+    }
     mk_local(0, local_ptr(d.addr0, d.flags & F_SRC_ARENA), desc_dtype(d),
              local_ptr(d.addr2, d.flags & F_DST_ARENA), desc_dtype(d), n);
     stamp(9);

@@ -106,6 +106,11 @@ enum CallFlags : u32 {
                               // segments from its own stream ring (reference:
                               // OP0_STREAM send/reduce-from-krnl-stream,
                               // dma_mover.cpp:497)
+  F_DST_PEER = 1u << 8,       // copy: addr2 is an arena OFFSET in rank
+                              // root_src_dst's arena — one-sided put over
+                              // xGMI (reference: copy into a p2p buffer,
+                              // test_copy_p2p; user synchronizes, e.g.
+                              // barrier)
 };
 
 // -------------------------------------------------------------------- errors

@@ -187,6 +187,18 @@ Request* ACCL::copy(BaseBuffer& src, BaseBuffer& dst, u64 count,
                 from_device ? nullptr : &src, count);
 }
 
+Request* ACCL::put(BaseBuffer& src, u64 count, u32 dst_rank,
+                   u64 peer_arena_offset, bool from_device, bool run_async) {
+  // one-sided xGMI write into a peer-resident buffer (reference:
+  // test_copy_p2p — CCLO copy into a p2p bo); synchronize with barrier()
+  CallDesc d = make_desc(Op::copy, count, src.dtype(), src.dtype());
+  d.addr0 = src.arena_offset();
+  d.addr2 = peer_arena_offset;
+  d.root_src_dst = dst_rank;
+  d.flags = F_SRC_ARENA | F_DST_PEER;
+  return finish(d, run_async, nullptr, 0, from_device ? nullptr : &src, count);
+}
+
 Request* ACCL::combine(u64 count, ReduceFunction f, BaseBuffer& op0,
                        BaseBuffer& op1, BaseBuffer& res, bool from_device,
                        bool to_device, bool run_async) {

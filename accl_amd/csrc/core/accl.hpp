@@ -114,6 +114,11 @@ class ACCL {
   Request* copy(BaseBuffer& src, BaseBuffer& dst, u64 count,
                 bool from_device = false, bool to_device = false,
                 bool run_async = false);
+  // one-sided put into rank dst_rank's arena at peer_arena_offset
+  // (reference: copy into a p2p buffer, test_copy_p2p); user synchronizes
+  Request* put(BaseBuffer& src, u64 count, u32 dst_rank,
+               u64 peer_arena_offset, bool from_device = false,
+               bool run_async = false);
   Request* combine(u64 count, ReduceFunction f, BaseBuffer& op0,
                    BaseBuffer& op1, BaseBuffer& res,
                    bool from_device = false, bool to_device = false,

@@ -1338,7 +1338,9 @@ def _fuzz(a, rank, n, seed):
                 s.write(pattern(cnt, src_r, seed=i))
                 r = a.send(s, cnt, dst=dst_r, tag=tag, run_async=asyn)
                 if asyn:
-                    pending.append((r, None, None, None))
+                    # keep the buffer alive until completion (its dtor frees
+                    # the heap block for reuse — MPI buffer-lifetime rule)
+                    pending.append((r, None, None, s))
             elif rank == dst_r:
                 d = _mk(a, cnt)
                 r = a.recv(d, cnt, src=src_r, tag=tag, run_async=asyn)
@@ -1363,6 +1365,7 @@ def _fuzz(a, rank, n, seed):
         assert r.wait() == 0
         if d is not None:
             assert np.array_equal(rd(d, cnt), exp), "async recv data"
+    del pending
     a.barrier()
 
 
@@ -1445,3 +1448,22 @@ def _tuning(a, rank, n):
 
 def test_tuning_registers():
     run_ranks(_tuning, 3, opts=SMALL)
+
+
+def _put(a, rank, n):
+    """One-sided put (reference copy_p2p): rank 0 writes straight into
+    rank 1's buffer over the peer mapping; barrier = the user-level sync."""
+    cnt = 4000
+    d = _mk(a, cnt)  # same alloc order on both ranks -> same arena offset
+    if rank == 0:
+        s = _mk(a, cnt)
+        s.write(pattern(cnt, 42))
+        a.put(s, cnt, 1, d.arena_offset)
+    a.barrier()
+    if rank == 1:
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 42))
+    a.barrier()
+
+
+def test_one_sided_put():
+    run_ranks(_put, 2)

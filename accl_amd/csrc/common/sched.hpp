@@ -1892,6 +1892,34 @@ struct Cclo {
       case CfgFunc::set_max_rendezvous_size:
         max_rndzv_bytes = desc_count(d);
         return E_OK;
+      case CfgFunc::dump_state: {
+        // summary at dbg_off+4096 (flow dump uses [0,~3.2K), handshake
+        // probes use the last 2 KB)
+        volatile u64* w =
+            (volatile u64*)(tv.arena[me()] + tv.hdr(me())->dbg_off + 4096);
+        w[1] = nparked;
+        u32 k = 2;
+        for (u32 i = 0; i < MAX_INFLIGHT && k < 2 + 2 * MAX_INFLIGHT; ++i) {
+          const ParkedCall& p = cold->parked[i];
+          if (!p.used) continue;
+          w[k++] = u64(p.d.scenario) | (u64(p.d.root_src_dst) << 8) |
+                   (u64(p.d.tag) << 16) | (u64(p.ps.step) << 48);
+          w[k++] = p.ring_idx;
+        }
+        u64 pa_n = 0, pd_n = 0, uq_n = 0;
+        for (u32 r = 0; r < cfg.nranks; ++r) {
+          for (u32 q = 0; q < RNDZV_PEND; ++q) {
+            if (cold->pa[r][q].valid) pa_n++;
+            if (cold->pd[r][q].valid) pd_n++;
+          }
+          uq_n += cold->uq_t[r] - cold->uq_h[r];
+        }
+        w[66] = pa_n | (pd_n << 16) | (uq_n << 32);
+        w[67] = spill_busy;
+        fence_release_sys();
+        st_sys((volatile u64*)&w[0], ld_sys((const volatile u64*)&w[0]) + 1);
+        return E_OK;
+      }
       case CfgFunc::set_tuning:
         // tuning registers (reference configure_tuning_parameters):
         // knob id in root_src_dst, value in count

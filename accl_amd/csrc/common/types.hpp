@@ -61,6 +61,9 @@ enum class CfgFunc : u32 {
   set_max_eager_size = 3,
   set_max_rendezvous_size = 4,
   set_tuning = 5,
+  dump_state = 6,   // write a parked/pending/spill summary into the dbg
+                    // region (observability; no reference analogue beyond
+                    // the exchange-memory dumps, accl.cpp:964-1048)
 };
 
 enum class ReduceFunction : u32 { SUM = 0, MAX = 1 };

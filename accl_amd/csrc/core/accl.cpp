@@ -420,7 +420,39 @@ std::string ACCL::dump_engine_status() {
                 (unsigned long long)v.retired,
                 (unsigned long long)v.heartbeat,
                 (unsigned long long)v.ncomms);
-  return buf;
+  std::string out = buf;
+  // live engine-internal state (parked/pending/spill): ask the engine to
+  // snapshot into the dbg region, then decode
+  if (v.engine_up) {
+    CallDesc d = make_desc(Op::config, 0, DataType::none, DataType::none);
+    d.function = u32(CfgFunc::dump_state);
+    be_->call(d);
+    ArenaLayout L = arena_layout(be_->cfg());
+    std::vector<u64> w(68);
+    be_->read_arena(L.dbg_off + 4096, w.data(), w.size() * sizeof(u64));
+    std::snprintf(buf, sizeof(buf),
+                  "parked=%llu pending_addr=%llu pending_done=%llu "
+                  "unexpected=%llu spill_busy=0x%llx\n",
+                  (unsigned long long)w[1],
+                  (unsigned long long)(w[66] & 0xFFFF),
+                  (unsigned long long)((w[66] >> 16) & 0xFFFF),
+                  (unsigned long long)(w[66] >> 32),
+                  (unsigned long long)w[67]);
+    out += buf;
+    for (u32 k = 2; k < 2 + 2 * 32 && w[1]; k += 2) {
+      if (!w[k]) break;
+      std::snprintf(buf, sizeof(buf),
+                    "  parked[%u]: op=%llu peer=%llu tag=0x%llx step=%llu "
+                    "ring_idx=%llu\n",
+                    (k - 2) / 2, (unsigned long long)(w[k] & 0xFF),
+                    (unsigned long long)((w[k] >> 8) & 0xFF),
+                    (unsigned long long)((w[k] >> 16) & 0xFFFFFFFF),
+                    (unsigned long long)(w[k] >> 48),
+                    (unsigned long long)w[k + 1]);
+      out += buf;
+    }
+  }
+  return out;
 }
 
 std::string ACCL::dump_eager_rx_buffers(bool verbose) {

@@ -499,3 +499,21 @@ def test_dlpack_tensor_view(acc1):
     d = a.create_buffer(cnt, DT.float32)
     a.copy(b, d, cnt, from_device=True)
     assert np.allclose(rd(d, cnt), 3.0)
+
+
+def _put2(a, rank, n):
+    """One-sided xGMI put (reference copy_p2p) on the GPU engine."""
+    cnt = 30_000
+    d = a.create_buffer(cnt, DT.float32)  # same alloc order -> same offset
+    if rank == 0:
+        s = a.create_buffer(cnt, DT.float32)
+        s.write(pattern(cnt, 42, seed=6))
+        a.put(s, cnt, 1, d.arena_offset)
+    a.barrier()
+    if rank == 1:
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 42, seed=6))
+    a.barrier()
+
+
+def test_two_ranks_one_sided_put():
+    run_ranks(_put2, 2, backend="gpu", timeout=180)

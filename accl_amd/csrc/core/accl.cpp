@@ -424,9 +424,16 @@ std::string ACCL::dump_engine_status() {
   // live engine-internal state (parked/pending/spill): ask the engine to
   // snapshot into the dbg region, then decode
   if (v.engine_up) {
+    // bounded: a WEDGED engine (the main diagnostic case) must not make
+    // the dump itself block — fall back to the basic line after 2 s
     CallDesc d = make_desc(Op::config, 0, DataType::none, DataType::none);
     d.function = u32(CfgFunc::dump_state);
-    be_->call(d);
+    try {
+      u64 seq = be_->submit(d);
+      be_->wait(seq, nullptr, 2000);
+    } catch (const accl_error&) {
+      return out + "(engine busy/wedged: internal-state snapshot skipped)\n";
+    }
     ArenaLayout L = arena_layout(be_->cfg());
     std::vector<u64> w(68);
     be_->read_arena(L.dbg_off + 4096, w.data(), w.size() * sizeof(u64));

@@ -11,6 +11,11 @@ import numpy as np
 import accl_amd as A
 import accl_amd._core as core
 
+if not A.ACCL._has_gpu():
+    print("device_stream demo needs an MI355X (device-initiated push runs "
+          "inside a HIP kernel) — skipping")
+    sys.exit(0)
+
 a = A.ACCL(nranks=1, rank=0, backend="gpu")
 try:
     n = 10000

@@ -1467,3 +1467,80 @@ def _put(a, rank, n):
 
 def test_one_sided_put():
     run_ranks(_put, 2)
+
+
+def _fuzz2(a, rank, n, seed):
+    """Adversarial generation-2 fuzz: pairwise bursts where the receiver
+    posts its (async) recvs in a PERMUTED order vs the send order — the
+    parking + OOO-matching stressor — mixed with subcomm and compressed
+    collectives."""
+    rng = np.random.default_rng(seed)
+    sub_members = [0, 1]
+    sub = a.split_communicator(sub_members) if rank in sub_members and n > 2 \
+        else None
+    for i in range(24):
+        kind = int(rng.integers(0, 4))
+        if kind == 0:
+            # pairwise burst with permuted posting order on the receiver
+            pair = rng.choice(n, size=2, replace=False)
+            src_r, dst_r = int(pair[0]), int(pair[1])
+            k = int(rng.integers(1, 5))
+            cnts = [int(rng.integers(16, 20_000)) for _ in range(k)]
+            tags = list(rng.choice(np.arange(1, 1000), size=k, replace=False))
+            perm = list(rng.permutation(k))
+            if rank == src_r:
+                reqs = []
+                for j in range(k):
+                    s = _mk(a, cnts[j])
+                    s.write(pattern(cnts[j], 1000 * i + j, seed=seed))
+                    reqs.append((a.send(s, cnts[j], dst=dst_r, tag=int(tags[j]),
+                                        run_async=True), s))
+                for r, _ in reqs:
+                    assert r.wait() == 0, f"burst{i} send"
+            elif rank == dst_r:
+                reqs = []
+                for j in perm:  # permuted posting order
+                    d = _mk(a, cnts[j])
+                    reqs.append((a.recv(d, cnts[j], src=src_r, tag=int(tags[j]),
+                                        run_async=True), d, j))
+                for r, d, j in reqs:
+                    assert r.wait() == 0, f"burst{i} recv{j}"
+                    assert np.array_equal(rd(d, cnts[j]),
+                                          pattern(cnts[j], 1000 * i + j,
+                                                  seed=seed)), \
+                        f"burst{i} msg{j} data"
+        elif kind == 1:
+            cnt = int(rng.integers(64, 10_000))
+            s, d = _mk(a, cnt), _mk(a, cnt)
+            s.write(pattern(cnt, rank, seed=i) / 16)
+            a.allreduce(s, d, cnt, RF.SUM, compress_dtype=DT.float16)
+            exp = np.stack([pattern(cnt, r, seed=i) / 16
+                            for r in range(n)]).sum(0)
+            assert np.allclose(rd(d, cnt), exp, atol=0.05), f"op{i} comp-ar"
+        elif kind == 2:
+            # draw BEFORE the participation gate: every rank must consume
+            # the same RNG stream or the scripts diverge
+            cnt = int(rng.integers(64, 8_000))
+            if sub is not None:
+                s, d = _mk(a, cnt), _mk(a, cnt)
+                s.write(pattern(cnt, rank, seed=i))
+                a.allreduce(s, d, cnt, RF.SUM, comm=sub)
+                exp = np.stack([pattern(cnt, r, seed=i)
+                                for r in sub_members]).sum(0)
+                assert np.allclose(rd(d, cnt), exp), f"op{i} sub-ar"
+        else:
+            cnt = int(rng.integers(64, 12_000))
+            s, d = _mk(a, cnt), _mk(a, cnt * n)
+            s.write(pattern(cnt, rank, seed=i))
+            a.allgather(s, d, cnt)
+            exp = np.concatenate([pattern(cnt, r, seed=i) for r in range(n)])
+            assert np.array_equal(rd(d, cnt * n), exp), f"op{i} ag"
+    a.barrier()
+
+
+@pytest.mark.parametrize("seed", [13, 29])
+def test_protocol_fuzz2(seed):
+    run_ranks(lambda a, r, n: _fuzz2(a, r, n, seed), 3, opts=SMALL,
+              timeout=240)
+    run_ranks(lambda a, r, n: _fuzz2(a, r, n, seed), 2, opts=DIRECT,
+              timeout=240)

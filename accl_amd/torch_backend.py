@@ -240,7 +240,10 @@ class AcclProcessGroup(dist.ProcessGroup):
     def alltoall_base(self, output, input, out_sizes, in_sizes, opts=None):
         if (out_sizes and len(set(out_sizes)) > 1) or \
                 (in_sizes and len(set(in_sizes)) > 1):
-            raise NotImplementedError("accl backend: uneven alltoall")
+            # uneven splits (MoE-style alltoallv): decompose into tagged
+            # async send/recv pairs — the engine's parking + out-of-order
+            # matching make the concurrent posting safe
+            return self._alltoallv(output, input, out_sizes, in_sizes)
         per = input.numel() // self.size()
         if self._rawable(output, input):
             self._raw(_core.Op.alltoall, per, input, output)
@@ -250,6 +253,74 @@ class AcclProcessGroup(dist.ProcessGroup):
         self._a.alltoall(s, d, per, from_device=True, to_device=True)
         self._download(output, d, dv)
         return _Work()
+
+    def _alltoallv(self, output, input, out_sizes, in_sizes):
+        P, me = self.size(), self.rank()
+        in_off = [0]
+        for v in in_sizes:
+            in_off.append(in_off[-1] + int(v))
+        out_off = [0]
+        for v in out_sizes:
+            out_off.append(out_off[-1] + int(v))
+        iflat, oflat = input.reshape(-1), output.reshape(-1)
+        # local block
+        oflat[out_off[me]:out_off[me + 1]].copy_(
+            iflat[in_off[me]:in_off[me + 1]])
+        reqs, keep = [], []
+        base_tag = 0x2A2A0000  # private tag block, peer-disambiguated
+        for p in range(P):
+            if p == me:
+                continue
+            n_in = int(in_sizes[p])
+            if n_in:
+                sb, _ = self._buf(n_in, input.dtype, f"a2av_s{p}")
+                self._upload_slice(iflat[in_off[p]:in_off[p] + n_in], sb)
+                reqs.append((self._a.send(sb, n_in, dst=p,
+                                          tag=base_tag + me,
+                                          from_device=self._gpu,
+                                          run_async=True), None, None, sb))
+            n_out = int(out_sizes[p])
+            if n_out:
+                db, dv = self._buf(n_out, output.dtype, f"a2av_r{p}")
+                reqs.append((self._a.recv(db, n_out, src=p,
+                                          tag=base_tag + p,
+                                          to_device=self._gpu,
+                                          run_async=True),
+                             (oflat, out_off[p], n_out, db, dv), None, db))
+        for r, dl, _, buf in reqs:
+            e = r.wait()
+            if e:
+                raise RuntimeError(
+                    f"alltoallv failed: {_core.error_to_string(e)}")
+            if dl is not None:
+                oflat, off, n_out, db, dv = dl
+                self._download_slice(oflat[off:off + n_out], db, dv)
+        return _Work()
+
+    def _upload_slice(self, flat, b):
+        if self._gpu:
+            v = self._a.tensor(b)
+            v[:flat.numel()].copy_(flat)
+            torch.cuda.current_stream().synchronize()
+        else:
+            arr = flat.detach().numpy() if flat.dtype != torch.bfloat16 \
+                else flat.detach().view(torch.int16).numpy()
+            b.write(np.ascontiguousarray(arr).view(np.int8))
+
+    def _download_slice(self, flat, b, view=None):
+        n = flat.numel()
+        if self._gpu:
+            v = view if view is not None else self._a.tensor(b)
+            flat.copy_(v[:n])
+            torch.cuda.current_stream().synchronize()
+        else:
+            raw = np.zeros(n * flat.element_size(), np.int8)
+            b.read(raw)
+            src = torch.from_numpy(raw).view(
+                torch.int16 if flat.dtype == torch.bfloat16 else flat.dtype)
+            if flat.dtype == torch.bfloat16:
+                src = src.view(torch.bfloat16)
+            flat.copy_(src[:n])
 
     def send(self, tensors, dst, tag):
         for t in tensors:

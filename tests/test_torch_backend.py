@@ -49,6 +49,21 @@ def _rank_main(rank, world, port, q):
 
         dist.barrier()
 
+        # uneven all_to_all_single (MoE-style alltoallv): rank r sends
+        # (p+1)*10*(r+1) elems to p; receives (r+1)*10*(p+1) from p
+        in_sizes = [(p + 1) * 10 * (rank + 1) for p in range(world)]
+        out_sizes = [(rank + 1) * 10 * (p + 1) for p in range(world)]
+        a2a_in = torch.cat([
+            torch.full((in_sizes[p],), float(100 * rank + p))
+            for p in range(world)])
+        a2a_out = torch.zeros(sum(out_sizes))
+        dist.all_to_all_single(a2a_out, a2a_in, out_sizes, in_sizes)
+        off = 0
+        for p in range(world):
+            seg = a2a_out[off:off + out_sizes[p]]
+            assert seg.eq(float(100 * p + rank)).all(), (rank, p, seg[:3])
+            off += out_sizes[p]
+
         # one real DDP training step over the backend
         from torch.nn.parallel import DistributedDataParallel as DDP
         torch.manual_seed(7)

@@ -1302,6 +1302,129 @@ struct Cclo {
     return E_OK;
   }
 
+  // ---- one-shot small-message fan-in (reference: flat-tree reduce for
+  // small messages, ccl_offload_control.c:1531-1602). Each contributor's
+  // whole message fits ONE eager slot; the consumer waits for all P-1 slot
+  // headers (pool-aware) and folds slot payloads + own src into dst with a
+  // single n-ary mover pass — one network round, no dst round-trips.
+  // Returns E_OK and fills extras/meta, or E_NOT_READY-free blocking wait.
+  struct SlotRef { const char* pay; u64 seq; u32 from_pool; u32 pool_qi; };
+  ACCL_HD bool collect_one_slot(u32 gpeer, u32 tag, DataType wdt, u64 n,
+                                SlotRef& out) {
+    const u32 wsz = dtype_size(wdt);
+    u64 deadline = deadline_now();
+    for (;;) {
+      // pool first (a parked recv's probe may have spilled it)
+      for (u32 qi = cold->uq_h[gpeer]; qi != cold->uq_t[gpeer]; ++qi) {
+        Unexpected& u = cold->uq[gpeer][qi % UQ_DEPTH];
+        if (u.bytes == 0 || u.tag != tag) continue;
+        if (u.arith != u32(wdt) || u.bytes / wsz != n) {
+          err |= E_SEGMENT;
+          return false;
+        }
+        out = SlotRef{spill_ptr(u.spare_slot), 0, 1, qi};
+        return true;
+      }
+      u64 seq = sq.eager_rx[gpeer] + 1;
+      u32 sl = u32((seq - 1) % cfg.n_slots);
+      SlotHdr* h = tv.slot_hdr(me(), gpeer, sl);
+      if (ld_sys(&h->seq) == seq) {
+        fence_acquire_sys();
+        if (h->tag == tag) {
+          if (h->arith != u32(wdt) || u64(h->bytes) / wsz != n) {
+            err |= E_SEGMENT;
+            return false;
+          }
+          out = SlotRef{tv.slot_payload(me(), gpeer, sl), seq, 0, 0};
+          return true;
+        }
+        if (spill_head(gpeer, h, sl, seq)) { deadline = deadline_now(); continue; }
+        if (err) return false;
+      }
+      if (!wait_pred_tick(deadline)) return false;
+    }
+  }
+  ACCL_HD void release_slot(u32 gpeer, const SlotRef& s) {
+    if (s.from_pool) {
+      Unexpected& u = cold->uq[gpeer][s.pool_qi % UQ_DEPTH];
+      spill_busy &= ~(1ull << u.spare_slot);
+      u.bytes = 0;
+      while (cold->uq_h[gpeer] != cold->uq_t[gpeer] &&
+             cold->uq[gpeer][cold->uq_h[gpeer] % UQ_DEPTH].bytes == 0)
+        cold->uq_h[gpeer]++;
+    } else {
+      sq.eager_rx[gpeer] = s.seq;
+      sq.credit_ret[gpeer] = s.seq;
+      st_sys(&tv.chan_ctl(gpeer, me())->credit, s.seq);
+    }
+  }
+
+  // small allreduce/reduce eligibility: whole message in one slot, no
+  // compression, fan-in within the n-ary move budget
+  ACCL_HD bool one_shot_ok(u64 n, DataType dt, DataType wdt, u32 P) const {
+    return dt == wdt && P >= 2 && P <= MOVE_MAX_SRC &&
+           n * dtype_size(dt) <= (64u << 10) &&
+           n * dtype_size(dt) <= cfg.slot_bytes;
+  }
+
+  // one-shot allreduce: everyone broadcasts its vector; every rank folds
+  // P-1 slot payloads + own src into dst with one n-ary pass
+  ACCL_HD u32 allreduce_one_shot(const CallDesc& d, const CommView& c,
+                                 u32 tag) {
+    u64 n = desc_count(d);
+    DataType dt = desc_dtype(d);
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    const u32 P = c.size, r = c.rank;
+    u32 nf = 0;
+    for (u32 p = 0; p < P; ++p)
+      if (p != r) mk_tx(nf++, c.global(p), src, dt, dt, n, tag);
+    u32 e = run_flows(nf);
+    if (e) return e;
+    SlotRef refs[MOVE_MAX_SRC];
+    const char* extras[MOVE_MAX_SRC];
+    u32 nx = 0;
+    for (u32 p = 0; p < P; ++p) {
+      if (p == r) continue;
+      if (!collect_one_slot(c.global(p), tag, dt, n, refs[nx])) return err;
+      extras[nx] = refs[nx].pay;
+      nx++;
+    }
+    e = nary_reduce(dst, src, dt, n, extras, nx, int(d.function));
+    u32 k = 0;
+    for (u32 p = 0; p < P; ++p)
+      if (p != r) release_slot(c.global(p), refs[k++]);
+    return e;
+  }
+
+  // one-shot reduce at root: leaves send one slot each; root folds them
+  ACCL_HD u32 reduce_one_shot(const CallDesc& d, const CommView& c, u32 root,
+                              u32 tag) {
+    u64 n = desc_count(d);
+    DataType dt = desc_dtype(d);
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    const u32 P = c.size;
+    if (c.rank != root) {
+      mk_tx(0, c.global(root), src, dt, dt, n, tag);
+      return run_flows(1);
+    }
+    char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    SlotRef refs[MOVE_MAX_SRC];
+    const char* extras[MOVE_MAX_SRC];
+    u32 nx = 0;
+    for (u32 p = 0; p < P; ++p) {
+      if (p == root) continue;
+      if (!collect_one_slot(c.global(p), tag, dt, n, refs[nx])) return err;
+      extras[nx] = refs[nx].pay;
+      nx++;
+    }
+    u32 e = nary_reduce(dst, src, dt, n, extras, nx, int(d.function));
+    u32 k = 0;
+    for (u32 p = 0; p < P; ++p)
+      if (p != root) release_slot(c.global(p), refs[k++]);
+    return e;
+  }
+
   // reduce at root: fan-in with a serialized reduce chain per segment
   // (reference: reduce, ccl_offload_control.c:1507-1744)
   ACCL_HD u32 op_reduce(const CallDesc& d, const CommView& c) {
@@ -1314,6 +1437,8 @@ struct Cclo {
       mk_local(0, src, dt, local_ptr(d.addr2, d.flags & F_DST_ARENA), dt, n);
       return run_flows(1);
     }
+    if (one_shot_ok(n, dt, wdt, c.size))
+      return reduce_one_shot(d, c, root, tag);
     if (coll_direct_ok(d, n, dt, wdt)) {
       if (c.rank != root) return reduce_direct_leaf(c, root, src, dt, n, tag);
       return reduce_direct_root(d, c, tag);
@@ -1417,6 +1542,10 @@ struct Cclo {
       mk_local(0, src, dt, dst, dt, total);
       return run_flows(1);
     }
+    // Small-message one-shot path: single round + single n-ary fold —
+    // halves latency vs RS+AG on the latency-bound end of the curve
+    if (one_shot_ok(total, dt, wdt, P))
+      return allreduce_one_shot(d, c, tag);
     // Large-message direct path: compose direct reduce_scatter (windowed
     // n-ary fan-in over xGMI stages) + direct allgather (single peer
     // writes) — no eager slot staging on either phase. (reference shape:

@@ -83,7 +83,10 @@ class ACCL:
 
     def __init__(self, nranks=None, rank=None, backend="auto", job=None,
                  device=None, heap_bytes=None, bootstrap="auto", ranks=None,
-                 **opts):
+                 opts=None, **extra):
+        # accept both ACCL(opts={...}) and ACCL(slot_bytes=..., ...)
+        opts = dict(opts or {})
+        opts.update(extra)
         if ranks is not None:
             # reference-style rank map (generate_ranks): world size + eager
             # slot size come from it (reference: ACCL ctor takes the rank

@@ -148,6 +148,7 @@ struct Cclo {
   // probe plumbing (set around run_call_inner)
   u32 probe_;
   ParkState* ps_;
+  u32 in_drain_;  // re-entrancy guard for drain_for_parked()
 
   ACCL_HD u32 me() const { return cfg.rank; }
 
@@ -180,6 +181,13 @@ struct Cclo {
     // slot header invisible for the whole 10 s deadline).
     __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
 #endif
+    // Flow-free progress for PARKED calls while this op spins: without it,
+    // two engines deadlock when a committed multi-segment send waits for
+    // credit that only a parked recv (stuck behind this very wait on the
+    // peer) would return — the reference's rxbuf-offload engines drain the
+    // wire independently of the consumer for exactly this reason
+    // (kernels/cclo/hls/rxbuf_offload/*).
+    drain_for_parked();
     cpu_pause();
     if (wallclock() > deadline) { err |= E_TIMEOUT; return false; }
     return true;
@@ -200,6 +208,183 @@ struct Cclo {
   // s's arena (chan_ctl(s, r)->credit), advanced by r.
   ACCL_HD u64 tx_credit(u32 peer) {  // slots consumed by peer (cumulative)
     return ld_sys(&tv.chan_ctl(me(), peer)->credit);
+  }
+
+  // Flow-free single-segment eager send (direct mover submit + header
+  // publish): the parked-send resume/drain path — safe to run from INSIDE
+  // another op's wait (never touches the flow table). Returns elements
+  // sent (0 = no credit / would block).
+  ACCL_HD u64 eager_send_segment(u32 peer, const char* src, DataType sdt,
+                                 DataType wdt, u64 off, u64 remaining,
+                                 u64 total, u32 tag) {
+    u64 next = sq.eager_tx[peer];
+    if (next - tx_credit(peer) >= cfg.n_slots) return 0;  // no credit
+    const u64 seg_cap = u64(cfg.slot_bytes) / dtype_size(wdt);
+    u64 n = min64(remaining, seg_cap);
+    u32 slot = u32(next % cfg.n_slots);
+    MoveDesc m{};
+    m.dst = (u64)tv.slot_payload(peer, me(), slot);
+    m.dst_dt = u8(wdt);
+    m.src[0] = (u64)(src + off * dtype_size(sdt));
+    m.src_dt[0] = u8(sdt);
+    m.nsrc = 1;
+    m.count = n;
+    u32 tok = mv->submit(m);
+    u64 deadline = deadline_now();
+    while (!mv->poll(tok))
+      if (!wait_pred_tick_nodrain(deadline)) return 0;
+    SlotHdr* h = tv.slot_hdr(peer, me(), slot);
+    h->tag = tag;
+    h->bytes = u32(n * dtype_size(wdt));
+    h->msg_count = total;
+    h->arith = u32(wdt);
+    h->flags = (off == 0 ? SEG_FIRST : 0) |
+               (off + n >= total ? SEG_LAST : 0);
+    fence_release_sys();
+    st_sys(&h->seq, next + 1);
+    sq.eager_tx[peer] = next + 1;
+    return n;
+  }
+
+  // flow-free synchronous mover copy (safe from inside another op's wait:
+  // never touches the flow table)
+  ACCL_HD bool copy_free(const void* src, DataType sdt, char* dstp,
+                         DataType ddt, u64 count) {
+    MoveDesc m{};
+    m.dst = (u64)dstp;
+    m.dst_dt = u8(ddt);
+    m.src[0] = (u64)src;
+    m.src_dt[0] = u8(sdt);
+    m.nsrc = 1;
+    m.count = count;
+    u32 tok = mv->submit(m);
+    u64 deadline = deadline_now();
+    while (!mv->poll(tok))
+      if (!wait_pred_tick_nodrain(deadline)) return false;
+    return true;
+  }
+
+  // wait tick WITHOUT the parked-drain hook (used inside the drain itself
+  // and other non-reentrant spots)
+  ACCL_HD bool wait_pred_tick_nodrain(u64& deadline) {
+#if defined(__HIP_DEVICE_COMPILE__)
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+#endif
+    cpu_pause();
+    if (wallclock() > deadline) { err |= E_TIMEOUT; return false; }
+    return true;
+  }
+
+  // Bounded flow-free progress for parked calls (called from in-op waits):
+  //  - parked eager recv: spill ONE arrived head segment to the unexpected
+  //    pool (returns the sender's credit — the deadlock breaker)
+  //  - parked eager send: push ONE segment when credit allows (progress
+  //    recorded in its ParkState; the retry resumes from there)
+  ACCL_HD void drain_for_parked() {
+    if (!nparked || in_drain_) return;
+    in_drain_ = 1;
+    for (u32 i = 0; i < MAX_INFLIGHT; ++i) {
+      ParkedCall& p = cold->parked[i];
+      if (!p.used) continue;
+      // per-(pair,tag) FIFO: skip if an older parked call shares the key
+      bool blocked = false;
+      for (u32 j = 0; j < MAX_INFLIGHT; ++j) {
+        const ParkedCall& q = cold->parked[j];
+        if (!q.used || j == i) continue;
+        if (q.ring_idx < p.ring_idx && q.d.scenario == p.d.scenario &&
+            q.d.root_src_dst == p.d.root_src_dst &&
+            q.d.comm_id == p.d.comm_id && q.d.tag == p.d.tag) {
+          blocked = true;
+          break;
+        }
+      }
+      if (blocked) continue;
+      if (p.d.comm_id >= ncomms) continue;
+      const CommView& c = comms[p.d.comm_id];
+      u32 peer = c.global(p.d.root_src_dst);
+      if (peer == me()) continue;
+      Op op = Op(p.d.scenario);
+      DataType dt = desc_dtype(p.d), wdt = desc_wire_dtype(p.d);
+      u64 n = desc_count(p.d);
+      if (op == Op::recv && !use_rndzv(n, dt, wdt)) {
+        // deliver matching segments straight into the parked recv's dst
+        // (progress in its ParkState — op_recv_eager's step-2 encoding)
+        // and spill mismatched heads. Direct delivery matters: the spill
+        // pool is BOUNDED (UQ_DEPTH), so a recv that only spills livelocks
+        // once the pool fills with its own message.
+        char* dst = local_ptr(p.d.addr2, p.d.flags & F_DST_ARENA);
+        u64 got = (p.ps.step == 2) ? p.ps.w[0] : 0;
+        u32 mtag = (p.ps.step == 2) ? u32(p.ps.w[1]) : p.d.tag;
+        const u32 wsz = dtype_size(wdt), dsz = dtype_size(dt);
+        bool prog = true;
+        while (prog && got < n) {
+          prog = false;
+          for (u32 qi = cold->uq_h[peer]; qi != cold->uq_t[peer]; ++qi) {
+            Unexpected& u = cold->uq[peer][qi % UQ_DEPTH];
+            if (u.bytes == 0) continue;
+            if (mtag != TAG_ANY && u.tag != mtag) continue;
+            if (got > 0 && u.tag != mtag) continue;
+            if (u.arith != u32(wdt)) break;      // leave for retry to error
+            u64 nseg = u.bytes / wsz;
+            if (nseg > n - got) break;           // leave for retry to error
+            if (got == 0) mtag = u.tag;
+            if (!copy_free(spill_ptr(u.spare_slot), wdt, dst + got * dsz,
+                           dt, nseg))
+              break;
+            spill_busy &= ~(1ull << u.spare_slot);
+            u.bytes = 0;
+            while (cold->uq_h[peer] != cold->uq_t[peer] &&
+                   cold->uq[peer][cold->uq_h[peer] % UQ_DEPTH].bytes == 0)
+              cold->uq_h[peer]++;
+            got += nseg;
+            prog = true;
+            break;
+          }
+          if (prog) continue;
+          u64 seq = sq.eager_rx[peer] + 1;
+          u32 sl = u32((seq - 1) % cfg.n_slots);
+          SlotHdr* h = tv.slot_hdr(me(), peer, sl);
+          if (ld_sys(&h->seq) != seq) break;
+          fence_acquire_sys();
+          bool match = (mtag == TAG_ANY) || (h->tag == mtag);
+          u64 nseg = u64(h->bytes) / wsz;
+          if (match && h->arith == u32(wdt) && nseg <= n - got) {
+            if (got == 0) mtag = h->tag;
+            if (!copy_free(tv.slot_payload(me(), peer, sl), wdt,
+                           dst + got * dsz, dt, nseg))
+              break;
+            sq.eager_rx[peer] = seq;
+            sq.credit_ret[peer] = seq;
+            st_sys(&tv.chan_ctl(peer, me())->credit, seq);
+            got += nseg;
+            prog = true;
+          } else if (!match) {
+            prog = spill_head(peer, h, sl, seq);
+          }  // mismatched arith / overlong segment: leave for the retry
+        }
+        p.ps.w[0] = got;
+        p.ps.w[1] = mtag;
+        p.ps.step = 2;
+      } else if (op == Op::send && !use_rndzv(n, dt, wdt)) {
+        u64 sent = (p.ps.step == 1) ? p.ps.w[0] : 0;
+        if (sent < n) {
+          const char* src =
+              local_ptr(p.d.addr0, p.d.flags & F_SRC_ARENA);
+          u64 did = eager_send_segment(peer, src, dt, wdt, sent, n - sent, n,
+                                       p.d.tag);
+          if (did) {
+            p.ps.w[0] = sent + did;
+            p.ps.step = 1;
+          }
+        }
+      } else if (op == Op::send) {
+        // parked rendezvous send: push whatever windows the receiver has
+        // posted (the receiver may be committed-blocking on them while OUR
+        // engine is blocked in an unrelated op — same cycle as eager)
+        (void)rndzv_send_push(peer, p.d, p.ps);
+      }
+    }
+    in_drain_ = 0;
   }
 
   // ---------------- flow stepping ----------------
@@ -598,23 +783,36 @@ struct Cclo {
       if (!wait_pred_tick(deadline)) return false;
     }
   }
-  // non-consuming probe: is a TAG-MATCHING addr record available right now?
-  // (a foreign-tag head record must NOT make a parked send commit — it
-  // would block the engine inside wait_addr while the matching send for
-  // that record sits queued behind it)
+  // probe: is a TAG-MATCHING addr record available right now? Foreign-tag
+  // head records are SPILLED into the pending set while probing (their
+  // owners match them from there), so a record queued BEHIND a foreign
+  // head is still discoverable — without this, a parked send whose sync
+  // caller blocks the next submission can deadlock (its record second in
+  // the ring, nothing ever consuming the first). A matching head must NOT
+  // make the probe commit blindly either: committing on a foreign head
+  // would block the engine inside wait_addr.
   ACCL_HD bool addr_ready(u32 gpeer, u32 want_tag) {
-    for (u32 k = 0; k < RNDZV_PEND; ++k) {
-      PendRndzv& p = cold->pa[gpeer][k];
-      if (p.valid && (want_tag == TAG_ANY || p.tag == want_tag ||
-                      p.tag == TAG_ANY))
-        return true;
+    for (;;) {
+      for (u32 k = 0; k < RNDZV_PEND; ++k) {
+        PendRndzv& p = cold->pa[gpeer][k];
+        if (p.valid && (want_tag == TAG_ANY || p.tag == want_tag ||
+                        p.tag == TAG_ANY))
+          return true;
+      }
+      u64 seq = sq.rndzv_addr_rx[gpeer] + 1;
+      RndzvRec* r = tv.rndzv_addr(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
+      if (ld_sys(&r->seq) != seq) return false;
+      fence_acquire_sys();
+      u32 t = r->tag;
+      if (want_tag == TAG_ANY || t == want_tag || t == TAG_ANY) return true;
+      // spill the foreign head to pending and keep looking
+      u32 k = 0;
+      while (k < RNDZV_PEND && cold->pa[gpeer][k].valid) ++k;
+      if (k >= RNDZV_PEND) return false;  // pending full: try again later
+      cold->pa[gpeer][k] =
+          PendRndzv{seq, r->offset, r->count, r->tag, r->arith, 1};
+      sq.rndzv_addr_rx[gpeer] = seq;
     }
-    u64 seq = sq.rndzv_addr_rx[gpeer] + 1;
-    RndzvRec* r = tv.rndzv_addr(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
-    if (ld_sys(&r->seq) != seq) return false;
-    fence_acquire_sys();
-    u32 t = r->tag;
-    return want_tag == TAG_ANY || t == want_tag || t == TAG_ANY;
   }
   ACCL_HD void post_done(u32 gpeer, u32 tag) {
     u64 seq = ++sq.rndzv_done_tx[gpeer];
@@ -706,29 +904,86 @@ struct Cclo {
     }
     if (use_rndzv(n, dt, wdt)) {
       // follow the receiver's posted windows (tag-matched, out of order
-      // w.r.t. other rendezvous ops on this pair)
-      if (probe_ && !addr_ready(peer, d.tag)) return E_NOT_READY;
-      u64 sent = 0;
-      while (sent < n) {
-        RndzvRec rec{};
-        if (!wait_addr(peer, d.tag, rec)) return err;
-        u64 w = min64(n - sent, rec.count);
-        mk_tx_direct(0, peer, src + sent * dtype_size(dt), dt,
-                     DataType(rec.arith ? rec.arith : u32(dt)), w, rec.offset,
-                     rec_slot(rec, cfg.n_rndzv));
-        u32 e = run_flows(1);
-        if (e) return e;
-        sent += w;
+      // w.r.t. other rendezvous ops on this pair). Flow-free resumable
+      // pushes in BOTH modes: a probe parks when no window is posted
+      // (committing to an in-op wait risks the cross-rank deadlock); the
+      // blocking fallback (park table full) resumes from the same
+      // ParkState — it must never resend windows a probe already pushed.
+      if (probe_) {
+        if (!rndzv_send_push(peer, d, *ps_)) return E_NOT_READY;
+        post_done(peer, d.tag);
+        return E_OK;
+      }
+      ParkState local{};
+      ParkState& ps = ps_ ? *ps_ : local;
+      u64 deadline = deadline_now();
+      while (!rndzv_send_push(peer, d, ps)) {
+        if (err) return err;
+        if (!wait_pred_tick(deadline)) return err;
       }
       post_done(peer, d.tag);
       return E_OK;
     }
-    // eager: park when the credit window is exhausted and nothing was sent
-    // (receiver absent or far behind)
-    if (probe_ && sq.eager_tx[peer] - tx_credit(peer) >= cfg.n_slots)
-      return E_NOT_READY;
-    mk_tx(0, peer, src, dt, wdt, n, d.tag);
-    return run_flows(1);
+    // eager: flow-free per-segment pushes (<= max_eager, so at most a
+    // couple of slot-sized segments on production geometry); probe mode
+    // parks on no-credit with progress saved, blocking mode waits
+    u64 sent = (ps_ && ps_->step == 1) ? ps_->w[0] : 0;
+    u64 deadline = deadline_now();
+    while (sent < n) {
+      u64 did = eager_send_segment(peer, src, dt, wdt, sent, n - sent, n,
+                                   d.tag);
+      if (err) return err;
+      if (!did) {
+        if (probe_) {
+          if (ps_) { ps_->w[0] = sent; ps_->step = 1; }
+          return E_NOT_READY;
+        }
+        if (!wait_pred_tick(deadline)) return err;
+        continue;
+      }
+      sent += did;
+      if (ps_) { ps_->w[0] = sent; ps_->step = 1; }
+      deadline = deadline_now();
+    }
+    return E_OK;
+  }
+
+  // flow-free rendezvous window pushes (parked-send resume/drain): true
+  // when the whole message has been written
+  ACCL_HD bool rndzv_send_push(u32 peer, const CallDesc& d, ParkState& ps) {
+    u64 n = desc_count(d);
+    DataType dt = desc_dtype(d);
+    const u32 esz = dtype_size(dt);
+    const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
+    u64 sent = (ps.step == 1) ? ps.w[0] : 0;
+    while (sent < n) {
+      if (!addr_ready(peer, d.tag)) {
+        ps.w[0] = sent;
+        ps.step = 1;
+        return false;
+      }
+      RndzvRec rec{};
+      if (!wait_addr(peer, d.tag, rec)) return false;  // immediate (ready)
+      u64 w = min64(n - sent, rec.count);
+      MoveDesc m{};
+      m.dst = (u64)(tv.arena[peer] + rec.offset);
+      m.dst_dt = u8(rec.arith ? rec.arith : u32(dt));
+      m.src[0] = (u64)(src + sent * esz);
+      m.src_dt[0] = u8(dt);
+      m.nsrc = 1;
+      m.count = w;
+      u32 tok = mv->submit(m);
+      u64 deadline = deadline_now();
+      while (!mv->poll(tok))
+        if (!wait_pred_tick_nodrain(deadline)) return false;
+      fence_release_sys();
+      st_sys(tv.direct_word(peer, me(), rec_slot(rec, cfg.n_rndzv)),
+             w * u64(dtype_size(DataType(m.dst_dt))));
+      sent += w;
+      ps.w[0] = sent;
+      ps.step = 1;
+    }
+    return true;
   }
 
   ACCL_HD u32 spill_slot_count() {
@@ -782,6 +1037,12 @@ struct Cclo {
                             u64 n, u32 want_tag) {
     u64 got = 0;
     u32 mtag = want_tag;
+    // resumable in probe mode: partial progress (and the matched tag, if
+    // TAG_ANY) survives a park — the retry continues where it left off
+    if (probe_ && ps_ && ps_->step == 2) {
+      got = ps_->w[0];
+      mtag = u32(ps_->w[1]);
+    }
     const u32 wsz = dtype_size(wdt), dsz = dtype_size(ddt);
     u64 deadline = deadline_now();
     while (got < n) {
@@ -837,8 +1098,13 @@ struct Cclo {
         if (spill_head(peer, hd, sl, seq)) { deadline = deadline_now(); continue; }
         if (err) return err;
       }
-      // park instead of spinning while nothing has matched yet
-      if (probe_ && got == 0) return E_NOT_READY;
+      // probe mode NEVER spins: park (with progress) whenever the channel
+      // has nothing for us right now — committing to an in-op wait here is
+      // the classic cross-rank deadlock (peer blocked behind its own op)
+      if (probe_) {
+        if (ps_) { ps_->w[0] = got; ps_->w[1] = mtag; ps_->step = 2; }
+        return E_NOT_READY;
+      }
       if (!wait_pred_tick(deadline)) return err;
     }
     return E_OK;
@@ -1066,11 +1332,17 @@ struct Cclo {
     bool direct = use_rndzv(n, dt, wdt) && (d.flags & F_DST_ARENA) &&
                   (d.flags & F_SRC_ARENA) && 2 * u64(c.size) - 1 <= MAX_FLOWS;
     if (direct) {
+      // to peer p I post the slot for P'S DATA in MY dst (addr2 + p*n) —
+      // the record's offset is where the CONSUMING sender writes in MY
+      // arena (bug history: posting my OWN block index here sent every
+      // peer's vector to the same region; it survived all small tests
+      // because an opts-plumbing bug kept max_eager at the 4 MiB default,
+      // so the direct branch never actually ran below 4 MiB)
       u32 myslot[MAX_RANKS];
       for (u32 p = 0; p < c.size; ++p)
         if (p != r)
           myslot[p] = post_addr(c.global(p),
-                                d.addr2 + u64(r) * n * dtype_size(dt), n, tag,
+                                d.addr2 + u64(p) * n * dtype_size(dt), n, tag,
                                 u32(dt));
       for (u32 p = 0; p < c.size; ++p) {
         if (p == r) continue;
@@ -1381,16 +1653,23 @@ struct Cclo {
       if (p != r) mk_tx(nf++, c.global(p), src, dt, dt, n, tag);
     u32 e = run_flows(nf);
     if (e) return e;
+    // fold in LOCAL-RANK order with own src in position r: every rank
+    // computes the identical fp reduction (replicas must stay bitwise in
+    // sync — DDP/c10d semantics)
     SlotRef refs[MOVE_MAX_SRC];
-    const char* extras[MOVE_MAX_SRC];
+    const char* contrib[MOVE_MAX_SRC];
     u32 nx = 0;
     for (u32 p = 0; p < P; ++p) {
-      if (p == r) continue;
+      if (p == r) {
+        contrib[p] = src;
+        continue;
+      }
       if (!collect_one_slot(c.global(p), tag, dt, n, refs[nx])) return err;
-      extras[nx] = refs[nx].pay;
+      contrib[p] = refs[nx].pay;
       nx++;
     }
-    e = nary_reduce(dst, src, dt, n, extras, nx, int(d.function));
+    e = nary_reduce(dst, contrib[0], dt, n, &contrib[1], P - 1,
+                    int(d.function));
     u32 k = 0;
     for (u32 p = 0; p < P; ++p)
       if (p != r) release_slot(c.global(p), refs[k++]);
@@ -1409,16 +1688,22 @@ struct Cclo {
       return run_flows(1);
     }
     char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
+    // local-rank fold order (matches allreduce_one_shot for reproducible
+    // numerics across the reduce family)
     SlotRef refs[MOVE_MAX_SRC];
-    const char* extras[MOVE_MAX_SRC];
+    const char* contrib[MOVE_MAX_SRC];
     u32 nx = 0;
     for (u32 p = 0; p < P; ++p) {
-      if (p == root) continue;
+      if (p == root) {
+        contrib[p] = src;
+        continue;
+      }
       if (!collect_one_slot(c.global(p), tag, dt, n, refs[nx])) return err;
-      extras[nx] = refs[nx].pay;
+      contrib[p] = refs[nx].pay;
       nx++;
     }
-    u32 e = nary_reduce(dst, src, dt, n, extras, nx, int(d.function));
+    u32 e = nary_reduce(dst, contrib[0], dt, n, &contrib[1], P - 1,
+                        int(d.function));
     u32 k = 0;
     for (u32 p = 0; p < P; ++p)
       if (p != root) release_slot(c.global(p), refs[k++]);
@@ -1829,6 +2114,14 @@ struct Cclo {
     w[2] = nd;
     w[3] = wallclock();
     w[4] = u64(cfg.rank) | (u64(cfg.nranks) << 32);
+    if (Op(scen) == Op::barrier && nd == 0) {
+      // barrier diagnosis: observed peer token vs expected epoch per pair
+      for (u32 g = 0; g < cfg.nranks && g < 24; ++g) {
+        w[8 + 2 * g] = g == cfg.rank ? ~0ull : ld_sys(tv.barrier_word(me(), g));
+        w[8 + 2 * g + 1] = sq.barrier_epoch[g];
+      }
+      w[2] = 0;
+    }
     fence_release_sys();
     st_sys((volatile u64*)&w[0], ld_sys((const volatile u64*)&w[0]) + 1);
   }

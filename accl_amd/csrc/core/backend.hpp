@@ -118,6 +118,15 @@ class Backend {
              (unsigned long long)(w[4] & 0xFFFFFFFF),
              (unsigned long long)(w[4] >> 32), (unsigned long long)w[2]);
     std::string out = buf;
+    if ((w[1] & 0xFFFFFFFF) == 13 && w[2] == 0) {  // barrier diagnosis
+      for (u32 g = 0; g < cfg_.nranks && g < 24; ++g) {
+        if (w[8 + 2 * g] == ~0ull) continue;
+        snprintf(buf, sizeof(buf), "  barrier pair %u: token=%llu expect=%llu\n",
+                 g, (unsigned long long)w[8 + 2 * g],
+                 (unsigned long long)w[8 + 2 * g + 1]);
+        out += buf;
+      }
+    }
     for (u64 i = 0; i < w[2] && i < 24; ++i) {
       const u64* f = &w[8 + i * 16];
       u32 kind = u32(f[0] & 0xFF);

@@ -1544,3 +1544,27 @@ def test_protocol_fuzz2(seed):
               timeout=240)
     run_ranks(lambda a, r, n: _fuzz2(a, r, n, seed), 2, opts=DIRECT,
               timeout=240)
+
+
+def _replica_identity(a, rank, n):
+    """fp32 allreduce must produce BITWISE-IDENTICAL results on every rank
+    (c10d/DDP replicas stay in sync) across all schedule variants:
+    one-shot (small), fullmesh RS+AG (medium), composed direct (large)."""
+    for cnt in (1024, 200_000, 1_572_864 // 4 * n):
+        s, d = _mk(a, cnt), _mk(a, cnt)
+        s.write((pattern(cnt, rank, seed=cnt % 97) * 1.7).astype(np.float32))
+        a.allreduce(s, d, cnt, RF.SUM)
+        mine = rd(d, cnt)
+        g = _mk(a, cnt * n)
+        a.allgather(d, g, cnt)
+        allv = rd(g, cnt * n).reshape(n, cnt)
+        for r in range(n):
+            assert np.array_equal(allv[r], mine), \
+                f"replica divergence cnt={cnt} vs rank {r}"
+    a.barrier()
+
+
+def test_allreduce_replica_identity():
+    run_ranks(_replica_identity, 2, opts=SMALL)
+    run_ranks(_replica_identity, 3, opts=SMALL)
+    run_ranks(_replica_identity, 3, opts=DIRECT)

@@ -1267,6 +1267,62 @@ def test_parked_recv_vs_collectives():
         run_ranks(_parked_vs_collectives, 2, opts=SMALL)
 
 
+def _parked_recv_pool_overflow(a, rank, n):
+    """Deadlock regression (fuzz seed 23): a parked recv whose message
+    overflows the bounded unexpected pool (UQ_DEPTH slot-sized spills)
+    while the engine is committed inside collectives. The drain must
+    deliver segments straight into the parked recv's destination, not
+    just spill — spilling alone livelocks once the pool fills."""
+    cnt = 12_000  # 48 KB = ~12 slot segments > UQ_DEPTH(8) + n_slots(4)
+    if rank == 1:
+        d = _mk(a, cnt)
+        req = a.recv(d, cnt, src=0, tag=88, run_async=True)
+    if rank == 0:
+        sv = _mk(a, cnt)
+        sv.write(pattern(cnt, 4, seed=5))
+        sreq = a.send(sv, cnt, dst=1, tag=88, run_async=True)
+    # both engines now grind through collectives; rank 1's recv is parked
+    # the whole time and only the drain hook can move its message
+    s, dr = _mk(a, 600), _mk(a, 600)
+    s.write(pattern(600, rank, seed=11))
+    for _ in range(8):
+        a.allreduce(s, dr, 600, RF.SUM)
+    exp = np.stack([pattern(600, r, seed=11) for r in range(n)]).sum(0)
+    assert np.allclose(rd(dr, 600), exp)
+    if rank == 0:
+        assert sreq.wait() == 0
+    else:
+        assert req.wait() == 0
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 4, seed=5))
+    a.barrier()
+
+
+def test_parked_recv_pool_overflow():
+    for _ in range(3):
+        run_ranks(_parked_recv_pool_overflow, 2, opts=SMALL)
+
+
+def _head_to_head_async(a, rank, n):
+    """Symmetric async exchange larger than the credit window in BOTH
+    directions at once: every send parks, and only drain/retry-driven
+    flow-free pushes can complete them (fuzz seed 11 regression)."""
+    cnt = 9_000
+    other = 1 - rank
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    s.write(pattern(cnt, rank, seed=2))
+    sreq = a.send(s, cnt, dst=other, tag=7, run_async=True)
+    rreq = a.recv(d, cnt, src=other, tag=7, run_async=True)
+    # engines serve a barrier while both transfers are parked
+    a.barrier()
+    assert sreq.wait() == 0 and rreq.wait() == 0
+    assert np.array_equal(rd(d, cnt), pattern(cnt, other, seed=2))
+
+
+def test_head_to_head_async():
+    for _ in range(3):
+        run_ranks(_head_to_head_async, 2, opts=SMALL)
+
+
 def _fuzz_script(seed, n, nops=40):
     """Deterministic op script shared by all ranks: mixes collectives,
     tagged pairwise send/recv (sync + async), sizes spanning inline/fleet/

@@ -124,12 +124,19 @@ class AcclProcessGroup(dist.ProcessGroup):
 
     # ---------------- staging ----------------
     def _buf(self, count, tdt, which):
-        key = (which, count, tdt)
-        hit = self._bufs.get(key)
+        # round capacity up to the next power of two so DDP's many bucket
+        # shapes share buffers, and keep a bounded LRU (arena heap blocks
+        # are freed when the Buffer is dropped)
+        cap = 1 << max(0, (count - 1).bit_length())
+        key = (which, cap, tdt)
+        hit = self._bufs.pop(key, None)
         if hit is not None:
+            self._bufs[key] = hit  # re-insert: most recently used
             return hit
-        b = self._a.create_buffer(count, _t2dt(tdt), device_only=self._gpu)
+        b = self._a.create_buffer(cap, _t2dt(tdt), device_only=self._gpu)
         view = self._a.tensor(b) if self._gpu else None
+        while len(self._bufs) >= 16:
+            self._bufs.pop(next(iter(self._bufs)))
         self._bufs[key] = (b, view)
         return b, view
 

@@ -160,7 +160,10 @@ struct alignas(64) RndzvRec {
   u32 arith;
   u64 offset;     // destination arena offset (addr ring) / echo (done ring)
   u64 count;      // elements
-  u64 _pad[4];
+  u64 prog_idx;   // addr ring: progress-word index the sender must write
+                  // (allocated from the receiver's per-pair pool — NOT the
+                  // ring slot, which recycles while windows are pending)
+  u64 _pad[3];
 };
 static_assert(sizeof(RndzvRec) == 64, "");
 
@@ -174,7 +177,14 @@ struct alignas(64) EagerChanCtl {
   u64 credit;     // number of slots this rank's messages to `peer`... see note
   u64 tx_ctr;     // STREAM lanes only: shared tx-seq allocator so the engine,
                   // the host and device kernels can produce into one channel
-  u64 _pad[6];
+  // rendezvous-ring flow control (same placement rule as `credit`): the
+  // POSTER of addr/done records polls these locally; the ring CONSUMER
+  // advances them (cumulative consumed seq). Without them a poster can
+  // overwrite unconsumed records once parked ops leave windows
+  // outstanding long enough for the n_rndzv ring to wrap.
+  u64 addr_ret;   // addr records I posted to `peer` that it has consumed
+  u64 done_ret;   // done records I posted to `peer` that it has consumed
+  u64 _pad[4];
 };
 // NOTE on credit placement: for channel (s -> r), slot headers+payload live in
 // r's arena at index [s]; the credit word lives in s's arena at index [r]
@@ -234,6 +244,11 @@ struct ArenaLayout {
 
 constexpr u64 DBG_DUMP_BYTES = 8192;  // flow-state dump region size
 
+// per-pair pool of window progress words. Sized so it can never exhaust
+// structurally: MAX_INFLIGHT parked recvs x 2 windows + an active direct
+// collective's 2 banks < 128.
+constexpr u32 N_PROG = 128;
+
 inline ArenaLayout arena_layout(const ProtoConfig& c, u64 spare_bytes = 32u << 20) {
   auto align_up = [](u64 x, u64 a) { return (x + a - 1) & ~(a - 1); };
   ArenaLayout L{};
@@ -252,7 +267,7 @@ inline ArenaLayout arena_layout(const ProtoConfig& c, u64 spare_bytes = 32u << 2
   L.barrier_off = off = align_up(off, 256);
   off += u64(c.nranks) * sizeof(u64);
   L.direct_off = off = align_up(off, 256);
-  off += u64(c.nranks) * c.n_rndzv * sizeof(u64);
+  off += u64(c.nranks) * N_PROG * sizeof(u64);
   L.devcall_off = off = align_up(off, 256);
   off += sizeof(DevCallRing) + u64(DEVCALL_RING) * sizeof(DevCallSlot) +
          u64(DEVCALL_RING) * sizeof(DevCallRet);
@@ -304,14 +319,16 @@ struct TransportView {
   ACCL_HD u64* barrier_word(u32 r, u32 peer) const {
     return (u64*)(arena[r] + hdr(r)->barrier_off) + peer;
   }
-  // per-WINDOW direct-write progress words: in r's arena, lane [s] slot [i]
-  // (i = the addr-ring slot of the posted window). The sender of a window
-  // writes cumulative bytes WITHIN the window; the receiver zeroes the word
-  // when posting and polls it locally. Window-scoped (not pair-cumulative)
-  // so out-of-order matched rendezvous ops never corrupt each other.
+  // per-WINDOW direct-write progress words: in r's arena, lane [s] word [i]
+  // (i = RndzvRec.prog_idx, allocated from r's per-pair pool when posting —
+  // decoupled from the addr-ring slot, which recycles while parked windows
+  // are still pending). The sender of a window writes cumulative bytes
+  // WITHIN the window; the receiver zeroes the word when posting and polls
+  // it locally. Window-scoped (not pair-cumulative) so out-of-order matched
+  // rendezvous ops never corrupt each other.
   ACCL_HD volatile u64* direct_word(u32 r, u32 s, u32 i) const {
     return (volatile u64*)(arena[r] + hdr(r)->direct_off) +
-           u64(s) * hdr(r)->n_rndzv + i;
+           u64(s) * N_PROG + i;
   }
   ACCL_HD char* heap_ptr(u32 r, u64 off) const { return arena[r] + off; }
 

@@ -75,6 +75,8 @@ struct Flow {
   // pending segment fifo:
   PendSeg pend[FLOW_INFLIGHT];
   u32 ph, pt;             // head/tail (pt-ph = in flight)
+  u32 win_slot;           // rx_direct: progress-word pool index (freed when
+                          // the window is fully consumed)
   u64 msg_count_hdr;      // value for SlotHdr.msg_count (tx)
 };
 
@@ -113,9 +115,9 @@ struct Cclo {
   // ---- rendezvous pending sets (the RNDZV_PENDING spill queue analogue,
   // reference ccl_offload_control.c:154-408): in-seq records whose tag does
   // not match the current wait are parked here for later out-of-order match.
-  static constexpr u32 RNDZV_PEND = 4;
+  static constexpr u32 RNDZV_PEND = 8;
   struct PendRndzv { u64 seq; u64 offset; u64 count; u32 tag; u32 arith;
-                     u32 valid; };
+                     u32 valid; u32 prog; };
 
   // ---- multi-call interleaving (the CMD_CALL_RETRY requeue analogue,
   // reference ccl_offload_control.c:2460-2478 + current_step resume
@@ -139,6 +141,7 @@ struct Cclo {
     PendRndzv pa[MAX_RANKS][RNDZV_PEND];  // addr records
     PendRndzv pd[MAX_RANKS][RNDZV_PEND];  // done records
     ParkedCall parked[MAX_INFLIGHT];
+    u64 prog_busy[MAX_RANKS][2];  // window progress-word pool (N_PROG bits)
   };
   ColdState* cold;
   u64 spill_busy;          // bitmap over spill slots (<= 64)
@@ -149,6 +152,15 @@ struct Cclo {
   u32 probe_;
   ParkState* ps_;
   u32 in_drain_;  // re-entrancy guard for drain_for_parked()
+  // last non-flow waitpoint (dump_timeout diagnosis): what a flows=0
+  // timeout was actually spinning on
+  u64 wp_kind_;   // 0 none, 1 wait_addr, 2 wait_done
+  u64 wp_info_;   // peer | tag<<32
+  u64 wp_seq_;    // awaited ring seq | observed head seq<<32
+  // pairs whose live channel head is HELD (unreleased SlotRef from a
+  // one-shot collect): the drain must not spill/consume that pair's ring
+  // or the held payload gets overwritten once its credit returns early
+  u64 drain_hold_;
 
   ACCL_HD u32 me() const { return cfg.rank; }
 
@@ -303,6 +315,10 @@ struct Cclo {
       const CommView& c = comms[p.d.comm_id];
       u32 peer = c.global(p.d.root_src_dst);
       if (peer == me()) continue;
+      // a one-shot collect holds this pair's live head unreleased — any
+      // spill/consume here returns its credit early and the sender
+      // overwrites the held payload
+      if ((drain_hold_ >> (peer & 63)) & 1) continue;
       Op op = Op(p.d.scenario);
       DataType dt = desc_dtype(p.d), wdt = desc_wire_dtype(p.d);
       u64 n = desc_count(p.d);
@@ -645,6 +661,9 @@ struct Cclo {
             fence_acquire_sys();
             f.done = avail_elems;
             f.submitted = avail_elems;
+            if (f.done >= f.count)  // window consumed: free its progress word
+              cold->prog_busy[f.gpeer & 63][(f.win_slot >> 6) & 1] &=
+                  ~(1ull << (f.win_slot & 63));
             any = true;
           }
           return any;
@@ -722,6 +741,7 @@ struct Cclo {
     Flow& f = fl(i);
     f.kind = FLOW_RX_DIRECT; f.gpeer = gpeer; f.ddt = u8(ddt); f.count = count;
     f.prog_addr = (u64)tv.direct_word(me(), gpeer, slot);
+    f.win_slot = slot;
   }
 
   // ---------------- rendezvous record rings ----------------
@@ -729,17 +749,88 @@ struct Cclo {
   // will write to us); zero the window's progress word first. Returns the
   // ring slot. reference: rendezvous_send_addr
   // (ccl_offload_control.c:142-150).
+  // ring flow control: cumulative consumed counts, published by the
+  // consumer into the POSTER's arena (same placement rule as eager credit)
+  ACCL_HD bool addr_room(u32 gpeer) {
+    return sq.rndzv_addr_tx[gpeer] -
+               ld_sys(&tv.chan_ctl(me(), gpeer)->addr_ret) < cfg.n_rndzv;
+  }
+  ACCL_HD bool done_room(u32 gpeer) {
+    return sq.rndzv_done_tx[gpeer] -
+               ld_sys(&tv.chan_ctl(me(), gpeer)->done_ret) < cfg.n_rndzv;
+  }
+  // one consumption step of MY incoming addr/done ring from gpeer: spill
+  // the head into the pending set. Run while blocked in post_addr/post_done
+  // so two ranks posting to each other symmetrically always drain.
+  ACCL_HD void consume_addr_tick(u32 gpeer) {
+    u64 seq = sq.rndzv_addr_rx[gpeer] + 1;
+    RndzvRec* r = tv.rndzv_addr(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
+    if (ld_sys(&r->seq) != seq) return;
+    fence_acquire_sys();
+    u32 k = 0;
+    while (k < RNDZV_PEND && cold->pa[gpeer][k].valid) ++k;
+    if (k >= RNDZV_PEND) return;
+    cold->pa[gpeer][k] = PendRndzv{seq, r->offset, r->count, r->tag, r->arith,
+                                   1, u32(r->prog_idx)};
+    sq.rndzv_addr_rx[gpeer] = seq;
+    st_sys(&tv.chan_ctl(gpeer, me())->addr_ret, seq);
+  }
+  ACCL_HD void consume_done_tick(u32 gpeer) {
+    u64 seq = sq.rndzv_done_rx[gpeer] + 1;
+    RndzvRec* r = tv.rndzv_done(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
+    if (ld_sys(&r->seq) != seq) return;
+    fence_acquire_sys();
+    u32 k = 0;
+    while (k < RNDZV_PEND && cold->pd[gpeer][k].valid) ++k;
+    if (k >= RNDZV_PEND) return;
+    cold->pd[gpeer][k] = PendRndzv{seq, 0, 0, r->tag, 0, 1};
+    sq.rndzv_done_rx[gpeer] = seq;
+    st_sys(&tv.chan_ctl(gpeer, me())->done_ret, seq);
+  }
+
   ACCL_HD u32 post_addr(u32 gpeer, u64 offset, u64 count, u32 tag, u32 arith) {
+    if (!addr_room(gpeer)) {
+      // ring full: records outlive ops (parked recvs), so wait for the
+      // consumer — draining our own inbound rings meanwhile so symmetric
+      // posters can't block each other. On deadline: post anyway (err is
+      // set; the op fails loud downstream instead of silently losing a
+      // record).
+      u64 deadline = deadline_now();
+      while (!addr_room(gpeer)) {
+        consume_addr_tick(gpeer);
+        consume_done_tick(gpeer);
+        if (!wait_pred_tick(deadline)) break;
+      }
+    }
+    // allocate a progress word (pool is sized to never exhaust; the wait
+    // is a belt for error-path leaks)
+    u64* pb = cold->prog_busy[gpeer & 63];
+    u32 idx = 0;
+    for (;;) {
+      idx = 0;
+      while (idx < N_PROG && ((pb[idx >> 6] >> (idx & 63)) & 1)) ++idx;
+      if (idx < N_PROG) break;
+      u64 deadline = deadline_now();
+      while (idx >= N_PROG) {
+        consume_addr_tick(gpeer);
+        consume_done_tick(gpeer);
+        if (!wait_pred_tick(deadline)) { idx = 0; break; }
+        idx = 0;
+        while (idx < N_PROG && ((pb[idx >> 6] >> (idx & 63)) & 1)) ++idx;
+      }
+      break;
+    }
+    pb[idx >> 6] |= 1ull << (idx & 63);
     u64 seq = ++sq.rndzv_addr_tx[gpeer];
     u32 slot = u32((seq - 1) % cfg.n_rndzv);
-    st_sys(tv.direct_word(me(), gpeer, slot), 0);  // reset window progress
+    st_sys(tv.direct_word(me(), gpeer, idx), 0);  // reset window progress
     RndzvRec* r = tv.rndzv_addr(gpeer, me(), slot);
-    u64 val[8] = {seq, (u64(arith) << 32) | tag, offset, count, 0, 0, 0, 0};
+    u64 val[8] = {seq, (u64(arith) << 32) | tag, offset, count, idx, 0, 0, 0};
     publish_rec((volatile u64*)r, val);
-    return slot;
+    return idx;
   }
-  ACCL_HD static u32 rec_slot(const RndzvRec& rec, u32 n_rndzv) {
-    return u32((rec.seq - 1) % n_rndzv);
+  ACCL_HD static u32 rec_slot(const RndzvRec& rec, u32 /*n_rndzv*/) {
+    return u32(rec.prog_idx);
   }
 
   // Wait for an addr record from gpeer MATCHING want_tag; non-matching
@@ -761,6 +852,7 @@ struct Cclo {
         PendRndzv& p = cold->pa[gpeer][best];
         out.seq = p.seq; out.tag = p.tag; out.arith = p.arith;
         out.offset = p.offset; out.count = p.count;
+        out.prog_idx = p.prog;
         p.valid = 0;
         return true;
       }
@@ -768,18 +860,28 @@ struct Cclo {
       RndzvRec* r = tv.rndzv_addr(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
       if (ld_sys(&r->seq) == seq) {
         fence_acquire_sys();
+        // copy the record BEFORE publishing consumption — the poster may
+        // reuse the slot the instant addr_ret advances
+        u32 rtag = r->tag, rarith = r->arith;
+        u64 roff = r->offset, rcnt = r->count, rprog = r->prog_idx;
         sq.rndzv_addr_rx[gpeer] = seq;
-        if (want_tag == TAG_ANY || r->tag == want_tag || r->tag == TAG_ANY) {
-          out.seq = seq; out.tag = r->tag; out.arith = r->arith;
-          out.offset = r->offset; out.count = r->count;
+        st_sys(&tv.chan_ctl(gpeer, me())->addr_ret, seq);
+        if (want_tag == TAG_ANY || rtag == want_tag || rtag == TAG_ANY) {
+          out.seq = seq; out.tag = rtag; out.arith = rarith;
+          out.offset = roff; out.count = rcnt;
+          out.prog_idx = rprog;
           return true;
         }
         u32 k = 0;
         while (k < RNDZV_PEND && cold->pa[gpeer][k].valid) ++k;
         if (k >= RNDZV_PEND) { err |= E_RNDZV; return false; }
-        cold->pa[gpeer][k] = PendRndzv{seq, r->offset, r->count, r->tag, r->arith, 1};
+        cold->pa[gpeer][k] = PendRndzv{seq, roff, rcnt, rtag, rarith, 1,
+                                       u32(rprog)};
         continue;
       }
+      wp_kind_ = 1;
+      wp_info_ = u64(gpeer) | (u64(want_tag) << 32);
+      wp_seq_ = (seq & 0xFFFFFFFFull) | (ld_sys(&r->seq) << 32);
       if (!wait_pred_tick(deadline)) return false;
     }
   }
@@ -809,12 +911,49 @@ struct Cclo {
       u32 k = 0;
       while (k < RNDZV_PEND && cold->pa[gpeer][k].valid) ++k;
       if (k >= RNDZV_PEND) return false;  // pending full: try again later
-      cold->pa[gpeer][k] =
-          PendRndzv{seq, r->offset, r->count, r->tag, r->arith, 1};
+      cold->pa[gpeer][k] = PendRndzv{seq, r->offset, r->count, r->tag,
+                                     r->arith, 1, u32(r->prog_idx)};
       sq.rndzv_addr_rx[gpeer] = seq;
+      st_sys(&tv.chan_ctl(gpeer, me())->addr_ret, seq);
     }
   }
+  // probe: is a tag-matching DONE record available right now? (mirror of
+  // addr_ready — foreign heads spill to the pending set). Needed because a
+  // probe-mode recv must never block in wait_done: the sender may be a
+  // PARKED call on the peer whose post_done only happens at ITS retry,
+  // and that retry needs the peer engine free (mutual-block otherwise).
+  ACCL_HD bool done_ready(u32 gpeer, u32 want_tag) {
+    for (;;) {
+      for (u32 k = 0; k < RNDZV_PEND; ++k) {
+        PendRndzv& p = cold->pd[gpeer][k];
+        if (p.valid && (want_tag == TAG_ANY || p.tag == want_tag ||
+                        p.tag == TAG_ANY))
+          return true;
+      }
+      u64 seq = sq.rndzv_done_rx[gpeer] + 1;
+      RndzvRec* r = tv.rndzv_done(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
+      if (ld_sys(&r->seq) != seq) return false;
+      fence_acquire_sys();
+      u32 t = r->tag;
+      if (want_tag == TAG_ANY || t == want_tag || t == TAG_ANY) return true;
+      u32 k = 0;
+      while (k < RNDZV_PEND && cold->pd[gpeer][k].valid) ++k;
+      if (k >= RNDZV_PEND) return false;  // pending full: try again later
+      cold->pd[gpeer][k] = PendRndzv{seq, 0, 0, r->tag, 0, 1};
+      sq.rndzv_done_rx[gpeer] = seq;
+      st_sys(&tv.chan_ctl(gpeer, me())->done_ret, seq);
+    }
+  }
+
   ACCL_HD void post_done(u32 gpeer, u32 tag) {
+    if (!done_room(gpeer)) {
+      u64 deadline = deadline_now();
+      while (!done_room(gpeer)) {
+        consume_addr_tick(gpeer);
+        consume_done_tick(gpeer);
+        if (!wait_pred_tick(deadline)) break;
+      }
+    }
     u64 seq = ++sq.rndzv_done_tx[gpeer];
     RndzvRec* r = tv.rndzv_done(gpeer, me(), u32((seq - 1) % cfg.n_rndzv));
     u64 val[8] = {seq, tag, 0, 0, 0, 0, 0, 0};
@@ -836,15 +975,20 @@ struct Cclo {
       RndzvRec* r = tv.rndzv_done(me(), gpeer, u32((seq - 1) % cfg.n_rndzv));
       if (ld_sys(&r->seq) == seq) {
         fence_acquire_sys();
+        u32 rtag = r->tag;  // copy before publishing consumption
         sq.rndzv_done_rx[gpeer] = seq;
-        if (want_tag == TAG_ANY || r->tag == want_tag || r->tag == TAG_ANY)
+        st_sys(&tv.chan_ctl(gpeer, me())->done_ret, seq);
+        if (want_tag == TAG_ANY || rtag == want_tag || rtag == TAG_ANY)
           return true;
         u32 k = 0;
         while (k < RNDZV_PEND && cold->pd[gpeer][k].valid) ++k;
         if (k >= RNDZV_PEND) { err |= E_RNDZV; return false; }
-        cold->pd[gpeer][k] = PendRndzv{seq, 0, 0, r->tag, 0, 1};
+        cold->pd[gpeer][k] = PendRndzv{seq, 0, 0, rtag, 0, 1};
         continue;
       }
+      wp_kind_ = 2;
+      wp_info_ = u64(gpeer) | (u64(want_tag) << 32);
+      wp_seq_ = (seq & 0xFFFFFFFFull) | (ld_sys(&r->seq) << 32);
       if (!wait_pred_tick(deadline)) return false;
     }
   }
@@ -911,6 +1055,7 @@ struct Cclo {
       // ParkState — it must never resend windows a probe already pushed.
       if (probe_) {
         if (!rndzv_send_push(peer, d, *ps_)) return E_NOT_READY;
+        if (!done_room(peer)) return E_NOT_READY;  // never block in probe
         post_done(peer, d.tag);
         return E_OK;
       }
@@ -1117,6 +1262,14 @@ struct Cclo {
     char* dst = local_ptr(d.addr2, d.flags & F_DST_ARENA);
     if (use_rndzv(n, dt, wdt) && peer != me()) {
       const u32 esz = dtype_size(dt);
+      // resumed park in the final phase: all data landed, only the done
+      // record outstanding (probe-mode ops must never block in wait_done —
+      // the sender may be a parked call whose post_done happens at ITS
+      // retry, which needs the peer engine free)
+      if (probe_ && ps_ && ps_->step == 3) {
+        if (!done_ready(peer, d.tag)) return E_NOT_READY;
+        return wait_done(peer, d.tag) ? E_OK : err;
+      }
       if (d.flags & F_DST_ARENA) {
         // window the posting by max_rndzv_bytes (reference:
         // set_max_rendezvous_size caps a single rendezvous transfer);
@@ -1135,14 +1288,16 @@ struct Cclo {
         }
         while (got < n) {
           while (posted < n && nw < 2) {
+            if (probe_ && !addr_room(peer)) break;  // never block in probe
             u64 w = min64(n - posted, wmax);
             wslot[(wi + nw) % 2] =
                 post_addr(peer, d.addr2 + posted * esz, w, d.tag, u32(dt));
             wcnt[(wi + nw) % 2] = w;
             posted += w; nw++;
           }
-          if (probe_ && got == 0 &&
-              ld_sys(tv.direct_word(me(), peer, wslot[wi])) == 0) {
+          if (probe_ &&
+              (nw == 0 || (got == 0 && ld_sys(tv.direct_word(
+                                           me(), peer, wslot[wi])) == 0))) {
             if (ps_) {
               ps_->w[0] = posted; ps_->w[1] = got;
               ps_->w[2] = wcnt[0]; ps_->w[3] = wcnt[1];
@@ -1157,6 +1312,10 @@ struct Cclo {
           if (e) return e;
           got += wcnt[wi];
           wi ^= 1; nw--;
+        }
+        if (probe_ && !done_ready(peer, d.tag)) {
+          if (ps_) ps_->step = 3;
+          return E_NOT_READY;
         }
         return wait_done(peer, d.tag) ? E_OK : err;
       }
@@ -1185,6 +1344,10 @@ struct Cclo {
         if ((e = run_flows(1))) return e;
         got += wcnt[wi];
         wi ^= 1; nw--;
+      }
+      if (probe_ && !done_ready(peer, d.tag)) {
+        if (ps_) ps_->step = 3;
+        return E_NOT_READY;
       }
       return wait_done(peer, d.tag) ? E_OK : err;
     }
@@ -1608,6 +1771,7 @@ struct Cclo {
             return false;
           }
           out = SlotRef{tv.slot_payload(me(), gpeer, sl), seq, 0, 0};
+          drain_hold_ |= 1ull << (gpeer & 63);
           return true;
         }
         if (spill_head(gpeer, h, sl, seq)) { deadline = deadline_now(); continue; }
@@ -1628,6 +1792,7 @@ struct Cclo {
       sq.eager_rx[gpeer] = s.seq;
       sq.credit_ret[gpeer] = s.seq;
       st_sys(&tv.chan_ctl(gpeer, me())->credit, s.seq);
+      drain_hold_ &= ~(1ull << (gpeer & 63));
     }
   }
 
@@ -2114,6 +2279,9 @@ struct Cclo {
     w[2] = nd;
     w[3] = wallclock();
     w[4] = u64(cfg.rank) | (u64(cfg.nranks) << 32);
+    w[5] = wp_kind_;
+    w[6] = wp_info_;
+    w[7] = wp_seq_;
     if (Op(scen) == Op::barrier && nd == 0) {
       // barrier diagnosis: observed peer token vs expected epoch per pair
       for (u32 g = 0; g < cfg.nranks && g < 24; ++g) {
@@ -2277,6 +2445,7 @@ struct Cclo {
 
   ACCL_HD u32 run_call_inner(const CallDesc& d) {
     err = 0;
+    drain_hold_ = 0;  // holds never span calls (belt for error paths)
     if (d.comm_id >= ncomms && Op(d.scenario) != Op::copy &&
         Op(d.scenario) != Op::combine && Op(d.scenario) != Op::config &&
         Op(d.scenario) != Op::nop)
@@ -2355,6 +2524,8 @@ struct Cclo {
         // survive (desynced pairs need a reset on both ends).
         for (u32 i = 0; i < MAX_FLOWS; ++i) flows[i] = Flow{};
         for (u32 r = 0; r < MAX_RANKS; ++r) { cold->uq_h[r] = cold->uq_t[r] = 0; }
+        for (u32 r = 0; r < MAX_RANKS; ++r)
+          cold->prog_busy[r][0] = cold->prog_busy[r][1] = 0;
         spill_busy = 0;
         err = 0;
         return E_OK;

@@ -118,6 +118,17 @@ class Backend {
              (unsigned long long)(w[4] & 0xFFFFFFFF),
              (unsigned long long)(w[4] >> 32), (unsigned long long)w[2]);
     std::string out = buf;
+    if (w[5]) {  // last non-flow waitpoint (wait_addr / wait_done spin)
+      snprintf(buf, sizeof(buf),
+               "\n  waitpoint: %s peer=%llu tag=0x%llx await_seq=%llu "
+               "head_seq=%llu",
+               w[5] == 1 ? "wait_addr" : w[5] == 2 ? "wait_done" : "?",
+               (unsigned long long)(w[6] & 0xFFFFFFFF),
+               (unsigned long long)(w[6] >> 32),
+               (unsigned long long)(w[7] & 0xFFFFFFFF),
+               (unsigned long long)(w[7] >> 32));
+      out += buf;
+    }
     if ((w[1] & 0xFFFFFFFF) == 13 && w[2] == 0) {  // barrier diagnosis
       for (u32 g = 0; g < cfg_.nranks && g < 24; ++g) {
         if (w[8 + 2 * g] == ~0ull) continue;

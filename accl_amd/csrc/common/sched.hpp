@@ -221,6 +221,16 @@ struct Cclo {
   ACCL_HD u64 tx_credit(u32 peer) {  // slots consumed by peer (cumulative)
     return ld_sys(&tv.chan_ctl(me(), peer)->credit);
   }
+  // Return credit MONOTONICALLY: consumers can finish out of order (a
+  // flow's pending segment copy retires after the drain already consumed
+  // and credited newer segments), and a stale write would shrink the
+  // sender's window below its outstanding count — a permanent stall.
+  ACCL_HD void ret_credit(u32 peer, u64 seq) {
+    if (seq > sq.credit_ret[peer]) {
+      sq.credit_ret[peer] = seq;
+      st_sys(&tv.chan_ctl(peer, me())->credit, seq);
+    }
+  }
 
   // Flow-free single-segment eager send (direct mover submit + header
   // publish): the parked-send resume/drain path — safe to run from INSIDE
@@ -370,8 +380,7 @@ struct Cclo {
                            dst + got * dsz, dt, nseg))
               break;
             sq.eager_rx[peer] = seq;
-            sq.credit_ret[peer] = seq;
-            st_sys(&tv.chan_ctl(peer, me())->credit, seq);
+            ret_credit(peer, seq);
             got += nseg;
             prog = true;
           } else if (!match) {
@@ -442,8 +451,7 @@ struct Cclo {
             break;
           }
           // payload consumed -> return credit to the sender (cumulative).
-          sq.credit_ret[f.gpeer] = p.seq;
-          st_sys(&tv.chan_ctl(f.gpeer, me())->credit, p.seq);
+          ret_credit(f.gpeer, p.seq);
           break;
         }
         case FLOW_TX_DIRECT: {
@@ -1169,8 +1177,7 @@ struct Cclo {
     spill_busy |= 1ull << sp;
     cold->uq_t[peer]++;
     sq.eager_rx[peer] = seq;
-    sq.credit_ret[peer] = seq;
-    st_sys(&tv.chan_ctl(peer, me())->credit, seq);
+    ret_credit(peer, seq);
     return true;
   }
 
@@ -1234,8 +1241,7 @@ struct Cclo {
           u32 e = run_flows(1);
           if (e) return e;
           sq.eager_rx[peer] = seq;
-          sq.credit_ret[peer] = seq;
-          st_sys(&tv.chan_ctl(peer, me())->credit, seq);
+          ret_credit(peer, seq);
           got += nseg;
           deadline = deadline_now();
           continue;
@@ -1790,8 +1796,7 @@ struct Cclo {
         cold->uq_h[gpeer]++;
     } else {
       sq.eager_rx[gpeer] = s.seq;
-      sq.credit_ret[gpeer] = s.seq;
-      st_sys(&tv.chan_ctl(gpeer, me())->credit, s.seq);
+      ret_credit(gpeer, s.seq);
       drain_hold_ &= ~(1ull << (gpeer & 63));
     }
   }

@@ -1436,6 +1436,30 @@ def test_protocol_fuzz_direct():
     run_ranks(lambda a, r, n: _fuzz(a, r, n, 51), 3, opts=DIRECT, timeout=240)
 
 
+# two-slot / 1 KB geometry: maximum ring-wrap and credit pressure. TINYD
+# additionally forces rendezvous everywhere with a TWO-deep addr ring —
+# the regression surface for ring flow control, pooled progress words,
+# monotonic credits and the probe-safe done wait (each found by campaign
+# seeds noted below).
+TINY = {"n_slots": 2, "slot_bytes": 1024, "timeout_us": 20_000_000}
+TINYD = {"n_slots": 2, "slot_bytes": 1024, "max_eager": 1024,
+         "n_rndzv": 2, "timeout_us": 20_000_000}
+
+
+@pytest.mark.parametrize("seed,P,opts", [
+    (100, 2, DIRECT),   # probe-mode wait_done mutual block
+    (133, 4, TINY),     # one-shot collect vs drain spill (data corruption)
+    (161, 2, SMALL),    # out-of-order credit regression
+    (127, 2, TINYD),    # progress-word reset under ring recycling
+    (110, 3, TINYD),    # addr-ring overwrite without flow control
+    (117, 4, TINYD),
+])
+def test_protocol_fuzz_tiny(seed, P, opts):
+    for _ in range(2):
+        run_ranks(lambda a, r, n: _fuzz(a, r, n, seed), P, opts=opts,
+                  timeout=240)
+
+
 def _ar_direct(a, rank, n):
     """Composed direct allreduce (direct RS + direct AG, no eager
     staging): counts divisible by P trigger the path under tiny max_eager."""

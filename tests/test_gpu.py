@@ -419,6 +419,51 @@ def test_two_ranks_parked_interleave():
     run_ranks(_parked2, 2, backend="gpu", timeout=240)
 
 
+SMALL_GPU = {"n_slots": 4, "slot_bytes": 4096, "timeout_us": 20_000_000}
+
+
+def _drain2(a, rank, n):
+    """Cross-rank park-deadlock regressions on the GPU engine under small
+    slots (same scenarios as the emulator's test_head_to_head_async and
+    test_parked_recv_pool_overflow: flow-free parked-send pushes and
+    drain-to-destination for parked recvs)."""
+    cnt = 9_000
+    other = 1 - rank
+    s, d = a.create_buffer(cnt, DT.float32), a.create_buffer(cnt, DT.float32)
+    s.write(pattern(cnt, rank, seed=2))
+    sreq = a.send(s, cnt, dst=other, tag=7, run_async=True)
+    rreq = a.recv(d, cnt, src=other, tag=7, run_async=True)
+    a.barrier()
+    assert sreq.wait() == 0 and rreq.wait() == 0
+    assert np.array_equal(rd(d, cnt), pattern(cnt, other, seed=2))
+    # parked recv whose message overflows the unexpected pool while the
+    # engine grinds through collectives
+    big = 12_000
+    if rank == 1:
+        db = a.create_buffer(big, DT.float32)
+        req = a.recv(db, big, src=0, tag=88, run_async=True)
+    else:
+        sv = a.create_buffer(big, DT.float32)
+        sv.write(pattern(big, 4, seed=5))
+        sq = a.send(sv, big, dst=1, tag=88, run_async=True)
+    s2, d2 = a.create_buffer(600, DT.float32), a.create_buffer(600, DT.float32)
+    s2.write(pattern(600, rank, seed=11))
+    for _ in range(8):
+        a.allreduce(s2, d2, 600, RF.SUM)
+    exp = np.stack([pattern(600, r, seed=11) for r in range(n)]).sum(0)
+    assert np.allclose(rd(d2, 600), exp)
+    if rank == 1:
+        assert req.wait() == 0
+        assert np.array_equal(rd(db, big), pattern(big, 4, seed=5))
+    else:
+        assert sq.wait() == 0
+    a.barrier()
+
+
+def test_two_ranks_drain_regressions():
+    run_ranks(_drain2, 2, backend="gpu", opts=SMALL_GPU, timeout=240)
+
+
 def _compressed_large(a, rank, n):
     """Fleet-path compression tiles: 4 MB f32 message on an f16 wire
     (vectorized cast tx, fused cast+reduce rx, up-cast phase 2)."""

@@ -1050,11 +1050,12 @@ struct Cclo {
     if (d.flags & F_SRC_STREAM)  // send-from-stream (OP0_STREAM analogue)
       return stream_fed(u32(d.addr0), nullptr, dt, n, i64(peer), wdt, d.tag);
     const char* src = local_ptr(d.addr0, d.flags & F_SRC_ARENA);
-    if (peer == me()) {  // self-send: defer to matching recv via loopback slot
-      mk_tx(0, peer, src, dt, wdt, n, d.tag);
-      return run_flows(1);
-    }
-    if (use_rndzv(n, dt, wdt)) {
+    // self-send uses the generic eager path (loopback slots): it is
+    // probe-aware, so a large self-send parks on credit instead of
+    // committing the engine to a wait only its own matching recv (queued
+    // BEHIND it) could satisfy. Rendezvous is peer-only — op_recv matches
+    // self traffic on the eager channel.
+    if (use_rndzv(n, dt, wdt) && peer != me()) {
       // follow the receiver's posted windows (tag-matched, out of order
       // w.r.t. other rendezvous ops on this pair). Flow-free resumable
       // pushes in BOTH modes: a probe parks when no window is posted

@@ -1323,6 +1323,24 @@ def test_head_to_head_async():
         run_ranks(_head_to_head_async, 2, opts=SMALL)
 
 
+def _self_sendrecv(a, rank, n):
+    """Large async self-send over the credit window: the send must park on
+    credit (probe-aware loopback) so the matching self-recv queued behind
+    it can be served — a committed self-send wedges the engine."""
+    cnt = 9_000  # 36 KB >> 4 slots x 4 KB
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    s.write(pattern(cnt, 3, seed=6))
+    sreq = a.send(s, cnt, dst=rank, tag=42, run_async=True)
+    rreq = a.recv(d, cnt, src=rank, tag=42, run_async=True)
+    assert sreq.wait() == 0 and rreq.wait() == 0
+    assert np.array_equal(rd(d, cnt), pattern(cnt, 3, seed=6))
+
+
+def test_self_sendrecv():
+    run_ranks(_self_sendrecv, 1, opts=SMALL)
+    run_ranks(_self_sendrecv, 2, opts=SMALL)
+
+
 def _fuzz_script(seed, n, nops=40):
     """Deterministic op script shared by all ranks: mixes collectives,
     tagged pairwise send/recv (sync + async), sizes spanning inline/fleet/

@@ -1341,6 +1341,38 @@ def test_self_sendrecv():
     run_ranks(_self_sendrecv, 2, opts=SMALL)
 
 
+def _batched_p40(a, rank, n):
+    """Large-P batched fallbacks (2P-1 > MAX_FLOWS=72): allreduce ring,
+    reduce_scatter_batched's paired tx/rx, allgather_batched."""
+    cnt = 256
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    s.write(pattern(cnt, rank, seed=1))
+    a.allreduce(s, d, cnt, RF.SUM)
+    exp = np.stack([pattern(cnt, r, seed=1) for r in range(n)]).sum(0)
+    assert np.allclose(rd(d, cnt), exp), "p40 allreduce"
+    per = 64
+    s2, d2 = _mk(a, per * n), _mk(a, per)
+    s2.write(np.concatenate([pattern(per, rank + j, seed=2)
+                             for j in range(n)]))
+    a.reduce_scatter(s2, d2, per, RF.SUM)
+    exp2 = np.stack([pattern(per, r + rank, seed=2) for r in range(n)]).sum(0)
+    assert np.allclose(rd(d2, per), exp2), "p40 rs"
+    s3, d3 = _mk(a, per), _mk(a, per * n)
+    s3.write(pattern(per, rank, seed=3))
+    a.allgather(s3, d3, per)
+    got = rd(d3, per * n)
+    for r in range(n):
+        assert np.array_equal(got[r * per:(r + 1) * per],
+                              pattern(per, r, seed=3)), f"p40 ag r{r}"
+    a.barrier()
+
+
+def test_batched_collectives_p40():
+    run_ranks(_batched_p40, 40,
+              opts={"n_slots": 4, "slot_bytes": 4096,
+                    "timeout_us": 60_000_000}, timeout=300)
+
+
 def _fuzz_script(seed, n, nops=40):
     """Deterministic op script shared by all ranks: mixes collectives,
     tagged pairwise send/recv (sync + async), sizes spanning inline/fleet/

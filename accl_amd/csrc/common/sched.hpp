@@ -152,6 +152,10 @@ struct Cclo {
   u32 probe_;
   ParkState* ps_;
   u32 in_drain_;  // re-entrancy guard for drain_for_parked()
+  // parked index currently being retried, +1 (0 = none): the drain must
+  // not touch an entry whose own probe is running above it — both would
+  // consume the same windows/segments through the same ParkState
+  u32 active_parked1_;
   // last non-flow waitpoint (dump_timeout diagnosis): what a flows=0
   // timeout was actually spinning on
   u64 wp_kind_;   // 0 none, 1 wait_addr, 2 wait_done
@@ -308,6 +312,7 @@ struct Cclo {
     for (u32 i = 0; i < MAX_INFLIGHT; ++i) {
       ParkedCall& p = cold->parked[i];
       if (!p.used) continue;
+      if (i + 1 == active_parked1_) continue;  // its own probe is running
       // per-(pair,tag) FIFO: skip if an older parked call shares the key
       bool blocked = false;
       for (u32 j = 0; j < MAX_INFLIGHT; ++j) {
@@ -390,6 +395,51 @@ struct Cclo {
         p.ps.w[0] = got;
         p.ps.w[1] = mtag;
         p.ps.step = 2;
+      } else if (op == Op::recv && (p.d.flags & F_DST_ARENA)) {
+        // windowed rendezvous recv: consume completed windows and rotate
+        // new ones in the background (data lands directly in dst — the
+        // sender writes the arena window; "consumption" is observing the
+        // progress word and freeing its pool slot)
+        ParkState& ps = p.ps;
+        if (ps.step == 3) continue;  // only the done record outstanding
+        const u32 esz = dtype_size(dt);
+        u64 wmax = max_rndzv_bytes ? max_rndzv_bytes / esz : n;
+        if (!wmax) wmax = 1;
+        u64 posted = 0, got = 0;
+        u32 wslot[2] = {0, 0};
+        u64 wcnt[2] = {0, 0};
+        u32 wi = 0, nw = 0;
+        if (ps.step == 1) {
+          posted = ps.w[0]; got = ps.w[1];
+          wcnt[0] = ps.w[2]; wcnt[1] = ps.w[3];
+          wslot[0] = u32(ps.w[4]); wslot[1] = u32(ps.w[4] >> 32);
+          wi = u32(ps.w[5]); nw = u32(ps.w[5] >> 32);
+        }
+        bool prog = true;
+        while (prog && got < n) {
+          prog = false;
+          while (posted < n && nw < 2 && addr_room(peer)) {
+            u64 w = min64(n - posted, wmax);
+            wslot[(wi + nw) % 2] =
+                post_addr(peer, p.d.addr2 + posted * esz, w, p.d.tag, u32(dt));
+            wcnt[(wi + nw) % 2] = w;
+            posted += w; nw++; prog = true;
+          }
+          if (nw &&
+              ld_sys(tv.direct_word(me(), peer, wslot[wi])) >=
+                  wcnt[wi] * u64(esz)) {
+            fence_acquire_sys();
+            cold->prog_busy[peer & 63][(wslot[wi] >> 6) & 1] &=
+                ~(1ull << (wslot[wi] & 63));
+            got += wcnt[wi];
+            wi ^= 1; nw--; prog = true;
+          }
+        }
+        ps.w[0] = posted; ps.w[1] = got;
+        ps.w[2] = wcnt[0]; ps.w[3] = wcnt[1];
+        ps.w[4] = u64(wslot[0]) | (u64(wslot[1]) << 32);
+        ps.w[5] = u64(wi) | (u64(nw) << 32);
+        ps.step = (got >= n) ? 3 : 1;
       } else if (op == Op::send && !use_rndzv(n, dt, wdt)) {
         u64 sent = (p.ps.step == 1) ? p.ps.w[0] : 0;
         if (sent < n) {
@@ -2396,7 +2446,9 @@ struct Cclo {
         }
       }
       if (blocked) continue;
+      active_parked1_ = i + 1;
       u32 e = run_call_probe(p.d, true, p.ps);
+      active_parked1_ = 0;
       if (e & E_NOT_READY) {
         if (wallclock() <= p.deadline) continue;
         err = E_TIMEOUT;

@@ -244,6 +244,51 @@ class AcclProcessGroup(dist.ProcessGroup):
             self._reduce_scatter_base(out, flat, opts)
         return _Work()
 
+    def reduce(self, tensors, opts=None):
+        root = opts.rootRank if opts is not None else 0
+        op = opts.reduceOp if opts is not None else dist.ReduceOp.SUM
+        for t in tensors:
+            n = t.numel()
+            s, _ = self._upload(t, n, "rd_s")
+            d, dv = self._buf(n, t.dtype, "rd_d")
+            self._a.reduce(s, d, n, root, _red(op), from_device=True,
+                           to_device=True)
+            if self.rank() == root:
+                self._download(t, d, dv)
+                if _is_avg(op):
+                    t.div_(self.size())
+        return _Work(list(tensors))
+
+    def gather(self, output_lists, input_list, opts=None):
+        root = opts.rootRank if opts is not None else 0
+        for i, inp in enumerate(input_list):
+            n = inp.numel()
+            s, _ = self._upload(inp, n, "ga_s")
+            d, dv = self._buf(n * self.size(), inp.dtype, "ga_d")
+            self._a.gather(s, d, n, root, from_device=True, to_device=True)
+            if self.rank() == root:
+                gathered = torch.empty(n * self.size(), dtype=inp.dtype)
+                if self._gpu:
+                    gathered = gathered.cuda()
+                self._download(gathered, d, dv)
+                for r, o in enumerate(output_lists[i]):
+                    o.reshape(-1).copy_(gathered[r * n:(r + 1) * n])
+        return _Work()
+
+    def scatter(self, output_list, input_lists, opts=None):
+        root = opts.rootRank if opts is not None else 0
+        for i, out in enumerate(output_list):
+            n = out.numel()
+            if self.rank() == root:
+                flat = torch.cat([t.reshape(-1) for t in input_lists[i]])
+                s, _ = self._upload(flat, n * self.size(), "sc_s")
+            else:
+                s, _ = self._buf(n, out.dtype, "sc_leaf")  # unused by leaves
+            d, dv = self._buf(n, out.dtype, "sc_d")
+            self._a.scatter(s, d, n, root, from_device=True, to_device=True)
+            self._download(out, d, dv)
+        return _Work()
+
     def alltoall_base(self, output, input, out_sizes, in_sizes, opts=None):
         if (out_sizes and len(set(out_sizes)) > 1) or \
                 (in_sizes and len(set(in_sizes)) > 1):

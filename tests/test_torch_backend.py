@@ -49,6 +49,23 @@ def _rank_main(rank, world, port, q):
 
         dist.barrier()
 
+        # rooted collectives: reduce / gather / scatter
+        rt = torch.full((96,), float(rank + 1))
+        dist.reduce(rt, dst=1)
+        if rank == 1:
+            assert torch.allclose(rt, torch.full((96,), 3.0)), rt[:4]
+        g_in = torch.full((48,), float(10 + rank))
+        g_out = [torch.zeros(48) for _ in range(world)] if rank == 0 else None
+        dist.gather(g_in, g_out, dst=0)
+        if rank == 0:
+            for r in range(world):
+                assert g_out[r].eq(float(10 + r)).all(), (r, g_out[r][:3])
+        sc_out = torch.zeros(40)
+        sc_in = ([torch.full((40,), float(20 + r)) for r in range(world)]
+                 if rank == 1 else None)
+        dist.scatter(sc_out, sc_in, src=1)
+        assert sc_out.eq(float(20 + rank)).all(), sc_out[:3]
+
         # uneven all_to_all_single (MoE-style alltoallv): rank r sends
         # (p+1)*10*(r+1) elems to p; receives (r+1)*10*(p+1) from p
         in_sizes = [(p + 1) * 10 * (rank + 1) for p in range(world)]

@@ -19,9 +19,24 @@ for kb in 64 128; do
     > gpurun_out/r3_inline$kb.log 2>&1
   echo "INLINE_KB=$kb:"; grep -E "^65536|^262144|^1048576" gpurun_out/r3_inline$kb.log
 done
-# 4) 2-proc direct + parked paths once more on fresh silicon
-timeout 300 python -m pytest "tests/test_gpu.py::test_two_ranks_direct_paths" \
+# 4) 2-proc direct + parked + drain-regression paths once more on fresh silicon
+timeout 400 python -m pytest "tests/test_gpu.py::test_two_ranks_direct_paths" \
   "tests/test_gpu.py::test_two_ranks_parked_interleave" \
+  "tests/test_gpu.py::test_two_ranks_drain_regressions" \
   "tests/test_gpu.py::test_four_ranks_nary_direct" -q -p no:cacheprovider \
   > gpurun_out/r3_direct.log 2>&1
 echo "DIRECT_RC=$?"; tail -2 gpurun_out/r3_direct.log
+# 5) protocol fuzz on the GPU engine: 2 procs / 1 GPU, small-slot geometry
+#    (the late-round-2 deadlock surface — seeds from test_protocol_fuzz_tiny)
+timeout 400 python - <<'EOF' > gpurun_out/r3_fuzz.log 2>&1
+import sys
+sys.path.insert(0, "/root/repo"); sys.path.insert(0, "/root/repo/tests")
+from emu_util import run_ranks
+import test_emulator as T
+SMALLG = {"n_slots": 4, "slot_bytes": 4096, "timeout_us": 20_000_000}
+for seed in (11, 23, 100, 161):
+    run_ranks(lambda a, r, n: T._fuzz(a, r, n, seed), 2, backend="gpu",
+              opts=SMALLG, timeout=120)
+    print("gpu fuzz seed", seed, "ok", flush=True)
+EOF
+echo "FUZZ_RC=$?"; tail -5 gpurun_out/r3_fuzz.log

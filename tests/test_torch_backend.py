@@ -144,6 +144,16 @@ def _gpu_rank(rank, world, port, q):
         dist.all_gather_into_tensor(out, torch.full((128,), float(rank + 5),
                                                     device="cuda"))
         assert out[:128].cpu().eq(5.0).all() and out[128:].cpu().eq(6.0).all()
+        rt = torch.full((96,), float(rank + 1), device="cuda")
+        dist.reduce(rt, dst=1)
+        if rank == 1:
+            assert rt.cpu().eq(3.0).all()
+        if rank == 0:
+            dist.send(torch.full((64,), 7.0, device="cuda"), dst=1, tag=9)
+        else:
+            rbuf = torch.zeros(64, device="cuda")
+            dist.recv(rbuf, src=0, tag=9)
+            assert rbuf.cpu().eq(7.0).all()
         dist.barrier()
         dist.destroy_process_group()
         q.put((rank, None))

@@ -2553,7 +2553,9 @@ struct Cclo {
           if (!p.used) continue;
           w[k++] = u64(p.d.scenario) | (u64(p.d.root_src_dst) << 8) |
                    (u64(p.d.tag) << 16) | (u64(p.ps.step) << 48);
-          w[k++] = p.ring_idx;
+          // progress cursor (w[0] of the ParkState: elements sent/got or
+          // windows posted) packed with the ring index for the host decode
+          w[k++] = (p.ring_idx & 0xFFFFFFFFull) | (p.ps.w[0] << 32);
         }
         u64 pa_n = 0, pd_n = 0, uq_n = 0;
         for (u32 r = 0; r < cfg.nranks; ++r) {

@@ -450,12 +450,13 @@ std::string ACCL::dump_engine_status() {
       if (!w[k]) break;
       std::snprintf(buf, sizeof(buf),
                     "  parked[%u]: op=%llu peer=%llu tag=0x%llx step=%llu "
-                    "ring_idx=%llu\n",
+                    "ring_idx=%llu progress=%llu\n",
                     (k - 2) / 2, (unsigned long long)(w[k] & 0xFF),
                     (unsigned long long)((w[k] >> 8) & 0xFF),
                     (unsigned long long)((w[k] >> 16) & 0xFFFFFFFF),
                     (unsigned long long)(w[k] >> 48),
-                    (unsigned long long)w[k + 1]);
+                    (unsigned long long)(w[k + 1] & 0xFFFFFFFF),
+                    (unsigned long long)(w[k + 1] >> 32));
       out += buf;
     }
   }

@@ -1341,6 +1341,33 @@ def test_self_sendrecv():
     run_ranks(_self_sendrecv, 2, opts=SMALL)
 
 
+def _compressed_head_to_head(a, rank, n):
+    """Compressed (f32 data, f16 wire) async exchange over the credit
+    window with a barrier forcing drain-driven progress: parked compressed
+    sends are always eager regardless of size, so the drain's flow-free
+    segment pushes must cast correctly in both directions."""
+    cnt = 9000
+    other = 1 - rank
+    rng = np.random.default_rng(5)
+    x = (rng.standard_normal(cnt).astype(np.float32) / 8) \
+        .astype(np.float16).astype(np.float32)
+    s, d = _mk(a, cnt), _mk(a, cnt)
+    s.write(x if rank == 0 else x * 2)
+    sreq = a.send(s, cnt, dst=other, tag=70, run_async=True,
+                  compress_dtype=A.DataType.float16)
+    rreq = a.recv(d, cnt, src=other, tag=70, run_async=True,
+                  compress_dtype=A.DataType.float16)
+    a.barrier()
+    assert sreq.wait() == 0 and rreq.wait() == 0
+    exp = (x * 2 if rank == 0 else x)
+    assert np.allclose(rd(d, cnt), exp, atol=0.01)
+
+
+def test_compressed_head_to_head():
+    for _ in range(3):
+        run_ranks(_compressed_head_to_head, 2, opts=SMALL)
+
+
 def _batched_p40(a, rank, n):
     """Large-P batched fallbacks (2P-1 > MAX_FLOWS=72): allreduce ring,
     reduce_scatter_batched's paired tx/rx, allgather_batched."""

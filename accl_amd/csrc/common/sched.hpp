@@ -355,9 +355,11 @@ struct Cclo {
             if (u.bytes == 0) continue;
             if (mtag != TAG_ANY && u.tag != mtag) continue;
             if (got > 0 && u.tag != mtag) continue;
-            if (u.arith != u32(wdt)) break;      // leave for retry to error
+            // arith/segment violations are the retry's to report
+            // (E_COMPRESSION/E_SEGMENT); stop pool delivery here
+            if (u.arith != u32(wdt)) break;
             u64 nseg = u.bytes / wsz;
-            if (nseg > n - got) break;           // leave for retry to error
+            if (nseg > n - got) break;
             if (got == 0) mtag = u.tag;
             if (!copy_free(spill_ptr(u.spare_slot), wdt, dst + got * dsz,
                            dt, nseg))
@@ -860,23 +862,20 @@ struct Cclo {
         if (!wait_pred_tick(deadline)) break;
       }
     }
-    // allocate a progress word (pool is sized to never exhaust; the wait
-    // is a belt for error-path leaks)
+    // allocate a progress word (the pool is sized to never exhaust
+    // structurally; the deadline wait is a belt for error-path leaks and
+    // fails loud via E_TIMEOUT)
     u64* pb = cold->prog_busy[gpeer & 63];
     u32 idx = 0;
+    u64 deadline = 0;
     for (;;) {
       idx = 0;
       while (idx < N_PROG && ((pb[idx >> 6] >> (idx & 63)) & 1)) ++idx;
       if (idx < N_PROG) break;
-      u64 deadline = deadline_now();
-      while (idx >= N_PROG) {
-        consume_addr_tick(gpeer);
-        consume_done_tick(gpeer);
-        if (!wait_pred_tick(deadline)) { idx = 0; break; }
-        idx = 0;
-        while (idx < N_PROG && ((pb[idx >> 6] >> (idx & 63)) & 1)) ++idx;
-      }
-      break;
+      if (!deadline) deadline = deadline_now();
+      consume_addr_tick(gpeer);
+      consume_done_tick(gpeer);
+      if (!wait_pred_tick(deadline)) { idx = 0; break; }
     }
     pb[idx >> 6] |= 1ull << (idx & 63);
     u64 seq = ++sq.rndzv_addr_tx[gpeer];
@@ -887,7 +886,9 @@ struct Cclo {
     publish_rec((volatile u64*)r, val);
     return idx;
   }
-  ACCL_HD static u32 rec_slot(const RndzvRec& rec, u32 /*n_rndzv*/) {
+  // progress-word pool index carried in the record (decoupled from the
+  // addr-ring slot, which recycles while windows are pending)
+  ACCL_HD static u32 rec_slot(const RndzvRec& rec) {
     return u32(rec.prog_idx);
   }
 
@@ -1181,7 +1182,7 @@ struct Cclo {
       while (!mv->poll(tok))
         if (!wait_pred_tick_nodrain(deadline)) return false;
       fence_release_sys();
-      st_sys(tv.direct_word(peer, me(), rec_slot(rec, cfg.n_rndzv)),
+      st_sys(tv.direct_word(peer, me(), rec_slot(rec)),
              w * u64(dtype_size(DataType(m.dst_dt))));
       sent += w;
       ps.w[0] = sent;
@@ -1431,7 +1432,7 @@ struct Cclo {
           RndzvRec rec{};
           if (!wait_addr(c.global(p), tag, rec)) return err;
           mk_tx_direct(nf++, c.global(p), src, dt, dt, n, rec.offset,
-                       rec_slot(rec, cfg.n_rndzv));
+                       rec_slot(rec));
         } else {
           mk_tx(nf++, c.global(p), src, dt, wdt, n, tag);
         }
@@ -1479,7 +1480,7 @@ struct Cclo {
           RndzvRec rec{};
           if (!wait_addr(c.global(p), tag, rec)) return err;
           mk_tx_direct(nf++, c.global(p), s, dt, dt, n, rec.offset,
-                       rec_slot(rec, cfg.n_rndzv));
+                       rec_slot(rec));
         } else {
           mk_tx(nf++, c.global(p), s, dt, wdt, n, tag);
         }
@@ -1529,7 +1530,7 @@ struct Cclo {
       RndzvRec rec{};
       if (!wait_addr(c.global(root), tag, rec)) return err;
       mk_tx_direct(0, c.global(root), src, dt, dt, n, rec.offset,
-                   rec_slot(rec, cfg.n_rndzv));
+                   rec_slot(rec));
       return run_flows(1);
     }
     mk_tx(0, c.global(root), src, dt, wdt, n, tag);
@@ -1569,7 +1570,7 @@ struct Cclo {
         RndzvRec rec{};
         if (!wait_addr(c.global(p), tag, rec)) return err;
         mk_tx_direct(nf++, c.global(p), src, dt, dt, n, rec.offset,
-                     rec_slot(rec, cfg.n_rndzv));
+                     rec_slot(rec));
         mk_rx_direct(nf++, c.global(p), n, dt, myslot[p]);
       }
       return run_flows(nf);
@@ -1668,7 +1669,7 @@ struct Cclo {
       if (!wait_addr(c.global(root), tag, rec)) return err;
       u64 w = min64(n - sent, rec.count);
       mk_tx_direct(0, c.global(root), src + sent * esz, dt, dt, w, rec.offset,
-                   rec_slot(rec, cfg.n_rndzv));
+                   rec_slot(rec));
       u32 e = run_flows(1);
       if (e) return e;
       sent += w;
@@ -1774,7 +1775,7 @@ struct Cclo {
         if (!wait_addr(c.global(p), tag, rec)) return err;
         mk_tx_direct(nf++, c.global(p), src + (u64(p) * n + off) * esz, dt,
                      dt, min64(wc, rec.count), rec.offset,
-                     rec_slot(rec, cfg.n_rndzv));
+                     rec_slot(rec));
       }
       for (u32 p = 0; p < P; ++p)
         if (p != r) mk_rx_direct(nf++, c.global(p), wc, dt, sp[bank][p]);
@@ -2179,7 +2180,7 @@ struct Cclo {
         RndzvRec rec{};
         if (!wait_addr(c.global(p), tag, rec)) return err;
         mk_tx_direct(nf++, c.global(p), src + u64(p) * n * esz, dt, dt, n,
-                     rec.offset, rec_slot(rec, cfg.n_rndzv));
+                     rec.offset, rec_slot(rec));
         mk_rx_direct(nf++, c.global(p), n, dt, sp[p]);
       }
       return run_flows(nf);

@@ -94,6 +94,7 @@ struct Cclo {
   u64 max_rndzv_bytes;     // window cap for a single posted rendezvous
                            // window (0 = unlimited); reference:
                            // set_max_rendezvous_size (accl.hpp:103-104)
+  u32 tune_oneshot_max;    // one-shot fan-in cutoff bytes (0 = 64 KB)
   u32 tune_fullmesh_max;   // allreduce fullmesh->ring cutoff (runtime
                            // tuning register; reference flat-tree caps,
                            // driver/xrt/src/accl.cpp:1198-1208); 0 = default
@@ -1856,9 +1857,9 @@ struct Cclo {
   // small allreduce/reduce eligibility: whole message in one slot, no
   // compression, fan-in within the n-ary move budget
   ACCL_HD bool one_shot_ok(u64 n, DataType dt, DataType wdt, u32 P) const {
+    u64 cap = tune_oneshot_max ? tune_oneshot_max : (64u << 10);
     return dt == wdt && P >= 2 && P <= MOVE_MAX_SRC &&
-           n * dtype_size(dt) <= (64u << 10) &&
-           n * dtype_size(dt) <= cfg.slot_bytes;
+           n * dtype_size(dt) <= cap && n * dtype_size(dt) <= cfg.slot_bytes;
   }
 
   // one-shot allreduce: everyone broadcasts its vector; every rank folds
@@ -2576,6 +2577,7 @@ struct Cclo {
         // tuning registers (reference configure_tuning_parameters):
         // knob id in root_src_dst, value in count
         if (d.root_src_dst == 0) tune_fullmesh_max = u32(desc_count(d));
+        if (d.root_src_dst == 1) tune_oneshot_max = u32(desc_count(d));
         return E_OK;
       case CfgFunc::reset: {
         // soft reset (reference: encore_soft_reset drains retry queue +

@@ -1601,6 +1601,19 @@ def _tuning(a, rank, n):
     exp = np.stack([pattern(cnt, r, seed=8) for r in range(n)]).sum(0)
     assert np.allclose(rd(d, cnt), exp)
     a.barrier()
+    # knob 1: one-shot fan-in cutoff. Force it OFF (1 byte) then generous;
+    # a small allreduce must be correct through either path.
+    for cap in (1, 1 << 20):
+        a.set_tuning(1, cap)
+        c2 = 500  # 2 KB: one-shot eligible only when cap allows
+        s2, d2 = _mk(a, c2), _mk(a, c2)
+        s2.write(pattern(c2, rank, seed=9 + cap))
+        a.allreduce(s2, d2, c2, RF.SUM)
+        exp2 = np.stack([pattern(c2, r, seed=9 + cap)
+                         for r in range(n)]).sum(0)
+        assert np.allclose(rd(d2, c2), exp2), f"cap={cap}"
+    a.set_tuning(1, 0)  # back to default
+    a.barrier()
 
 
 def test_tuning_registers():

@@ -168,6 +168,10 @@ struct Cclo {
   u64 drain_hold_;
 
   ACCL_HD u32 me() const { return cfg.rank; }
+  // wildcard (TAG_ANY) matching must never capture engine-internal
+  // collective traffic: TAG_COLL-tagged segments/records belong to the
+  // collective state machines, not to user receives
+  ACCL_HD static bool any_ok(u32 tag) { return !(tag & TAG_COLL); }
 
   // device-only micro-timeline (GPU: GpuMover::dbg; emulator mover: no-op)
   template <class M>
@@ -354,7 +358,7 @@ struct Cclo {
           for (u32 qi = cold->uq_h[peer]; qi != cold->uq_t[peer]; ++qi) {
             Unexpected& u = cold->uq[peer][qi % UQ_DEPTH];
             if (u.bytes == 0) continue;
-            if (mtag != TAG_ANY && u.tag != mtag) continue;
+            if (mtag == TAG_ANY ? !any_ok(u.tag) : u.tag != mtag) continue;
             if (got > 0 && u.tag != mtag) continue;
             // arith/segment violations are the retry's to report
             // (E_COMPRESSION/E_SEGMENT); stop pool delivery here
@@ -380,7 +384,7 @@ struct Cclo {
           SlotHdr* h = tv.slot_hdr(me(), peer, sl);
           if (ld_sys(&h->seq) != seq) break;
           fence_acquire_sys();
-          bool match = (mtag == TAG_ANY) || (h->tag == mtag);
+          bool match = mtag == TAG_ANY ? any_ok(h->tag) : (h->tag == mtag);
           u64 nseg = u64(h->bytes) / wsz;
           if (match && h->arith == u32(wdt) && nseg <= n - got) {
             if (got == 0) mtag = h->tag;
@@ -604,7 +608,7 @@ struct Cclo {
               Unexpected& u = cold->uq[f.gpeer][qi % UQ_DEPTH];
               if (u.bytes == 0) continue;
               u32 want = (f.submitted == 0) ? f.tag : f.matched_tag;
-              if (want != TAG_ANY && u.tag != want) continue;
+              if (want == TAG_ANY ? !any_ok(u.tag) : u.tag != want) continue;
               if (u.arith != u32(f.wdt)) { err |= E_COMPRESSION; return any; }
               u32 wsz0 = dtype_size(DataType(f.wdt));
               u64 n0 = u.bytes / wsz0;
@@ -654,14 +658,18 @@ struct Cclo {
           // forever. (reference analogue: rxbuf_seek matching by
           // (tag, src, seqn), rxbuf_seek.cpp:53-72)
           u32 want = (f.submitted == 0) ? f.tag : f.matched_tag;
-          if (want != TAG_ANY && h->tag != want) {
+          bool wmatch = want == TAG_ANY ? any_ok(h->tag) : (h->tag == want);
+          if (!wmatch) {
             bool sibling = false;
             for (u32 j = 0; j < nflows_; ++j) {
               const Flow& g = flows[j];
               if (g.kind != FLOW_RX || flow_done(g) || g.gpeer != f.gpeer)
                 continue;
               u32 gw = (g.submitted == 0) ? g.tag : g.matched_tag;
-              if (gw == TAG_ANY || gw == h->tag) { sibling = true; break; }
+              if ((gw == TAG_ANY && any_ok(h->tag)) || gw == h->tag) {
+                sibling = true;
+                break;
+              }
             }
             if (!sibling) {
               if (spill_head(f.gpeer, h, slot, next + 1)) { any = true; continue; }
@@ -904,8 +912,10 @@ struct Cclo {
       for (u32 k = 0; k < RNDZV_PEND; ++k) {
         PendRndzv& p = cold->pa[gpeer][k];
         if (!p.valid) continue;
-        if (want_tag != TAG_ANY && p.tag != want_tag && p.tag != TAG_ANY)
-          continue;
+        bool m = want_tag == TAG_ANY ? any_ok(p.tag)
+                 : (p.tag == want_tag ||
+                    (p.tag == TAG_ANY && any_ok(want_tag)));
+        if (!m) continue;
         if (best < 0 || p.seq < cold->pa[gpeer][best].seq) best = int(k);
       }
       if (best >= 0) {
@@ -926,7 +936,9 @@ struct Cclo {
         u64 roff = r->offset, rcnt = r->count, rprog = r->prog_idx;
         sq.rndzv_addr_rx[gpeer] = seq;
         st_sys(&tv.chan_ctl(gpeer, me())->addr_ret, seq);
-        if (want_tag == TAG_ANY || rtag == want_tag || rtag == TAG_ANY) {
+        bool m = want_tag == TAG_ANY ? any_ok(rtag)
+                 : (rtag == want_tag || (rtag == TAG_ANY && any_ok(want_tag)));
+        if (m) {
           out.seq = seq; out.tag = rtag; out.arith = rarith;
           out.offset = roff; out.count = rcnt;
           out.prog_idx = rprog;
@@ -957,8 +969,10 @@ struct Cclo {
     for (;;) {
       for (u32 k = 0; k < RNDZV_PEND; ++k) {
         PendRndzv& p = cold->pa[gpeer][k];
-        if (p.valid && (want_tag == TAG_ANY || p.tag == want_tag ||
-                        p.tag == TAG_ANY))
+        if (p.valid && (want_tag == TAG_ANY
+                            ? any_ok(p.tag)
+                            : (p.tag == want_tag ||
+                               (p.tag == TAG_ANY && any_ok(want_tag)))))
           return true;
       }
       u64 seq = sq.rndzv_addr_rx[gpeer] + 1;
@@ -966,7 +980,10 @@ struct Cclo {
       if (ld_sys(&r->seq) != seq) return false;
       fence_acquire_sys();
       u32 t = r->tag;
-      if (want_tag == TAG_ANY || t == want_tag || t == TAG_ANY) return true;
+      if (want_tag == TAG_ANY ? any_ok(t)
+                              : (t == want_tag ||
+                                 (t == TAG_ANY && any_ok(want_tag))))
+        return true;
       // spill the foreign head to pending and keep looking
       u32 k = 0;
       while (k < RNDZV_PEND && cold->pa[gpeer][k].valid) ++k;
@@ -986,8 +1003,10 @@ struct Cclo {
     for (;;) {
       for (u32 k = 0; k < RNDZV_PEND; ++k) {
         PendRndzv& p = cold->pd[gpeer][k];
-        if (p.valid && (want_tag == TAG_ANY || p.tag == want_tag ||
-                        p.tag == TAG_ANY))
+        if (p.valid && (want_tag == TAG_ANY
+                            ? any_ok(p.tag)
+                            : (p.tag == want_tag ||
+                               (p.tag == TAG_ANY && any_ok(want_tag)))))
           return true;
       }
       u64 seq = sq.rndzv_done_rx[gpeer] + 1;
@@ -995,7 +1014,10 @@ struct Cclo {
       if (ld_sys(&r->seq) != seq) return false;
       fence_acquire_sys();
       u32 t = r->tag;
-      if (want_tag == TAG_ANY || t == want_tag || t == TAG_ANY) return true;
+      if (want_tag == TAG_ANY ? any_ok(t)
+                              : (t == want_tag ||
+                                 (t == TAG_ANY && any_ok(want_tag))))
+        return true;
       u32 k = 0;
       while (k < RNDZV_PEND && cold->pd[gpeer][k].valid) ++k;
       if (k >= RNDZV_PEND) return false;  // pending full: try again later
@@ -1026,8 +1048,10 @@ struct Cclo {
       for (u32 k = 0; k < RNDZV_PEND; ++k) {
         PendRndzv& p = cold->pd[gpeer][k];
         if (!p.valid) continue;
-        if (want_tag != TAG_ANY && p.tag != want_tag && p.tag != TAG_ANY)
-          continue;
+        bool m = want_tag == TAG_ANY ? any_ok(p.tag)
+                 : (p.tag == want_tag ||
+                    (p.tag == TAG_ANY && any_ok(want_tag)));
+        if (!m) continue;
         if (best < 0 || p.seq < cold->pd[gpeer][best].seq) best = int(k);
       }
       if (best >= 0) { cold->pd[gpeer][best].valid = 0; return true; }
@@ -1038,7 +1062,9 @@ struct Cclo {
         u32 rtag = r->tag;  // copy before publishing consumption
         sq.rndzv_done_rx[gpeer] = seq;
         st_sys(&tv.chan_ctl(gpeer, me())->done_ret, seq);
-        if (want_tag == TAG_ANY || rtag == want_tag || rtag == TAG_ANY)
+        if (want_tag == TAG_ANY ? any_ok(rtag)
+                                : (rtag == want_tag ||
+                                   (rtag == TAG_ANY && any_ok(want_tag))))
           return true;
         u32 k = 0;
         while (k < RNDZV_PEND && cold->pd[gpeer][k].valid) ++k;
@@ -1256,7 +1282,7 @@ struct Cclo {
       for (u32 qi = cold->uq_h[peer]; qi != cold->uq_t[peer]; ++qi) {
         Unexpected& u = cold->uq[peer][qi % UQ_DEPTH];
         if (u.bytes == 0) continue;  // consumed hole
-        if (mtag != TAG_ANY && u.tag != mtag) continue;
+        if (mtag == TAG_ANY ? !any_ok(u.tag) : u.tag != mtag) continue;
         if (u.arith != u32(wdt)) { err |= E_COMPRESSION; return err; }
         if (got == 0) mtag = u.tag;
         else if (u.tag != mtag) continue;
@@ -1283,7 +1309,7 @@ struct Cclo {
       SlotHdr* hd = tv.slot_hdr(me(), peer, sl);
       if (ld_sys(&hd->seq) == seq) {
         fence_acquire_sys();
-        bool match = (mtag == TAG_ANY) || (hd->tag == mtag);
+        bool match = mtag == TAG_ANY ? any_ok(hd->tag) : (hd->tag == mtag);
         if (match) {
           if (hd->arith != u32(wdt)) { err |= E_COMPRESSION; return err; }
           if (got == 0) mtag = hd->tag;

@@ -1341,6 +1341,38 @@ def test_self_sendrecv():
     run_ranks(_self_sendrecv, 2, opts=SMALL)
 
 
+def _parked_tag_any(a, rank, n):
+    """TAG_ANY recvs parked behind collectives: the drain must lock onto
+    the first-arriving tag per recv (MPI wildcard semantics) while
+    delivering directly to the destination."""
+    cnt = 6000
+    if rank == 1:
+        d1, d2 = _mk(a, cnt), _mk(a, cnt)
+        r1 = a.recv(d1, cnt, src=0, tag=A.TAG_ANY, run_async=True)
+        r2 = a.recv(d2, cnt, src=0, tag=A.TAG_ANY, run_async=True)
+    s, dr = _mk(a, 400), _mk(a, 400)
+    s.write(pattern(400, rank, seed=13))
+    for _ in range(6):
+        a.allreduce(s, dr, 400, RF.SUM)
+    if rank == 0:
+        s1, s2 = _mk(a, cnt), _mk(a, cnt)
+        s1.write(pattern(cnt, 21, seed=1))
+        s2.write(pattern(cnt, 22, seed=1))
+        a.send(s1, cnt, dst=1, tag=101)
+        a.send(s2, cnt, dst=1, tag=202)
+    else:
+        assert r1.wait() == 0 and r2.wait() == 0
+        # wildcard recvs match in arrival order
+        assert np.array_equal(rd(d1, cnt), pattern(cnt, 21, seed=1))
+        assert np.array_equal(rd(d2, cnt), pattern(cnt, 22, seed=1))
+    a.barrier()
+
+
+def test_parked_tag_any():
+    for _ in range(3):
+        run_ranks(_parked_tag_any, 2, opts=SMALL)
+
+
 def _compressed_head_to_head(a, rank, n):
     """Compressed (f32 data, f16 wire) async exchange over the credit
     window with a barrier forcing drain-driven progress: parked compressed

@@ -1434,10 +1434,16 @@ def test_batched_collectives_p40():
 
 def _fuzz_script(seed, n, nops=40):
     """Deterministic op script shared by all ranks: mixes collectives,
-    tagged pairwise send/recv (sync + async), sizes spanning inline/fleet/
-    rendezvous. Every rank derives the same script from the seed."""
+    tagged pairwise send/recv (sync + async + wildcard), sizes spanning
+    inline/fleet/rendezvous. Every rank derives the same script from the
+    seed. Wildcard (TAG_ANY) recvs are only generated for SYNC transfers
+    on pairs with no async history, where matching is deterministic."""
     rng = np.random.default_rng(seed)
+    # wildcard flags come from a SEPARATE stream so historical seeds keep
+    # reproducing the exact scripts that caught their bugs
+    rng_any = np.random.default_rng(seed + 987_654_321)
     ops = []
+    dirty = set()  # (src, dst) pairs with in-flight async traffic
     for _ in range(nops):
         k = int(rng.integers(0, 7))
         cnt = int(rng.integers(16, 30_000))
@@ -1452,8 +1458,13 @@ def _fuzz_script(seed, n, nops=40):
             ops.append(("bcast", cnt, int(rng.integers(0, n))))
         elif k == 4:
             a_, b_ = rng.choice(n, size=2, replace=False)
-            ops.append(("sendrecv", cnt, int(a_), int(b_), tag,
-                        bool(rng.integers(0, 2))))
+            asyn = bool(rng.integers(0, 2))
+            anytag = (not asyn and (int(a_), int(b_)) not in dirty and
+                      bool(rng_any.integers(0, 3) == 0))
+            if asyn:
+                dirty.add((int(a_), int(b_)))
+            ops.append(("sendrecv", cnt, int(a_), int(b_), tag, asyn,
+                        anytag))
         elif k == 5:
             ops.append(("alltoall", cnt))
         else:
@@ -1497,7 +1508,7 @@ def _fuzz(a, rank, n, seed):
             assert np.array_equal(rd(b, cnt), pattern(cnt, 55, seed=i)), \
                 f"op{i} bcast"
         elif op[0] == "sendrecv":
-            cnt, src_r, dst_r, tag, asyn = op[1:]
+            cnt, src_r, dst_r, tag, asyn, anytag = op[1:]
             if rank == src_r:
                 s = _mk(a, cnt)
                 s.write(pattern(cnt, src_r, seed=i))
@@ -1508,7 +1519,8 @@ def _fuzz(a, rank, n, seed):
                     pending.append((r, None, None, s))
             elif rank == dst_r:
                 d = _mk(a, cnt)
-                r = a.recv(d, cnt, src=src_r, tag=tag, run_async=asyn)
+                rtag = A.TAG_ANY if anytag else tag
+                r = a.recv(d, cnt, src=src_r, tag=rtag, run_async=asyn)
                 if asyn:
                     pending.append((r, d, cnt, pattern(cnt, src_r, seed=i)))
                 else:

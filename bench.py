@@ -67,8 +67,10 @@ def main():
     if backend == "auto":
         backend = "gpu" if A.ACCL._has_gpu() else "emu"
     slot_mb = 4 if backend == "gpu" else 1
-    heap = max(4 * args.bytes + (64 << 20), 1 << 30) if backend == "gpu" else None
-    if heap is not None and args.sweep:
+    # size the arena heap for src+dst (+sweep buffers) on BOTH backends —
+    # the emulator arena is plain shm, so large default-bytes runs work too
+    heap = max(4 * args.bytes + (64 << 20), 1 << 30)
+    if args.sweep:
         heap += 2 << 30  # dedicated 1 GiB sweep src+dst (full-range sweep)
     a = A.ACCL(nranks=world, rank=rank, backend=backend,
                heap_bytes=heap,

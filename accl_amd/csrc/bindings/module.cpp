@@ -493,7 +493,10 @@ PYBIND11_MODULE(_core, m) {
     d["features"] = py::make_tuple(
         "multi_call_interleaving", "ooo_rendezvous_matching",
         "windowed_nary_fan_in", "compression_f16_bf16_wire",
-        "device_initiated_calls", "stream_rings", "torch_backend");
+        "device_initiated_calls", "stream_rings", "torch_backend",
+        "nonblocking_probe_ops", "parked_drain_progress",
+        "flow_controlled_rndzv_rings", "pooled_progress_words",
+        "one_shot_small_fan_in", "wildcard_user_tags");
     return d;
   });
 

@@ -1341,6 +1341,72 @@ def test_self_sendrecv():
     run_ranks(_self_sendrecv, 2, opts=SMALL)
 
 
+def _parked_vs_subcomm(a, rank, n):
+    """A parked GLOBAL-comm recv while a SUBGROUP communicator's
+    collectives grind the same pair channels: cross-comm tag demux (tags
+    differ only in comm_id bits) plus drain delivery to the parked recv."""
+    cnt = 8000
+    if rank == 2:
+        d = _mk(a, cnt)
+        req = a.recv(d, cnt, src=0, tag=33, run_async=True)
+    half = [0, 1]
+    cid = a.split_communicator(half) if rank in half else None
+    if rank in half:
+        s, dr = _mk(a, 700), _mk(a, 700)
+        s.write(pattern(700, rank, seed=17))
+        for _ in range(8):
+            a.allreduce(s, dr, 700, RF.SUM, comm=cid)
+        exp = np.stack([pattern(700, r, seed=17) for r in half]).sum(0)
+        assert np.allclose(rd(dr, 700), exp)
+    if rank == 0:
+        sv = _mk(a, cnt)
+        sv.write(pattern(cnt, 8, seed=18))
+        a.send(sv, cnt, dst=2, tag=33)
+    elif rank == 2:
+        assert req.wait() == 0
+        assert np.array_equal(rd(d, cnt), pattern(cnt, 8, seed=18))
+    a.barrier()
+
+
+def test_parked_vs_subcomm():
+    for _ in range(3):
+        run_ranks(_parked_vs_subcomm, 3, opts=SMALL)
+
+
+def _async_collectives(a, rank, n):
+    """Async collectives queue on the ring and the engine serializes them;
+    the host may fire several before waiting any (plus an async send/recv
+    parked across the whole batch)."""
+    if rank == 1:
+        dx = _mk(a, 3000)
+        rx = a.recv(dx, 3000, src=0, tag=55, run_async=True)
+    reqs = []
+    bufs = []
+    for i in range(4):
+        s, d = _mk(a, 1000), _mk(a, 1000)
+        s.write(pattern(1000, rank + i, seed=20 + i))
+        reqs.append(a.allreduce(s, d, 1000, RF.SUM, run_async=True))
+        bufs.append((s, d, i))
+    for (s, d, i), r in zip(bufs, reqs):
+        assert r.wait() == 0
+        exp = np.stack([pattern(1000, rr + i, seed=20 + i)
+                        for rr in range(n)]).sum(0)
+        assert np.allclose(rd(d, 1000), exp), f"async ar {i}"
+    if rank == 0:
+        sv = _mk(a, 3000)
+        sv.write(pattern(3000, 77, seed=2))
+        a.send(sv, 3000, dst=1, tag=55)
+    elif rank == 1:
+        assert rx.wait() == 0
+        assert np.array_equal(rd(dx, 3000), pattern(3000, 77, seed=2))
+    a.barrier()
+
+
+def test_async_collectives():
+    for _ in range(3):
+        run_ranks(_async_collectives, 2, opts=SMALL)
+
+
 def _parked_tag_any(a, rank, n):
     """TAG_ANY recvs parked behind collectives: the drain must lock onto
     the first-arriving tag per recv (MPI wildcard semantics) while

@@ -346,6 +346,10 @@ void ACCL::set_timeout_ms(u64 ms) {
   be_->call(d);
 }
 void ACCL::set_max_eager_size(u64 bytes) {
+  // protocol-selection config: a PARKED call that chose its eager/rndzv
+  // path under the old threshold must not be re-interpreted under the new
+  // one (its ParkState encoding is path-specific) — drain the queue first
+  be_->quiesce();
   CallDesc d = make_desc(Op::config, bytes, DataType::none, DataType::none);
   d.function = u32(CfgFunc::set_max_eager_size);
   be_->call(d);
@@ -357,6 +361,7 @@ void ACCL::set_tuning(u32 knob, u64 value) {
   be_->call(d);
 }
 void ACCL::set_max_rendezvous_size(u64 bytes) {
+  be_->quiesce();  // same rule as set_max_eager_size (window sizing)
   CallDesc d = make_desc(Op::config, bytes, DataType::none, DataType::none);
   d.function = u32(CfgFunc::set_max_rendezvous_size);
   be_->call(d);

@@ -231,7 +231,7 @@ class ACCL:
                     "info", "set_timeout_ms", "set_max_eager_size",
                     "set_max_rendezvous_size", "set_tuning",
                     "dump_communicator", "dump_eager_rx_buffers",
-                    "dump_streams", "dump_engine_status",
+                    "dump_streams", "dump_engine_status", "dump_rendezvous",
                     "create_communicator", "split_communicator",
                     "free_request", "deinit"):
             return getattr(self._a, name)

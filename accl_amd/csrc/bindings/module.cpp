@@ -336,6 +336,7 @@ PYBIND11_MODULE(_core, m) {
            py::arg("verbose") = false)
       .def("dump_streams", &ACCL::dump_streams)
       .def("dump_engine_status", &ACCL::dump_engine_status)
+      .def("dump_rendezvous", &ACCL::dump_rendezvous)
       .def("barrier", &ACCL::barrier, py::arg("comm") = GLOBAL_COMM,
            py::arg("run_async") = false, py::return_value_policy::reference, py::keep_alive<0, 1>(),
            py::call_guard<py::gil_scoped_release>())

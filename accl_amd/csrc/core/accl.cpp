@@ -468,6 +468,57 @@ std::string ACCL::dump_engine_status() {
   return out;
 }
 
+std::string ACCL::dump_rendezvous() {
+  // Host-readable snapshot of MY arena's rendezvous surface: inbound addr
+  // records (windows receivers posted to me as a sender), inbound done
+  // records, and the flow-control credits my own posts run against.
+  // (reference analogue: exchange-memory debug dumps, accl.cpp:964-1048)
+  const ProtoConfig& c = be_->cfg();
+  ArenaLayout L = arena_layout(c);
+  std::string out = "rendezvous rings (" + std::to_string(c.n_rndzv) +
+                    " records/pair, prog pool " + std::to_string(N_PROG) +
+                    "):\n";
+  char b[200];
+  for (u32 p = 0; p < c.nranks; ++p) {
+    u64 a_new = 0, d_new = 0;
+    std::string recs;
+    for (u32 i = 0; i < c.n_rndzv; ++i) {
+      RndzvRec r{};
+      be_->read_arena(L.rndzv_addr_off +
+                          (u64(p) * c.n_rndzv + i) * sizeof(RndzvRec),
+                      &r, sizeof(r));
+      if (r.seq > a_new) a_new = r.seq;
+      if (r.seq) {
+        std::snprintf(b, sizeof(b),
+                      "    addr[%u]: seq=%llu tag=0x%x count=%llu "
+                      "prog_idx=%llu\n",
+                      i, (unsigned long long)r.seq, r.tag,
+                      (unsigned long long)r.count,
+                      (unsigned long long)r.prog_idx);
+        recs += b;
+      }
+      RndzvRec d{};
+      be_->read_arena(L.rndzv_done_off +
+                          (u64(p) * c.n_rndzv + i) * sizeof(RndzvRec),
+                      &d, sizeof(d));
+      if (d.seq > d_new) d_new = d.seq;
+    }
+    EagerChanCtl ctl{};
+    u64 lane_bytes = sizeof(EagerChanCtl) + u64(c.n_slots) * sizeof(SlotHdr);
+    be_->read_arena(L.eager_off + u64(p) * lane_bytes, &ctl, sizeof(ctl));
+    std::snprintf(b, sizeof(b),
+                  "  pair %u: newest_inbound_addr_seq=%llu "
+                  "newest_inbound_done_seq=%llu my_posts_consumed: "
+                  "addr=%llu done=%llu\n",
+                  p, (unsigned long long)a_new, (unsigned long long)d_new,
+                  (unsigned long long)ctl.addr_ret,
+                  (unsigned long long)ctl.done_ret);
+    out += b;
+    out += recs;
+  }
+  return out;
+}
+
 std::string ACCL::dump_eager_rx_buffers(bool verbose) {
   const ProtoConfig& c = be_->cfg();
   ArenaLayout L = arena_layout(c);

@@ -211,6 +211,7 @@ class ACCL {
   // dump_exchange_memory, driver/xrt/src/accl.cpp:964-1048) ---
   std::string dump_communicator(u32 comm = GLOBAL_COMM);
   std::string dump_eager_rx_buffers(bool verbose = false);
+  std::string dump_rendezvous();
   std::string dump_streams();
   std::string dump_engine_status();
   u32 comm_size(u32 comm) const { return comm_sizes_.at(comm); }

@@ -1848,3 +1848,13 @@ def test_allreduce_replica_identity():
     run_ranks(_replica_identity, 2, opts=SMALL)
     run_ranks(_replica_identity, 3, opts=SMALL)
     run_ranks(_replica_identity, 3, opts=DIRECT)
+
+
+def _dump_rndzv(a, rank, n):
+    _rndzv_window(a, rank, n)
+    s = a.dump_rendezvous()
+    assert "rendezvous rings" in s and "my_posts_consumed" in s
+
+
+def test_dump_rendezvous():
+    run_ranks(_dump_rndzv, 2, opts=DIRECT)
